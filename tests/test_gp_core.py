@@ -1,0 +1,178 @@
+"""Numerics tests for the GP core (CPU torch path = kernel oracle)."""
+
+import math
+
+import numpy as np
+import pytest
+import torch
+
+from vizier_amd._src.gp import acquisitions as acq_lib
+from vizier_amd._src.gp import gp_model, lbfgs, output_warpers
+from vizier_amd._src.gp.matern import gram_matern52, matern52
+
+
+class TestMatern:
+
+  def test_gram_against_naive_loop(self):
+    rng = np.random.default_rng(0)
+    x = torch.tensor(rng.uniform(0, 1, (12, 4)), dtype=torch.float64)
+    ls = torch.tensor(rng.uniform(0.3, 2.0, (4,)), dtype=torch.float64)
+    amp = torch.tensor(1.7, dtype=torch.float64)
+    K = gram_matern52(x, None, ls, amp)
+    for i in range(12):
+      for j in range(12):
+        r = math.sqrt(float((((x[i] - x[j]) / ls) ** 2).sum()))
+        sr = math.sqrt(5) * r
+        expected = 1.7 ** 2 * (1 + sr + sr * sr / 3) * math.exp(-sr)
+        assert float(K[i, j]) == pytest.approx(expected, rel=1e-9)
+
+  def test_gram_psd_and_diag(self):
+    rng = np.random.default_rng(1)
+    x = torch.tensor(rng.uniform(0, 1, (50, 8)), dtype=torch.float64)
+    ls = torch.full((8,), 0.5, dtype=torch.float64)
+    K = gram_matern52(x, None, ls, torch.tensor(2.0, dtype=torch.float64))
+    assert torch.allclose(torch.diagonal(K),
+                          torch.full((50,), 4.0, dtype=torch.float64))
+    eigs = torch.linalg.eigvalsh(K)
+    assert float(eigs.min()) > -1e-8
+
+  def test_cross_gram_shape(self):
+    x1 = torch.rand(7, 3)
+    x2 = torch.rand(9, 3)
+    K = gram_matern52(x1, x2, torch.ones(3), torch.tensor(1.0))
+    assert K.shape == (7, 9)
+
+
+class TestLBFGS:
+
+  def test_batched_quadratic(self):
+    A = torch.tensor([[3.0, 1.0], [1.0, 2.0]])
+    targets = torch.tensor([[1.0, -2.0], [0.5, 3.0], [-1.0, 0.0]])
+
+    def loss(x):
+      d = x - targets
+      return torch.einsum('ri,ij,rj->r', d, A, d)
+
+    x0 = torch.zeros(3, 2)
+    x, f = lbfgs.minimize_batched(loss, x0, max_iters=50)
+    assert torch.allclose(x, targets, atol=1e-4)
+    assert float(f.max()) < 1e-7
+
+  def test_rosenbrock_batch(self):
+    def loss(x):
+      a, b = x[:, 0], x[:, 1]
+      return (1 - a) ** 2 + 100 * (b - a * a) ** 2
+
+    x0 = torch.tensor([[-1.0, 1.0], [0.0, 0.0], [2.0, 2.0]])
+    x, f = lbfgs.minimize_batched(loss, x0, max_iters=300)
+    assert float(f.min()) < 1e-4
+
+
+class TestGPTraining:
+
+  def _make_data(self, n=40, d=3, noise=0.01, seed=0):
+    rng = np.random.default_rng(seed)
+    x = rng.uniform(0, 1, (n, d))
+    y = np.sin(3 * x[:, 0]) + 0.5 * x[:, 1] ** 2 + \
+        noise * rng.standard_normal(n)
+    return (torch.tensor(x, dtype=torch.float32),
+            torch.tensor(y, dtype=torch.float32))
+
+  def test_fit_reduces_nll_and_interpolates(self):
+    x, y = self._make_data()
+    post = gp_model.train_gp(x, y, num_restarts=3, max_iters=40, seed=1)
+    mean, stddev = post.predict(x)
+    # In-sample predictions close to targets, small stddev.
+    assert float((mean - y).abs().mean()) < 0.1
+    assert float(stddev.mean()) < 0.3
+
+  def test_posterior_variance_grows_off_data(self):
+    x, y = self._make_data()
+    post = gp_model.train_gp(x, y, num_restarts=2, max_iters=30, seed=2)
+    _, stddev_on = post.predict(x[:5])
+    far = torch.full((1, 3), 5.0)
+    _, stddev_far = post.predict(far)
+    assert float(stddev_far[0]) > float(stddev_on.mean()) * 2
+
+  def test_kinv_matches_cholesky_solve_path(self):
+    x, y = self._make_data(n=25)
+    post = gp_model.train_gp(x, y, num_restarts=2, max_iters=30, seed=3)
+    xq = torch.rand(10, 3)
+    mean_a, std_a = post.predict(xq)
+    post_nokinv = gp_model.GPPosterior(
+        x=post.x, params=post.params, L=post.L, alpha=post.alpha,
+        K_inv=None, nll=post.nll)
+    mean_b, std_b = post_nokinv.predict(xq)
+    assert torch.allclose(mean_a, mean_b, atol=1e-4)
+    assert torch.allclose(std_a, std_b, atol=1e-3)
+
+  def test_cholesky_jitter_recovers_singular(self):
+    K = torch.ones(5, 5)  # rank-1, singular
+    L = gp_model.cholesky_with_jitter(K, torch.tensor(1.0))
+    assert torch.isfinite(L).all()
+
+
+class TestAcquisitions:
+
+  def test_ei_positive_and_monotone_in_mean(self):
+    ei = acq_lib.EI(best_value=0.0)
+    mean = torch.tensor([-1.0, 0.0, 1.0])
+    stddev = torch.ones(3)
+    vals = ei(mean, stddev)
+    assert (vals > 0).all()
+    assert vals[0] < vals[1] < vals[2]
+
+  def test_ucb_lcb(self):
+    mean, stddev = torch.tensor([1.0]), torch.tensor([0.5])
+    assert float(acq_lib.UCB(1.8)(mean, stddev)) == pytest.approx(1.9)
+    assert float(acq_lib.LCB(1.8)(mean, stddev)) == pytest.approx(0.1)
+
+  def test_trust_region_radius_and_penalty(self):
+    trusted = torch.rand(10, 4) * 0.1
+    tr = acq_lib.TrustRegion(trusted)
+    expected = 0.2 + 0.3 * 10 / (5 * 5)
+    assert tr.trust_radius == pytest.approx(expected)
+    near = trusted[0] + 0.01
+    far = torch.full((4,), 0.99)
+    xs = torch.stack([near, far])
+    scores = torch.tensor([1.0, 100.0])
+    out = tr.apply(xs, scores)
+    assert float(out[0]) == 1.0
+    assert float(out[1]) < -1e4 + 1
+
+  def test_hv_scalarization(self):
+    s = acq_lib.create_hv_scalarization(100, 2, seed=0)
+    ys = torch.tensor([[1.0, 1.0], [0.1, 0.1]])
+    vals = s(ys).mean(dim=0)
+    assert float(vals[0]) > float(vals[1])
+
+
+class TestOutputWarpers:
+
+  def test_default_pipeline_finite(self):
+    labels = np.array([[1.0], [2.0], [np.nan], [100.0], [-50.0]])
+    warper = output_warpers.create_default_warper()
+    out = warper.warp(labels)
+    assert np.isfinite(out).all()
+    # Order of finite labels preserved.
+    assert out[0, 0] < out[1, 0] < out[3, 0]
+    # NaN (infeasible) mapped below everything finite.
+    assert out[2, 0] < out[0, 0]
+
+  def test_halfrank_preserves_good_half(self):
+    labels = np.arange(10, dtype=np.float64)[:, None]
+    warped = output_warpers.HalfRankComponent().warp(labels.copy())
+    np.testing.assert_allclose(warped[5:, 0], labels[5:, 0])
+    assert (np.diff(warped[:, 0]) > 0).all()
+
+  def test_log_warper_roundtrip(self):
+    labels = np.array([[0.5], [1.0], [7.0], [3.0]])
+    w = output_warpers.LogWarperComponent()
+    warped = w.warp(labels.copy())
+    back = w.unwarp(warped)
+    np.testing.assert_allclose(back, labels, rtol=1e-10)
+
+  def test_zscore(self):
+    labels = np.array([[1.0], [2.0], [3.0]])
+    out = output_warpers.ZScoreLabels().warp(labels)
+    assert abs(out.mean()) < 1e-9

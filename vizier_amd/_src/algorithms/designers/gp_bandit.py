@@ -1,0 +1,320 @@
+"""GP-Bandit designer: the flagship GPU path.
+
+Capability parity with vizier/_src/algorithms/designers/gp_bandit.py
+(VizierGPBandit :88): seed phase (center + quasi-random), label warping
+(HalfRank+Log+Infeasible), Matern-5/2 ARD GP fit with restarted L-BFGS,
+UCB acquisition (coef 1.8) optimized by the vectorized Eagle strategy
+with a 75k-evaluation budget and batch 25 (:60-66), trust region (:142),
+q-EI parallel suggestions, and multi-objective support via hypervolume
+scalarization (:155,:220-242).
+
+On MI355X the converter's feature matrix, the GP state and the Eagle
+pool all live on `cuda:0`; the acquisition sweep can also be sharded
+across GPUs via vizier_amd/_src/parallel (config 4).
+"""
+
+from __future__ import annotations
+
+import dataclasses
+import time
+from typing import List, Optional, Sequence
+
+import numpy as np
+import torch
+
+from vizier_amd import pyvizier as vz
+from vizier_amd.converters.core import TrialToArrayConverter
+from vizier_amd._src.algorithms.core.abstractions import (
+    ActiveTrials,
+    CompletedTrials,
+    Designer,
+    Prediction,
+    Predictor,
+)
+from vizier_amd._src.algorithms.designers.quasi_random import (
+    QuasiRandomDesigner,
+)
+from vizier_amd._src.algorithms.optimizers.eagle import (
+    CandidateBatch,
+    EagleStrategyConfig,
+)
+from vizier_amd._src.algorithms.optimizers.vectorized import (
+    EagleFeatureCodec,
+    VectorizedOptimizerFactory,
+    trials_to_sorted_features,
+)
+from vizier_amd._src.gp import acquisitions as acq_lib
+from vizier_amd._src.gp import gp_model, output_warpers
+from vizier_amd._src.pythia import suggest_default
+
+
+def default_device() -> str:
+  return 'cuda' if torch.cuda.is_available() else 'cpu'
+
+
+@dataclasses.dataclass
+class GPBanditConfig:
+  """Tunables (defaults = reference budgets, gp_bandit.py:60-68)."""
+
+  acquisition: str = 'ucb'      # 'ucb' | 'ei' | 'pi' | 'qei' | 'thompson'
+  ucb_coefficient: float = 1.8
+  max_evaluations: int = 75000
+  suggestion_batch_size: int = 25
+  num_seed_trials: int = 2
+  ard_restarts: int = 4
+  ard_max_iters: int = 50
+  use_trust_region: bool = True
+  num_scalarizations: int = 1000  # multi-objective
+  device: Optional[str] = None
+  dtype: torch.dtype = torch.float32
+
+
+class VizierGPBandit(Designer, Predictor):
+  """GP surrogate + Eagle acquisition sweep."""
+
+  def __init__(self, problem: vz.ProblemStatement,
+               config: Optional[GPBanditConfig] = None, *, seed: int = 0):
+    self._problem = problem
+    self._config = config or GPBanditConfig()
+    self._seed = seed
+    self._trials: List[vz.Trial] = []
+    self._converter = TrialToArrayConverter(problem)
+    self._codec = EagleFeatureCodec(self._converter)
+    self._quasi_random = QuasiRandomDesigner(problem.search_space, seed=seed) \
+        if not problem.search_space.is_conditional else None
+    self._device = self._config.device or default_device()
+    self._posteriors: List[gp_model.GPPosterior] = []
+    self._last_fit_count = -1
+
+  @classmethod
+  def from_problem(cls, problem: vz.ProblemStatement,
+                   seed: int = 0) -> 'VizierGPBandit':
+    return cls(problem, seed=seed)
+
+  # -- Designer API ---------------------------------------------------------
+
+  def update(self, completed: CompletedTrials, all_active: ActiveTrials
+             ) -> None:
+    del all_active
+    self._trials.extend(completed.trials)
+
+  def suggest(self, count: Optional[int] = None
+              ) -> Sequence[vz.TrialSuggestion]:
+    count = count or 1
+    start = time.monotonic()
+    if len(self._trials) < self._config.num_seed_trials:
+      return self._seed_suggestions(count)
+    suggestions = self._gp_suggestions(count)
+    for s in suggestions:
+      s.metadata.ns('gp_bandit')['time_spent'] = \
+          f'{time.monotonic() - start:.4f}s'
+    return suggestions
+
+  def _seed_suggestions(self, count: int) -> List[vz.TrialSuggestion]:
+    out: List[vz.TrialSuggestion] = []
+    if not self._trials:
+      params = suggest_default.get_default_parameters(
+          self._problem.search_space)
+      out.append(vz.TrialSuggestion(
+          params, metadata=vz.Metadata({'seeded': 'center'})))
+    remaining = count - len(out)
+    if remaining > 0:
+      if self._quasi_random is not None:
+        out.extend(self._quasi_random.suggest(remaining))
+      else:
+        from vizier_amd._src.algorithms.designers.random import (
+            RandomDesigner,
+        )
+        out.extend(RandomDesigner(self._problem.search_space,
+                                  seed=self._seed).suggest(remaining))
+    return out
+
+  # -- GP path --------------------------------------------------------------
+
+  def _prepare_labels(self, raw_labels: np.ndarray) -> np.ndarray:
+    """Per-metric default warping -> (N, M) finite labels."""
+    warped = np.zeros_like(raw_labels)
+    for m in range(raw_labels.shape[1]):
+      warper = output_warpers.create_default_warper()
+      warped[:, m] = warper.warp(raw_labels[:, m:m + 1]).flatten()
+    return warped
+
+  def _fit(self) -> None:
+    if self._last_fit_count == len(self._trials) and self._posteriors:
+      return  # Cached: no new trials since the last fit (gp_bandit.py:464).
+    cfg = self._config
+    x_np, y_np = self._converter.to_xy(self._trials)
+    y_np = self._prepare_labels(y_np)
+    x = torch.as_tensor(x_np, dtype=cfg.dtype, device=self._device)
+    self._x = x
+    self._posteriors = []
+    for m in range(y_np.shape[1]):
+      y = torch.as_tensor(y_np[:, m], dtype=cfg.dtype, device=self._device)
+      self._posteriors.append(gp_model.train_gp(
+          x, y, num_restarts=cfg.ard_restarts,
+          max_iters=cfg.ard_max_iters, seed=self._seed))
+    self._warped_labels = torch.as_tensor(y_np, dtype=cfg.dtype,
+                                          device=self._device)
+    self._last_fit_count = len(self._trials)
+
+  def _make_trust_region(self) -> Optional[acq_lib.TrustRegion]:
+    if not self._config.use_trust_region:
+      return None
+    onehot = torch.zeros(self._converter.n_features, dtype=torch.bool,
+                         device=self._device)
+    for col in self._converter.output_specs:
+      if col.is_onehot:
+        onehot[col.start:col.start + col.width] = True
+    return acq_lib.TrustRegion(self._x, onehot)
+
+  def _score_factory(self, count: int):
+    """Returns (score_fn over CandidateBatch, n_parallel)."""
+    cfg = self._config
+    trust_region = self._make_trust_region()
+    multi_objective = len(self._posteriors) > 1
+
+    if multi_objective:
+      scalarizer = acq_lib.create_hv_scalarization(
+          cfg.num_scalarizations, len(self._posteriors), seed=self._seed,
+          reference_point=acq_lib.get_reference_point(self._warped_labels))
+
+      def score_fn(batch: CandidateBatch) -> torch.Tensor:
+        dense = self._codec.decode(batch)
+        flat = dense.reshape(-1, dense.shape[-1])
+        per_metric = []
+        for post in self._posteriors:
+          mean, stddev = post.predict(flat)
+          per_metric.append(mean + cfg.ucb_coefficient * stddev)
+        ys = torch.stack(per_metric, dim=-1)       # (B*q, M)
+        scores = scalarizer(ys).mean(dim=0)        # (B*q,)
+        scores = scores.reshape(dense.shape[0], dense.shape[1]).amax(dim=1)
+        if trust_region is not None:
+          scores = trust_region.apply(flat.reshape(dense.shape)[:, 0, :],
+                                      scores)
+        return scores
+      return score_fn, 1
+
+    posterior = self._posteriors[0]
+    best_value = float(self._warped_labels[:, 0].max())
+
+    if cfg.acquisition == 'qei' and count > 1:
+      qei = acq_lib.QEI(best_value=best_value, seed=self._seed)
+
+      def score_fn(batch: CandidateBatch) -> torch.Tensor:
+        dense = self._codec.decode(batch)          # (B, q, D)
+        mean, cov = posterior_batched_cov(posterior, dense)
+        q = dense.shape[1]
+        cov = cov + 1e-8 * torch.eye(q, dtype=cov.dtype,
+                                     device=cov.device)
+        L = torch.linalg.cholesky(cov)             # (B, q, q)
+        g = torch.Generator(device='cpu').manual_seed(self._seed)
+        eps = torch.randn(128, 1, q, generator=g).to(dense.device,
+                                                     dense.dtype)
+        samples = mean.unsqueeze(0) + torch.einsum(
+            'sbq,bqr->sbr', eps.expand(128, dense.shape[0], q), L)
+        scores = (samples - best_value).clamp_min(0).amax(-1).mean(0)
+        if trust_region is not None:
+          flat_scores = trust_region.apply(dense[:, 0, :], scores)
+          scores = flat_scores
+        return scores
+      return score_fn, count
+
+    if cfg.acquisition == 'ei':
+      acquisition = acq_lib.EI(best_value=best_value)
+    elif cfg.acquisition == 'pi':
+      acquisition = acq_lib.PI(best_value=best_value)
+    elif cfg.acquisition == 'thompson':
+      acquisition = acq_lib.Sample(seed=self._seed)
+    else:
+      acquisition = acq_lib.UCB(coefficient=cfg.ucb_coefficient)
+    scoring = acq_lib.ScoringFunction(posterior, acquisition, trust_region)
+
+    def score_fn(batch: CandidateBatch) -> torch.Tensor:
+      dense = self._codec.decode(batch)[:, 0, :]
+      return scoring(dense)
+    return score_fn, 1
+
+  def _gp_suggestions(self, count: int) -> List[vz.TrialSuggestion]:
+    cfg = self._config
+    self._fit()
+    score_fn, n_parallel = self._score_factory(count)
+
+    factory = VectorizedOptimizerFactory(
+        eagle_config=EagleStrategyConfig(),
+        max_evaluations=cfg.max_evaluations,
+        suggestion_batch_size=cfg.suggestion_batch_size)
+    optimizer = factory(
+        n_continuous=self._codec.n_continuous,
+        categorical_sizes=self._codec.categorical_sizes,
+        n_parallel=n_parallel, seed=self._seed + len(self._trials),
+        device=self._device, dtype=cfg.dtype)
+
+    rewards_np = self._warped_labels[:, 0].cpu().numpy() \
+        if len(self._posteriors) == 1 else \
+        self._warped_labels.mean(dim=1).cpu().numpy()
+    prior_features, prior_rewards = trials_to_sorted_features(
+        self._converter, self._codec, self._trials, rewards_np,
+        device=self._device, dtype=cfg.dtype)
+    # Expand priors to the optimizer's q axis.
+    q = n_parallel
+    if q > 1:
+      prior_features = CandidateBatch(
+          prior_features.continuous.expand(-1, q, -1).contiguous(),
+          prior_features.categorical.expand(-1, q, -1).contiguous())
+
+    results = optimizer.optimize(
+        score_fn, count=1 if n_parallel > 1 else count,
+        prior_features=prior_features, prior_rewards=prior_rewards)
+
+    dense = self._codec.decode(results.features)   # (k, q, D)
+    if n_parallel > 1:
+      rows = dense[0]                               # (q, D) -> q suggestions
+    else:
+      rows = dense[:, 0, :]                         # (k, D)
+    params = self._converter.to_parameters(rows.detach().cpu().numpy())
+    return [vz.TrialSuggestion(p) for p in params][:count]
+
+  # -- Predictor API --------------------------------------------------------
+
+  def predict(self, trials: Sequence[vz.TrialSuggestion],
+              rng: Optional[np.random.Generator] = None,
+              num_samples: Optional[int] = None) -> Prediction:
+    del rng, num_samples
+    self._fit()
+    x = torch.as_tensor(self._converter.to_features(trials),
+                        dtype=self._config.dtype, device=self._device)
+    means, stddevs = [], []
+    for post in self._posteriors:
+      mean, stddev = post.predict(x)
+      means.append(mean.cpu().numpy())
+      stddevs.append(stddev.cpu().numpy())
+    return Prediction(mean=np.stack(means, axis=-1),
+                      stddev=np.stack(stddevs, axis=-1))
+
+
+def posterior_batched_cov(posterior: gp_model.GPPosterior,
+                          dense: torch.Tensor):
+  """Batched joint predictive (mean, cov) for groups: dense (B, q, D)."""
+  from vizier_amd._src.ops import dispatch as ops
+  B, q, D = dense.shape
+  flat = dense.reshape(B * q, D)
+  k = ops.gram_matern52(flat, posterior.x, posterior.params.lengthscales,
+                        posterior.params.amplitude).reshape(B, q, -1)
+  mean = posterior.params.mean + k @ posterior.alpha
+  kqq = gram_batched(dense, posterior.params.lengthscales,
+                     posterior.params.amplitude)
+  if posterior.K_inv is not None:
+    t = torch.einsum('bqn,nm->bqm', k, posterior.K_inv)
+    cov = kqq - torch.einsum('bqn,brn->bqr', t, k)
+  else:
+    v = torch.linalg.solve_triangular(
+        posterior.L, k.reshape(B * q, -1).T, upper=False)
+    v = v.T.reshape(B, q, -1)
+    cov = kqq - torch.einsum('bqn,brn->bqr', v, v)
+  return mean, cov
+
+
+def gram_batched(x: torch.Tensor, lengthscales: torch.Tensor,
+                 amplitude: torch.Tensor) -> torch.Tensor:
+  from vizier_amd._src.gp.matern import gram_matern52
+  return gram_matern52(x, None, lengthscales, amplitude)

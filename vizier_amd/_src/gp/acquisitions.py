@@ -1,0 +1,249 @@
+"""Acquisition functions + trust region (PyTorch).
+
+Capability parity with vizier/_src/algorithms/designers/gp/acquisitions.py
+(UCB :214 w/ default coef 1.8, LCB :229, EI :244, PI :261, Sample :278,
+q-family :496-569, create_hv_scalarization :571, TrustRegion :691-820
+with the -1e4-distance outside-penalty :152-174).
+"""
+
+from __future__ import annotations
+
+import dataclasses
+import math
+from typing import Callable, List, Optional, Sequence
+
+import torch
+
+from vizier_amd._src.gp.gp_model import GPPosterior
+
+
+def _normal():
+  return torch.distributions.Normal(0.0, 1.0)
+
+
+class AcquisitionFunction:
+  """Maps posterior (mean, stddev) at candidates to scores."""
+
+  def __call__(self, mean: torch.Tensor, stddev: torch.Tensor
+               ) -> torch.Tensor:
+    raise NotImplementedError
+
+
+@dataclasses.dataclass
+class UCB(AcquisitionFunction):
+  coefficient: float = 1.8
+
+  def __call__(self, mean, stddev):
+    return mean + self.coefficient * stddev
+
+
+@dataclasses.dataclass
+class LCB(AcquisitionFunction):
+  coefficient: float = 1.8
+
+  def __call__(self, mean, stddev):
+    return mean - self.coefficient * stddev
+
+
+@dataclasses.dataclass
+class EI(AcquisitionFunction):
+  """Expected improvement over `best_value` (labels are maximized)."""
+
+  best_value: float = 0.0
+
+  def __call__(self, mean, stddev):
+    z = (mean - self.best_value) / stddev
+    n = _normal()
+    return stddev * (z * n.cdf(z) + n.log_prob(z).exp())
+
+
+@dataclasses.dataclass
+class PI(AcquisitionFunction):
+  best_value: float = 0.0
+
+  def __call__(self, mean, stddev):
+    z = (mean - self.best_value) / stddev
+    return _normal().cdf(z)
+
+
+@dataclasses.dataclass
+class Sample(AcquisitionFunction):
+  """Thompson-style single posterior sample."""
+
+  seed: int = 0
+
+  def __call__(self, mean, stddev):
+    g = torch.Generator(device='cpu').manual_seed(self.seed)
+    eps = torch.randn(mean.shape, generator=g).to(mean.device, mean.dtype)
+    return mean + stddev * eps
+
+
+# -- q-family: joint acquisition of a parallel suggestion batch -------------
+
+
+def _posterior_samples(posterior: GPPosterior, xq: torch.Tensor,
+                       num_samples: int, seed: int) -> torch.Tensor:
+  """(S, Q) joint samples of f at xq (uses the full predictive cov)."""
+  mean, cov = posterior.predict_cov(xq)
+  q = xq.shape[0]
+  cov = cov + 1e-8 * torch.eye(q, dtype=cov.dtype, device=cov.device)
+  L = torch.linalg.cholesky(cov)
+  g = torch.Generator(device='cpu').manual_seed(seed)
+  eps = torch.randn(num_samples, q, generator=g).to(mean.device, mean.dtype)
+  return mean.unsqueeze(0) + eps @ L.T
+
+
+@dataclasses.dataclass
+class QEI:
+  """Monte-Carlo q-EI: mean over samples of max-improvement in the batch."""
+
+  best_value: float = 0.0
+  num_samples: int = 128
+  seed: int = 0
+
+  def __call__(self, posterior: GPPosterior, xq: torch.Tensor
+               ) -> torch.Tensor:
+    samples = _posterior_samples(posterior, xq, self.num_samples, self.seed)
+    improvement = (samples - self.best_value).clamp_min(0.0)
+    return improvement.amax(dim=1).mean()
+
+
+@dataclasses.dataclass
+class QPI:
+  best_value: float = 0.0
+  num_samples: int = 128
+  seed: int = 0
+
+  def __call__(self, posterior, xq):
+    samples = _posterior_samples(posterior, xq, self.num_samples, self.seed)
+    return (samples.amax(dim=1) > self.best_value).float().mean()
+
+
+@dataclasses.dataclass
+class QUCB:
+  """q-UCB via |deviation| trick: mean + c * sqrt(pi/2) |eps| stddev."""
+
+  coefficient: float = 1.8
+  num_samples: int = 128
+  seed: int = 0
+
+  def __call__(self, posterior, xq):
+    mean, cov = posterior.predict_cov(xq)
+    q = xq.shape[0]
+    cov = cov + 1e-8 * torch.eye(q, dtype=cov.dtype, device=cov.device)
+    L = torch.linalg.cholesky(cov)
+    g = torch.Generator(device='cpu').manual_seed(self.seed)
+    eps = torch.randn(self.num_samples, q, generator=g).to(mean.device,
+                                                           mean.dtype)
+    dev = (eps @ L.T).abs() * math.sqrt(math.pi / 2.0)
+    return (mean.unsqueeze(0) + self.coefficient * dev).amax(dim=1).mean()
+
+
+# -- scalarization (multi-objective) ----------------------------------------
+
+
+@dataclasses.dataclass
+class HyperVolumeScalarization:
+  """Chebyshev-style hypervolume scalarization (acquisitions.py:571).
+
+  score(y) = min_m (y_m - ref_m)/w_m over positive weights w.
+  """
+
+  weights: torch.Tensor      # (V, M) positive directions
+  reference_point: Optional[torch.Tensor] = None  # (M,)
+
+  def __call__(self, ys: torch.Tensor) -> torch.Tensor:
+    """ys: (..., M) -> (V, ...) scalarized values."""
+    if self.reference_point is not None:
+      ys = ys - self.reference_point
+    w = self.weights.reshape((-1,) + (1,) * (ys.dim() - 1) + (ys.shape[-1],))
+    return (ys.unsqueeze(0) / w).amin(dim=-1)
+
+
+def create_hv_scalarization(num_scalarizations: int, num_metrics: int,
+                            seed: int = 0,
+                            reference_point: Optional[torch.Tensor] = None
+                            ) -> HyperVolumeScalarization:
+  g = torch.Generator(device='cpu').manual_seed(seed)
+  w = torch.randn(num_scalarizations, num_metrics, generator=g).abs()
+  w = w / w.norm(dim=-1, keepdim=True)
+  return HyperVolumeScalarization(weights=w.clamp_min(1e-6),
+                                  reference_point=reference_point)
+
+
+def get_reference_point(labels: torch.Tensor, scale: float = 0.01
+                        ) -> torch.Tensor:
+  """worst - scale * range per metric (acquisitions.py:~130)."""
+  best = labels.amax(dim=0)
+  worst = labels.amin(dim=0)
+  return worst - scale * (best - worst)
+
+
+# -- trust region ------------------------------------------------------------
+
+
+class TrustRegion:
+  """Union of L-inf balls around observed points (acquisitions.py:691).
+
+  radius = 0.2 + (0.5 - 0.2) * num_obs / (5 * (dof + 1)); a radius > 0.5
+  disables the constraint. One-hot (categorical) feature columns are
+  excluded from the distance but counted in dof.
+  """
+
+  MIN_RADIUS = 0.2
+  DIMENSION_FACTOR = 5.0
+
+  def __init__(self, trusted: torch.Tensor,
+               onehot_column_mask: Optional[torch.Tensor] = None):
+    """trusted: (N, D) observed features in [0,1]."""
+    self._trusted = trusted
+    d = trusted.shape[-1]
+    if onehot_column_mask is None:
+      onehot_column_mask = torch.zeros(d, dtype=torch.bool,
+                                       device=trusted.device)
+    self._onehot = onehot_column_mask
+    num_obs = trusted.shape[0]
+    dof = d  # continuous dims + categorical columns
+    trust_level = num_obs / (self.DIMENSION_FACTOR * (dof + 1))
+    if num_obs == 0:
+      self.trust_radius = 1.0
+    else:
+      self.trust_radius = self.MIN_RADIUS + (0.5 - self.MIN_RADIUS) * \
+          trust_level
+
+  def min_linf_distance(self, xs: torch.Tensor) -> torch.Tensor:
+    """xs: (..., D) -> (...) L-inf distance to the nearest trusted point."""
+    if self._trusted.shape[0] == 0:
+      return torch.full(xs.shape[:-1], -float('inf'), dtype=xs.dtype,
+                        device=xs.device)
+    diff = (xs.unsqueeze(-2) - self._trusted).abs()   # (..., N, D)
+    diff = torch.where(self._onehot, torch.zeros_like(diff), diff)
+    return diff.amax(dim=-1).amin(dim=-1)
+
+  def apply(self, xs: torch.Tensor, scores: torch.Tensor) -> torch.Tensor:
+    """Penalizes scores outside the region: -1e4 - distance."""
+    if self.trust_radius > 0.5:
+      return scores
+    distance = self.min_linf_distance(xs)
+    return torch.where(distance <= self.trust_radius, scores,
+                       -1e4 - distance)
+
+
+@dataclasses.dataclass
+class ScoringFunction:
+  """Posterior + acquisition + optional trust region, over a batch.
+
+  The callable the acquisition optimizer evaluates: xs (B, D) -> (B,).
+  On GPU this routes through the fused HIP posterior_scores kernel.
+  """
+
+  posterior: GPPosterior
+  acquisition: AcquisitionFunction
+  trust_region: Optional[TrustRegion] = None
+
+  def __call__(self, xs: torch.Tensor) -> torch.Tensor:
+    mean, stddev = self.posterior.predict(xs)
+    scores = self.acquisition(mean, stddev)
+    if self.trust_region is not None:
+      scores = self.trust_region.apply(xs, scores)
+    return scores
