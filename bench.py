@@ -1,0 +1,172 @@
+"""Flagship benchmark: GP-Bandit suggest() latency, 20D @ N=1000 trials.
+
+Measures the BASELINE.json headline metric: wall-clock of a full
+`suggest()` (incremental GP refit at N~1000 + 75,000-evaluation Eagle
+acquisition sweep, batch 25 — the reference's own budgets) on a
+synthetic BBOB-style objective with random-init GP hyperparameters.
+
+Single GPU by default. With torchrun (--gpus N), the acquisition sweep
+runs data-parallel: each rank sweeps its own 75k-candidate shard (weak
+scaling) and the per-shard top-k is all-gathered over RCCL/xGMI, so the
+whole job evaluates N x 75k candidates per suggest.
+
+Usage: python bench.py [--gpus N] [--steps K] [--warmup W]
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+import torch
+
+from vizier_amd import pyvizier as vz
+from vizier_amd._src.algorithms.core.abstractions import (
+    ActiveTrials,
+    CompletedTrials,
+)
+from vizier_amd._src.algorithms.designers.gp_bandit import (
+    GPBanditConfig,
+    VizierGPBandit,
+)
+
+DIM = 20
+N_TRIALS = 1000
+MAX_EVALS = 75000
+BATCH = 25
+
+
+def bbob_objective(x: np.ndarray) -> float:
+  """Shifted sphere with mild ellipsoid conditioning (BBOB-style)."""
+  shift = np.linspace(-2.0, 2.0, DIM)
+  weights = 10.0 ** np.linspace(0, 2, DIM)
+  z = x - shift
+  return float(-np.sum(weights * z * z) / weights.sum())
+
+
+def make_problem() -> vz.ProblemStatement:
+  problem = vz.ProblemStatement()
+  for i in range(DIM):
+    problem.search_space.root.add_float_param(f'x{i}', -5.0, 5.0)
+  problem.metric_information.append(vz.MetricInformation(
+      name='obj', goal=vz.ObjectiveMetricGoal.MAXIMIZE))
+  return problem
+
+
+def trial_from(params: dict, uid: int) -> vz.Trial:
+  t = vz.Trial(params, id=uid)
+  x = np.array([params[f'x{i}'] for i in range(DIM)])
+  t.complete(vz.Measurement(metrics={'obj': bbob_objective(x)}))
+  return t
+
+
+def main() -> None:
+  parser = argparse.ArgumentParser()
+  parser.add_argument('--gpus', type=int, default=1)
+  parser.add_argument('--steps', type=int, default=5)
+  parser.add_argument('--warmup', type=int, default=2)
+  args = parser.parse_args()
+
+  world_size = int(os.environ.get('WORLD_SIZE', '1'))
+  rank = int(os.environ.get('RANK', '0'))
+  local_rank = int(os.environ.get('LOCAL_RANK', '0'))
+  distributed = world_size > 1
+
+  use_gpu = torch.cuda.is_available()
+  if distributed:
+    backend = 'nccl' if use_gpu else 'gloo'
+    if use_gpu:
+      torch.cuda.set_device(local_rank)
+    torch.distributed.init_process_group(backend=backend)
+  device = f'cuda:{local_rank}' if use_gpu else 'cpu'
+
+  def sync():
+    if use_gpu:
+      torch.cuda.synchronize()
+    if distributed:
+      torch.distributed.barrier()
+
+  problem = make_problem()
+  config = GPBanditConfig(
+      max_evaluations=MAX_EVALS, suggestion_batch_size=BATCH,
+      ard_restarts=4, ard_max_iters=50, device=device,
+      data_parallel=distributed)
+  designer = VizierGPBandit(problem, config, seed=0)
+
+  # Pre-populate N=1000 synthetic trials (identical on every rank).
+  rng = np.random.default_rng(0)
+  trials = []
+  for uid in range(1, N_TRIALS + 1):
+    params = {f'x{i}': float(v)
+              for i, v in enumerate(rng.uniform(-5, 5, DIM))}
+    trials.append(trial_from(params, uid))
+  designer.update(CompletedTrials(trials), ActiveTrials())
+  next_uid = N_TRIALS + 1
+
+  def one_step(uid: int) -> None:
+    # A step = one full suggest (GP refit + 75k-eval sweep) and the
+    # completion of its trial, so every timed step does the whole
+    # pipeline at N ~ 1000.
+    suggestions = designer.suggest(1)
+    params = dict(suggestions[0].parameters.as_dict())
+    designer.update(CompletedTrials([trial_from(params, uid)]),
+                    ActiveTrials())
+
+  for w in range(args.warmup):
+    one_step(next_uid)
+    next_uid += 1
+
+  sync()
+  t0 = time.perf_counter()
+  for k in range(args.steps):
+    one_step(next_uid)
+    next_uid += 1
+  sync()
+  elapsed = time.perf_counter() - t0
+
+  # MAX over ranks of the measured wall-clock.
+  if distributed:
+    t = torch.tensor([elapsed], dtype=torch.float64,
+                     device=device if use_gpu else 'cpu')
+    torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+    elapsed = float(t.item())
+
+  ms_per_step = elapsed / args.steps * 1000.0
+  value = ms_per_step
+
+  if rank == 0:
+    result = {
+        'metric': 'suggest_wall_clock_ms_gp_bandit_20d_n1000',
+        'value': value,
+        'unit': 'ms',
+        'n_gpus': world_size,
+        'steps': args.steps,
+        'warmup': args.warmup,
+        'ms_per_step': ms_per_step,
+        'higher_is_better': False,
+        'scaling': 'weak',
+        'vs_baseline': None,
+        'dtype': 'fp32',
+        'data': 'synthetic',
+        'config': {
+            'model': 'gp_bandit_matern52_ucb_eagle',
+            'dim': DIM,
+            'n_trials': N_TRIALS,
+            'acquisition_evals_per_gpu': MAX_EVALS,
+            'eagle_batch': BATCH,
+            'ard': 'lbfgs_50it_x5_restarts',
+            'global_batch': 1,
+            'seq_len': None,
+            'parallelism': f'dp{world_size}_sharded_sweep',
+        },
+    }
+    print(json.dumps(result))
+
+  if distributed:
+    torch.distributed.destroy_process_group()
+
+
+if __name__ == '__main__':
+  main()

@@ -45,6 +45,7 @@ from vizier_amd._src.algorithms.optimizers.vectorized import (
 )
 from vizier_amd._src.gp import acquisitions as acq_lib
 from vizier_amd._src.gp import gp_model, output_warpers
+from vizier_amd._src.parallel import sharded_sweep
 from vizier_amd._src.pythia import suggest_default
 
 
@@ -65,6 +66,7 @@ class GPBanditConfig:
   ard_max_iters: int = 50
   use_trust_region: bool = True
   num_scalarizations: int = 1000  # multi-objective
+  data_parallel: bool = False     # shard the sweep across dist ranks
   device: Optional[str] = None
   dtype: torch.dtype = torch.float32
 
@@ -150,9 +152,12 @@ class VizierGPBandit(Designer, Predictor):
     self._posteriors = []
     for m in range(y_np.shape[1]):
       y = torch.as_tensor(y_np[:, m], dtype=cfg.dtype, device=self._device)
-      self._posteriors.append(gp_model.train_gp(
+      post = gp_model.train_gp(
           x, y, num_restarts=cfg.ard_restarts,
-          max_iters=cfg.ard_max_iters, seed=self._seed))
+          max_iters=cfg.ard_max_iters, seed=self._seed)
+      if cfg.data_parallel:
+        sharded_sweep.broadcast_posterior(post)
+      self._posteriors.append(post)
     self._warped_labels = torch.as_tensor(y_np, dtype=cfg.dtype,
                                           device=self._device)
     self._last_fit_count = len(self._trials)
@@ -243,10 +248,12 @@ class VizierGPBandit(Designer, Predictor):
         eagle_config=EagleStrategyConfig(),
         max_evaluations=cfg.max_evaluations,
         suggestion_batch_size=cfg.suggestion_batch_size)
+    shard_rank = sharded_sweep.rank() if cfg.data_parallel else 0
     optimizer = factory(
         n_continuous=self._codec.n_continuous,
         categorical_sizes=self._codec.categorical_sizes,
-        n_parallel=n_parallel, seed=self._seed + len(self._trials),
+        n_parallel=n_parallel,
+        seed=self._seed + len(self._trials) + 7919 * shard_rank,
         device=self._device, dtype=cfg.dtype)
 
     rewards_np = self._warped_labels[:, 0].cpu().numpy() \
@@ -267,6 +274,11 @@ class VizierGPBandit(Designer, Predictor):
         prior_features=prior_features, prior_rewards=prior_rewards)
 
     dense = self._codec.decode(results.features)   # (k, q, D)
+    if cfg.data_parallel and sharded_sweep.is_initialized():
+      # All-gather each shard's top-k; every rank selects the identical
+      # global winners (KB-scale payload over xGMI).
+      dense, _ = sharded_sweep.allgather_topk(
+          dense, results.rewards, dense.shape[0])
     if n_parallel > 1:
       rows = dense[0]                               # (q, D) -> q suggestions
     else:

@@ -113,9 +113,19 @@ class VectorizedEagleStrategy:
     self.batch_size = min(batch_size, self.pool_size)
     self._gen = torch.Generator(device=self.device)
     self._gen.manual_seed(seed)
+    self._seed = seed
+    self._cat_sizes_long = torch.tensor(self.categorical_sizes,
+                                        dtype=torch.long,
+                                        device=self.device)
     if self.n_categorical:
       self._cat_sizes_t = torch.tensor(self.categorical_sizes,
                                        device=self.device)
+    # The fused HIP path replaces the ~30-launch torch step with 2 launches.
+    if self.device.type == 'cuda':
+      from vizier_amd._src.ops import dispatch as ops
+      self._ext = ops.require_ext()
+    else:
+      self._ext = None
 
   # -- random sampling ------------------------------------------------------
 
@@ -151,8 +161,8 @@ class VectorizedEagleStrategy:
         perturbations=torch.full((self.pool_size,),
                                  self.config.perturbation,
                                  device=self.device, dtype=self.dtype),
-        best_reward=torch.tensor(-float('inf'), device=self.device,
-                                 dtype=self.dtype))
+        best_reward=torch.full((1,), -float('inf'), device=self.device,
+                               dtype=self.dtype))
 
   def _dist2(self, a: CandidateBatch, b: CandidateBatch) -> torch.Tensor:
     """Squared distances (nA, nB): continuous L2 + categorical Hamming."""
@@ -207,6 +217,19 @@ class VectorizedEagleStrategy:
     if state.iterations < n_batches:
       return CandidateBatch(batch.continuous.clone(),
                             batch.categorical.clone())
+    if self._ext is not None:
+      cfg = self.config
+      cat_factor = (cfg.pure_categorical_perturbation_factor
+                    if self.n_continuous == 0
+                    else cfg.categorical_perturbation_factor)
+      out_cont, out_cat = self._ext.eagle_suggest(
+          state.continuous, state.categorical, state.rewards,
+          state.perturbations, self._cat_sizes_long, start,
+          self.batch_size, cfg.visibility, cfg.gravity,
+          cfg.negative_gravity, cfg.normalization_scale, cat_factor,
+          cfg.prob_same_category_without_perturbation, self._seed,
+          state.iterations)
+      return CandidateBatch(out_cont, out_cat)
     return self._mutate(state, batch, state.rewards[sl],
                         state.perturbations[sl])
 
@@ -311,6 +334,17 @@ class VectorizedEagleStrategy:
       state.continuous[sl] = batch.continuous
       state.categorical[sl] = batch.categorical
       state.rewards[sl] = batch_rewards
+    elif self._ext is not None:
+      state.best_reward = new_best  # kernel recomputes it in place too
+      self._ext.eagle_update(
+          state.continuous, state.categorical, state.rewards,
+          state.perturbations, batch.continuous.contiguous(),
+          batch.categorical.contiguous(), batch_rewards.contiguous(),
+          self._cat_sizes_long, state.best_reward, start,
+          cfg.penalize_factor, cfg.perturbation_lower_bound,
+          cfg.perturbation, self._seed ^ 0xABCDEF, state.iterations)
+      state.iterations += 1
+      return state
     else:
       prev_rewards = state.rewards[sl]
       perturbations = state.perturbations[sl]

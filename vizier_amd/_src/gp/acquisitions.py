@@ -229,19 +229,56 @@ class TrustRegion:
                        -1e4 - distance)
 
 
-@dataclasses.dataclass
 class ScoringFunction:
   """Posterior + acquisition + optional trust region, over a batch.
 
   The callable the acquisition optimizer evaluates: xs (B, D) -> (B,).
-  On GPU this routes through the fused HIP posterior_scores kernel.
+  On GPU (with the trust region anchored at the training points, the
+  GP-Bandit case) this is ONE fused HIP launch (posterior_score.hip);
+  otherwise it composes torch ops.
   """
 
-  posterior: GPPosterior
-  acquisition: AcquisitionFunction
-  trust_region: Optional[TrustRegion] = None
+  _FUSED = {UCB: 'ucb', LCB: 'lcb', EI: 'ei', PI: 'pi'}
+
+  def __init__(self, posterior: GPPosterior,
+               acquisition: AcquisitionFunction,
+               trust_region: Optional[TrustRegion] = None):
+    self.posterior = posterior
+    self.acquisition = acquisition
+    self.trust_region = trust_region
+    # Pre-extract scalars once so the hot loop never syncs the device.
+    self._amp = float(posterior.params.amplitude)
+    self._mean_c = float(posterior.params.mean)
+    self._acq_name = self._FUSED.get(type(acquisition))
+    self._coef = getattr(acquisition, 'coefficient', 0.0)
+    self._best = getattr(acquisition, 'best_value', 0.0)
+    self._onehot_u8 = None
+    if trust_region is not None:
+      self._onehot_u8 = trust_region._onehot.to(torch.uint8)
+      self._tr_radius = float(trust_region.trust_radius)
+      self._tr_anchored = trust_region._trusted is posterior.x
+    else:
+      self._tr_radius = 0.0
+      self._tr_anchored = True
+
+  def _can_fuse(self, xs: torch.Tensor) -> bool:
+    return (xs.is_cuda and self._acq_name is not None and
+            self.posterior.K_inv is not None and self._tr_anchored)
 
   def __call__(self, xs: torch.Tensor) -> torch.Tensor:
+    if self._can_fuse(xs):
+      from vizier_amd._src.ops import dispatch as ops
+      onehot = self._onehot_u8
+      if onehot is None:
+        onehot = torch.zeros(xs.shape[-1], dtype=torch.uint8,
+                             device=xs.device)
+        self._onehot_u8 = onehot
+      return ops.fused_posterior_scores(
+          xs, self.posterior.x, self.posterior.params.lengthscales,
+          self._amp, self._mean_c, self.posterior.alpha,
+          self.posterior.K_inv, onehot, self._acq_name, self._coef,
+          self._best, self._tr_radius if self.trust_region is not None
+          else 0.0)
     mean, stddev = self.posterior.predict(xs)
     scores = self.acquisition(mean, stddev)
     if self.trust_region is not None:

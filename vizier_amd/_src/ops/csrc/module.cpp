@@ -1,0 +1,162 @@
+// Python bindings for the vizier_amd gfx950 kernels (torch extension).
+
+#include <ATen/cuda/CUDAContext.h>
+#include <torch/extension.h>
+
+#include <hip/hip_runtime.h>
+
+extern "C" void launch_gram_matern52(const float* x1, const float* x2,
+                                     const float* inv_ls, float* out,
+                                     int n, int m, int d, float amp2,
+                                     int sym, hipStream_t stream);
+
+extern "C" void launch_posterior_score(
+    const float* xq, const float* x, const float* inv_ls,
+    const float* alpha, const float* kinv, const unsigned char* onehot,
+    float* out, int b, int n, int d, float amp2, float mean_c, int acq,
+    float coef, float best_value, float tr_radius, hipStream_t stream);
+
+extern "C" void launch_eagle_suggest(
+    const float* pool_cont, const long* pool_cat, const float* rewards,
+    const float* perturbations, const long* cat_sizes, float* out_cont,
+    long* out_cat, int batch_start, int batch_size, int pool_size, int q,
+    int dc, int dcat, int max_cat, float visibility, float gravity,
+    float neg_gravity, float norm_scale, float cat_factor, float p_same,
+    unsigned long long seed, unsigned long long offset,
+    hipStream_t stream);
+
+extern "C" void launch_eagle_update(
+    float* pool_cont, long* pool_cat, float* rewards, float* perturbations,
+    const float* batch_cont, const long* batch_cat,
+    const float* batch_rewards, const long* cat_sizes, float* best_reward,
+    int batch_start, int batch_size, int q, int dc, int dcat,
+    float penalize_factor, float perturbation_lower_bound,
+    float base_perturbation, unsigned long long seed,
+    unsigned long long offset, hipStream_t stream);
+
+namespace {
+
+void check_f32(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be a GPU tensor");
+  TORCH_CHECK(t.scalar_type() == torch::kFloat32, name, " must be fp32");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+hipStream_t current_stream() {
+  return at::cuda::getCurrentCUDAStream().stream();
+}
+
+torch::Tensor gram_matern52(torch::Tensor x1, torch::Tensor x2,
+                            torch::Tensor lengthscales, double amplitude) {
+  check_f32(x1, "x1");
+  check_f32(x2, "x2");
+  check_f32(lengthscales, "lengthscales");
+  const int n = x1.size(0), m = x2.size(0), d = x1.size(1);
+  TORCH_CHECK(x2.size(1) == d && lengthscales.numel() == d,
+              "dimension mismatch");
+  auto inv_ls = 1.0f / lengthscales;
+  auto out = torch::empty({n, m}, x1.options());
+  const int sym = (x1.data_ptr() == x2.data_ptr() && n == m) ? 1 : 0;
+  launch_gram_matern52(x1.data_ptr<float>(), x2.data_ptr<float>(),
+                       inv_ls.data_ptr<float>(), out.data_ptr<float>(), n,
+                       m, d, (float)(amplitude * amplitude), sym,
+                       current_stream());
+  return out;
+}
+
+torch::Tensor posterior_scores(torch::Tensor xq, torch::Tensor x,
+                               torch::Tensor lengthscales, double amplitude,
+                               double mean_c, torch::Tensor alpha,
+                               torch::Tensor kinv, torch::Tensor onehot,
+                               int64_t acq, double coef, double best_value,
+                               double tr_radius) {
+  check_f32(xq, "xq");
+  check_f32(x, "x");
+  check_f32(lengthscales, "lengthscales");
+  check_f32(alpha, "alpha");
+  check_f32(kinv, "kinv");
+  TORCH_CHECK(onehot.scalar_type() == torch::kUInt8 && onehot.is_cuda(),
+              "onehot must be a uint8 GPU tensor");
+  const int b = xq.size(0), d = xq.size(1), n = x.size(0);
+  TORCH_CHECK(d <= 512, "posterior_scores supports D <= 512");
+  TORCH_CHECK(n <= 36000, "posterior_scores supports N <= 36000 (LDS)");
+  TORCH_CHECK(x.size(1) == d && alpha.numel() == n &&
+              kinv.size(0) == n && kinv.size(1) == n, "shape mismatch");
+  auto inv_ls = 1.0f / lengthscales;
+  auto out = torch::empty({b}, xq.options());
+  launch_posterior_score(
+      xq.data_ptr<float>(), x.data_ptr<float>(), inv_ls.data_ptr<float>(),
+      alpha.data_ptr<float>(), kinv.data_ptr<float>(),
+      onehot.data_ptr<unsigned char>(), out.data_ptr<float>(), b, n, d,
+      (float)(amplitude * amplitude), (float)mean_c, (int)acq, (float)coef,
+      (float)best_value, (float)tr_radius, current_stream());
+  return out;
+}
+
+std::vector<torch::Tensor> eagle_suggest(
+    torch::Tensor pool_cont, torch::Tensor pool_cat, torch::Tensor rewards,
+    torch::Tensor perturbations, torch::Tensor cat_sizes,
+    int64_t batch_start, int64_t batch_size, double visibility,
+    double gravity, double neg_gravity, double norm_scale,
+    double cat_factor, double p_same, int64_t seed, int64_t offset) {
+  check_f32(pool_cont, "pool_cont");
+  check_f32(rewards, "rewards");
+  check_f32(perturbations, "perturbations");
+  const int pool_size = pool_cont.size(0);
+  const int q = pool_cont.size(1);
+  const int dc = pool_cont.size(2);
+  const int dcat = pool_cat.size(2);
+  int max_cat = 0;
+  if (dcat > 0) max_cat = cat_sizes.max().item<int64_t>();
+  auto out_cont = torch::empty({batch_size, q, dc}, pool_cont.options());
+  auto out_cat = torch::empty({batch_size, q, dcat}, pool_cat.options());
+  launch_eagle_suggest(
+      pool_cont.data_ptr<float>(),
+      dcat ? pool_cat.data_ptr<long>() : nullptr,
+      rewards.data_ptr<float>(), perturbations.data_ptr<float>(),
+      dcat ? cat_sizes.data_ptr<long>() : nullptr,
+      out_cont.data_ptr<float>(),
+      dcat ? out_cat.data_ptr<long>() : nullptr, (int)batch_start,
+      (int)batch_size, pool_size, q, dc, dcat, max_cat, (float)visibility,
+      (float)gravity, (float)neg_gravity, (float)norm_scale,
+      (float)cat_factor, (float)p_same, (unsigned long long)seed,
+      (unsigned long long)offset, current_stream());
+  return {out_cont, out_cat};
+}
+
+void eagle_update(torch::Tensor pool_cont, torch::Tensor pool_cat,
+                  torch::Tensor rewards, torch::Tensor perturbations,
+                  torch::Tensor batch_cont, torch::Tensor batch_cat,
+                  torch::Tensor batch_rewards, torch::Tensor cat_sizes,
+                  torch::Tensor best_reward, int64_t batch_start,
+                  double penalize_factor, double perturbation_lower_bound,
+                  double base_perturbation, int64_t seed, int64_t offset) {
+  const int batch_size = batch_cont.size(0);
+  const int q = batch_cont.size(1);
+  const int dc = batch_cont.size(2);
+  const int dcat = batch_cat.size(2);
+  launch_eagle_update(
+      pool_cont.data_ptr<float>(),
+      dcat ? pool_cat.data_ptr<long>() : nullptr,
+      rewards.data_ptr<float>(), perturbations.data_ptr<float>(),
+      batch_cont.data_ptr<float>(),
+      dcat ? batch_cat.data_ptr<long>() : nullptr,
+      batch_rewards.data_ptr<float>(),
+      dcat ? cat_sizes.data_ptr<long>() : nullptr,
+      best_reward.data_ptr<float>(), (int)batch_start, batch_size, q, dc,
+      dcat, (float)penalize_factor, (float)perturbation_lower_bound,
+      (float)base_perturbation, (unsigned long long)seed,
+      (unsigned long long)offset, current_stream());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("gram_matern52", &gram_matern52,
+        "Fused Matern-5/2 ARD Gram matrix (gfx950)");
+  m.def("posterior_scores", &posterior_scores,
+        "Fused GP posterior + acquisition + trust region (gfx950)");
+  m.def("eagle_suggest", &eagle_suggest,
+        "Fused Eagle suggest step (gfx950)");
+  m.def("eagle_update", &eagle_update, "Fused Eagle update step (gfx950)");
+}

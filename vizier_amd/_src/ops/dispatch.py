@@ -2,14 +2,13 @@
 
 Policy: on a GPU (ROCm) machine the HIP extension MUST be present and is
 always used for GPU tensors — if it fails to import, GPU ops raise
-loudly rather than silently falling back to eager PyTorch (the judge's
-"native code not loaded" check). On CPU the PyTorch reference
-implementations run (they are also the numerics oracle for the kernels).
+loudly rather than silently falling back to eager PyTorch. On CPU the
+PyTorch reference implementations run (they are also the numerics oracle
+for the kernels, compared in tests/test_gpu_ops.py).
 """
 
 from __future__ import annotations
 
-import os
 from typing import Optional
 
 import torch
@@ -18,6 +17,8 @@ from vizier_amd._src.gp import matern as _torch_matern
 
 _EXT = None
 _EXT_ERROR: Optional[str] = None
+
+ACQ_CODES = {'ucb': 0, 'lcb': 1, 'ei': 2, 'pi': 3, 'mean': 4, 'stddev': 5}
 
 
 def _load_extension():
@@ -36,7 +37,7 @@ def extension_available() -> bool:
   return _load_extension() is not None
 
 
-def _require_ext():
+def require_ext():
   ext = _load_extension()
   if ext is None:
     raise RuntimeError(
@@ -51,56 +52,23 @@ def gram_matern52(x1: torch.Tensor, x2: Optional[torch.Tensor],
                   amplitude: torch.Tensor) -> torch.Tensor:
   """Matern-5/2 ARD Gram / cross-Gram matrix."""
   if x1.is_cuda:
-    ext = _require_ext()
-    return ext.gram_matern52(x1, x1 if x2 is None else x2,
-                             lengthscales, amplitude)
+    ext = require_ext()
+    x2t = x1 if x2 is None else x2
+    return ext.gram_matern52(x1.contiguous(), x2t.contiguous(),
+                             lengthscales.contiguous(), float(amplitude))
   return _torch_matern.gram_matern52(x1, x2, lengthscales, amplitude)
 
 
-def fused_predict(k_qn: None = None, **kwargs):  # placeholder, see ops.py
-  raise NotImplementedError
-
-
-def ucb_score(mean: torch.Tensor, stddev: torch.Tensor,
-              coefficient: float) -> torch.Tensor:
-  return mean + coefficient * stddev
-
-
-def posterior_scores(xq: torch.Tensor, posterior, acquisition: str,
-                     **kw) -> torch.Tensor:
-  """Fused posterior + acquisition evaluation for a candidate batch.
-
-  GPU path: one HIP launch computing k-vectors (MFMA), mean/variance
-  (GEMM quadform vs K^-1) and the acquisition, without materializing
-  intermediates in HBM. CPU path composes torch ops.
-  """
-  if xq.is_cuda:
-    ext = _require_ext()
-    return ext.posterior_scores(
-        xq, posterior.x, posterior.params.lengthscales,
-        posterior.params.amplitude, posterior.params.noise,
-        posterior.params.mean, posterior.alpha, posterior.K_inv,
-        acquisition, kw.get('coefficient', 1.8), kw.get('best_value', 0.0))
-  mean, stddev = posterior.predict(xq)
-  if acquisition == 'ucb':
-    return mean + kw.get('coefficient', 1.8) * stddev
-  if acquisition == 'lcb':
-    return mean - kw.get('coefficient', 1.8) * stddev
-  if acquisition == 'ei':
-    import math
-    best = kw.get('best_value', 0.0)
-    z = (mean - best) / stddev
-    normal = torch.distributions.Normal(torch.zeros_like(z),
-                                        torch.ones_like(z))
-    return stddev * (z * normal.cdf(z) + normal.log_prob(z).exp())
-  if acquisition == 'pi':
-    best = kw.get('best_value', 0.0)
-    z = (mean - best) / stddev
-    normal = torch.distributions.Normal(torch.zeros_like(z),
-                                        torch.ones_like(z))
-    return normal.cdf(z)
-  if acquisition == 'mean':
-    return mean
-  if acquisition == 'stddev':
-    return stddev
-  raise ValueError(f'Unknown acquisition {acquisition}')
+def fused_posterior_scores(xq: torch.Tensor, x: torch.Tensor,
+                           lengthscales: torch.Tensor, amplitude: float,
+                           mean_c: float, alpha: torch.Tensor,
+                           k_inv: torch.Tensor, onehot_u8: torch.Tensor,
+                           acq: str, coef: float, best_value: float,
+                           tr_radius: float) -> torch.Tensor:
+  """One-launch GP posterior + acquisition + trust region (GPU only)."""
+  ext = require_ext()
+  return ext.posterior_scores(
+      xq.contiguous(), x.contiguous(), lengthscales.contiguous(),
+      float(amplitude), float(mean_c), alpha.contiguous(),
+      k_inv.contiguous(), onehot_u8.contiguous(), ACQ_CODES[acq],
+      float(coef), float(best_value), float(tr_radius))
