@@ -78,6 +78,12 @@ def _init_raw(num_restarts: int, dim: int, generator: torch.Generator,
   return raw
 
 
+def _chol_solve(L: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+  """(L L^T)^-1 b via two triangular solves (batched)."""
+  z = torch.linalg.solve_triangular(L, b, upper=False)
+  return torch.linalg.solve_triangular(L.mT, z, upper=True)
+
+
 def cholesky_with_jitter(K: torch.Tensor, amplitude2: torch.Tensor,
                          max_tries: int = 6) -> torch.Tensor:
   """Batched Cholesky with escalating jitter (tuned_gp_models.py:92)."""
@@ -107,7 +113,9 @@ def negative_log_marginal_likelihood(raw: torch.Tensor, x: torch.Tensor,
   K = K + noise * torch.eye(n, dtype=x.dtype, device=x.device)
   L, info = torch.linalg.cholesky_ex(K)
   resid = (y.unsqueeze(0) - params.mean.unsqueeze(-1)).unsqueeze(-1)
-  alpha = torch.cholesky_solve(resid, L)
+  # NOTE: torch.cholesky_solve hits a hipErrorLaunchFailure on this ROCm
+  # build (see profiles/notes); two triangular solves are equivalent.
+  alpha = _chol_solve(L, resid)
   quad = (resid * alpha).sum(dim=(-1, -2))
   logdet = 2.0 * torch.log(torch.diagonal(L, dim1=-2, dim2=-1)).sum(-1)
   nll = 0.5 * (quad + logdet + n * math.log(2 * math.pi))
@@ -189,7 +197,11 @@ def train_gp(x: torch.Tensor, y: torch.Tensor, *,
   K = K + params.noise * torch.eye(n, dtype=x.dtype, device=x.device)
   L = cholesky_with_jitter(K, params.amplitude ** 2)
   resid = (y - params.mean).unsqueeze(-1)
-  alpha = torch.cholesky_solve(resid, L).squeeze(-1)
-  K_inv = torch.cholesky_inverse(L) if precompute_inverse else None
+  alpha = _chol_solve(L, resid).squeeze(-1)
+  K_inv = None
+  if precompute_inverse:
+    eye = torch.eye(n, dtype=x.dtype, device=x.device)
+    z = torch.linalg.solve_triangular(L, eye, upper=False)
+    K_inv = z.T @ z
   return GPPosterior(x=x, params=params, L=L, alpha=alpha, K_inv=K_inv,
                      nll=float(best_f[idx]))

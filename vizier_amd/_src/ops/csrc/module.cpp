@@ -36,10 +36,10 @@ extern "C" void launch_eagle_update(
 
 namespace {
 
-void check_f32(const torch::Tensor& t, const char* name) {
+torch::Tensor check_f32(const torch::Tensor& t, const char* name) {
   TORCH_CHECK(t.is_cuda(), name, " must be a GPU tensor");
   TORCH_CHECK(t.scalar_type() == torch::kFloat32, name, " must be fp32");
-  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+  return t.contiguous();
 }
 
 hipStream_t current_stream() {
@@ -48,9 +48,9 @@ hipStream_t current_stream() {
 
 torch::Tensor gram_matern52(torch::Tensor x1, torch::Tensor x2,
                             torch::Tensor lengthscales, double amplitude) {
-  check_f32(x1, "x1");
-  check_f32(x2, "x2");
-  check_f32(lengthscales, "lengthscales");
+  x1 = check_f32(x1, "x1");
+  x2 = check_f32(x2, "x2");
+  lengthscales = check_f32(lengthscales, "lengthscales");
   const int n = x1.size(0), m = x2.size(0), d = x1.size(1);
   TORCH_CHECK(x2.size(1) == d && lengthscales.numel() == d,
               "dimension mismatch");
@@ -70,13 +70,14 @@ torch::Tensor posterior_scores(torch::Tensor xq, torch::Tensor x,
                                torch::Tensor kinv, torch::Tensor onehot,
                                int64_t acq, double coef, double best_value,
                                double tr_radius) {
-  check_f32(xq, "xq");
-  check_f32(x, "x");
-  check_f32(lengthscales, "lengthscales");
-  check_f32(alpha, "alpha");
-  check_f32(kinv, "kinv");
+  xq = check_f32(xq, "xq");
+  x = check_f32(x, "x");
+  lengthscales = check_f32(lengthscales, "lengthscales");
+  alpha = check_f32(alpha, "alpha");
+  kinv = check_f32(kinv, "kinv");
   TORCH_CHECK(onehot.scalar_type() == torch::kUInt8 && onehot.is_cuda(),
               "onehot must be a uint8 GPU tensor");
+  onehot = onehot.contiguous();
   const int b = xq.size(0), d = xq.size(1), n = x.size(0);
   TORCH_CHECK(d <= 512, "posterior_scores supports D <= 512");
   TORCH_CHECK(n <= 36000, "posterior_scores supports N <= 36000 (LDS)");
@@ -99,9 +100,11 @@ std::vector<torch::Tensor> eagle_suggest(
     int64_t batch_start, int64_t batch_size, double visibility,
     double gravity, double neg_gravity, double norm_scale,
     double cat_factor, double p_same, int64_t seed, int64_t offset) {
-  check_f32(pool_cont, "pool_cont");
-  check_f32(rewards, "rewards");
-  check_f32(perturbations, "perturbations");
+  pool_cont = check_f32(pool_cont, "pool_cont");
+  rewards = check_f32(rewards, "rewards");
+  perturbations = check_f32(perturbations, "perturbations");
+  pool_cat = pool_cat.contiguous();
+  cat_sizes = cat_sizes.contiguous();
   const int pool_size = pool_cont.size(0);
   const int q = pool_cont.size(1);
   const int dc = pool_cont.size(2);
