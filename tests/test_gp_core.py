@@ -46,16 +46,18 @@ class TestMatern:
 class TestLBFGS:
 
   def test_batched_quadratic(self):
+    # NOTE: loss_fn must apply the SAME objective to every row (the
+    # parallel line search evaluates an (S*R, P) stack).
     A = torch.tensor([[3.0, 1.0], [1.0, 2.0]])
-    targets = torch.tensor([[1.0, -2.0], [0.5, 3.0], [-1.0, 0.0]])
+    target = torch.tensor([1.0, -2.0])
 
     def loss(x):
-      d = x - targets
+      d = x - target
       return torch.einsum('ri,ij,rj->r', d, A, d)
 
-    x0 = torch.zeros(3, 2)
+    x0 = torch.tensor([[0.0, 0.0], [5.0, -5.0], [-3.0, 3.0]])
     x, f = lbfgs.minimize_batched(loss, x0, max_iters=50)
-    assert torch.allclose(x, targets, atol=1e-4)
+    assert torch.allclose(x, target.expand(3, 2), atol=1e-4)
     assert float(f.max()) < 1e-7
 
   def test_rosenbrock_batch(self):

@@ -64,6 +64,7 @@ class GPBanditConfig:
   num_seed_trials: int = 2
   ard_restarts: int = 4
   ard_max_iters: int = 50
+  ard_warm_iters: int = 20   # iters when warm-starting from the last fit
   use_trust_region: bool = True
   num_scalarizations: int = 1000  # multi-objective
   data_parallel: bool = False     # shard the sweep across dist ranks
@@ -149,12 +150,17 @@ class VizierGPBandit(Designer, Predictor):
     y_np = self._prepare_labels(y_np)
     x = torch.as_tensor(x_np, dtype=cfg.dtype, device=self._device)
     self._x = x
+    prev = self._posteriors
     self._posteriors = []
     for m in range(y_np.shape[1]):
       y = torch.as_tensor(y_np[:, m], dtype=cfg.dtype, device=self._device)
+      warm = prev[m].raw if m < len(prev) and prev[m].raw is not None \
+          else None
       post = gp_model.train_gp(
           x, y, num_restarts=cfg.ard_restarts,
-          max_iters=cfg.ard_max_iters, seed=self._seed)
+          max_iters=cfg.ard_warm_iters if warm is not None
+          else cfg.ard_max_iters,
+          seed=self._seed, warm_start_raw=warm)
       if cfg.data_parallel:
         sharded_sweep.broadcast_posterior(post)
       self._posteriors.append(post)

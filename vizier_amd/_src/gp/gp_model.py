@@ -137,6 +137,7 @@ class GPPosterior:
   alpha: torch.Tensor        # (N,) (K + noise I)^-1 (y - mean)
   K_inv: Optional[torch.Tensor]  # (N, N), for the GEMM variance path
   nll: float                 # training loss of the selected restart
+  raw: Optional[torch.Tensor] = None  # unconstrained params (warm starts)
 
   def predict(self, xq: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
     """Returns (mean, stddev) of the latent f at query points (Q, D)."""
@@ -169,7 +170,8 @@ class GPPosterior:
 
 def train_gp(x: torch.Tensor, y: torch.Tensor, *,
              num_restarts: int = 4, max_iters: int = 50,
-             seed: int = 0, precompute_inverse: bool = True
+             seed: int = 0, precompute_inverse: bool = True,
+             warm_start_raw: Optional[torch.Tensor] = None
              ) -> GPPosterior:
   """Fits GP hyperparameters by restarting batched L-BFGS on the NLL.
 
@@ -183,6 +185,9 @@ def train_gp(x: torch.Tensor, y: torch.Tensor, *,
   generator = torch.Generator(device='cpu').manual_seed(seed)
   raw0 = _init_raw(num_restarts + 1, d, generator, 'cpu',
                    torch.float32).to(device=x.device, dtype=x.dtype)
+  if warm_start_raw is not None and warm_start_raw.numel() == d + 3:
+    # Warm start from the previous fit's optimum (incremental refits).
+    raw0 = torch.cat([warm_start_raw.reshape(1, -1).to(raw0), raw0], 0)
 
   def loss_fn(raw: torch.Tensor) -> torch.Tensor:
     return negative_log_marginal_likelihood(raw, x, y)
@@ -204,4 +209,4 @@ def train_gp(x: torch.Tensor, y: torch.Tensor, *,
     z = torch.linalg.solve_triangular(L, eye, upper=False)
     K_inv = z.T @ z
   return GPPosterior(x=x, params=params, L=L, alpha=alpha, K_inv=K_inv,
-                     nll=float(best_f[idx]))
+                     nll=float(best_f[idx]), raw=raw.detach())
