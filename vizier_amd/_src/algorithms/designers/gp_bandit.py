@@ -268,12 +268,21 @@ class VizierGPBandit(Designer, Predictor):
     prior_features, prior_rewards = trials_to_sorted_features(
         self._converter, self._codec, self._trials, rewards_np,
         device=self._device, dtype=cfg.dtype)
-    # Expand priors to the optimizer's q axis.
+    # For q-acquisitions, group priors into batches of q distinct trials
+    # (vectorized_base.py:390-429's parallel-batch reshape).
     q = n_parallel
     if q > 1:
-      prior_features = CandidateBatch(
-          prior_features.continuous.expand(-1, q, -1).contiguous(),
-          prior_features.categorical.expand(-1, q, -1).contiguous())
+      n_groups = prior_features.continuous.shape[0] // q
+      if n_groups == 0:
+        prior_features, prior_rewards = None, None
+      else:
+        cont = prior_features.continuous[:n_groups * q, 0, :]
+        cat = prior_features.categorical[:n_groups * q, 0, :]
+        prior_features = CandidateBatch(
+            cont.reshape(n_groups, q, -1).contiguous(),
+            cat.reshape(n_groups, q, -1).contiguous())
+        prior_rewards = prior_rewards[:n_groups * q].reshape(
+            n_groups, q).amax(dim=1)
 
     results = optimizer.optimize(
         score_fn, count=1 if n_parallel > 1 else count,
