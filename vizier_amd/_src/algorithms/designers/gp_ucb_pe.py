@@ -1,0 +1,265 @@
+"""GP-UCB-PE designer: the DEFAULT algorithm.
+
+Capability parity with vizier/_src/algorithms/designers/gp_ucb_pe.py
+(VizierGPUCBPEBandit :609, UCBPEConfig :80, UCBScoreFunction :282,
+PEScoreFunction :384, suggest flow :1356-1445): the first suggestion of
+a batch exploits with UCB (coef 1.8); the remaining batch members run
+Pure Exploration — maximize the predictive stddev conditioned on
+completed AND pending/hallucinated points, linearly penalized (coef 10)
+for leaving the promising region {x : UCB_explore(x) >= threshold},
+where threshold is the predicted mean at the observed point with the
+highest UCB. Probabilistic UCB/PE overwrites and the high-noise
+exploration boost mirror the reference's UCBPEConfig defaults.
+"""
+
+from __future__ import annotations
+
+import copy
+import dataclasses
+from typing import List, Optional, Sequence
+
+import numpy as np
+import torch
+
+from vizier_amd import pyvizier as vz
+from vizier_amd.converters.core import TrialToArrayConverter
+from vizier_amd._src.algorithms.core.abstractions import (
+    ActiveTrials,
+    CompletedTrials,
+    Designer,
+)
+from vizier_amd._src.algorithms.designers import gp_bandit
+from vizier_amd._src.algorithms.optimizers.eagle import (
+    CandidateBatch,
+    EagleStrategyConfig,
+)
+from vizier_amd._src.algorithms.optimizers.vectorized import (
+    EagleFeatureCodec,
+    VectorizedOptimizerFactory,
+    trials_to_sorted_features,
+)
+from vizier_amd._src.gp import acquisitions as acq_lib
+from vizier_amd._src.gp import gp_model, output_warpers
+
+
+@dataclasses.dataclass
+class UCBPEConfig:
+  """Reference defaults (gp_ucb_pe.py:80-130)."""
+
+  ucb_coefficient: float = 1.8
+  explore_region_ucb_coefficient: float = 0.5
+  cb_violation_penalty_coefficient: float = 10.0
+  ucb_overwrite_probability: float = 0.25
+  pe_overwrite_probability: float = 0.1
+  pe_overwrite_probability_in_high_noise: float = 0.7
+  signal_to_noise_threshold: float = 0.7
+  max_evaluations: int = 75000
+  suggestion_batch_size: int = 25
+  num_seed_trials: int = 2
+  ard_restarts: int = 4
+  ard_max_iters: int = 50
+  ard_warm_iters: int = 20
+  use_trust_region: bool = True
+  device: Optional[str] = None
+  dtype: torch.dtype = torch.float32
+
+
+class VizierGPUCBPEBandit(Designer):
+  """UCB exploitation + Pure-Exploration batch fill."""
+
+  def __init__(self, problem: vz.ProblemStatement,
+               config: Optional[UCBPEConfig] = None, *, seed: int = 0):
+    self._problem = problem
+    self._config = config or UCBPEConfig()
+    self._seed = seed
+    self._rng = np.random.default_rng(seed)
+    self._completed: List[vz.Trial] = []
+    self._active: List[vz.Trial] = []
+    self._converter = TrialToArrayConverter(problem)
+    self._codec = EagleFeatureCodec(self._converter)
+    self._device = self._config.device or gp_bandit.default_device()
+    self._posterior: Optional[gp_model.GPPosterior] = None
+    self._last_fit_count = -1
+    self._last_suggest_completed = 0
+    from vizier_amd._src.algorithms.designers.quasi_random import (
+        QuasiRandomDesigner,
+    )
+    self._quasi_random = (
+        QuasiRandomDesigner(problem.search_space, seed=seed)
+        if not problem.search_space.is_conditional else None)
+
+  @classmethod
+  def from_problem(cls, problem: vz.ProblemStatement,
+                   seed: int = 0) -> 'VizierGPUCBPEBandit':
+    return cls(problem, seed=seed)
+
+  # -- Designer API ---------------------------------------------------------
+
+  def update(self, completed: CompletedTrials, all_active: ActiveTrials
+             ) -> None:
+    self._completed.extend(completed.trials)
+    self._active = list(all_active.trials)
+
+  def suggest(self, count: Optional[int] = None
+              ) -> Sequence[vz.TrialSuggestion]:
+    count = count or 1
+    cfg = self._config
+    if len(self._completed) < cfg.num_seed_trials:
+      return self._seed_suggestions(count)
+
+    self._fit()
+    has_new = len(self._completed) > self._last_suggest_completed
+    self._last_suggest_completed = len(self._completed)
+
+    # Hallucinated feature set starts as completed + currently-active.
+    x_all = [self._posterior.x]
+    if self._active:
+      x_all.append(torch.as_tensor(
+          self._converter.to_features(self._active), dtype=cfg.dtype,
+          device=self._device))
+
+    suggestions: List[vz.TrialSuggestion] = []
+    for i in range(count):
+      use_ucb = self._choose_ucb(has_new and i == 0 and not self._active)
+      dense = self._optimize_one(use_ucb, torch.cat(x_all, dim=0))
+      x_all.append(dense.reshape(1, -1))
+      params = self._converter.to_parameters(
+          dense.detach().cpu().numpy())[0]
+      s = vz.TrialSuggestion(params)
+      s.metadata.ns('gp_ucb_pe')['acquisition'] = \
+          'ucb' if use_ucb else 'pe'
+      suggestions.append(s)
+    return suggestions
+
+  # -- internals ------------------------------------------------------------
+
+  def _seed_suggestions(self, count: int) -> List[vz.TrialSuggestion]:
+    from vizier_amd._src.pythia import suggest_default
+    out: List[vz.TrialSuggestion] = []
+    if not self._completed and not self._active:
+      out.append(vz.TrialSuggestion(suggest_default.get_default_parameters(
+          self._problem.search_space)))
+    remaining = count - len(out)
+    if remaining > 0:
+      if self._quasi_random is not None:
+        out.extend(self._quasi_random.suggest(remaining))
+      else:
+        from vizier_amd._src.algorithms.designers.random import (
+            RandomDesigner,
+        )
+        out.extend(RandomDesigner(self._problem.search_space,
+                                  seed=self._seed).suggest(remaining))
+    return out
+
+  def _fit(self) -> None:
+    if self._last_fit_count == len(self._completed) and self._posterior:
+      return
+    cfg = self._config
+    x_np, y_np = self._converter.to_xy(self._completed)
+    warper = output_warpers.create_default_warper()
+    y_np = warper.warp(y_np[:, :1]).flatten()
+    x = torch.as_tensor(x_np, dtype=cfg.dtype, device=self._device)
+    y = torch.as_tensor(y_np, dtype=cfg.dtype, device=self._device)
+    warm = self._posterior.raw if self._posterior is not None else None
+    self._posterior = gp_model.train_gp(
+        x, y, num_restarts=cfg.ard_restarts,
+        max_iters=cfg.ard_warm_iters if warm is not None
+        else cfg.ard_max_iters, seed=self._seed, warm_start_raw=warm)
+    self._warped_labels = y
+    self._last_fit_count = len(self._completed)
+
+  def _choose_ucb(self, fresh_first: bool) -> bool:
+    cfg = self._config
+    params = self._posterior.params
+    signal = float(params.amplitude) ** 2
+    noise = float(params.noise)
+    high_noise = (cfg.signal_to_noise_threshold > 0 and
+                  signal / max(noise, 1e-12)
+                  < cfg.signal_to_noise_threshold)
+    if fresh_first:
+      # New completed trials: exploit, unless PE-overwrite fires.
+      p_pe = (cfg.pe_overwrite_probability_in_high_noise if high_noise
+              else cfg.pe_overwrite_probability)
+      return self._rng.random() >= p_pe
+    # No new data (or later batch members): explore, unless UCB-overwrite.
+    if high_noise:
+      return False
+    return self._rng.random() < cfg.ucb_overwrite_probability
+
+  def _variance_posterior(self, x_all: torch.Tensor
+                          ) -> gp_model.GPPosterior:
+    """GP conditioned on all (completed+hallucinated) features.
+
+    Only the predictive variance is used, so alpha is zeros.
+    """
+    params = self._posterior.params
+    from vizier_amd._src.gp.matern import gram_matern52
+    n = x_all.shape[0]
+    K = gram_matern52(x_all, None, params.lengthscales, params.amplitude)
+    K = K + params.noise * torch.eye(n, dtype=x_all.dtype,
+                                     device=x_all.device)
+    L = gp_model.cholesky_with_jitter(K, params.amplitude ** 2)
+    eye = torch.eye(n, dtype=x_all.dtype, device=x_all.device)
+    z = torch.linalg.solve_triangular(L, eye, upper=False)
+    K_inv = z.T @ z
+    return gp_model.GPPosterior(
+        x=x_all, params=params, L=L,
+        alpha=torch.zeros(n, dtype=x_all.dtype, device=x_all.device),
+        K_inv=K_inv, nll=0.0)
+
+  def _optimize_one(self, use_ucb: bool, x_all: torch.Tensor
+                    ) -> torch.Tensor:
+    cfg = self._config
+    posterior = self._posterior
+    onehot = torch.zeros(self._converter.n_features, dtype=torch.bool,
+                         device=self._device)
+    for col in self._converter.output_specs:
+      if col.is_onehot:
+        onehot[col.start:col.start + col.width] = True
+    trust_region = acq_lib.TrustRegion(x_all, onehot) \
+        if cfg.use_trust_region else None
+
+    if use_ucb:
+      scoring = acq_lib.ScoringFunction(
+          posterior, acq_lib.UCB(cfg.ucb_coefficient), trust_region)
+
+      def score_fn(batch: CandidateBatch) -> torch.Tensor:
+        return scoring(self._codec.decode(batch)[:, 0, :])
+    else:
+      # Promising-region threshold: predicted mean at the observed point
+      # with the highest UCB (gp_ucb_pe.py:175-205).
+      mean_obs, stddev_obs = posterior.predict(posterior.x)
+      ucb_obs = mean_obs + cfg.ucb_coefficient * stddev_obs
+      threshold = mean_obs[int(torch.argmax(ucb_obs))]
+      var_post = self._variance_posterior(x_all)
+
+      def score_fn(batch: CandidateBatch) -> torch.Tensor:
+        xs = self._codec.decode(batch)[:, 0, :]
+        mean, stddev = posterior.predict(xs)
+        explore_ucb = mean + cfg.explore_region_ucb_coefficient * stddev
+        _, stddev_all = var_post.predict(xs)
+        penalty = cfg.cb_violation_penalty_coefficient * torch.minimum(
+            explore_ucb - threshold, torch.zeros_like(explore_ucb))
+        scores = stddev_all + penalty
+        if trust_region is not None:
+          scores = trust_region.apply(xs, scores)
+        return scores
+
+    factory = VectorizedOptimizerFactory(
+        eagle_config=EagleStrategyConfig(),
+        max_evaluations=cfg.max_evaluations,
+        suggestion_batch_size=cfg.suggestion_batch_size)
+    optimizer = factory(
+        n_continuous=self._codec.n_continuous,
+        categorical_sizes=self._codec.categorical_sizes,
+        seed=self._seed + len(self._completed) + x_all.shape[0],
+        device=self._device, dtype=cfg.dtype)
+
+    rewards = self._warped_labels.cpu().numpy()
+    prior_features, prior_rewards = trials_to_sorted_features(
+        self._converter, self._codec, self._completed, rewards,
+        device=self._device, dtype=cfg.dtype)
+    results = optimizer.optimize(score_fn, count=1,
+                                 prior_features=prior_features,
+                                 prior_rewards=prior_rewards)
+    return self._codec.decode(results.features)[0, 0, :]
