@@ -1,0 +1,195 @@
+"""Tests for BBOB, experimenters, runners, analyzers and test harnesses."""
+
+import numpy as np
+import pytest
+
+from vizier_amd import benchmarks as vzb
+from vizier_amd import pyvizier as vz
+from vizier_amd._src.algorithms.designers.random import RandomDesigner
+from vizier_amd._src.algorithms.designers.quasi_random import (
+    QuasiRandomDesigner,
+)
+from vizier_amd._src.algorithms.testing.comparator_runner import (
+    FailedComparisonTestError,
+    SimpleRegretComparisonTester,
+)
+from vizier_amd._src.algorithms.testing.simplekd_runner import (
+    SimpleKDConvergenceTester,
+)
+from vizier_amd._src.benchmarks.experimenters.synthetic import bbob
+from vizier_amd._src.benchmarks.experimenters.synthetic.simplekd import (
+    SimpleKDExperimenter,
+)
+
+
+class TestBBOB:
+
+  @pytest.mark.parametrize('fn', bbob.BBOB_FUNCTIONS,
+                           ids=lambda f: f.__name__)
+  def test_function_finite_and_scalar(self, fn):
+    rng = np.random.default_rng(0)
+    for dim in (2, 5):
+      for _ in range(3):
+        x = rng.uniform(-5, 5, dim)
+        v = fn(x, seed=1)
+        assert np.isfinite(v), (fn.__name__, x)
+
+  def test_sphere_minimum_at_origin(self):
+    assert bbob.Sphere(np.zeros(4)) == 0.0
+    assert bbob.Sphere(np.ones(4)) == 4.0
+
+  def test_default_problem_statement(self):
+    problem = bbob.DefaultBBOBProblemStatement(6)
+    assert len(problem.search_space.parameters) == 6
+    assert problem.metric_information.item().goal.is_minimize
+
+  def test_deterministic_given_seed(self):
+    x = np.array([1.0, -2.0, 0.5])
+    assert bbob.Rastrigin(x, seed=4) == bbob.Rastrigin(x, seed=4)
+    assert bbob.Rastrigin(x, seed=4) != bbob.Rastrigin(x, seed=5)
+
+
+class TestExperimenters:
+
+  def _base(self, dim=3):
+    return vzb.NumpyExperimenter(
+        bbob.Sphere, bbob.DefaultBBOBProblemStatement(dim))
+
+  def _trial(self, values):
+    return vz.Trial({f'x{i}': v for i, v in enumerate(values)}, id=1)
+
+  def test_numpy_experimenter(self):
+    exptr = self._base()
+    t = self._trial([1.0, 2.0, 2.0])
+    exptr.evaluate([t])
+    assert t.final_measurement.metrics['bbob_eval'].value == 9.0
+
+  def test_shifting(self):
+    exptr = vzb.ShiftingExperimenter(self._base(), np.array([1.0, 0, 0]))
+    t = self._trial([0.0, 0.0, 0.0])
+    exptr.evaluate([t])
+    assert t.final_measurement.metrics['bbob_eval'].value == 1.0
+
+  def test_noisy(self):
+    exptr = vzb.NoisyExperimenter(self._base(), noise_std=0.1, seed=0)
+    t = self._trial([0.0, 0.0, 0.0])
+    exptr.evaluate([t])
+    assert t.final_measurement.metrics['bbob_eval'].value != 0.0
+
+  def test_sign_flip(self):
+    exptr = vzb.SignFlipExperimenter(self._base())
+    assert exptr.problem_statement().metric_information.item(
+    ).goal.is_maximize
+    t = self._trial([1.0, 0.0, 0.0])
+    exptr.evaluate([t])
+    assert t.final_measurement.metrics['bbob_eval'].value == -1.0
+
+  def test_discretizing(self):
+    exptr = vzb.DiscretizingExperimenter(self._base(),
+                                         {'x0': [-1.0, 0.0, 1.0]})
+    space = exptr.problem_statement().search_space
+    assert space.get('x0').type == vz.ParameterType.DISCRETE
+
+  def test_multiobjective(self):
+    exptr = vzb.MultiObjectiveExperimenter(
+        {'m1': self._base(), 'm2': vzb.NumpyExperimenter(
+            bbob.NegativeSphere, bbob.DefaultBBOBProblemStatement(3))})
+    problem = exptr.problem_statement()
+    assert len(problem.metric_information) == 2
+    t = self._trial([1.0, 0.0, 0.0])
+    exptr.evaluate([t])
+    assert t.final_measurement.metrics['m1'].value == 1.0
+    assert t.final_measurement.metrics['m2'].value == 99.0
+
+  def test_factories(self):
+    factory = vzb.SingleObjectiveExperimenterFactory(
+        vzb.BBOBExperimenterFactory('Sphere', 4),
+        shift=np.full(4, 0.5), noise_std=0.01)
+    exptr = factory()
+    t = vz.Trial({f'x{i}': 0.0 for i in range(4)}, id=1)
+    exptr.evaluate([t])
+    assert np.isfinite(t.final_measurement.metrics['bbob_eval'].value)
+
+
+class TestRunnerAndAnalyzers:
+
+  def test_benchmark_runner_loop(self):
+    exptr = vzb.NumpyExperimenter(bbob.Sphere,
+                                  bbob.DefaultBBOBProblemStatement(2))
+    state = vzb.BenchmarkState.from_designer_factory(
+        lambda p: RandomDesigner(p.search_space, seed=0), exptr)
+    vzb.BenchmarkRunner([vzb.GenerateAndEvaluate(3)],
+                        num_repeats=4).run(state)
+    trials = state.algorithm.supporter.GetTrials()
+    assert len(trials) == 12
+    assert all(t.status == vz.TrialStatus.COMPLETED for t in trials)
+
+  def test_convergence_curve(self):
+    mi = vz.MetricInformation(name='m',
+                              goal=vz.ObjectiveMetricGoal.MINIMIZE)
+    trials = []
+    for i, v in enumerate([5.0, 3.0, 4.0, 1.0]):
+      t = vz.Trial(id=i + 1)
+      t.complete(vz.Measurement(metrics={'m': v}))
+      trials.append(t)
+    curve = vzb.ConvergenceCurveConverter(mi).convert(trials)
+    np.testing.assert_allclose(curve.ys[0], [5.0, 3.0, 3.0, 1.0])
+
+  def test_hypervolume_curve_monotone(self):
+    metrics = [vz.MetricInformation(name='a', goal=1),
+               vz.MetricInformation(name='b', goal=1)]
+    rng = np.random.default_rng(0)
+    trials = []
+    for i in range(10):
+      t = vz.Trial(id=i + 1)
+      t.complete(vz.Measurement(metrics={'a': rng.random(),
+                                         'b': rng.random()}))
+      trials.append(t)
+    curve = vzb.HypervolumeCurveConverter(metrics).convert(trials)
+    assert np.all(np.diff(curve.ys[0]) >= -1e-9)
+
+  def test_comparators(self):
+    xs = np.arange(1, 11)
+    slow = vzb.ConvergenceCurve(xs, np.linspace(0, 0.8, 10)[None, :])
+    fast = vzb.ConvergenceCurve(xs, np.sqrt(np.linspace(0, 1, 10))[None,
+                                                                   :])
+    le = vzb.LogEfficiencyConvergenceCurveComparator(slow).score(fast)
+    assert le > 0
+    pb = vzb.PercentageBetterConvergenceCurveComparator(slow).score(fast)
+    assert pb > 0.5
+    wr = vzb.WinRateComparator(slow).score(fast)
+    assert wr >= 0.0
+
+
+class TestStatisticalGates:
+
+  def test_quasi_random_beats_nothing_burns(self):
+    # Eagle-vs-random style gate: quasi-random should NOT be confidently
+    # better than itself (sanity of the test machinery).
+    exptr = vzb.NumpyExperimenter(bbob.Sphere,
+                                  bbob.DefaultBBOBProblemStatement(2))
+    tester = SimpleRegretComparisonTester(
+        baseline_num_trials=10, candidate_num_trials=10,
+        baseline_num_repeats=3, candidate_num_repeats=3, alpha=0.01)
+    with pytest.raises(FailedComparisonTestError):
+      tester.assert_benchmark_state_better_simple_regret(
+          exptr,
+          lambda p, s: RandomDesigner(p.search_space, seed=s),
+          lambda p, s: RandomDesigner(p.search_space, seed=100 + s))
+
+  def test_simplekd_gate_random_converges_with_budget(self):
+    tester = SimpleKDConvergenceTester(
+        best_category='corner',
+        designer_factory=lambda p, s: RandomDesigner(p.search_space,
+                                                     seed=s),
+        num_trials=400, max_relative_error=0.12,
+        num_seeds=3, num_required=2)
+    tester.assert_convergence()
+
+  def test_simplekd_experimenter(self):
+    exptr = SimpleKDExperimenter('center')
+    t = vz.Trial({'categorical': 'center', 'discrete': 0.7, 'int': 2,
+                  'float': 1.0}, id=1)
+    exptr.evaluate([t])
+    assert t.final_measurement.metrics['value'].value == pytest.approx(
+        exptr.optimal_value)

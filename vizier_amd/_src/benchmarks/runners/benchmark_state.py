@@ -1,0 +1,82 @@
+"""Benchmark state: experimenter + algorithm runner.
+
+Capability parity with vizier/_src/benchmarks/runners/benchmark_state.py
+(BenchmarkState :92, PolicySuggester/DesignerSuggester factories
+:42-165).
+"""
+
+from __future__ import annotations
+
+import dataclasses
+from typing import Callable, Optional
+
+from vizier_amd import pyvizier as vz
+from vizier_amd._src.algorithms.core.abstractions import Designer
+from vizier_amd._src.algorithms.policies.designer_policy import (
+    DesignerPolicy,
+)
+from vizier_amd._src.pythia.local_policy_supporters import (
+    InRamPolicySupporter,
+)
+from vizier_amd._src.pythia.policy import Policy
+from vizier_amd._src.benchmarks.experimenters.experimenter import (
+    Experimenter,
+)
+
+
+class PolicySuggester:
+  """Drives a Pythia policy against an in-RAM study."""
+
+  def __init__(self, policy: Policy, supporter: InRamPolicySupporter):
+    self._policy = policy
+    self._supporter = supporter
+
+  @property
+  def supporter(self) -> InRamPolicySupporter:
+    return self._supporter
+
+  def suggest(self, batch_size: Optional[int] = None):
+    return self._supporter.SuggestTrials(self._policy, batch_size or 1)
+
+  @classmethod
+  def from_designer_factory(
+      cls, problem: vz.ProblemStatement,
+      designer_factory: Callable[[vz.ProblemStatement], Designer],
+      seed: Optional[int] = None) -> 'PolicySuggester':
+    supporter = InRamPolicySupporter(problem)
+    policy = DesignerPolicy(supporter, designer_factory)
+    del seed
+    return cls(policy, supporter)
+
+
+@dataclasses.dataclass
+class BenchmarkState:
+  """The moving parts of one benchmark run."""
+
+  experimenter: Experimenter
+  algorithm: PolicySuggester
+
+  @classmethod
+  def from_designer_factory(
+      cls, designer_factory: Callable[[vz.ProblemStatement], Designer],
+      experimenter: Experimenter,
+      seed: Optional[int] = None) -> 'BenchmarkState':
+    return cls(
+        experimenter=experimenter,
+        algorithm=PolicySuggester.from_designer_factory(
+            experimenter.problem_statement(), designer_factory, seed))
+
+
+BenchmarkStateFactory = Callable[[], BenchmarkState]
+
+
+@dataclasses.dataclass
+class DesignerBenchmarkStateFactory:
+  """Factory capturing (experimenter, designer_factory)."""
+
+  experimenter: Experimenter
+  designer_factory: Callable[[vz.ProblemStatement], Designer]
+
+  def __call__(self, seed: Optional[int] = None) -> BenchmarkState:
+    return BenchmarkState.from_designer_factory(
+        self.designer_factory, self.experimenter, seed)

@@ -47,6 +47,10 @@ class Namespace(abc.Sequence):
   def __setattr__(self, *_):
     raise AttributeError('Namespace is immutable')
 
+  def __reduce__(self):
+    # Supports copy/deepcopy/pickle despite the immutability guard.
+    return (Namespace, (self._parts,))
+
   @classmethod
   def decode(cls, s: str) -> 'Namespace':
     if not s:

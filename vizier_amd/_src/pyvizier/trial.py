@@ -93,6 +93,10 @@ class ParameterValue:
   def __setattr__(self, *_):
     raise AttributeError('ParameterValue is immutable')
 
+  def __reduce__(self):
+    # Supports copy/deepcopy/pickle despite the immutability guard.
+    return (ParameterValue, (self._value,))
+
   @property
   def value(self) -> ParameterValueTypes:
     return self._value
