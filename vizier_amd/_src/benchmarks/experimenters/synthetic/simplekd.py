@@ -43,8 +43,8 @@ class SimpleKDExperimenter(Experimenter):
 
   @property
   def optimal_value(self) -> float:
-    # category bonus 2 + discrete 0.7^2 + int (2/2)^2 + float 1.
-    return 2.0 + 0.49 + 1.0 + 1.0
+    # category bonus 2 + discrete (-0.8)^2 + int (2/2)^2 + float 1.
+    return 2.0 + 0.64 + 1.0 + 1.0
 
   def _value(self, trial: vz.Trial) -> float:
     cat = trial.parameters.get_value('categorical')
