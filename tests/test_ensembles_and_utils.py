@@ -1,0 +1,226 @@
+"""Tests: scalarizers, scheduled/ensemble/meta designers, utils."""
+
+import datetime
+import time
+
+import numpy as np
+import pytest
+
+from vizier_amd import pyvizier as vz
+from vizier_amd.utils import json_utils, profiler
+from vizier_amd._src.algorithms.core.abstractions import (
+    ActiveTrials,
+    CompletedTrials,
+)
+from vizier_amd._src.algorithms.designers.meta_learning import (
+    MetaLearningConfig,
+    MetaLearningDesigner,
+)
+from vizier_amd._src.algorithms.designers.random import RandomDesigner
+from vizier_amd._src.algorithms.designers import scalarization as sc
+from vizier_amd._src.algorithms.designers.scalarizing_designer import (
+    ScalarizingDesigner,
+)
+from vizier_amd._src.algorithms.designers.scheduled_designer import (
+    ExponentialScheduledParam,
+    LinearScheduledParam,
+    ScheduledDesigner,
+)
+from vizier_amd._src.algorithms.designers.unsafe_as_infeasible_designer import (
+    UnsafeAsInfeasibleDesigner,
+)
+from vizier_amd._src.algorithms.ensemble.ensemble_design import (
+    AdaptiveEnsembleDesign,
+    EXP3IXEnsembleDesign,
+    EXP3UniformEnsembleDesign,
+)
+from vizier_amd._src.algorithms.ensemble.ensemble_designer import (
+    EnsembleDesigner,
+)
+
+
+def problem_2obj():
+  problem = vz.ProblemStatement()
+  problem.search_space.root.add_float_param('x', 0.0, 1.0)
+  problem.metric_information.extend([
+      vz.MetricInformation(name='a', goal=vz.ObjectiveMetricGoal.MAXIMIZE),
+      vz.MetricInformation(name='b', goal=vz.ObjectiveMetricGoal.MINIMIZE),
+  ])
+  return problem
+
+
+class TestScalarization:
+
+  def test_linear_and_chebyshev(self):
+    w = np.array([1.0, 2.0])
+    ys = np.array([[1.0, 1.0], [2.0, 0.5]])
+    np.testing.assert_allclose(sc.LinearScalarization(w)(ys), [3.0, 3.0])
+    np.testing.assert_allclose(sc.ChebyshevScalarization(w)(ys),
+                               [1.0, 1.0])
+
+  def test_hypervolume_scalarization(self):
+    s = sc.HyperVolumeScalarization(np.array([1.0, 1.0]),
+                                    reference_point=np.zeros(2))
+    assert s(np.array([2.0, 3.0])) == 2.0
+
+  def test_scalarizing_designer(self):
+    problem = problem_2obj()
+    designer = ScalarizingDesigner(
+        problem, lambda p: RandomDesigner(p.search_space, seed=0),
+        sc.LinearScalarization(np.array([1.0, 1.0])))
+    t = vz.Trial({'x': 0.5}, id=1)
+    t.complete(vz.Measurement(metrics={'a': 1.0, 'b': 2.0}))
+    designer.update(CompletedTrials([t]), ActiveTrials())
+    assert designer.suggest(2)
+
+
+class TestScheduled:
+
+  def test_params(self):
+    lin = LinearScheduledParam(4.0, 1.0)
+    assert lin.value(0.0) == 4.0 and lin.value(1.0) == 1.0
+    exp = ExponentialScheduledParam(4.0, 1.0)
+    assert exp.value(0.0) == pytest.approx(4.0)
+    assert exp.value(1.0) == pytest.approx(1.0)
+    assert 1.0 < exp.value(0.5) < 4.0
+
+  def test_scheduled_designer_progress(self):
+    problem = vz.ProblemStatement()
+    problem.search_space.root.add_float_param('x', 0.0, 1.0)
+    problem.metric_information.append(vz.MetricInformation(name='m',
+                                                           goal=1))
+    captured = []
+
+    def factory(p, coef):
+      captured.append(coef)
+      return RandomDesigner(p.search_space, seed=1)
+
+    designer = ScheduledDesigner(
+        problem, factory, {'coef': LinearScheduledParam(10.0, 0.0)},
+        expected_total_num_trials=10)
+    designer.update(CompletedTrials([]), ActiveTrials())
+    designer.suggest(1)
+    trials = []
+    for i in range(5):
+      t = vz.Trial({'x': 0.5}, id=i + 1)
+      t.complete(vz.Measurement(metrics={'m': 0.0}))
+      trials.append(t)
+    designer.update(CompletedTrials(trials), ActiveTrials())
+    designer.suggest(1)
+    assert captured[0] == 10.0
+    assert captured[1] == pytest.approx(5.0)
+
+
+class TestEnsembles:
+
+  @pytest.mark.parametrize('cls', [EXP3UniformEnsembleDesign,
+                                   EXP3IXEnsembleDesign,
+                                   AdaptiveEnsembleDesign])
+  def test_strategy_shifts_towards_winner(self, cls):
+    strat = cls([0, 1], seed=0)
+    for _ in range(50):
+      strat.update(0, 1.0)
+      strat.update(1, 0.0)
+    probs = strat.ensemble_probs
+    assert probs[0] > probs[1]
+    assert probs.sum() == pytest.approx(1.0)
+
+  def test_ensemble_designer_attributes_rewards(self):
+    problem = vz.ProblemStatement()
+    problem.search_space.root.add_float_param('x', 0.0, 1.0)
+    problem.metric_information.append(vz.MetricInformation(name='m',
+                                                           goal=1))
+    designer = EnsembleDesigner(
+        {'r1': RandomDesigner(problem.search_space, seed=1),
+         'r2': RandomDesigner(problem.search_space, seed=2)})
+    uid = 0
+    for _ in range(10):
+      for s in designer.suggest(1):
+        uid += 1
+        t = s.to_trial(uid)
+        t.complete(vz.Measurement(
+            metrics={'m': float(s.parameters.get_value('x'))}))
+        designer.update(CompletedTrials([t]), ActiveTrials())
+    assert uid == 10
+
+
+class TestMetaLearning:
+
+  def test_hyperparameter_epochs(self):
+    problem = vz.ProblemStatement()
+    problem.search_space.root.add_float_param('x', 0.0, 1.0)
+    problem.metric_information.append(vz.MetricInformation(name='m',
+                                                           goal=1))
+    meta_space = vz.SearchSpace()
+    meta_space.root.add_float_param('exploration', 0.1, 1.0)
+    seen = []
+
+    def tuned_factory(p, hparams):
+      seen.append(dict(hparams))
+      return RandomDesigner(p.search_space, seed=0)
+
+    designer = MetaLearningDesigner(
+        problem, tuned_factory, meta_space,
+        config=MetaLearningConfig(num_trials_per_update=3))
+    uid = 0
+    for _ in range(7):
+      s = designer.suggest(1)[0]
+      uid += 1
+      t = s.to_trial(uid)
+      t.complete(vz.Measurement(metrics={'m': float(uid)}))
+      designer.update(CompletedTrials([t]), ActiveTrials())
+    assert len(seen) >= 3  # initial + >= 2 epochs
+    assert all(0.1 <= h['exploration'] <= 1.0 for h in seen)
+
+
+class TestSafetyWrapper:
+
+  def test_unsafe_marked_infeasible(self):
+    problem = vz.ProblemStatement()
+    problem.search_space.root.add_float_param('x', 0.0, 1.0)
+    problem.metric_information.extend([
+        vz.MetricInformation(name='obj', goal=1),
+        vz.MetricInformation(name='safe', goal=1, safety_threshold=0.5)])
+    received = []
+
+    class Spy(RandomDesigner):
+      def update(self, completed, all_active):
+        received.extend(completed.trials)
+
+    designer = UnsafeAsInfeasibleDesigner(
+        problem, lambda p: Spy(p.search_space, seed=0))
+    t1 = vz.Trial({'x': 0.1}, id=1)
+    t1.complete(vz.Measurement(metrics={'obj': 1.0, 'safe': 0.9}))
+    t2 = vz.Trial({'x': 0.2}, id=2)
+    t2.complete(vz.Measurement(metrics={'obj': 1.0, 'safe': 0.1}))
+    designer.update(CompletedTrials([t1, t2]), ActiveTrials())
+    assert not received[0].infeasible
+    assert received[1].infeasible
+
+
+class TestUtils:
+
+  def test_profiler_collects(self):
+    with profiler.collect_events() as events:
+      with profiler.timeit('scope_a'):
+        time.sleep(0.01)
+
+      @profiler.record_runtime
+      def fn():
+        time.sleep(0.005)
+      fn()
+    latencies = profiler.get_latencies_dict(events)
+    assert 'scope_a' in latencies and 'fn' in latencies
+    assert latencies['scope_a'][0] >= datetime.timedelta(seconds=0.01)
+    # Outside collect_events nothing is recorded.
+    with profiler.timeit('scope_b'):
+      pass
+    assert 'scope_b' not in profiler.get_latencies_dict(events)
+
+  def test_json_numpy_roundtrip(self):
+    obj = {'a': np.arange(6, dtype=np.int32).reshape(2, 3),
+           'b': [np.float64(1.5)]}
+    s = json_utils.dumps(obj)
+    back = json_utils.loads(s)
+    np.testing.assert_array_equal(back['a'], obj['a'])
+    assert back['a'].dtype == np.int32
