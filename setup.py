@@ -13,6 +13,7 @@ from torch.utils.cpp_extension import BuildExtension, CUDAExtension
 _SRC = [
     'vizier_amd/_src/ops/csrc/module.cpp',
     'vizier_amd/_src/ops/csrc/gram_matern52.hip',
+    'vizier_amd/_src/ops/csrc/gram_matern52_bf16.hip',
     'vizier_amd/_src/ops/csrc/posterior_score.hip',
     'vizier_amd/_src/ops/csrc/eagle_step.hip',
 ]

@@ -212,3 +212,34 @@ class TestGPUDesignerEndToEnd:
         t.complete(vz.Measurement(metrics={'obj': val}))
         designer.update(CompletedTrials([t]), ActiveTrials())
     assert best > -0.15, f'GPU GP-Bandit failed to converge: {best}'  # 12 iters
+
+
+class TestGramBf16MFMA:
+
+  @pytest.mark.parametrize('n,m,d', [(64, 64, 20), (500, 500, 8),
+                                     (257, 130, 51)])
+  def test_bf16_mfma_matches_fp64_reference(self, ext, n, m, d):
+    from vizier_amd._src.gp.matern import gram_matern52
+    g = torch.Generator().manual_seed(7)
+    x1 = torch.rand(n, d, generator=g).cuda()
+    x2 = torch.rand(m, d, generator=g).cuda()
+    ls = (torch.rand(d, generator=g) * 1.5 + 0.2).cuda()
+    amp = 1.4
+    got = ext.gram_matern52_bf16(x1, x2, ls, amp)
+    want = gram_matern52(x1.cpu().double(), x2.cpu().double(),
+                         ls.cpu().double(), torch.tensor(amp).double())
+    err = (got.cpu().double() - want).abs().max()
+    # bf16 cross-term: ~3 decimal digits; kernel values are O(amp^2).
+    assert float(err) < 3e-2 * amp * amp, f'max err {err}'
+    # Asymmetric inputs catch operand/output transposes (guide G9): the
+    # error against the TRANSPOSED reference must be much larger.
+    err_t = (got.cpu().double() - want.T).abs().max() if n == m else None
+    if err_t is not None:
+      assert float(err_t) > 10 * float(err)
+
+  def test_bf16_diag_is_amp2(self, ext):
+    x = torch.rand(100, 16).cuda()
+    ls = torch.full((16,), 0.5).cuda()
+    K = ext.gram_matern52_bf16(x, x, ls, 2.0)
+    assert torch.allclose(torch.diagonal(K),
+                          torch.full((100,), 4.0).cuda(), atol=0.05)

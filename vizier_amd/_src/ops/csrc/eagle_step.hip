@@ -53,11 +53,17 @@ eagle_suggest_kernel(const float* __restrict__ pool_cont,  // (P, q, Dc)
                      const long* __restrict__ cat_sizes,   // (Dcat,)
                      float* __restrict__ out_cont,         // (B, q, Dc)
                      long* __restrict__ out_cat,           // (B, q, Dcat)
-                     int batch_start, int batch_size, int pool_size, int q,
+                     const unsigned long long* __restrict__ iter_ptr,
+                     int n_batches, int batch_size, int pool_size, int q,
                      int dc, int dcat, int max_cat, float visibility,
                      float gravity, float neg_gravity, float norm_scale,
                      float cat_factor, float p_same,
-                     unsigned long long seed, unsigned long long offset) {
+                     unsigned long long seed) {
+  // The iteration counter lives in device memory so the whole step is
+  // hipGraph-capturable (no host-varying kernel arguments).
+  const unsigned long long offset = *iter_ptr;
+  const int batch_start = (int)(offset % (unsigned long long)n_batches) *
+                          batch_size;
   __shared__ float scale[MAX_POOL];
   __shared__ float red[8];
   __shared__ float scale_sum_s;
@@ -200,11 +206,14 @@ eagle_update_kernel(float* __restrict__ pool_cont,
                     const long* __restrict__ batch_cat,
                     const float* __restrict__ batch_rewards,
                     const long* __restrict__ cat_sizes,
-                    float* __restrict__ best_reward, int batch_start,
-                    int batch_size, int q, int dc, int dcat,
+                    float* __restrict__ best_reward,
+                    unsigned long long* __restrict__ iter_ptr,
+                    int n_batches, int batch_size, int q, int dc, int dcat,
                     float penalize_factor, float perturbation_lower_bound,
-                    float base_perturbation, unsigned long long seed,
-                    unsigned long long offset) {
+                    float base_perturbation, unsigned long long seed) {
+  const unsigned long long offset = *iter_ptr;
+  const int batch_start = (int)(offset % (unsigned long long)n_batches) *
+                          batch_size;
   __shared__ float red[8];
   __shared__ float new_best_s;
   const int tid = threadIdx.x;
@@ -264,6 +273,9 @@ eagle_update_kernel(float* __restrict__ pool_cont,
     }
     __syncthreads();
   }
+  if (tid == 0) {
+    *iter_ptr = offset + 1;  // advance the device iteration counter
+  }
 }
 
 // -- launchers ---------------------------------------------------------------
@@ -271,31 +283,31 @@ eagle_update_kernel(float* __restrict__ pool_cont,
 extern "C" void launch_eagle_suggest(
     const float* pool_cont, const long* pool_cat, const float* rewards,
     const float* perturbations, const long* cat_sizes, float* out_cont,
-    long* out_cat, int batch_start, int batch_size, int pool_size, int q,
+    long* out_cat, const unsigned long long* iter_ptr, int n_batches,
+    int batch_size, int pool_size, int q,
     int dc, int dcat, int max_cat, float visibility, float gravity,
     float neg_gravity, float norm_scale, float cat_factor, float p_same,
-    unsigned long long seed, unsigned long long offset,
-    hipStream_t stream) {
+    unsigned long long seed, hipStream_t stream) {
   hipLaunchKernelGGL(eagle_suggest_kernel, dim3(batch_size), dim3(BLOCK),
                      0, stream, pool_cont, pool_cat, rewards,
                      perturbations, cat_sizes, out_cont, out_cat,
-                     batch_start, batch_size, pool_size, q, dc, dcat,
-                     max_cat, visibility, gravity, neg_gravity, norm_scale,
-                     cat_factor, p_same, seed, offset);
+                     iter_ptr, n_batches, batch_size, pool_size, q, dc,
+                     dcat, max_cat, visibility, gravity, neg_gravity,
+                     norm_scale, cat_factor, p_same, seed);
 }
 
 extern "C" void launch_eagle_update(
     float* pool_cont, long* pool_cat, float* rewards, float* perturbations,
     const float* batch_cont, const long* batch_cat,
     const float* batch_rewards, const long* cat_sizes, float* best_reward,
-    int batch_start, int batch_size, int q, int dc, int dcat,
+    unsigned long long* iter_ptr, int n_batches, int batch_size, int q,
+    int dc, int dcat,
     float penalize_factor, float perturbation_lower_bound,
-    float base_perturbation, unsigned long long seed,
-    unsigned long long offset, hipStream_t stream) {
+    float base_perturbation, unsigned long long seed, hipStream_t stream) {
   hipLaunchKernelGGL(eagle_update_kernel, dim3(1), dim3(BLOCK), 0, stream,
                      pool_cont, pool_cat, rewards, perturbations,
                      batch_cont, batch_cat, batch_rewards, cat_sizes,
-                     best_reward, batch_start, batch_size, q, dc, dcat,
-                     penalize_factor, perturbation_lower_bound,
-                     base_perturbation, seed, offset);
+                     best_reward, iter_ptr, n_batches, batch_size, q, dc,
+                     dcat, penalize_factor, perturbation_lower_bound,
+                     base_perturbation, seed);
 }

@@ -10,6 +10,11 @@ extern "C" void launch_gram_matern52(const float* x1, const float* x2,
                                      int n, int m, int d, float amp2,
                                      int sym, hipStream_t stream);
 
+extern "C" void launch_gram_matern52_bf16(
+    const unsigned short* z1, const unsigned short* z2, const float* n1,
+    const float* n2, float* out, int n, int m, int dp, float amp2,
+    hipStream_t stream);
+
 extern "C" void launch_posterior_score(
     const float* xq, const float* x, const float* inv_ls,
     const float* alpha, const float* kinv, const unsigned char* onehot,
@@ -19,20 +24,21 @@ extern "C" void launch_posterior_score(
 extern "C" void launch_eagle_suggest(
     const float* pool_cont, const long* pool_cat, const float* rewards,
     const float* perturbations, const long* cat_sizes, float* out_cont,
-    long* out_cat, int batch_start, int batch_size, int pool_size, int q,
+    long* out_cat, const unsigned long long* iter_ptr, int n_batches,
+    int batch_size, int pool_size, int q,
     int dc, int dcat, int max_cat, float visibility, float gravity,
     float neg_gravity, float norm_scale, float cat_factor, float p_same,
-    unsigned long long seed, unsigned long long offset,
-    hipStream_t stream);
+    unsigned long long seed, hipStream_t stream);
 
 extern "C" void launch_eagle_update(
     float* pool_cont, long* pool_cat, float* rewards, float* perturbations,
     const float* batch_cont, const long* batch_cat,
     const float* batch_rewards, const long* cat_sizes, float* best_reward,
-    int batch_start, int batch_size, int q, int dc, int dcat,
+    unsigned long long* iter_ptr, int n_batches, int batch_size, int q,
+    int dc, int dcat,
     float penalize_factor, float perturbation_lower_bound,
     float base_perturbation, unsigned long long seed,
-    unsigned long long offset, hipStream_t stream);
+    hipStream_t stream);
 
 namespace {
 
@@ -61,6 +67,37 @@ torch::Tensor gram_matern52(torch::Tensor x1, torch::Tensor x2,
                        inv_ls.data_ptr<float>(), out.data_ptr<float>(), n,
                        m, d, (float)(amplitude * amplitude), sym,
                        current_stream());
+  return out;
+}
+
+torch::Tensor gram_matern52_bf16(torch::Tensor x1, torch::Tensor x2,
+                                 torch::Tensor lengthscales,
+                                 double amplitude) {
+  x1 = check_f32(x1, "x1");
+  x2 = check_f32(x2, "x2");
+  lengthscales = check_f32(lengthscales, "lengthscales");
+  const int n = x1.size(0), m = x2.size(0), d = x1.size(1);
+  TORCH_CHECK(x2.size(1) == d && lengthscales.numel() == d,
+              "dimension mismatch");
+  auto z1 = x1 / lengthscales;
+  auto z2 = (x1.data_ptr() == x2.data_ptr()) ? z1 : x2 / lengthscales;
+  auto n1 = (z1 * z1).sum(-1);
+  auto n2 = (z2 * z2).sum(-1);
+  const int dp = (d + 31) / 32 * 32;
+  auto z1b = torch::zeros({n, dp},
+                          x1.options().dtype(torch::kBFloat16));
+  auto z2b = torch::zeros({m, dp},
+                          x1.options().dtype(torch::kBFloat16));
+  z1b.index_put_({torch::indexing::Slice(),
+                  torch::indexing::Slice(0, d)}, z1.to(torch::kBFloat16));
+  z2b.index_put_({torch::indexing::Slice(),
+                  torch::indexing::Slice(0, d)}, z2.to(torch::kBFloat16));
+  auto out = torch::empty({n, m}, x1.options());
+  launch_gram_matern52_bf16(
+      (const unsigned short*)z1b.data_ptr(),
+      (const unsigned short*)z2b.data_ptr(), n1.data_ptr<float>(),
+      n2.data_ptr<float>(), out.data_ptr<float>(), n, m, dp,
+      (float)(amplitude * amplitude), current_stream());
   return out;
 }
 
@@ -97,33 +134,34 @@ torch::Tensor posterior_scores(torch::Tensor xq, torch::Tensor x,
 std::vector<torch::Tensor> eagle_suggest(
     torch::Tensor pool_cont, torch::Tensor pool_cat, torch::Tensor rewards,
     torch::Tensor perturbations, torch::Tensor cat_sizes,
-    int64_t batch_start, int64_t batch_size, double visibility,
-    double gravity, double neg_gravity, double norm_scale,
-    double cat_factor, double p_same, int64_t seed, int64_t offset) {
+    torch::Tensor iter_counter, int64_t n_batches, int64_t batch_size,
+    double visibility, double gravity, double neg_gravity,
+    double norm_scale, double cat_factor, double p_same, int64_t seed,
+    torch::Tensor out_cont, torch::Tensor out_cat, int64_t max_cat) {
   pool_cont = check_f32(pool_cont, "pool_cont");
   rewards = check_f32(rewards, "rewards");
   perturbations = check_f32(perturbations, "perturbations");
+  TORCH_CHECK(iter_counter.scalar_type() == torch::kInt64 &&
+              iter_counter.is_cuda(), "iter_counter must be int64 cuda");
   pool_cat = pool_cat.contiguous();
   cat_sizes = cat_sizes.contiguous();
   const int pool_size = pool_cont.size(0);
   const int q = pool_cont.size(1);
   const int dc = pool_cont.size(2);
   const int dcat = pool_cat.size(2);
-  int max_cat = 0;
-  if (dcat > 0) max_cat = cat_sizes.max().item<int64_t>();
-  auto out_cont = torch::empty({batch_size, q, dc}, pool_cont.options());
-  auto out_cat = torch::empty({batch_size, q, dcat}, pool_cat.options());
   launch_eagle_suggest(
       pool_cont.data_ptr<float>(),
       dcat ? pool_cat.data_ptr<long>() : nullptr,
       rewards.data_ptr<float>(), perturbations.data_ptr<float>(),
       dcat ? cat_sizes.data_ptr<long>() : nullptr,
       out_cont.data_ptr<float>(),
-      dcat ? out_cat.data_ptr<long>() : nullptr, (int)batch_start,
-      (int)batch_size, pool_size, q, dc, dcat, max_cat, (float)visibility,
+      dcat ? out_cat.data_ptr<long>() : nullptr,
+      (const unsigned long long*)iter_counter.data_ptr<int64_t>(),
+      (int)n_batches, (int)batch_size, pool_size, q, dc, dcat,
+      (int)max_cat, (float)visibility,
       (float)gravity, (float)neg_gravity, (float)norm_scale,
       (float)cat_factor, (float)p_same, (unsigned long long)seed,
-      (unsigned long long)offset, current_stream());
+      current_stream());
   return {out_cont, out_cat};
 }
 
@@ -131,9 +169,10 @@ void eagle_update(torch::Tensor pool_cont, torch::Tensor pool_cat,
                   torch::Tensor rewards, torch::Tensor perturbations,
                   torch::Tensor batch_cont, torch::Tensor batch_cat,
                   torch::Tensor batch_rewards, torch::Tensor cat_sizes,
-                  torch::Tensor best_reward, int64_t batch_start,
+                  torch::Tensor best_reward, torch::Tensor iter_counter,
+                  int64_t n_batches,
                   double penalize_factor, double perturbation_lower_bound,
-                  double base_perturbation, int64_t seed, int64_t offset) {
+                  double base_perturbation, int64_t seed) {
   const int batch_size = batch_cont.size(0);
   const int q = batch_cont.size(1);
   const int dc = batch_cont.size(2);
@@ -146,10 +185,12 @@ void eagle_update(torch::Tensor pool_cont, torch::Tensor pool_cat,
       dcat ? batch_cat.data_ptr<long>() : nullptr,
       batch_rewards.data_ptr<float>(),
       dcat ? cat_sizes.data_ptr<long>() : nullptr,
-      best_reward.data_ptr<float>(), (int)batch_start, batch_size, q, dc,
+      best_reward.data_ptr<float>(),
+      (unsigned long long*)iter_counter.data_ptr<int64_t>(),
+      (int)n_batches, batch_size, q, dc,
       dcat, (float)penalize_factor, (float)perturbation_lower_bound,
       (float)base_perturbation, (unsigned long long)seed,
-      (unsigned long long)offset, current_stream());
+      current_stream());
 }
 
 }  // namespace
@@ -157,6 +198,8 @@ void eagle_update(torch::Tensor pool_cont, torch::Tensor pool_cat,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gram_matern52", &gram_matern52,
         "Fused Matern-5/2 ARD Gram matrix (gfx950)");
+  m.def("gram_matern52_bf16", &gram_matern52_bf16,
+        "bf16 MFMA Matern-5/2 Gram matrix (gfx950 matrix cores)");
   m.def("posterior_scores", &posterior_scores,
         "Fused GP posterior + acquisition + trust region (gfx950)");
   m.def("eagle_suggest", &eagle_suggest,
