@@ -52,12 +52,18 @@ class VectorizedOptimizer:
     batch_size = strategy.batch_size
     iterations = max(1, (self.max_evaluations - 1) // batch_size + 1)
 
+    self.last_used_graph = False
+    self.last_graph_error = None
     if strategy._ext is not None:
       try:
-        return self._optimize_hipgraph(score_fn, count, state, iterations)
-      except Exception:
+        result = self._optimize_hipgraph(score_fn, count, state,
+                                         iterations)
+        self.last_used_graph = True
+        return result
+      except Exception as e:
         # Score function not graph-capturable (e.g. host RNG inside
         # q-acquisitions): fall back to the eager loop from scratch.
+        self.last_graph_error = repr(e)
         state = strategy.init_state(prior_features, prior_rewards)
 
     # Accumulate every evaluated candidate; one top-k at the end. At the

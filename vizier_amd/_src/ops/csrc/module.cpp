@@ -81,8 +81,6 @@ torch::Tensor gram_matern52_bf16(torch::Tensor x1, torch::Tensor x2,
               "dimension mismatch");
   auto z1 = x1 / lengthscales;
   auto z2 = (x1.data_ptr() == x2.data_ptr()) ? z1 : x2 / lengthscales;
-  auto n1 = (z1 * z1).sum(-1);
-  auto n2 = (z2 * z2).sum(-1);
   const int dp = (d + 31) / 32 * 32;
   auto z1b = torch::zeros({n, dp},
                           x1.options().dtype(torch::kBFloat16));
@@ -92,6 +90,11 @@ torch::Tensor gram_matern52_bf16(torch::Tensor x1, torch::Tensor x2,
                   torch::indexing::Slice(0, d)}, z1.to(torch::kBFloat16));
   z2b.index_put_({torch::indexing::Slice(),
                   torch::indexing::Slice(0, d)}, z2.to(torch::kBFloat16));
+  // Norms of the bf16-ROUNDED vectors: bf16 x bf16 products accumulate
+  // exactly in fp32, so d^2 = n1 + n2 - 2 z1b.z2b is the exact squared
+  // distance of the rounded inputs (diagonal exactly 0).
+  auto n1 = (z1b.to(torch::kFloat32) * z1b.to(torch::kFloat32)).sum(-1);
+  auto n2 = (z2b.to(torch::kFloat32) * z2b.to(torch::kFloat32)).sum(-1);
   auto out = torch::empty({n, m}, x1.options());
   launch_gram_matern52_bf16(
       (const unsigned short*)z1b.data_ptr(),
