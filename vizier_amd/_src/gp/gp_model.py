@@ -193,7 +193,8 @@ def train_gp(x: torch.Tensor, y: torch.Tensor, *,
     return negative_log_marginal_likelihood(raw, x, y)
 
   best_raw, best_f = lbfgs.minimize_batched(loss_fn, raw0,
-                                            max_iters=max_iters)
+                                            max_iters=max_iters,
+                                            check_every=5)
   idx = int(torch.argmin(best_f))
   raw = best_raw[idx]
   params = GPParams.from_raw(raw)
