@@ -262,23 +262,27 @@ class ScoringFunction:
       self._tr_anchored = True
 
   def _can_fuse(self, xs: torch.Tensor) -> bool:
-    return (xs.is_cuda and self._acq_name is not None and
-            self.posterior.K_inv is not None and self._tr_anchored)
+    """True when the GPU K^-1 quadform path applies (any acquisition)."""
+    return xs.is_cuda and self.posterior.K_inv is not None
 
   def __call__(self, xs: torch.Tensor) -> torch.Tensor:
     if self._can_fuse(xs):
+      # GPU composed path: hand-written HIP gram kernel for the
+      # k-vectors + rocBLAS for the plain K^-1 quadform GEMM. Measured
+      # 2.8x faster than the single fused kernel at B=25/N=1000
+      # (profiles/scorebench) because the GEMM fills the chip; all ops
+      # are hipGraph-capturable so launch count does not matter.
       from vizier_amd._src.ops import dispatch as ops
-      onehot = self._onehot_u8
-      if onehot is None:
-        onehot = torch.zeros(xs.shape[-1], dtype=torch.uint8,
-                             device=xs.device)
-        self._onehot_u8 = onehot
-      return ops.fused_posterior_scores(
-          xs, self.posterior.x, self.posterior.params.lengthscales,
-          self._amp, self._mean_c, self.posterior.alpha,
-          self.posterior.K_inv, onehot, self._acq_name, self._coef,
-          self._best, self._tr_radius if self.trust_region is not None
-          else 0.0)
+      post = self.posterior
+      k = ops.gram_matern52(xs, post.x, post.params.lengthscales,
+                            post.params.amplitude)
+      amp2 = self._amp * self._amp
+      var = (amp2 - (k * (k @ post.K_inv)).sum(-1)).clamp_min(1e-12)
+      mean = self._mean_c + k @ post.alpha
+      scores = self.acquisition(mean, var.sqrt())
+      if self.trust_region is not None:
+        scores = self.trust_region.apply(xs, scores)
+      return scores
     mean, stddev = self.posterior.predict(xs)
     scores = self.acquisition(mean, stddev)
     if self.trust_region is not None:
