@@ -40,3 +40,17 @@ a, b = fused(), composed()
 # fused includes trust region; composed here does not -> compare masked
 print('max |fused-composed| (TR off):',
       float((fused().cpu() - composed().cpu()).abs().max()))
+
+def chunked():
+    return ext.posterior_scores_chunked(xq, post.x, ls, amp, mean_c,
+                                        alpha, kinv, onehot, 0, 1.8, 0.0,
+                                        0.0)
+
+for _ in range(20): chunked()
+torch.cuda.synchronize(); t0 = time.perf_counter()
+for _ in range(200): chunked()
+torch.cuda.synchronize()
+print(f'chunked: {(time.perf_counter()-t0)/200*1e6:.1f} us/call',
+      flush=True)
+print('max |chunked-composed|:',
+      float((chunked().cpu() - composed().cpu()).abs().max()))

@@ -267,13 +267,26 @@ class ScoringFunction:
 
   def __call__(self, xs: torch.Tensor) -> torch.Tensor:
     if self._can_fuse(xs):
-      # GPU composed path: hand-written HIP gram kernel for the
-      # k-vectors + rocBLAS for the plain K^-1 quadform GEMM. Measured
-      # 2.8x faster than the single fused kernel at B=25/N=1000
-      # (profiles/scorebench) because the GEMM fills the chip; all ops
-      # are hipGraph-capturable so launch count does not matter.
-      from vizier_amd._src.ops import dispatch as ops
       post = self.posterior
+      from vizier_amd._src.ops import dispatch as ops
+      if self._acq_name is not None and self._tr_anchored:
+        # Primary GPU path: the 3-kernel chunked HIP scorer — one
+        # Python op; k-vectors + chip-filling K^-1 quadform chunks +
+        # acquisition/trust-region finalize (posterior_score.hip).
+        onehot = self._onehot_u8
+        if onehot is None:
+          onehot = torch.zeros(xs.shape[-1], dtype=torch.uint8,
+                               device=xs.device)
+          self._onehot_u8 = onehot
+        ext = ops.require_ext()
+        return ext.posterior_scores_chunked(
+            xs, post.x, post.params.lengthscales, self._amp,
+            self._mean_c, post.alpha, post.K_inv, onehot,
+            ops.ACQ_CODES[self._acq_name], self._coef, self._best,
+            self._tr_radius if self.trust_region is not None else 0.0)
+      # Composed path: hand-written HIP gram kernel for the k-vectors +
+      # rocBLAS for the plain K^-1 quadform GEMM (used for exotic
+      # acquisitions / unanchored trust regions).
       k = ops.gram_matern52(xs, post.x, post.params.lengthscales,
                             post.params.amplitude)
       amp2 = self._amp * self._amp

@@ -21,6 +21,13 @@ extern "C" void launch_posterior_score(
     float* out, int b, int n, int d, float amp2, float mean_c, int acq,
     float coef, float best_value, float tr_radius, hipStream_t stream);
 
+extern "C" void launch_posterior_score_chunked(
+    const float* xq, const float* x, const float* inv_ls,
+    const float* alpha, const float* kinv, const unsigned char* onehot,
+    float* k_ws, float* mu_ws, float* dist_ws, float* var_ws, float* out,
+    int b, int n, int d, float amp2, float mean_c, int acq, float coef,
+    float best_value, float tr_radius, hipStream_t stream);
+
 extern "C" void launch_eagle_suggest(
     const float* pool_cont, const long* pool_cat, const float* rewards,
     const float* perturbations, const long* cat_sizes, float* out_cont,
@@ -134,6 +141,37 @@ torch::Tensor posterior_scores(torch::Tensor xq, torch::Tensor x,
   return out;
 }
 
+torch::Tensor posterior_scores_chunked(
+    torch::Tensor xq, torch::Tensor x, torch::Tensor lengthscales,
+    double amplitude, double mean_c, torch::Tensor alpha,
+    torch::Tensor kinv, torch::Tensor onehot, int64_t acq, double coef,
+    double best_value, double tr_radius) {
+  xq = check_f32(xq, "xq");
+  x = check_f32(x, "x");
+  lengthscales = check_f32(lengthscales, "lengthscales");
+  alpha = check_f32(alpha, "alpha");
+  kinv = check_f32(kinv, "kinv");
+  onehot = onehot.contiguous();
+  const int b = xq.size(0), d = xq.size(1), n = x.size(0);
+  TORCH_CHECK(d <= 512, "posterior_scores supports D <= 512");
+  auto inv_ls = 1.0f / lengthscales;
+  auto k_ws = torch::empty({b, n}, xq.options());
+  auto mu_ws = torch::empty({b}, xq.options());
+  auto dist_ws = torch::empty({b}, xq.options());
+  auto var_ws = torch::empty({b, 10}, xq.options());
+  auto out = torch::empty({b}, xq.options());
+  launch_posterior_score_chunked(
+      xq.data_ptr<float>(), x.data_ptr<float>(), inv_ls.data_ptr<float>(),
+      alpha.data_ptr<float>(), kinv.data_ptr<float>(),
+      onehot.data_ptr<unsigned char>(), k_ws.data_ptr<float>(),
+      mu_ws.data_ptr<float>(), dist_ws.data_ptr<float>(),
+      var_ws.data_ptr<float>(), out.data_ptr<float>(), b, n, d,
+      (float)(amplitude * amplitude), (float)mean_c, (int)acq,
+      (float)coef, (float)best_value, (float)tr_radius,
+      current_stream());
+  return out;
+}
+
 std::vector<torch::Tensor> eagle_suggest(
     torch::Tensor pool_cont, torch::Tensor pool_cat, torch::Tensor rewards,
     torch::Tensor perturbations, torch::Tensor cat_sizes,
@@ -205,6 +243,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "bf16 MFMA Matern-5/2 Gram matrix (gfx950 matrix cores)");
   m.def("posterior_scores", &posterior_scores,
         "Fused GP posterior + acquisition + trust region (gfx950)");
+  m.def("posterior_scores_chunked", &posterior_scores_chunked,
+        "3-kernel chunked GP posterior scorer (chip-filling, gfx950)");
   m.def("eagle_suggest", &eagle_suggest,
         "Fused Eagle suggest step (gfx950)");
   m.def("eagle_update", &eagle_update, "Fused Eagle update step (gfx950)");

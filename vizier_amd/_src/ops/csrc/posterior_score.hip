@@ -148,3 +148,143 @@ extern "C" void launch_posterior_score(
                      xq, x, inv_ls, alpha, kinv, onehot, out, b, n, d,
                      amp2, mean_c, acq, coef, best_value, tr_radius);
 }
+
+// -- Chunked 3-kernel variant -------------------------------------------
+//
+// The single-workgroup-per-candidate kernel above underfills the chip at
+// small batch sizes (B=25 -> 25 of 256 CUs). This variant splits the
+// K^-1 quadform over NCHUNK j-ranges so B x NCHUNK workgroups run the
+// heavy phase; the three kernels are launched back-to-back from one
+// binding call (still a single Python-level op, hipGraph-capturable).
+
+#define NCHUNK 10
+
+extern "C" __global__ __launch_bounds__(BLOCK) void
+ps_kvec_kernel(const float* __restrict__ xq, const float* __restrict__ x,
+               const float* __restrict__ inv_ls,
+               const float* __restrict__ alpha,
+               const unsigned char* __restrict__ onehot,
+               float* __restrict__ k_out,    // (B, N)
+               float* __restrict__ mu_out,   // (B,)
+               float* __restrict__ dist_out, // (B,)
+               int b, int n, int d, float amp2) {
+  __shared__ float red[8];
+  __shared__ float xq_lds[512];
+  const int q = blockIdx.x;
+  if (q >= b) return;
+  const int tid = threadIdx.x;
+  for (int j = tid; j < d; j += BLOCK) xq_lds[j] = xq[q * d + j];
+  __syncthreads();
+  float mu_acc = 0.0f;
+  float min_linf = INFINITY;
+  for (int row = tid; row < n; row += BLOCK) {
+    const float* xr = x + (long)row * d;
+    float d2 = 0.0f, linf = 0.0f;
+    for (int j = 0; j < d; ++j) {
+      const float diff = xq_lds[j] - xr[j];
+      const float z = diff * inv_ls[j];
+      d2 = fmaf(z, z, d2);
+      if (!onehot[j]) linf = fmaxf(linf, fabsf(diff));
+    }
+    const float kv = amp2 * matern52_of_d2(d2);
+    k_out[(long)q * n + row] = kv;
+    mu_acc = fmaf(kv, alpha[row], mu_acc);
+    min_linf = fminf(min_linf, linf);
+  }
+  auto fsum = [](float a, float c) { return a + c; };
+  auto fmin_ = [](float a, float c) { return fminf(a, c); };
+  float mu = block_reduce(mu_acc, red, fsum, 0.0f);
+  if (tid == 0) mu_out[q] = mu;
+  __syncthreads();
+  float dist = block_reduce(min_linf, red, fmin_, INFINITY);
+  if (tid == 0) dist_out[q] = dist;
+}
+
+extern "C" __global__ __launch_bounds__(BLOCK) void
+ps_quadform_kernel(const float* __restrict__ k_in,   // (B, N)
+                   const float* __restrict__ kinv,   // (N, N)
+                   float* __restrict__ var_part,     // (B, NCHUNK)
+                   int b, int n) {
+  __shared__ float red[8];
+  const int q = blockIdx.x;
+  const int chunk = blockIdx.y;
+  const int tid = threadIdx.x;
+  const int j0 = (int)((long)chunk * n / NCHUNK);
+  const int j1 = (int)((long)(chunk + 1) * n / NCHUNK);
+  const float* k = k_in + (long)q * n;
+  float acc = 0.0f;
+  for (int j = j0 + tid; j < j1; j += BLOCK) {
+    const float* row = kinv + (long)j * n;
+    float t_j = 0.0f;
+    int i = 0;
+    for (; i + 4 <= n; i += 4) {
+      t_j = fmaf(row[i], k[i], t_j);
+      t_j = fmaf(row[i + 1], k[i + 1], t_j);
+      t_j = fmaf(row[i + 2], k[i + 2], t_j);
+      t_j = fmaf(row[i + 3], k[i + 3], t_j);
+    }
+    for (; i < n; ++i) t_j = fmaf(row[i], k[i], t_j);
+    acc = fmaf(k[j], t_j, acc);
+  }
+  auto fsum = [](float a, float c) { return a + c; };
+  float v = block_reduce(acc, red, fsum, 0.0f);
+  if (tid == 0) var_part[q * NCHUNK + chunk] = v;
+}
+
+extern "C" __global__ void
+ps_finalize_kernel(const float* __restrict__ mu_in,
+                   const float* __restrict__ dist_in,
+                   const float* __restrict__ var_part,
+                   float* __restrict__ out, int b, float amp2,
+                   float mean_c, int acq, float coef, float best_value,
+                   float tr_radius) {
+  const int q = blockIdx.x * blockDim.x + threadIdx.x;
+  if (q >= b) return;
+  float var = 0.0f;
+  for (int c = 0; c < NCHUNK; ++c) var += var_part[q * NCHUNK + c];
+  var = fmaxf(amp2 - var, 1e-12f);
+  const float sd = sqrtf(var);
+  const float mu = mu_in[q] + mean_c;
+  float score;
+  switch (acq) {
+    case ACQ_LCB: score = mu - coef * sd; break;
+    case ACQ_EI: {
+      const float z = (mu - best_value) / sd;
+      score = sd * (z * normal_cdf(z) + normal_pdf(z));
+      break;
+    }
+    case ACQ_PI: {
+      const float z = (mu - best_value) / sd;
+      score = normal_cdf(z);
+      break;
+    }
+    case ACQ_MEAN: score = mu; break;
+    case ACQ_STDDEV: score = sd; break;
+    case ACQ_UCB:
+    default: score = mu + coef * sd; break;
+  }
+  const float dist = dist_in[q];
+  if (tr_radius > 0.0f && tr_radius <= 0.5f && dist > tr_radius) {
+    score = -1e4f - dist;
+  }
+  out[q] = score;
+}
+
+extern "C" void launch_posterior_score_chunked(
+    const float* xq, const float* x, const float* inv_ls,
+    const float* alpha, const float* kinv, const unsigned char* onehot,
+    float* k_ws, float* mu_ws, float* dist_ws, float* var_ws, float* out,
+    int b, int n, int d, float amp2, float mean_c, int acq, float coef,
+    float best_value, float tr_radius, hipStream_t stream) {
+  hipLaunchKernelGGL(ps_kvec_kernel, dim3(b), dim3(BLOCK), 0, stream, xq,
+                     x, inv_ls, alpha, onehot, k_ws, mu_ws, dist_ws, b, n,
+                     d, amp2);
+  hipLaunchKernelGGL(ps_quadform_kernel, dim3(b, NCHUNK), dim3(BLOCK), 0,
+                     stream, k_ws, kinv, var_ws, b, n);
+  const int fin_block = 256;
+  hipLaunchKernelGGL(ps_finalize_kernel,
+                     dim3((b + fin_block - 1) / fin_block),
+                     dim3(fin_block), 0, stream, mu_ws, dist_ws, var_ws,
+                     out, b, amp2, mean_c, acq, coef, best_value,
+                     tr_radius);
+}
