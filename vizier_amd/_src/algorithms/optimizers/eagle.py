@@ -188,31 +188,44 @@ class VectorizedEagleStrategy:
   def _seed_with_priors(self, pool: CandidateBatch,
                         prior: CandidateBatch,
                         prior_rewards: torch.Tensor) -> CandidateBatch:
-    """Fills most of the pool with (recent, good) prior trials."""
+    """Fills most of the pool with (recent, good) prior trials.
+
+    The sequential closest-replacement loop (reference
+    eagle_strategy.py:652-690) runs on host NumPy: per step it is a tiny
+    (pool x D) distance computation, and doing it on-device would cost a
+    host sync per prior trial (~0.4s per suggest at N=1000 priors).
+    """
+    import numpy as np
     # Most recent first (reference flips the ordering).
-    order = torch.arange(prior_rewards.numel() - 1, -1, -1,
-                         device=self.device)
-    prior = prior.index(order)
-    prior_rewards = prior_rewards[order]
+    cont = prior.continuous.flip(0).cpu().numpy()
+    cat = prior.categorical.flip(0).cpu().numpy()
+    rewards = prior_rewards.flip(0).cpu().numpy()
     n_random = int(self.pool_size * (1 - self.config.prior_trials_pool_pct))
     space = self.pool_size - n_random
 
-    chosen = prior.index(slice(0, space))
-    chosen_rewards = prior_rewards[:space].clone()
-    # Remaining prior trials replace their nearest chosen neighbor if better.
-    for i in range(space, prior_rewards.numel()):
-      one = prior.index(slice(i, i + 1))
-      ind = int(torch.argmin(self._dist2(one, chosen)[0]))
-      if float(chosen_rewards[ind]) < float(prior_rewards[i]):
-        chosen.continuous[ind] = one.continuous[0]
-        chosen.categorical[ind] = one.categorical[0]
-        chosen_rewards[ind] = prior_rewards[i]
+    chosen_cont = cont[:space].reshape(min(space, len(rewards)), -1).copy()
+    chosen_cat = cat[:space].copy()
+    chosen_rewards = rewards[:space].copy()
+    for i in range(space, len(rewards)):
+      flat = cont[i].reshape(-1)
+      d2 = ((chosen_cont - flat) ** 2).sum(axis=1)
+      if self.n_categorical:
+        d2 = d2 + (chosen_cat.reshape(chosen_cat.shape[0], -1) !=
+                   cat[i].reshape(-1)).sum(axis=1)
+      ind = int(np.argmin(d2))
+      if chosen_rewards[ind] < rewards[i]:
+        chosen_cont[ind] = flat
+        chosen_cat[ind] = cat[i]
+        chosen_rewards[ind] = rewards[i]
 
-    n_chosen = chosen.continuous.shape[0]
+    n_chosen = chosen_cont.shape[0]
     out_cont = pool.continuous.clone()
     out_cat = pool.categorical.clone()
-    out_cont[n_random:n_random + n_chosen] = chosen.continuous
-    out_cat[n_random:n_random + n_chosen] = chosen.categorical
+    out_cont[n_random:n_random + n_chosen] = torch.as_tensor(
+        chosen_cont.reshape(n_chosen, self.n_parallel, self.n_continuous),
+        dtype=self.dtype, device=self.device)
+    out_cat[n_random:n_random + n_chosen] = torch.as_tensor(
+        chosen_cat, dtype=torch.long, device=self.device)
     return CandidateBatch(out_cont, out_cat)
 
   # -- suggest --------------------------------------------------------------

@@ -63,6 +63,9 @@ class VectorizedOptimizer:
       except Exception as e:
         # Score function not graph-capturable (e.g. host RNG inside
         # q-acquisitions): fall back to the eager loop from scratch.
+        import logging
+        logging.getLogger(__name__).warning(
+            'hipGraph sweep fell back to eager: %r', e)
         self.last_graph_error = repr(e)
         state = strategy.init_state(prior_features, prior_rewards)
 
