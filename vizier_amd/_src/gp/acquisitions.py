@@ -256,7 +256,10 @@ class ScoringFunction:
     if trust_region is not None:
       self._onehot_u8 = trust_region._onehot.to(torch.uint8)
       self._tr_radius = float(trust_region.trust_radius)
-      self._tr_anchored = trust_region._trusted is posterior.x
+      trusted = trust_region._trusted
+      self._tr_anchored = (
+          trusted.shape == posterior.x.shape and
+          trusted.data_ptr() == posterior.x.data_ptr())
     else:
       self._tr_radius = 0.0
       self._tr_anchored = True
