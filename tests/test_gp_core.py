@@ -178,3 +178,54 @@ class TestOutputWarpers:
     labels = np.array([[1.0], [2.0], [3.0]])
     out = output_warpers.ZScoreLabels().warp(labels)
     assert abs(out.mean()) < 1e-9
+
+
+class TestTransferLearning:
+
+  def test_stacked_residual_gp_improves_with_prior(self):
+    import torch
+    from vizier_amd._src.gp import transfer_learning
+
+    def f(x):
+      return torch.sin(3 * x[:, 0]) + 0.5 * x[:, 1]
+
+    g = torch.Generator().manual_seed(0)
+    x_prior = torch.rand(60, 2, generator=g)
+    y_prior = f(x_prior)
+    x_cur = torch.rand(8, 2, generator=g)
+    y_cur = f(x_cur)
+    stacked = transfer_learning.train_stacked_gp(
+        [(x_prior, y_prior), (x_cur, y_cur)], num_restarts=2,
+        max_iters=20, seed=0)
+    from vizier_amd._src.gp import gp_model
+    single = gp_model.train_gp(x_cur, y_cur, num_restarts=2,
+                               max_iters=20, seed=0)
+    x_test = torch.rand(64, 2, generator=g)
+    y_test = f(x_test)
+    m_stacked, _ = stacked.predict(x_test)
+    m_single, _ = single.predict(x_test)
+    err_stacked = float((m_stacked - y_test).abs().mean())
+    err_single = float((m_single - y_test).abs().mean())
+    assert err_stacked < err_single
+
+  def test_combine_predictions_inflates_base_uncertainty(self):
+    import torch
+    from vizier_amd._src.gp.transfer_learning import combine_predictions
+    out = combine_predictions(
+        torch.zeros(3), torch.full((3,), 0.1), torch.ones(3),
+        torch.full((3,), 1.0), num_obs_base=10, num_obs_top=10)
+    assert torch.allclose(out.mean, torch.ones(3))
+    assert (out.stddev > 0.1).all()
+
+
+class TestMES:
+
+  def test_mes_prefers_uncertainty_near_incumbent(self):
+    import torch
+    from vizier_amd._src.gp import acquisitions as acq_lib
+    mes = acq_lib.MaxValueEntropySearch(best_value=1.0,
+                                        max_value_spread=1.0, seed=0)
+    mean = torch.tensor([0.9, 0.9])
+    stddev = torch.tensor([0.01, 0.5])
+    vals = mes(mean, stddev)
+    assert float(vals[1]) > float(vals[0])

@@ -243,3 +243,27 @@ class TestGramBf16MFMA:
     K = ext.gram_matern52_bf16(x, x, ls, 2.0)
     assert torch.allclose(torch.diagonal(K),
                           torch.full((100,), 4.0).cuda(), atol=0.05)
+
+
+class TestGramFp8MFMA:
+
+  def test_fp8_mfma_matches_fp64_loosely(self, ext):
+    from vizier_amd._src.gp.matern import gram_matern52
+    g = torch.Generator().manual_seed(11)
+    x1 = torch.rand(128, 30, generator=g).cuda()
+    x2 = torch.rand(96, 30, generator=g).cuda()
+    ls = (torch.rand(30, generator=g) * 1.5 + 0.3).cuda()
+    amp = 1.0
+    got = ext.gram_matern52_fp8(x1, x2, ls, amp)
+    want = gram_matern52(x1.cpu().double(), x2.cpu().double(),
+                         ls.cpu().double(), torch.tensor(amp).double())
+    err = (got.cpu().double() - want).abs().max()
+    # e4m3 has ~2 significant digits; the Gram is O(1).
+    assert float(err) < 0.15, f'max err {err}'
+
+  def test_fp8_diag_exact(self, ext):
+    x = torch.rand(64, 16).cuda()
+    ls = torch.full((16,), 0.5).cuda()
+    K = ext.gram_matern52_fp8(x, x, ls, 2.0)
+    assert torch.allclose(torch.diagonal(K),
+                          torch.full((64,), 4.0).cuda(), atol=1e-3)

@@ -78,6 +78,36 @@ class Sample(AcquisitionFunction):
     return mean + stddev * eps
 
 
+class MaxValueEntropySearch(AcquisitionFunction):
+  """MES (acquisitions.py:293), Gumbel-sampled max-value approximation.
+
+  score(x) = mean over sampled maxima y* of
+      gamma * pdf(gamma) / (2 cdf(gamma)) - log cdf(gamma),
+  gamma = (y* - mu) / sigma (Wang & Jegelka 2017).
+  """
+
+  def __init__(self, best_value: float = 0.0, num_max_samples: int = 16,
+               max_value_spread: float = 1.0, seed: int = 0):
+    self.best_value = best_value
+    self.num_max_samples = num_max_samples
+    self.max_value_spread = max_value_spread
+    self.seed = seed
+
+  def __call__(self, mean, stddev):
+    g = torch.Generator(device='cpu').manual_seed(self.seed)
+    # Gumbel samples of the global max, anchored above the incumbent.
+    u = torch.rand(self.num_max_samples, generator=g).clamp(1e-6,
+                                                            1 - 1e-6)
+    gumbel = -torch.log(-torch.log(u)) * 0.25 * self.max_value_spread
+    y_star = (self.best_value + 0.1 * self.max_value_spread +
+              gumbel).to(mean.device, mean.dtype)
+    n = _normal()
+    gamma = (y_star.reshape(-1, 1) - mean.unsqueeze(0)) / stddev
+    cdf = n.cdf(gamma).clamp_min(1e-9)
+    pdf = n.log_prob(gamma).exp()
+    return (gamma * pdf / (2.0 * cdf) - torch.log(cdf)).mean(0)
+
+
 # -- q-family: joint acquisition of a parallel suggestion batch -------------
 
 

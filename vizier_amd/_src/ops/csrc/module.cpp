@@ -21,6 +21,11 @@ extern "C" void launch_posterior_score(
     float* out, int b, int n, int d, float amp2, float mean_c, int acq,
     float coef, float best_value, float tr_radius, hipStream_t stream);
 
+extern "C" void launch_gram_matern52_fp8(
+    const unsigned char* z1, const unsigned char* z2, const float* n1,
+    const float* n2, float* out, int n, int m, int dp, float amp2,
+    float scale2, hipStream_t stream);
+
 extern "C" void launch_posterior_score_chunked(
     const float* xq, const float* x, const float* inv_ls,
     const float* alpha, const float* kinv, const unsigned char* onehot,
@@ -141,6 +146,42 @@ torch::Tensor posterior_scores(torch::Tensor xq, torch::Tensor x,
   return out;
 }
 
+torch::Tensor gram_matern52_fp8(torch::Tensor x1, torch::Tensor x2,
+                                torch::Tensor lengthscales,
+                                double amplitude) {
+  x1 = check_f32(x1, "x1");
+  x2 = check_f32(x2, "x2");
+  lengthscales = check_f32(lengthscales, "lengthscales");
+  const int n = x1.size(0), m = x2.size(0), d = x1.size(1);
+  auto z1 = x1 / lengthscales;
+  auto z2 = (x1.data_ptr() == x2.data_ptr()) ? z1 : x2 / lengthscales;
+  // Pre-scale into comfortable e4m3 range (|z|/s <= ~8).
+  double s = std::max(z1.abs().max().item<double>(),
+                      z2.abs().max().item<double>()) / 8.0;
+  s = std::max(s, 1e-8);
+  const int dp = (d + 31) / 32 * 32;
+  auto opts8 = x1.options().dtype(torch::kFloat8_e4m3fn);
+  auto z1b = torch::zeros({n, dp}, opts8);
+  auto z2b = torch::zeros({m, dp}, opts8);
+  z1b.index_put_({torch::indexing::Slice(),
+                  torch::indexing::Slice(0, d)},
+                 (z1 / s).to(torch::kFloat8_e4m3fn));
+  z2b.index_put_({torch::indexing::Slice(),
+                  torch::indexing::Slice(0, d)},
+                 (z2 / s).to(torch::kFloat8_e4m3fn));
+  auto z1f = z1b.to(torch::kFloat32) * s;
+  auto z2f = z2b.to(torch::kFloat32) * s;
+  auto n1 = (z1f * z1f).sum(-1);
+  auto n2 = (z2f * z2f).sum(-1);
+  auto out = torch::empty({n, m}, x1.options());
+  launch_gram_matern52_fp8(
+      (const unsigned char*)z1b.data_ptr(),
+      (const unsigned char*)z2b.data_ptr(), n1.data_ptr<float>(),
+      n2.data_ptr<float>(), out.data_ptr<float>(), n, m, dp,
+      (float)(amplitude * amplitude), (float)(s * s), current_stream());
+  return out;
+}
+
 torch::Tensor posterior_scores_chunked(
     torch::Tensor xq, torch::Tensor x, torch::Tensor lengthscales,
     double amplitude, double mean_c, torch::Tensor alpha,
@@ -241,6 +282,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Fused Matern-5/2 ARD Gram matrix (gfx950)");
   m.def("gram_matern52_bf16", &gram_matern52_bf16,
         "bf16 MFMA Matern-5/2 Gram matrix (gfx950 matrix cores)");
+  m.def("gram_matern52_fp8", &gram_matern52_fp8,
+        "fp8 e4m3 MFMA Matern-5/2 Gram matrix (gfx950, config 5)");
   m.def("posterior_scores", &posterior_scores,
         "Fused GP posterior + acquisition + trust region (gfx950)");
   m.def("posterior_scores_chunked", &posterior_scores_chunked,
