@@ -98,8 +98,10 @@ class MaxValueEntropySearch(AcquisitionFunction):
     # Gumbel samples of the global max, anchored above the incumbent.
     u = torch.rand(self.num_max_samples, generator=g).clamp(1e-6,
                                                             1 - 1e-6)
-    gumbel = -torch.log(-torch.log(u)) * 0.25 * self.max_value_spread
-    y_star = (self.best_value + 0.1 * self.max_value_spread +
+    # The global max is at least the incumbent: clamp the Gumbel tail.
+    gumbel = (-torch.log(-torch.log(u))).clamp_min(0.0) * 0.25 * \
+        self.max_value_spread
+    y_star = (self.best_value + 0.05 * self.max_value_spread +
               gumbel).to(mean.device, mean.dtype)
     n = _normal()
     gamma = (y_star.reshape(-1, 1) - mean.unsqueeze(0)) / stddev
