@@ -64,8 +64,10 @@ class VectorizedOptimizer:
         # Score function not graph-capturable (e.g. host RNG inside
         # q-acquisitions): fall back to the eager loop from scratch.
         import logging
+        import traceback
         logging.getLogger(__name__).warning(
-            'hipGraph sweep fell back to eager: %r', e)
+            'hipGraph sweep fell back to eager: %r\n%s', e,
+            traceback.format_exc())
         self.last_graph_error = repr(e)
         state = strategy.init_state(prior_features, prior_rewards)
 

@@ -25,7 +25,7 @@ def minimize_batched(
     max_iters: int = 50,
     history: int = 10,
     grad_tol: float = 1e-7,
-    ls_steps: Tuple[float, ...] = (1.0, 0.3, 0.1, 0.03, 0.01),
+    ls_steps: Tuple[float, ...] = (1.0, 0.3, 0.08, 0.02),
     check_every: int = 10,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
   """Minimizes loss_fn over a batch of R independent parameter vectors.

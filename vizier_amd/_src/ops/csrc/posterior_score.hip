@@ -205,27 +205,42 @@ ps_quadform_kernel(const float* __restrict__ k_in,   // (B, N)
                    const float* __restrict__ kinv,   // (N, N)
                    float* __restrict__ var_part,     // (B, NCHUNK)
                    int b, int n) {
+  // Wave-per-row: all 64 lanes of a wave stream ONE Kinv row with
+  // coalesced float4 loads and shuffle-reduce the dot with k; a
+  // per-thread row walk here was 64 divergent streams (measured 65us,
+  // ~60 GB/s effective — see profiles/bench_kernel_stats.csv).
   __shared__ float red[8];
   const int q = blockIdx.x;
   const int chunk = blockIdx.y;
   const int tid = threadIdx.x;
+  const int wave = tid / WAVE_SIZE;
+  const int lane = tid % WAVE_SIZE;
+  const int waves = BLOCK / WAVE_SIZE;
   const int j0 = (int)((long)chunk * n / NCHUNK);
   const int j1 = (int)((long)(chunk + 1) * n / NCHUNK);
   const float* k = k_in + (long)q * n;
   float acc = 0.0f;
-  for (int j = j0 + tid; j < j1; j += BLOCK) {
-    const float* row = kinv + (long)j * n;
+  const int n4 = n / 4;
+  for (int j = j0 + wave; j < j1; j += waves) {
+    const float4* row4 = reinterpret_cast<const float4*>(
+        kinv + (long)j * n);
+    const float4* k4 = reinterpret_cast<const float4*>(k);
     float t_j = 0.0f;
-    int i = 0;
-    for (; i + 4 <= n; i += 4) {
-      t_j = fmaf(row[i], k[i], t_j);
-      t_j = fmaf(row[i + 1], k[i + 1], t_j);
-      t_j = fmaf(row[i + 2], k[i + 2], t_j);
-      t_j = fmaf(row[i + 3], k[i + 3], t_j);
+    for (int i4 = lane; i4 < n4; i4 += WAVE_SIZE) {
+      const float4 r = row4[i4];
+      const float4 kv = k4[i4];
+      t_j = fmaf(r.x, kv.x, t_j);
+      t_j = fmaf(r.y, kv.y, t_j);
+      t_j = fmaf(r.z, kv.z, t_j);
+      t_j = fmaf(r.w, kv.w, t_j);
     }
-    for (; i < n; ++i) t_j = fmaf(row[i], k[i], t_j);
-    acc = fmaf(k[j], t_j, acc);
+    for (int i = 4 * n4 + lane; i < n; i += WAVE_SIZE) {
+      t_j = fmaf(kinv[(long)j * n + i], k[i], t_j);
+    }
+    t_j = wave_reduce_sum(t_j);
+    if (lane == 0) acc = fmaf(k[j], t_j, acc);
   }
+  // acc lives in lane 0 of each wave; combine across waves.
   auto fsum = [](float a, float c) { return a + c; };
   float v = block_reduce(acc, red, fsum, 0.0f);
   if (tid == 0) var_part[q * NCHUNK + chunk] = v;
