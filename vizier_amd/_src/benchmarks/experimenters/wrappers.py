@@ -230,3 +230,76 @@ class MultiObjectiveExperimenter(Experimenter):
 
   def problem_statement(self) -> vz.ProblemStatement:
     return copy.deepcopy(self._problem)
+
+
+class PermutingExperimenter(Experimenter):
+  """Permutes parameter values between named parameters before evaluating."""
+
+  def __init__(self, exptr: Experimenter, parameters_to_permute:
+               Sequence[str], *, seed: Optional[int] = None):
+    self._exptr = exptr
+    names = list(parameters_to_permute)
+    rng = np.random.default_rng(seed)
+    self._mapping = dict(zip(names,
+                             [names[i] for i in rng.permutation(len(names))]))
+
+  def evaluate(self, suggestions: Sequence[vz.Trial]) -> None:
+    permuted = []
+    for trial in suggestions:
+      t = copy.deepcopy(trial)
+      values = {src: t.parameters.get_value(src)
+                for src in self._mapping}
+      for src, dst in self._mapping.items():
+        t.parameters[dst] = values[src]
+      permuted.append(t)
+    self._exptr.evaluate(permuted)
+    for trial, t in zip(suggestions, permuted):
+      if t.final_measurement is not None:
+        trial.complete(t.final_measurement,
+                       infeasibility_reason=t.infeasibility_reason)
+
+  def problem_statement(self) -> vz.ProblemStatement:
+    return self._exptr.problem_statement()
+
+
+class SparseExperimenter(Experimenter):
+  """Embeds the problem in a higher-dimensional space of inert parameters."""
+
+  def __init__(self, exptr: Experimenter, num_dummy_dimensions: int,
+               *, prefix: str = 'dummy'):
+    self._exptr = exptr
+    self._problem = exptr.problem_statement()
+    for i in range(num_dummy_dimensions):
+      self._problem.search_space.root.add_float_param(
+          f'{prefix}{i}', 0.0, 1.0)
+
+  def evaluate(self, suggestions: Sequence[vz.Trial]) -> None:
+    self._exptr.evaluate(suggestions)
+
+  def problem_statement(self) -> vz.ProblemStatement:
+    return copy.deepcopy(self._problem)
+
+
+class SurrogateExperimenter(Experimenter):
+  """Uses a trained regression model as the objective (HPO-B style).
+
+  The reference's HPO-B / NAS-Bench handlers evaluate offline-trained
+  surrogates; this wrapper provides the same mechanism for any
+  sklearn-style `.predict(features)` model over the converter features.
+  """
+
+  def __init__(self, model, problem: vz.ProblemStatement):
+    from vizier_amd.converters.core import TrialToArrayConverter
+    self._model = model
+    self._problem = problem
+    self._converter = TrialToArrayConverter(problem)
+    self._metric = problem.metric_information.item().name
+
+  def evaluate(self, suggestions: Sequence[vz.Trial]) -> None:
+    feats = self._converter.to_features(suggestions)
+    values = np.asarray(self._model.predict(feats)).reshape(-1)
+    for trial, v in zip(suggestions, values):
+      trial.complete(vz.Measurement(metrics={self._metric: float(v)}))
+
+  def problem_statement(self) -> vz.ProblemStatement:
+    return copy.deepcopy(self._problem)
