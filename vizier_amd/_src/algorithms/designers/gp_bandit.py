@@ -64,7 +64,7 @@ class GPBanditConfig:
   num_seed_trials: int = 2
   ard_restarts: int = 4
   ard_max_iters: int = 50
-  ard_warm_iters: int = 20   # iters when warm-starting from the last fit
+  ard_warm_iters: int = 12   # iters when warm-starting from the last fit
   use_trust_region: bool = True
   num_scalarizations: int = 1000  # multi-objective
   data_parallel: bool = False     # shard the sweep across dist ranks

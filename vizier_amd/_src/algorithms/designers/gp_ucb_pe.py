@@ -58,7 +58,7 @@ class UCBPEConfig:
   num_seed_trials: int = 2
   ard_restarts: int = 4
   ard_max_iters: int = 50
-  ard_warm_iters: int = 20
+  ard_warm_iters: int = 12
   use_trust_region: bool = True
   device: Optional[str] = None
   dtype: torch.dtype = torch.float32
