@@ -216,6 +216,7 @@ class VizierGPBandit(Designer, Predictor):
           scores = trust_region.apply(flat.reshape(dense.shape)[:, 0, :],
                                       scores)
         return scores
+      score_fn.graph_safe = False  # multi-posterior path uses rocBLAS
       return score_fn, 1
 
     posterior = self._posteriors[0]
@@ -241,6 +242,7 @@ class VizierGPBandit(Designer, Predictor):
           flat_scores = trust_region.apply(dense[:, 0, :], scores)
           scores = flat_scores
         return scores
+      score_fn.graph_safe = False  # host RNG + rocBLAS inside
       return score_fn, count
 
     if cfg.acquisition == 'ei':

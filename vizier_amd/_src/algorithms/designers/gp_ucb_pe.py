@@ -225,6 +225,10 @@ class VizierGPUCBPEBandit(Designer):
 
       def score_fn(batch: CandidateBatch) -> torch.Tensor:
         return scoring(self._codec.decode(batch)[:, 0, :])
+      # The trust region anchors at completed+pending points (not the
+      # GP train set), so scoring uses rocBLAS — not hipGraph-capturable
+      # on this ROCm build.
+      score_fn.graph_safe = False
     else:
       # Promising-region threshold: predicted mean at the observed point
       # with the highest UCB (gp_ucb_pe.py:175-205).
@@ -244,6 +248,7 @@ class VizierGPUCBPEBandit(Designer):
         if trust_region is not None:
           scores = trust_region.apply(xs, scores)
         return scores
+      score_fn.graph_safe = False  # two-posterior path uses rocBLAS
 
     factory = VectorizedOptimizerFactory(
         eagle_config=EagleStrategyConfig(),

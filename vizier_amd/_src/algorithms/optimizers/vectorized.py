@@ -54,7 +54,8 @@ class VectorizedOptimizer:
 
     self.last_used_graph = False
     self.last_graph_error = None
-    if strategy._ext is not None:
+    if strategy._ext is not None and getattr(score_fn, 'graph_safe',
+                                             True):
       try:
         result = self._optimize_hipgraph(score_fn, count, state,
                                          iterations)
