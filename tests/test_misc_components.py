@@ -216,3 +216,16 @@ class TestParetoTorchAndWarping:
     w = KumaraswamyInputWarper(a=2.0, b=0.7)
     x = np.linspace(0, 1, 11)
     np.testing.assert_allclose(w.unwarp(w.warp(x)), x, atol=1e-12)
+
+
+class TestVizierAlias:
+
+  def test_reference_style_imports(self):
+    from vizier import pyvizier as vz2
+    from vizier.service import clients as _clients
+    from vizier import pythia as _pythia
+    from vizier import benchmarks as _benchmarks
+    assert vz2.StudyConfig is not None
+    assert _clients.Study is not None
+    assert _pythia.Policy is not None
+    assert _benchmarks.BenchmarkRunner is not None
