@@ -124,7 +124,7 @@ class VectorizedEagleStrategy:
     if self.device.type == 'cuda':
       from vizier_amd._src.ops import dispatch as ops
       self._ext = ops.require_ext()
-      self._iter_t = torch.zeros(1, dtype=torch.long, device=self.device)
+      self._iter_t = torch.zeros(2, dtype=torch.long, device=self.device)
       self._out_cont = torch.empty(self.batch_size, n_parallel,
                                    n_continuous, dtype=dtype,
                                    device=self.device)

@@ -59,11 +59,18 @@ eagle_suggest_kernel(const float* __restrict__ pool_cont,  // (P, q, Dc)
                      float gravity, float neg_gravity, float norm_scale,
                      float cat_factor, float p_same,
                      unsigned long long seed) {
-  // The iteration counter lives in device memory so the whole step is
-  // hipGraph-capturable (no host-varying kernel arguments).
-  const unsigned long long offset = *iter_ptr;
+  // Counter handshake (hipGraph-capturable, race-free): suggest reads
+  // slot 0 and publishes the iteration to slot 1 for the update kernel;
+  // update reads slot 1 and advances slot 0 for the next suggest. All
+  // cross-kernel ordering comes from stream order; within a kernel only
+  // one block writes, and readers never read the slot their own grid
+  // writes.
+  const unsigned long long offset = iter_ptr[0];
   const int batch_start = (int)(offset % (unsigned long long)n_batches) *
                           batch_size;
+  if (blockIdx.x == 0 && threadIdx.x == 0) {
+    ((unsigned long long*)iter_ptr)[1] = offset;
+  }
   __shared__ float scale[MAX_POOL];
   __shared__ float red[8];
   __shared__ float scale_sum_s;
@@ -211,70 +218,69 @@ eagle_update_kernel(float* __restrict__ pool_cont,
                     int n_batches, int batch_size, int q, int dc, int dcat,
                     float penalize_factor, float perturbation_lower_bound,
                     float base_perturbation, unsigned long long seed) {
-  const unsigned long long offset = *iter_ptr;
+  // One block per batch member. Every block recomputes the (tiny)
+  // batch max + old best, so new_best is identical everywhere and
+  // writes are idempotent — no cross-block ordering needed.
+  const unsigned long long offset = iter_ptr[1];
   const int batch_start = (int)(offset % (unsigned long long)n_batches) *
                           batch_size;
   __shared__ float red[8];
   __shared__ float new_best_s;
   const int tid = threadIdx.x;
+  const int i = blockIdx.x;
+  if (i >= batch_size) return;
 
-  // Single workgroup handles the whole batch (it is tiny).
   float local_max = -INFINITY;
-  for (int i = tid; i < batch_size; i += BLOCK) {
-    local_max = fmaxf(local_max, batch_rewards[i]);
+  for (int t = tid; t < batch_size; t += BLOCK) {
+    local_max = fmaxf(local_max, batch_rewards[t]);
   }
   auto fmax_ = [](float a, float c) { return fmaxf(a, c); };
   float batch_max = block_reduce(local_max, red, fmax_, -INFINITY);
   if (tid == 0) {
     new_best_s = fmaxf(best_reward[0], batch_max);
-    best_reward[0] = new_best_s;
+    if (i == 0) best_reward[0] = new_best_s;
   }
   __syncthreads();
   const float new_best = new_best_s;
   const int flat_c = q * dc;
   const int flat_k = q * dcat;
 
-  for (int i = 0; i < batch_size; ++i) {
-    const int me = batch_start + i;
-    const float new_r = batch_rewards[i];
-    const float old_r = rewards[me];
-    const bool improved = new_r > old_r;
-    float pert = improved ? perturbations[me]
-                          : perturbations[me] * penalize_factor;
-    float reward = improved ? new_r : old_r;
-    const bool dead = (pert < perturbation_lower_bound) &&
-                      (reward != new_best);
-    if (improved) {
-      for (int j = tid; j < flat_c; j += BLOCK) {
-        pool_cont[me * flat_c + j] = batch_cont[i * flat_c + j];
-      }
-      for (int j = tid; j < flat_k; j += BLOCK) {
-        pool_cat[me * flat_k + j] = batch_cat[i * flat_k + j];
-      }
+  const int me = batch_start + i;
+  const float new_r = batch_rewards[i];
+  const float old_r = rewards[me];
+  const bool improved = new_r > old_r;
+  float pert = improved ? perturbations[me]
+                        : perturbations[me] * penalize_factor;
+  float reward = improved ? new_r : old_r;
+  const bool dead = (pert < perturbation_lower_bound) &&
+                    (reward != new_best);
+  if (improved) {
+    for (int j = tid; j < flat_c; j += BLOCK) {
+      pool_cont[me * flat_c + j] = batch_cont[i * flat_c + j];
     }
-    if (dead) {
-      for (int j = tid; j < flat_c; j += BLOCK) {
-        pool_cont[me * flat_c + j] = rng_uniform(
-            seed, offset ^ 0x5151ull, (unsigned)(me * flat_c + j));
-      }
-      for (int j = tid; j < flat_k; j += BLOCK) {
-        const long size = cat_sizes[j % dcat];
-        const float u = rng_uniform(seed, offset ^ 0x1234ull,
-                                    (unsigned)(me * flat_k + j));
-        long c = (long)(u * size);
-        pool_cat[me * flat_k + j] = min(c, size - 1);
-      }
-      reward = -INFINITY;
-      pert = base_perturbation;
+    for (int j = tid; j < flat_k; j += BLOCK) {
+      pool_cat[me * flat_k + j] = batch_cat[i * flat_k + j];
     }
-    if (tid == 0) {
-      rewards[me] = reward;
-      perturbations[me] = pert;
+  }
+  if (dead) {
+    for (int j = tid; j < flat_c; j += BLOCK) {
+      pool_cont[me * flat_c + j] = rng_uniform(
+          seed, offset ^ 0x5151ull, (unsigned)(me * flat_c + j));
     }
-    __syncthreads();
+    for (int j = tid; j < flat_k; j += BLOCK) {
+      const long size = cat_sizes[j % dcat];
+      const float u = rng_uniform(seed, offset ^ 0x1234ull,
+                                  (unsigned)(me * flat_k + j));
+      long c = (long)(u * size);
+      pool_cat[me * flat_k + j] = min(c, size - 1);
+    }
+    reward = -INFINITY;
+    pert = base_perturbation;
   }
   if (tid == 0) {
-    *iter_ptr = offset + 1;  // advance the device iteration counter
+    rewards[me] = reward;
+    perturbations[me] = pert;
+    if (i == 0) iter_ptr[0] = offset + 1;  // next suggest's iteration
   }
 }
 
@@ -304,7 +310,8 @@ extern "C" void launch_eagle_update(
     int dc, int dcat,
     float penalize_factor, float perturbation_lower_bound,
     float base_perturbation, unsigned long long seed, hipStream_t stream) {
-  hipLaunchKernelGGL(eagle_update_kernel, dim3(1), dim3(BLOCK), 0, stream,
+  hipLaunchKernelGGL(eagle_update_kernel, dim3(batch_size), dim3(BLOCK),
+                     0, stream,
                      pool_cont, pool_cat, rewards, perturbations,
                      batch_cont, batch_cat, batch_rewards, cat_sizes,
                      best_reward, iter_ptr, n_batches, batch_size, q, dc,
