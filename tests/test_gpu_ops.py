@@ -237,6 +237,22 @@ class TestGramBf16MFMA:
     if err_t is not None:
       assert float(err_t) > 10 * float(err)
 
+  @pytest.mark.parametrize('n,m,d', [(128, 130, 32), (640, 520, 20),
+                                     (1111, 777, 90)])
+  def test_bf16_lds_tiled_matches_fp64_reference(self, ext, n, m, d):
+    """The 128x128 LDS-tiled MFMA path (incl. ragged tile edges)."""
+    from vizier_amd._src.gp.matern import gram_matern52
+    g = torch.Generator().manual_seed(3)
+    x1 = torch.rand(n, d, generator=g).cuda()
+    x2 = torch.rand(m, d, generator=g).cuda()
+    ls = (torch.rand(d, generator=g) * 1.5 + 0.2).cuda()
+    amp = 1.3
+    got = ext.gram_matern52_bf16_tiled(x1, x2, ls, amp)
+    want = gram_matern52(x1.cpu().double(), x2.cpu().double(),
+                         ls.cpu().double(), torch.tensor(amp).double())
+    err = (got.cpu().double() - want).abs().max()
+    assert float(err) < 3e-2 * amp * amp, f'max err {err}'
+
   def test_bf16_diag_is_amp2(self, ext):
     x = torch.rand(100, 16).cuda()
     ls = torch.full((16,), 0.5).cuda()

@@ -15,6 +15,11 @@ extern "C" void launch_gram_matern52_bf16(
     const float* n2, float* out, int n, int m, int dp, float amp2,
     hipStream_t stream);
 
+extern "C" void launch_gram_matern52_bf16_tiled(
+    const unsigned short* z1, const unsigned short* z2, const float* n1,
+    const float* n2, float* out, int n, int m, int dp, float amp2,
+    hipStream_t stream);
+
 extern "C" void launch_posterior_score(
     const float* xq, const float* x, const float* inv_ls,
     const float* alpha, const float* kinv, const unsigned char* onehot,
@@ -82,9 +87,9 @@ torch::Tensor gram_matern52(torch::Tensor x1, torch::Tensor x2,
   return out;
 }
 
-torch::Tensor gram_matern52_bf16(torch::Tensor x1, torch::Tensor x2,
-                                 torch::Tensor lengthscales,
-                                 double amplitude) {
+torch::Tensor gram_matern52_bf16_impl(torch::Tensor x1, torch::Tensor x2,
+                                      torch::Tensor lengthscales,
+                                      double amplitude, bool tiled) {
   x1 = check_f32(x1, "x1");
   x2 = check_f32(x2, "x2");
   lengthscales = check_f32(lengthscales, "lengthscales");
@@ -108,12 +113,28 @@ torch::Tensor gram_matern52_bf16(torch::Tensor x1, torch::Tensor x2,
   auto n1 = (z1b.to(torch::kFloat32) * z1b.to(torch::kFloat32)).sum(-1);
   auto n2 = (z2b.to(torch::kFloat32) * z2b.to(torch::kFloat32)).sum(-1);
   auto out = torch::empty({n, m}, x1.options());
-  launch_gram_matern52_bf16(
-      (const unsigned short*)z1b.data_ptr(),
-      (const unsigned short*)z2b.data_ptr(), n1.data_ptr<float>(),
-      n2.data_ptr<float>(), out.data_ptr<float>(), n, m, dp,
-      (float)(amplitude * amplitude), current_stream());
+  auto launch = tiled ? launch_gram_matern52_bf16_tiled
+                      : launch_gram_matern52_bf16;
+  launch((const unsigned short*)z1b.data_ptr(),
+         (const unsigned short*)z2b.data_ptr(), n1.data_ptr<float>(),
+         n2.data_ptr<float>(), out.data_ptr<float>(), n, m, dp,
+         (float)(amplitude * amplitude), current_stream());
   return out;
+}
+
+torch::Tensor gram_matern52_bf16(torch::Tensor x1, torch::Tensor x2,
+                                 torch::Tensor lengthscales,
+                                 double amplitude) {
+  // LDS-tiled MFMA path once both dims fill 128x128 tiles; the
+  // register-direct strip kernel has less overhead for small grams.
+  const bool tiled = x1.size(0) >= 512 && x2.size(0) >= 512;
+  return gram_matern52_bf16_impl(x1, x2, lengthscales, amplitude, tiled);
+}
+
+torch::Tensor gram_matern52_bf16_tiled(torch::Tensor x1, torch::Tensor x2,
+                                       torch::Tensor lengthscales,
+                                       double amplitude) {
+  return gram_matern52_bf16_impl(x1, x2, lengthscales, amplitude, true);
 }
 
 torch::Tensor posterior_scores(torch::Tensor xq, torch::Tensor x,
@@ -282,6 +303,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Fused Matern-5/2 ARD Gram matrix (gfx950)");
   m.def("gram_matern52_bf16", &gram_matern52_bf16,
         "bf16 MFMA Matern-5/2 Gram matrix (gfx950 matrix cores)");
+  m.def("gram_matern52_bf16_tiled", &gram_matern52_bf16_tiled,
+        "128x128 LDS-tiled bf16 MFMA Matern-5/2 Gram (gfx950)");
   m.def("gram_matern52_fp8", &gram_matern52_fp8,
         "fp8 e4m3 MFMA Matern-5/2 Gram matrix (gfx950, config 5)");
   m.def("posterior_scores", &posterior_scores,
