@@ -149,3 +149,48 @@ class TestGPBandit:
     designer.update(CompletedTrials([bad]), ActiveTrials())
     out = designer.suggest(1)
     assert out
+
+  def test_transfer_learning_priors(self):
+    """set_priors stacks a prior-study GP under the current study's GP
+    (reference gp_bandit.py:289, gp/gp_models.py:245-365)."""
+    problem = make_problem()
+    rng = np.random.default_rng(5)
+    prior_trials = []
+    for uid in range(1, 41):
+      x = rng.uniform(0, 1, 4)
+      t = vz.Trial({f'x{i}': float(x[i]) for i in range(4)}, id=uid)
+      t.complete(vz.Measurement(
+          metrics={'obj': float(-((x - 0.7) ** 2).sum())}))
+      prior_trials.append(t)
+
+    designer = VizierGPBandit(problem, small_config(), seed=0)
+    designer.set_priors([CompletedTrials(prior_trials)])
+    assert designer._prior_stack is not None
+    run_loop(designer, 5)
+    # Residual stacking active: scoring + prediction go through the
+    # stacked GP chain.
+    assert designer._stacked is not None
+    pred = designer.predict(
+        [vz.TrialSuggestion({f'x{i}': 0.7 for i in range(4)})])
+    assert pred.mean.shape == (1, 1) and np.isfinite(pred.mean).all()
+    assert (pred.stddev > 0).all()
+
+  def test_transfer_learning_resilient_to_bad_prior(self):
+    """A misleading prior must not break convergence (the top GP learns
+    the residual; reference 'resilient to bad priors')."""
+    problem = make_problem()
+    rng = np.random.default_rng(9)
+    bad_prior = []
+    for uid in range(1, 31):
+      x = rng.uniform(0, 1, 4)
+      # Prior points AWAY from the true optimum (optimum at 0.0).
+      t = vz.Trial({f'x{i}': float(x[i]) for i in range(4)}, id=uid)
+      t.complete(vz.Measurement(
+          metrics={'obj': float(-((x - 0.0) ** 2).sum())}))
+      bad_prior.append(t)
+
+    designer = VizierGPBandit(problem, small_config(), seed=1)
+    designer.set_priors([CompletedTrials(bad_prior)])
+    trials = run_loop(designer, 8)
+    best = max(t.final_measurement.metrics['obj'].value for t in trials)
+    assert best > -0.4, f'transfer designer failed to adapt: {best}'

@@ -45,6 +45,7 @@ from vizier_amd._src.algorithms.optimizers.vectorized import (
 )
 from vizier_amd._src.gp import acquisitions as acq_lib
 from vizier_amd._src.gp import gp_model, output_warpers
+from vizier_amd._src.gp import transfer_learning
 from vizier_amd._src.parallel import sharded_sweep
 from vizier_amd._src.pythia import suggest_default
 
@@ -87,6 +88,8 @@ class VizierGPBandit(Designer, Predictor):
         if not problem.search_space.is_conditional else None
     self._device = self._config.device or default_device()
     self._posteriors: List[gp_model.GPPosterior] = []
+    self._prior_stack: Optional[transfer_learning.StackedResidualGP] = None
+    self._stacked: Optional[transfer_learning.StackedResidualGP] = None
     self._last_fit_count = -1
     self._x_cache = None
     self._y_cache = None
@@ -147,6 +150,36 @@ class VizierGPBandit(Designer, Predictor):
 
   # -- GP path --------------------------------------------------------------
 
+  def set_priors(self, prior_studies: Sequence[CompletedTrials]) -> None:
+    """Transfer learning: stack prior-study GPs under the current GP.
+
+    Reference parity: gp_bandit.py:289 set_priors / gp/gp_models.py
+    :245-365. Each prior study trains a GP on its own (warped) labels;
+    the current study's GP is then trained on the RESIDUALS of the
+    prior chain's mean, and predictions combine mean = base + top with
+    dof-weighted stddev inflation (gp/transfer_learning.py:46).
+    Single-objective only (as in the reference).
+    """
+    cfg = self._config
+    datasets = []
+    for study in prior_studies:
+      trials = list(study.trials)
+      if not trials:
+        continue
+      x_np = self._converter.to_features(trials)
+      y_np = self._prepare_labels(self._converter.to_labels(trials))
+      x = torch.as_tensor(x_np, dtype=cfg.dtype, device=self._device)
+      y = torch.as_tensor(y_np[:, 0], dtype=cfg.dtype, device=self._device)
+      datasets.append((x, y))
+    if datasets:
+      self._prior_stack = transfer_learning.train_stacked_gp(
+          datasets, num_restarts=cfg.ard_restarts,
+          max_iters=cfg.ard_max_iters, seed=self._seed + 101)
+    else:
+      self._prior_stack = None
+    self._stacked = None
+    self._last_fit_count = -1   # force refit against the new priors
+
   def _prepare_labels(self, raw_labels: np.ndarray) -> np.ndarray:
     """Per-metric default warping -> (N, M) finite labels."""
     warped = np.zeros_like(raw_labels)
@@ -165,8 +198,13 @@ class VizierGPBandit(Designer, Predictor):
     self._x = x
     prev = self._posteriors
     self._posteriors = []
+    transfer = self._prior_stack is not None and y_np.shape[1] == 1
     for m in range(y_np.shape[1]):
       y = torch.as_tensor(y_np[:, m], dtype=cfg.dtype, device=self._device)
+      if transfer and m == 0:
+        with torch.no_grad():
+          prior_mean, _ = self._prior_stack.predict(x)
+        y = y - prior_mean
       warm = prev[m].raw if m < len(prev) and prev[m].raw is not None \
           else None
       post = gp_model.train_gp(
@@ -177,6 +215,11 @@ class VizierGPBandit(Designer, Predictor):
       if cfg.data_parallel:
         sharded_sweep.broadcast_posterior(post)
       self._posteriors.append(post)
+    if transfer:
+      self._stacked = transfer_learning.StackedResidualGP(
+          self._posteriors[0], self._prior_stack, num_obs=x.shape[0])
+    else:
+      self._stacked = None
     self._warped_labels = torch.as_tensor(y_np, dtype=cfg.dtype,
                                           device=self._device)
     self._last_fit_count = len(self._trials)
@@ -222,6 +265,28 @@ class VizierGPBandit(Designer, Predictor):
     posterior = self._posteriors[0]
     best_value = float(self._warped_labels[:, 0].max())
 
+    if self._stacked is not None and cfg.acquisition in (
+        'ucb', 'ei', 'pi'):
+      # Transfer learning: score through the stacked residual GP
+      # (mean = base + top, dof-inflated stddev). Composed torch path.
+      if cfg.acquisition == 'ei':
+        acquisition = acq_lib.EI(best_value=best_value)
+      elif cfg.acquisition == 'pi':
+        acquisition = acq_lib.PI(best_value=best_value)
+      else:
+        acquisition = acq_lib.UCB(coefficient=cfg.ucb_coefficient)
+      stacked = self._stacked
+
+      def score_fn(batch: CandidateBatch) -> torch.Tensor:
+        dense = self._codec.decode(batch)[:, 0, :]
+        mean, stddev = stacked.predict(dense)
+        scores = acquisition(mean, stddev)
+        if trust_region is not None:
+          scores = trust_region.apply(dense, scores)
+        return scores
+      score_fn.graph_safe = False  # multi-GP chain uses rocBLAS
+      return score_fn, 1
+
     if cfg.acquisition == 'qei' and count > 1:
       qei = acq_lib.QEI(best_value=best_value, seed=self._seed)
 
@@ -229,9 +294,18 @@ class VizierGPBandit(Designer, Predictor):
         dense = self._codec.decode(batch)          # (B, q, D)
         mean, cov = posterior_batched_cov(posterior, dense)
         q = dense.shape[1]
-        cov = cov + 1e-8 * torch.eye(q, dtype=cov.dtype,
-                                     device=cov.device)
-        L = torch.linalg.cholesky(cov)             # (B, q, q)
+        eye = torch.eye(q, dtype=cov.dtype, device=cov.device)
+        # Near-duplicate candidate rows make cov singular; jitter
+        # relative to the diagonal scale, then fall back to a diagonal
+        # factor for any batch element that still fails.
+        diag = cov.diagonal(dim1=-2, dim2=-1)
+        jitter = 1e-4 * diag.mean(-1, keepdim=True).clamp_min(1e-10)
+        cov = cov + jitter.unsqueeze(-1) * eye
+        L, info = torch.linalg.cholesky_ex(cov)    # (B, q, q)
+        bad = (info > 0)
+        if bool(bad.any()):
+          L_diag = torch.diag_embed(diag.clamp_min(1e-12).sqrt())
+          L = torch.where(bad.view(-1, 1, 1), L_diag, L)
         g = torch.Generator(device='cpu').manual_seed(self._seed)
         eps = torch.randn(128, 1, q, generator=g).to(dense.device,
                                                      dense.dtype)
@@ -331,10 +405,15 @@ class VizierGPBandit(Designer, Predictor):
     x = torch.as_tensor(self._converter.to_features(trials),
                         dtype=self._config.dtype, device=self._device)
     means, stddevs = [], []
-    for post in self._posteriors:
-      mean, stddev = post.predict(x)
+    if self._stacked is not None:
+      mean, stddev = self._stacked.predict(x)
       means.append(mean.cpu().numpy())
       stddevs.append(stddev.cpu().numpy())
+    else:
+      for post in self._posteriors:
+        mean, stddev = post.predict(x)
+        means.append(mean.cpu().numpy())
+        stddevs.append(stddev.cpu().numpy())
     return Prediction(mean=np.stack(means, axis=-1),
                       stddev=np.stack(stddevs, axis=-1))
 
