@@ -213,6 +213,68 @@ class TestGPUDesignerEndToEnd:
         designer.update(CompletedTrials([t]), ActiveTrials())
     assert best > -0.25, f'GPU GP-Bandit failed to converge: {best}'
 
+  def test_multi_objective_and_qei_on_gpu(self, ext):
+    """MO hypervolume-scalarized scoring and q-EI batches run the full
+    GPU path (catches device-placement regressions in the composed
+    scorers that the fast single-objective path doesn't exercise)."""
+    from vizier_amd import pyvizier as vz
+    from vizier_amd._src.algorithms.core.abstractions import (
+        ActiveTrials,
+        CompletedTrials,
+    )
+    from vizier_amd._src.algorithms.designers.gp_bandit import (
+        GPBanditConfig,
+        VizierGPBandit,
+    )
+    rng = np.random.default_rng(0)
+
+    # Multi-objective: two conflicting objectives.
+    problem = vz.ProblemStatement()
+    for i in range(4):
+      problem.search_space.root.add_float_param(f'x{i}', 0.0, 1.0)
+    problem.metric_information = vz.MetricsConfig([
+        vz.MetricInformation(name=n, goal=vz.ObjectiveMetricGoal.MAXIMIZE)
+        for n in ('f1', 'f2')])
+    designer = VizierGPBandit(problem, GPBanditConfig(
+        max_evaluations=1500, ard_restarts=1, ard_max_iters=10,
+        device='cuda'), seed=0)
+    uid = 0
+    for _ in range(6):
+      for s in designer.suggest(1):
+        uid += 1
+        x = np.array([s.parameters.get_value(f'x{i}') for i in range(4)])
+        t = s.to_trial(uid)
+        t.complete(vz.Measurement(metrics={
+            'f1': float(-((x - 0.2) ** 2).sum()),
+            'f2': float(-((x - 0.8) ** 2).sum())}))
+        designer.update(CompletedTrials([t]), ActiveTrials())
+
+    # q-EI: batches of 3 distinct suggestions.
+    problem2 = vz.ProblemStatement()
+    for i in range(4):
+      problem2.search_space.root.add_float_param(f'x{i}', 0.0, 1.0)
+    problem2.metric_information.append(vz.MetricInformation(
+        name='obj', goal=vz.ObjectiveMetricGoal.MAXIMIZE))
+    d2 = VizierGPBandit(problem2, GPBanditConfig(
+        max_evaluations=1500, acquisition='qei', ard_restarts=1,
+        ard_max_iters=10, device='cuda'), seed=1)
+    uid = 0
+    for _ in range(4):
+      batch = d2.suggest(3)
+      assert len(batch) == 3
+      rows = [tuple(round(s.parameters.get_value(f'x{i}'), 6)
+                    for i in range(4)) for s in batch]
+      trials = []
+      for s in batch:
+        uid += 1
+        x = np.array([s.parameters.get_value(f'x{i}') for i in range(4)])
+        t = s.to_trial(uid)
+        t.complete(vz.Measurement(
+            metrics={'obj': float(-((x - 0.5) ** 2).sum())}))
+        trials.append(t)
+      d2.update(CompletedTrials(trials), ActiveTrials())
+    assert len(set(rows)) > 1, 'qEI produced identical batch points'
+
 
 class TestGramBf16MFMA:
 

@@ -112,10 +112,13 @@ def main():
   parser.add_argument('--evals', type=int, default=10000)
   parser.add_argument('--seeds', type=int, default=2)
   parser.add_argument('--out', default=None)
+  parser.add_argument('--only', choices=['qei', 'mo'], default=None)
   args = parser.parse_args()
 
   results = {'qei': {}, 'mo': {}}
   fns = {'sphere': bbob.Sphere, 'rastrigin': bbob.Rastrigin}
+  if args.only == 'mo':
+    fns = {}
   for name, fn in fns.items():
     for algo, use_gp in (('gp_qei_q4', True), ('quasi_random', False)):
       vals = []
@@ -128,8 +131,9 @@ def main():
               f'({time.time() - t0:.1f}s)', flush=True)
       results['qei'][f'{name}/{algo}'] = vals
 
-  for algo, use_gp in (('gp_hv_scalarized', True),
-                       ('quasi_random', False)):
+  mo_algos = [] if args.only == 'qei' else [
+      ('gp_hv_scalarized', True), ('quasi_random', False)]
+  for algo, use_gp in mo_algos:
     vals = []
     for seed in range(args.seeds):
       t0 = time.time()

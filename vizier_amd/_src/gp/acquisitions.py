@@ -186,6 +186,10 @@ class HyperVolumeScalarization:
 
   def __call__(self, ys: torch.Tensor) -> torch.Tensor:
     """ys: (..., M) -> (V, ...) scalarized values."""
+    if self.weights.device != ys.device:
+      self.weights = self.weights.to(ys.device, ys.dtype)
+      if self.reference_point is not None:
+        self.reference_point = self.reference_point.to(ys.device, ys.dtype)
     if self.reference_point is not None:
       ys = ys - self.reference_point
     w = self.weights.reshape((-1,) + (1,) * (ys.dim() - 1) + (ys.shape[-1],))
