@@ -106,7 +106,11 @@ class TestGPTraining:
         K_inv=None, nll=post.nll)
     mean_b, std_b = post_nokinv.predict(xq)
     assert torch.allclose(mean_a, mean_b, atol=1e-4)
-    assert torch.allclose(std_a, std_b, atol=1e-3)
+    # The GEMM quadform's fp32 cancellation error is bounded by the
+    # posterior cache's noise floor (~6e-5*amp^2 in variance, see
+    # train_gp) — a few percent of amp in stddev at worst.
+    amp = float(post.params.amplitude)
+    assert torch.allclose(std_a, std_b, atol=0.05 * amp)
 
   def test_cholesky_jitter_recovers_singular(self):
     K = torch.ones(5, 5)  # rank-1, singular

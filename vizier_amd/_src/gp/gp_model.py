@@ -199,15 +199,32 @@ def train_gp(x: torch.Tensor, y: torch.Tensor, *,
   raw = best_raw[idx]
   params = GPParams.from_raw(raw)
 
-  K = gram_matern52(x, None, params.lengthscales, params.amplitude)
-  K = K + params.noise * torch.eye(n, dtype=x.dtype, device=x.device)
-  L = cholesky_with_jitter(K, params.amplitude ** 2)
-  resid = (y - params.mean).unsqueeze(-1)
-  alpha = _chol_solve(L, resid).squeeze(-1)
+  # Posterior caches in fp64, stored fp32: once the fit drives noise
+  # toward its lower bound, cond(K) ~ amp^2/noise exceeds fp32's
+  # 1/eps and the explicit-inverse quadform (the HIP sweep's variance
+  # path) loses all accuracy. One fp64 factorization per fit is
+  # microseconds-scale on MI355X (CDNA4 FP64 matrix cores) and makes
+  # the cached alpha / K_inv correct to fp32 resolution.
+  x64 = x.double()
+  p64 = GPParams.from_raw(raw.double())
+  K = gram_matern52(x64, None, p64.lengthscales, p64.amplitude)
+  # fp32-consumable conditioning floor: the NLL fit may drive noise to
+  # ~1e-10 (the reference runs in float64 — jax x64), but the cached
+  # K_inv feeds fp32 GEMM quadforms whose cancellation error scales as
+  # eps * amp^4 / noise. Flooring the CACHE's noise at 1e-3*amp^2
+  # bounds that error to ~6e-5*amp^2 (a ~3% stddev error at the floor)
+  # at the cost of a ~0.03*amp stddev floor near training points —
+  # negligible against UCB-scale acquisition scores.
+  noise_eff = torch.maximum(p64.noise, 1e-3 * p64.amplitude ** 2)
+  K = K + noise_eff * torch.eye(n, dtype=x64.dtype, device=x.device)
+  L64 = cholesky_with_jitter(K, p64.amplitude ** 2)
+  resid = (y.double() - p64.mean).unsqueeze(-1)
+  alpha = _chol_solve(L64, resid).squeeze(-1)
   K_inv = None
   if precompute_inverse:
-    eye = torch.eye(n, dtype=x.dtype, device=x.device)
-    z = torch.linalg.solve_triangular(L, eye, upper=False)
-    K_inv = z.T @ z
-  return GPPosterior(x=x, params=params, L=L, alpha=alpha, K_inv=K_inv,
+    eye = torch.eye(n, dtype=x64.dtype, device=x.device)
+    z = torch.linalg.solve_triangular(L64, eye, upper=False)
+    K_inv = (z.T @ z).to(x.dtype)
+  return GPPosterior(x=x, params=params, L=L64.to(x.dtype),
+                     alpha=alpha.to(x.dtype), K_inv=K_inv,
                      nll=float(best_f[idx]), raw=raw.detach())

@@ -51,7 +51,11 @@ def gram_matern52(x1: torch.Tensor, x2: Optional[torch.Tensor],
     lengthscales = lengthscales.unsqueeze(-2)
   z1 = x1 / lengthscales
   z2 = z1 if x2 is None else x2 / lengthscales
-  r = pairwise_sqdist(z1, z2).sqrt()
+  # Epsilon below fp32 resolution: sqrt(0) has a NaN gradient, which
+  # would silently freeze an entire L-BFGS restart when two rows
+  # coincide (the diagonal does, whenever the GEMM form cancels
+  # exactly). matern52(1e-9) == 1.0f exactly, so values are unchanged.
+  r = pairwise_sqdist(z1, z2).clamp_min_(1e-18).sqrt()
   k = matern52(r)
   amp2 = (amplitude * amplitude)
   if amp2.dim() > 0:
