@@ -16,6 +16,7 @@ _SRC = [
     'vizier_amd/_src/ops/csrc/gram_matern52_bf16.hip',
     'vizier_amd/_src/ops/csrc/gram_matern52_bf16_tiled.hip',
     'vizier_amd/_src/ops/csrc/gram_matern52_fp8.hip',
+    'vizier_amd/_src/ops/csrc/gram_matern52_fp8_tiled.hip',
     'vizier_amd/_src/ops/csrc/posterior_score.hip',
     'vizier_amd/_src/ops/csrc/eagle_step.hip',
 ]

@@ -325,6 +325,19 @@ class TestGramBf16MFMA:
 
 class TestGramFp8MFMA:
 
+  @pytest.mark.parametrize('n,m,d', [(128, 130, 32), (1111, 777, 90)])
+  def test_fp8_lds_tiled_matches_fp64_loosely(self, ext, n, m, d):
+    from vizier_amd._src.gp.matern import gram_matern52
+    g = torch.Generator().manual_seed(13)
+    x1 = torch.rand(n, d, generator=g).cuda()
+    x2 = torch.rand(m, d, generator=g).cuda()
+    ls = (torch.rand(d, generator=g) * 1.5 + 0.3).cuda()
+    got = ext.gram_matern52_fp8_tiled(x1, x2, ls, 1.0)
+    want = gram_matern52(x1.cpu().double(), x2.cpu().double(),
+                         ls.cpu().double(), torch.tensor(1.0).double())
+    err = (got.cpu().double() - want).abs().max()
+    assert float(err) < 0.15, f'max err {err}'  # e4m3: ~2 digits
+
   def test_fp8_mfma_matches_fp64_loosely(self, ext):
     from vizier_amd._src.gp.matern import gram_matern52
     g = torch.Generator().manual_seed(11)

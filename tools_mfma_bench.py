@@ -39,6 +39,7 @@ def main():
     t16 = bench(lambda: ext.gram_matern52_bf16(x1, x2, ls, 1.0))
     tt = bench(lambda: ext.gram_matern52_bf16_tiled(x1, x2, ls, 1.0))
     t8 = bench(lambda: ext.gram_matern52_fp8(x1, x2, ls, 1.0))
+    t8t = bench(lambda: ext.gram_matern52_fp8_tiled(x1, x2, ls, 1.0))
     k32 = ext.gram_matern52(x1, x2, ls, 1.0)
     k16 = ext.gram_matern52_bf16(x1, x2, ls, 1.0)
     kt = ext.gram_matern52_bf16_tiled(x1, x2, ls, 1.0)
@@ -48,6 +49,7 @@ def main():
           f'bf16 MFMA {t16*1e3:.2f}ms ({flops/t16/1e12:.1f} TF) | '
           f'bf16 LDS-tiled {tt*1e3:.2f}ms ({flops/tt/1e12:.1f} TF) | '
           f'fp8 MFMA {t8*1e3:.2f}ms ({flops/t8/1e12:.1f} TF) | '
+          f'fp8 LDS-tiled {t8t*1e3:.2f}ms ({flops/t8t/1e12:.1f} TF) | '
           f'bf16 err {float((k16-k32).abs().max()):.4f} '
           f'tiled-vs-strip err {float((kt-k16).abs().max()):.2e} '
           f'fp8 err {float((k8-k32).abs().max()):.4f}', flush=True)

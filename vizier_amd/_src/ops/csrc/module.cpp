@@ -31,6 +31,11 @@ extern "C" void launch_gram_matern52_fp8(
     const float* n2, float* out, int n, int m, int dp, float amp2,
     float scale2, hipStream_t stream);
 
+extern "C" void launch_gram_matern52_fp8_tiled(
+    const unsigned char* z1, const unsigned char* z2, const float* n1,
+    const float* n2, float* out, int n, int m, int dp, float amp2,
+    float scale2, hipStream_t stream);
+
 extern "C" void launch_posterior_score_chunked(
     const float* xq, const float* x, const float* inv_ls,
     const float* alpha, const float* kinv, const unsigned char* onehot,
@@ -167,9 +172,9 @@ torch::Tensor posterior_scores(torch::Tensor xq, torch::Tensor x,
   return out;
 }
 
-torch::Tensor gram_matern52_fp8(torch::Tensor x1, torch::Tensor x2,
-                                torch::Tensor lengthscales,
-                                double amplitude) {
+torch::Tensor gram_matern52_fp8_impl(torch::Tensor x1, torch::Tensor x2,
+                                     torch::Tensor lengthscales,
+                                     double amplitude, bool tiled) {
   x1 = check_f32(x1, "x1");
   x2 = check_f32(x2, "x2");
   lengthscales = check_f32(lengthscales, "lengthscales");
@@ -195,12 +200,27 @@ torch::Tensor gram_matern52_fp8(torch::Tensor x1, torch::Tensor x2,
   auto n1 = (z1f * z1f).sum(-1);
   auto n2 = (z2f * z2f).sum(-1);
   auto out = torch::empty({n, m}, x1.options());
-  launch_gram_matern52_fp8(
-      (const unsigned char*)z1b.data_ptr(),
-      (const unsigned char*)z2b.data_ptr(), n1.data_ptr<float>(),
-      n2.data_ptr<float>(), out.data_ptr<float>(), n, m, dp,
-      (float)(amplitude * amplitude), (float)(s * s), current_stream());
+  auto launch = tiled ? launch_gram_matern52_fp8_tiled
+                      : launch_gram_matern52_fp8;
+  launch((const unsigned char*)z1b.data_ptr(),
+         (const unsigned char*)z2b.data_ptr(), n1.data_ptr<float>(),
+         n2.data_ptr<float>(), out.data_ptr<float>(), n, m, dp,
+         (float)(amplitude * amplitude), (float)(s * s),
+         current_stream());
   return out;
+}
+
+torch::Tensor gram_matern52_fp8(torch::Tensor x1, torch::Tensor x2,
+                                torch::Tensor lengthscales,
+                                double amplitude) {
+  const bool tiled = x1.size(0) >= 512 && x2.size(0) >= 512;
+  return gram_matern52_fp8_impl(x1, x2, lengthscales, amplitude, tiled);
+}
+
+torch::Tensor gram_matern52_fp8_tiled(torch::Tensor x1, torch::Tensor x2,
+                                      torch::Tensor lengthscales,
+                                      double amplitude) {
+  return gram_matern52_fp8_impl(x1, x2, lengthscales, amplitude, true);
 }
 
 torch::Tensor posterior_scores_chunked(
@@ -307,6 +327,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "128x128 LDS-tiled bf16 MFMA Matern-5/2 Gram (gfx950)");
   m.def("gram_matern52_fp8", &gram_matern52_fp8,
         "fp8 e4m3 MFMA Matern-5/2 Gram matrix (gfx950, config 5)");
+  m.def("gram_matern52_fp8_tiled", &gram_matern52_fp8_tiled,
+        "128x128 LDS-tiled fp8 e4m3 MFMA Matern-5/2 Gram (gfx950)");
   m.def("posterior_scores", &posterior_scores,
         "Fused GP posterior + acquisition + trust region (gfx950)");
   m.def("posterior_scores_chunked", &posterior_scores_chunked,
