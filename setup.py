@@ -19,6 +19,7 @@ _SRC = [
     'vizier_amd/_src/ops/csrc/gram_matern52_fp8_tiled.hip',
     'vizier_amd/_src/ops/csrc/posterior_score.hip',
     'vizier_amd/_src/ops/csrc/eagle_step.hip',
+    'vizier_amd/_src/ops/csrc/eagle_sweep.hip',
 ]
 
 setup(

@@ -180,6 +180,64 @@ class TestEagleKernels:
     assert (state.perturbations[sl] < 0.16).all()
 
 
+class TestEagleSweepMegakernel:
+
+  def _make_problem(self, n=300, d=12, seed=0):
+    from vizier_amd._src.gp import acquisitions as acq_lib
+    from vizier_amd._src.gp import gp_model
+    g = torch.Generator().manual_seed(seed)
+    x = torch.rand(n, d, generator=g).cuda()
+    y = (-((x - 0.4) ** 2).sum(-1) +
+         0.01 * torch.randn(n, generator=g).cuda())
+    post = gp_model.train_gp(x, y, num_restarts=2, max_iters=15)
+    onehot = torch.zeros(d, dtype=torch.bool, device='cuda')
+    tr = acq_lib.TrustRegion(post.x, onehot)
+    scoring = acq_lib.ScoringFunction(
+        post, acq_lib.UCB(coefficient=1.8), tr)
+    return post, scoring, d
+
+  def _run(self, scoring, d, *, force_graph, evals=5000, seed=7):
+    from vizier_amd._src.algorithms.optimizers.eagle import (
+        EagleStrategyConfig)
+    from vizier_amd._src.algorithms.optimizers.vectorized import (
+        VectorizedOptimizerFactory)
+    factory = VectorizedOptimizerFactory(
+        eagle_config=EagleStrategyConfig(), max_evaluations=evals,
+        suggestion_batch_size=25)
+    opt = factory(n_continuous=d, categorical_sizes=[], n_parallel=1,
+                  seed=seed, device='cuda', dtype=torch.float32)
+    if force_graph:
+      opt._megakernel_applicable = lambda score_fn: False
+
+    def score_fn(batch):
+      return scoring(batch.continuous[:, 0, :])
+    score_fn.scoring = scoring
+    score_fn.codec_identity = True
+    res = opt.optimize(score_fn, count=3)
+    return opt, res
+
+  def test_megakernel_bitwise_matches_hipgraph(self, ext):
+    """The persistent cooperative sweep must produce EXACTLY the same
+    pool trajectory as the hipGraph path (same RNG streams, same
+    NCHUNK partition, same reduce orders)."""
+    post, scoring, d = self._make_problem()
+    opt_g, res_g = self._run(scoring, d, force_graph=True)
+    assert opt_g.last_used_graph, opt_g.last_graph_error
+    opt_m, res_m = self._run(scoring, d, force_graph=False)
+    assert opt_m.last_used_megakernel, opt_m.last_graph_error
+    assert torch.equal(res_m.rewards, res_g.rewards)
+    assert torch.equal(res_m.features.continuous,
+                       res_g.features.continuous)
+
+  def test_megakernel_finds_optimum_region(self, ext):
+    post, scoring, d = self._make_problem()
+    opt, res = self._run(scoring, d, force_graph=False, evals=10000)
+    assert opt.last_used_megakernel
+    best = res.features.continuous[0, 0, :]
+    # UCB over a GP fit to -(x-0.4)^2 peaks near 0.4.
+    assert float((best - 0.4).abs().mean()) < 0.2
+
+
 class TestGPUDesignerEndToEnd:
 
   def test_gp_bandit_on_gpu(self, ext):

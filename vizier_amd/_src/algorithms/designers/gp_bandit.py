@@ -332,6 +332,10 @@ class VizierGPBandit(Designer, Predictor):
     def score_fn(batch: CandidateBatch) -> torch.Tensor:
       dense = self._codec.decode(batch)[:, 0, :]
       return scoring(dense)
+    # Expose the scorer internals so the optimizer can run the whole
+    # sweep in the persistent cooperative megakernel (continuous-only).
+    score_fn.scoring = scoring
+    score_fn.codec_identity = self._codec.identity
     return score_fn, 1
 
   def _gp_suggestions(self, count: int) -> List[vz.TrialSuggestion]:

@@ -53,7 +53,22 @@ class VectorizedOptimizer:
     iterations = max(1, (self.max_evaluations - 1) // batch_size + 1)
 
     self.last_used_graph = False
+    self.last_used_megakernel = False
     self.last_graph_error = None
+    if self._megakernel_applicable(score_fn):
+      try:
+        result = self._optimize_megakernel(score_fn, count, state,
+                                           iterations)
+        self.last_used_megakernel = True
+        return result
+      except Exception as e:
+        import logging
+        import traceback
+        logging.getLogger(__name__).warning(
+            'megakernel sweep fell back: %r\n%s', e,
+            traceback.format_exc())
+        self.last_graph_error = repr(e)
+        state = strategy.init_state(prior_features, prior_rewards)
     if strategy._ext is not None and getattr(score_fn, 'graph_safe',
                                              True):
       try:
@@ -101,6 +116,84 @@ class VectorizedOptimizer:
         features=CandidateBatch(all_cont[top.indices],
                                 all_cat[top.indices]),
         rewards=top.values)
+
+  def _megakernel_applicable(self, score_fn: ScoreFn) -> bool:
+    """Persistent-megakernel preconditions: GPU ext, continuous-only,
+    q == 1, and a fused-able ScoringFunction (same conditions as the
+    chunked HIP scorer fast path)."""
+    strategy = self.strategy
+    scoring = getattr(score_fn, 'scoring', None)
+    if (strategy._ext is None or scoring is None or
+        not getattr(score_fn, 'codec_identity', False)):
+      return False
+    if strategy.categorical_sizes or strategy.n_parallel != 1:
+      return False
+    post = scoring.posterior
+    return (scoring._acq_name is not None and scoring._tr_anchored and
+            post.K_inv is not None and post.x.is_cuda and
+            strategy.pool_size <= 128)
+
+  def _optimize_megakernel(self, score_fn: ScoreFn, count: int, state,
+                           iterations: int) -> VectorizedStrategyResults:
+    """Runs the steady-state sweep inside ONE cooperatively-launched
+    persistent kernel (eagle_sweep.hip): suggest -> GP score -> update
+    with grid-wide barriers, zero per-iteration dispatches. The
+    initialization phase (+2 steady iterations, aligning the device
+    counter) runs eagerly like the hipGraph path; the device phases
+    replicate the standalone kernels verbatim, so results are
+    bit-identical to the hipGraph path for the same seeds."""
+    import torch as _torch
+
+    strategy = self.strategy
+    scoring = score_fn.scoring
+    post = scoring.posterior
+    cfg = strategy.config
+    n_batches = strategy.pool_size // strategy.batch_size
+
+    warmup_steady = 2
+    eager_iters = min(iterations, n_batches + warmup_steady)
+    for _ in range(eager_iters):
+      batch = strategy.suggest(state)
+      rewards = score_fn(batch).detach()
+      strategy.update(state, batch, rewards)
+    remaining = iterations - eager_iters
+
+    if remaining > 0:
+      b = strategy.batch_size
+      n = post.x.shape[0]
+      dev = post.x.device
+      inv_ls = (1.0 / post.params.lengthscales).contiguous()
+      k_ws = _torch.empty(b, n, dtype=_torch.float32, device=dev)
+      mu_ws = _torch.empty(b, dtype=_torch.float32, device=dev)
+      dist_ws = _torch.empty(b, dtype=_torch.float32, device=dev)
+      var_ws = _torch.empty(b, 10, dtype=_torch.float32, device=dev)
+      scores = _torch.empty(b, dtype=_torch.float32, device=dev)
+      from vizier_amd._src.ops import dispatch as ops
+      amp2 = scoring._amp * scoring._amp
+      tr_radius = (scoring._tr_radius
+                   if scoring.trust_region is not None else 0.0)
+      strategy._ext.eagle_sweep(
+          state.continuous, state.rewards, state.perturbations,
+          state.best_reward.reshape(1), strategy._iter_t, post.x,
+          inv_ls, post.alpha, post.K_inv,
+          strategy._out_cont, k_ws, mu_ws, dist_ws, var_ws, scores,
+          n_batches, b, strategy.pool_size, state.iterations, remaining,
+          cfg.visibility, cfg.gravity, cfg.negative_gravity,
+          cfg.normalization_scale, cfg.penalize_factor,
+          cfg.perturbation_lower_bound, cfg.perturbation,
+          strategy._seed, strategy._seed ^ 0xABCDEF, amp2,
+          scoring._mean_c, ops.ACQ_CODES[scoring._acq_name],
+          scoring._coef, scoring._best, tr_radius)
+      state.iterations += remaining
+
+    rewards = torch.where(torch.isfinite(state.rewards), state.rewards,
+                          torch.full_like(state.rewards, -float('inf')))
+    k = min(count, rewards.numel())
+    top = torch.topk(rewards, k)
+    return VectorizedStrategyResults(
+        features=CandidateBatch(state.continuous[top.indices].clone(),
+                                state.categorical[top.indices].clone()),
+        rewards=top.values.clone())
 
   def _optimize_hipgraph(self, score_fn: ScoreFn, count: int, state,
                          iterations: int) -> VectorizedStrategyResults:

@@ -58,3 +58,36 @@ __device__ __forceinline__ float block_reduce(float v, float* lds4,
   }
   return v;  // valid in wave 0 lane 0
 }
+
+// Counter-based RNG (splitmix64 -> uniform/laplace). Shared by the
+// standalone Eagle kernels and the persistent sweep megakernel so the
+// two paths draw IDENTICAL streams for the same (seed, offset, idx).
+__device__ __forceinline__ unsigned long long vz_splitmix64(
+    unsigned long long z) {
+  z += 0x9e3779b97f4a7c15ull;
+  z = (z ^ (z >> 30)) * 0xbf58476d1ce4e5b9ull;
+  z = (z ^ (z >> 27)) * 0x94d049bb133111ebull;
+  return z ^ (z >> 31);
+}
+
+__device__ __forceinline__ float vz_rng_uniform(unsigned long long seed,
+                                                unsigned long long offset,
+                                                unsigned int idx) {
+  unsigned long long h = vz_splitmix64(seed ^ vz_splitmix64(offset ^ idx));
+  return ((h >> 40) + 0.5f) * (1.0f / 16777216.0f);  // 24 bits -> (0,1)
+}
+
+__device__ __forceinline__ float vz_rng_laplace(unsigned long long seed,
+                                                unsigned long long offset,
+                                                unsigned int idx) {
+  const float u = vz_rng_uniform(seed, offset, idx) - 0.5f;
+  const float a = fminf(fabsf(u), 0.499999f);
+  return (u >= 0.0f ? -1.0f : 1.0f) * log1pf(-2.0f * a);
+}
+
+__device__ __forceinline__ float vz_normal_cdf(float z) {
+  return 0.5f * erfcf(-z * 0.70710678118654752f);
+}
+__device__ __forceinline__ float vz_normal_pdf(float z) {
+  return 0.3989422804014327f * __expf(-0.5f * z * z);
+}

@@ -19,29 +19,9 @@
 
 // -- counter-based RNG (splitmix64 -> uniform in (0,1)) ---------------------
 
-__device__ __forceinline__ unsigned long long splitmix64(
-    unsigned long long z) {
-  z += 0x9e3779b97f4a7c15ull;
-  z = (z ^ (z >> 30)) * 0xbf58476d1ce4e5b9ull;
-  z = (z ^ (z >> 27)) * 0x94d049bb133111ebull;
-  return z ^ (z >> 31);
-}
-
-__device__ __forceinline__ float rng_uniform(unsigned long long seed,
-                                             unsigned long long offset,
-                                             unsigned int idx) {
-  unsigned long long h = splitmix64(seed ^ splitmix64(offset ^ idx));
-  // 24 high bits -> (0, 1)
-  return ((h >> 40) + 0.5f) * (1.0f / 16777216.0f);
-}
-
-__device__ __forceinline__ float rng_laplace(unsigned long long seed,
-                                             unsigned long long offset,
-                                             unsigned int idx) {
-  const float u = rng_uniform(seed, offset, idx) - 0.5f;
-  const float a = fminf(fabsf(u), 0.499999f);
-  return (u >= 0.0f ? -1.0f : 1.0f) * log1pf(-2.0f * a);
-}
+#define splitmix64 vz_splitmix64
+#define rng_uniform vz_rng_uniform
+#define rng_laplace vz_rng_laplace
 
 // -- suggest -----------------------------------------------------------------
 
