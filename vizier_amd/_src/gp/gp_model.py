@@ -27,6 +27,17 @@ from vizier_amd._src.gp import lbfgs
 from vizier_amd._src.gp.matern import gram_matern52
 from vizier_amd._src.ops import dispatch as ops
 
+# ROCm 7.2 / gfx950: MAGMA's batched Cholesky hits hipErrorLaunchFailure
+# for some (batch, N) shapes (e.g. 3x300x300 — see
+# profiles/cholprobe.log); torch's default heuristic routes certain
+# batched sizes there. hipSOLVER handles every shape we probed, so pin
+# the linalg backend once at import.
+try:
+  if torch.version.hip is not None:
+    torch.backends.cuda.preferred_linalg_library('cusolver')
+except Exception:  # pragma: no cover - older torch without the knob
+  pass
+
 # SoftClip-style bounds (log-space), mirroring tuned_gp_models.py:147-199.
 _LOG_AMP_BOUNDS = (math.log(1e-3), math.log(10.0))
 _LOG_NOISE_BOUNDS = (math.log(1e-10), math.log(1.0))

@@ -54,16 +54,17 @@ extern "C" void launch_eagle_suggest(
 
 extern "C" int launch_eagle_sweep(
     float* pool_cont, float* rewards, float* perturbations,
-    float* best_reward, unsigned long long* iter_ptr, const float* x,
-    const float* inv_ls, const float* alpha, const float* kinv,
-    float* out_cont, float* k_ws, float* mu_ws, float* dist_ws,
-    float* var_ws, float* scores, int n_batches, int batch_size,
-    int pool_size, int dc, int n, long long it_start, long long iterations,
-    float visibility, float gravity, float neg_gravity, float norm_scale,
-    float penalize_factor, float perturbation_lower_bound,
-    float base_perturbation, unsigned long long seed_suggest,
-    unsigned long long seed_update, float amp2, float mean_c, int acq,
-    float coef, float best_value, float tr_radius, hipStream_t stream);
+    float* best_reward, unsigned long long* iter_ptr,
+    unsigned int* barrier_buf, const float* x, const float* inv_ls,
+    const float* alpha, const float* kinv, float* out_cont, float* k_ws,
+    float* mu_ws, float* dist_ws, float* var_ws, int n_batches,
+    int batch_size, int pool_size, int dc, int n, long long it_start,
+    long long iterations, float visibility, float gravity,
+    float neg_gravity, float norm_scale, float penalize_factor,
+    float perturbation_lower_bound, float base_perturbation,
+    unsigned long long seed_suggest, unsigned long long seed_update,
+    float amp2, float mean_c, int acq, float coef, float best_value,
+    float tr_radius, hipStream_t stream);
 
 extern "C" void launch_eagle_update(
     float* pool_cont, long* pool_cat, float* rewards, float* perturbations,
@@ -332,10 +333,11 @@ void eagle_update(torch::Tensor pool_cont, torch::Tensor pool_cat,
 int64_t eagle_sweep(
     torch::Tensor pool_cont, torch::Tensor rewards,
     torch::Tensor perturbations, torch::Tensor best_reward,
-    torch::Tensor iter_counter, torch::Tensor x, torch::Tensor inv_ls,
+    torch::Tensor iter_counter, torch::Tensor barrier_buf,
+    torch::Tensor x, torch::Tensor inv_ls,
     torch::Tensor alpha, torch::Tensor kinv, torch::Tensor out_cont,
     torch::Tensor k_ws, torch::Tensor mu_ws, torch::Tensor dist_ws,
-    torch::Tensor var_ws, torch::Tensor scores, int64_t n_batches,
+    torch::Tensor var_ws, int64_t n_batches,
     int64_t batch_size, int64_t pool_size, int64_t it_start,
     int64_t iterations, double visibility, double gravity,
     double neg_gravity, double norm_scale, double penalize_factor,
@@ -350,20 +352,24 @@ int64_t eagle_sweep(
   inv_ls = check_f32(inv_ls, "inv_ls");
   alpha = check_f32(alpha, "alpha");
   kinv = check_f32(kinv, "kinv");
+  TORCH_CHECK(barrier_buf.scalar_type() == torch::kInt32 &&
+              barrier_buf.numel() >= 2, "barrier_buf must be int32 (2,)");
   const int dc = x.size(1);
   const int n = x.size(0);
   TORCH_CHECK(pool_cont.numel() == pool_size * dc,
               "pool/feature shape mismatch (continuous-only, q=1)");
   TORCH_CHECK(pool_size <= 128, "eagle_sweep supports pool <= 128");
+  TORCH_CHECK(n <= 8192, "eagle_sweep stages k in LDS: N <= 8192");
   const int ret = launch_eagle_sweep(
       pool_cont.data_ptr<float>(), rewards.data_ptr<float>(),
       perturbations.data_ptr<float>(), best_reward.data_ptr<float>(),
       (unsigned long long*)iter_counter.data_ptr<int64_t>(),
+      (unsigned int*)barrier_buf.data_ptr<int>(),
       x.data_ptr<float>(), inv_ls.data_ptr<float>(),
       alpha.data_ptr<float>(), kinv.data_ptr<float>(),
       out_cont.data_ptr<float>(), k_ws.data_ptr<float>(),
       mu_ws.data_ptr<float>(), dist_ws.data_ptr<float>(),
-      var_ws.data_ptr<float>(), scores.data_ptr<float>(),
+      var_ws.data_ptr<float>(),
       (int)n_batches, (int)batch_size, (int)pool_size, dc, n,
       (long long)it_start, (long long)iterations, (float)visibility,
       (float)gravity, (float)neg_gravity, (float)norm_scale,
