@@ -115,15 +115,16 @@ def probe_d():
   mu_ws = torch.empty(b, device='cuda')
   dist_ws = torch.empty(b, device='cuda')
   var_ws = torch.empty(b, 10, device='cuda')
-  scores = torch.empty(b, device='cuda')
+  barrier_buf = torch.zeros(2, dtype=torch.int32, device='cuda')
   cfg = strategy.config
 
   def run(iters):
     return ext.eagle_sweep(
         state.continuous, state.rewards, state.perturbations,
-        state.best_reward.reshape(1), strategy._iter_t, post.x, inv_ls,
+        state.best_reward.reshape(1), strategy._iter_t, barrier_buf,
+        post.x, inv_ls,
         post.alpha, post.K_inv, strategy._out_cont, k_ws, mu_ws,
-        dist_ws, var_ws, scores, strategy.pool_size // 25, b,
+        dist_ws, var_ws, strategy.pool_size // 25, b,
         strategy.pool_size, state.iterations, iters, cfg.visibility,
         cfg.gravity, cfg.negative_gravity, cfg.normalization_scale,
         cfg.penalize_factor, cfg.perturbation_lower_bound,
