@@ -66,6 +66,7 @@ class GPBanditConfig:
   ard_restarts: int = 4
   ard_max_iters: int = 50
   ard_warm_iters: int = 12   # iters when warm-starting from the last fit
+  ard_warm_restarts: int = 2  # random restarts kept on warm refits
   use_trust_region: bool = True
   num_scalarizations: int = 1000  # multi-objective
   data_parallel: bool = False     # shard the sweep across dist ranks
@@ -207,8 +208,13 @@ class VizierGPBandit(Designer, Predictor):
         y = y - prior_mean
       warm = prev[m].raw if m < len(prev) and prev[m].raw is not None \
           else None
+      # Warm refits: the previous optimum is almost always the winner,
+      # so keep only a couple of random restarts — the restart batch
+      # multiplies the per-iteration Cholesky cost (R x N^3).
       post = gp_model.train_gp(
-          x, y, num_restarts=cfg.ard_restarts,
+          x, y,
+          num_restarts=cfg.ard_warm_restarts if warm is not None
+          else cfg.ard_restarts,
           max_iters=cfg.ard_warm_iters if warm is not None
           else cfg.ard_max_iters,
           seed=self._seed, warm_start_raw=warm)

@@ -59,6 +59,7 @@ class UCBPEConfig:
   ard_restarts: int = 4
   ard_max_iters: int = 50
   ard_warm_iters: int = 12
+  ard_warm_restarts: int = 2
   use_trust_region: bool = True
   device: Optional[str] = None
   dtype: torch.dtype = torch.float32
@@ -162,7 +163,9 @@ class VizierGPUCBPEBandit(Designer):
     y = torch.as_tensor(y_np, dtype=cfg.dtype, device=self._device)
     warm = self._posterior.raw if self._posterior is not None else None
     self._posterior = gp_model.train_gp(
-        x, y, num_restarts=cfg.ard_restarts,
+        x, y,
+        num_restarts=cfg.ard_warm_restarts if warm is not None
+        else cfg.ard_restarts,
         max_iters=cfg.ard_warm_iters if warm is not None
         else cfg.ard_max_iters, seed=self._seed, warm_start_raw=warm)
     self._warped_labels = y
