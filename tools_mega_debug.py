@@ -146,7 +146,7 @@ def probe_d():
 if __name__ == '__main__':
   which = sys.argv[1] if len(sys.argv) > 1 else 'all'
   if which == 'all':
-    for p in 'abcd':
+    for p in 'abcde':
       r = subprocess.run(
           ['timeout', '180', sys.executable, __file__, p],
           capture_output=True, text=True)
@@ -156,3 +156,83 @@ if __name__ == '__main__':
         print('STDERR:', r.stderr[-1500:])
   else:
     globals()[f'probe_{which}']()
+
+
+def probe_e():
+  """Bisect: first in-kernel iteration where megakernel diverges from
+  the eager ext path (bitwise — eager steady-state uses the same
+  standalone kernels the graph captures)."""
+  import torch
+  from vizier_amd._src.ops import dispatch
+  ext = dispatch.require_ext()
+  from vizier_amd._src.gp import acquisitions as acq_lib
+  from vizier_amd._src.gp import gp_model
+  from vizier_amd._src.algorithms.optimizers.eagle import (
+      EagleStrategyConfig, VectorizedEagleStrategy)
+  from vizier_amd._src.ops import dispatch as ops
+  g = torch.Generator().manual_seed(0)
+  x = torch.rand(300, 12, generator=g).cuda()
+  y = (-((x - 0.4) ** 2).sum(-1) + 0.01 * torch.randn(300, generator=g).cuda())
+  post = gp_model.train_gp(x, y, num_restarts=2, max_iters=15)
+  onehot = torch.zeros(12, dtype=torch.bool, device='cuda')
+  tr = acq_lib.TrustRegion(post.x, onehot)
+  scoring = acq_lib.ScoringFunction(post, acq_lib.UCB(coefficient=1.8), tr)
+
+  def score_fn(batch):
+    return scoring(batch.continuous[:, 0, :])
+
+  def make(seed=7):
+    s = VectorizedEagleStrategy(
+        n_continuous=12, categorical_sizes=[], batch_size=25,
+        config=EagleStrategyConfig(), n_parallel=1, seed=seed,
+        device='cuda')
+    st = s.init_state()
+    nb = s.pool_size // s.batch_size
+    for _ in range(nb + 2):
+      b = s.suggest(st)
+      r = score_fn(b).detach()
+      s.update(st, b, r)
+    return s, st
+
+  for T in (1, 2, 3, 5, 10, 30, 100):
+    sA, stA = make()
+    for _ in range(T):
+      b = sA.suggest(stA)
+      r = score_fn(b).detach()
+      sA.update(stA, b, r)
+    sB, stB = make()
+    n = post.x.shape[0]
+    bsz = sB.batch_size
+    inv_ls = (1.0 / post.params.lengthscales).contiguous()
+    k_ws = torch.empty(bsz, n, device='cuda')
+    mu_ws = torch.empty(bsz, device='cuda')
+    dist_ws = torch.empty(bsz, device='cuda')
+    var_ws = torch.empty(bsz, 10, device='cuda')
+    bar = torch.zeros(2, dtype=torch.int32, device='cuda')
+    cfg = sB.config
+    ext.eagle_sweep(
+        stB.continuous, stB.rewards, stB.perturbations,
+        stB.best_reward.reshape(1), sB._iter_t, bar, post.x, inv_ls,
+        post.alpha, post.K_inv, sB._out_cont, k_ws, mu_ws, dist_ws,
+        var_ws, sB.pool_size // bsz, bsz, sB.pool_size,
+        stB.iterations, T, cfg.visibility, cfg.gravity,
+        cfg.negative_gravity, cfg.normalization_scale,
+        cfg.penalize_factor, cfg.perturbation_lower_bound,
+        cfg.perturbation, sB._seed, sB._seed ^ 0xABCDEF,
+        scoring._amp * scoring._amp, scoring._mean_c,
+        ops.ACQ_CODES['ucb'], scoring._coef, scoring._best, 0.24)
+    torch.cuda.synchronize()
+    pe = torch.equal(stA.continuous, stB.continuous)
+    re = torch.equal(stA.rewards, stB.rewards)
+    if pe and re:
+      print(f'T={T}: identical', flush=True)
+    else:
+      dr = (stA.rewards - stB.rewards).abs()
+      dp = (stA.continuous - stB.continuous).abs().amax(dim=(1, 2))
+      bad = torch.nonzero(dr + dp > 0).flatten().tolist()
+      print(f'T={T}: DIVERGED slots={bad[:8]} '
+            f'dr={[round(float(dr[i]),6) for i in bad[:4]]} '
+            f'dp={[round(float(dp[i]),6) for i in bad[:4]]} ', flush=True)
+      print('  A rewards:', [round(float(stA.rewards[i]), 6) for i in bad[:4]])
+      print('  B rewards:', [round(float(stB.rewards[i]), 6) for i in bad[:4]])
+      break
