@@ -197,6 +197,8 @@ class TestEagleSweepMegakernel:
     return post, scoring, d
 
   def _run(self, scoring, d, *, force_graph, evals=5000, seed=7):
+    import os
+    os.environ['VIZIER_AMD_MEGAKERNEL'] = '1'
     from vizier_amd._src.algorithms.optimizers.eagle import (
         EagleStrategyConfig)
     from vizier_amd._src.algorithms.optimizers.vectorized import (

@@ -121,6 +121,11 @@ class VectorizedOptimizer:
     """Persistent-megakernel preconditions: GPU ext, continuous-only,
     q == 1, and a fused-able ScoringFunction (same conditions as the
     chunked HIP scorer fast path)."""
+    import os
+    if os.environ.get('VIZIER_AMD_MEGAKERNEL', '0') != '1':
+      # Experimental until it beats the hipGraph path (see DESIGN.md);
+      # opt in with VIZIER_AMD_MEGAKERNEL=1.
+      return False
     strategy = self.strategy
     scoring = getattr(score_fn, 'scoring', None)
     if (strategy._ext is None or scoring is None or
