@@ -143,21 +143,6 @@ def probe_d():
           flush=True)
 
 
-if __name__ == '__main__':
-  which = sys.argv[1] if len(sys.argv) > 1 else 'all'
-  if which == 'all':
-    for p in 'abcde':
-      r = subprocess.run(
-          ['timeout', '180', sys.executable, __file__, p],
-          capture_output=True, text=True)
-      print(f'--- probe {p} (rc={r.returncode}) ---')
-      print(r.stdout[-2000:])
-      if r.returncode != 0:
-        print('STDERR:', r.stderr[-1500:])
-  else:
-    globals()[f'probe_{which}']()
-
-
 def probe_e():
   """Bisect: first in-kernel iteration where megakernel diverges from
   the eager ext path (bitwise — eager steady-state uses the same
@@ -236,3 +221,18 @@ def probe_e():
       print('  A rewards:', [round(float(stA.rewards[i]), 6) for i in bad[:4]])
       print('  B rewards:', [round(float(stB.rewards[i]), 6) for i in bad[:4]])
       break
+
+
+if __name__ == '__main__':
+  which = sys.argv[1] if len(sys.argv) > 1 else 'all'
+  if which == 'all':
+    for p in 'abcde':
+      r = subprocess.run(
+          ['timeout', '180', sys.executable, __file__, p],
+          capture_output=True, text=True)
+      print(f'--- probe {p} (rc={r.returncode}) ---')
+      print(r.stdout[-2000:])
+      if r.returncode != 0:
+        print('STDERR:', r.stderr[-1500:])
+  else:
+    globals()[f'probe_{which}']()
