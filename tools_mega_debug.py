@@ -205,7 +205,8 @@ def probe_e():
         cfg.penalize_factor, cfg.perturbation_lower_bound,
         cfg.perturbation, sB._seed, sB._seed ^ 0xABCDEF,
         scoring._amp * scoring._amp, scoring._mean_c,
-        ops.ACQ_CODES['ucb'], scoring._coef, scoring._best, 0.24)
+        ops.ACQ_CODES['ucb'], scoring._coef, scoring._best,
+        scoring._tr_radius)
     torch.cuda.synchronize()
     pe = torch.equal(stA.continuous, stB.continuous)
     re = torch.equal(stA.rewards, stB.rewards)
@@ -221,6 +222,89 @@ def probe_e():
       print('  A rewards:', [round(float(stA.rewards[i]), 6) for i in bad[:4]])
       print('  B rewards:', [round(float(stB.rewards[i]), 6) for i in bad[:4]])
       break
+
+
+def probe_f():
+  """One in-kernel iteration; compare suggest output and scores against
+  the eager ext path phase by phase."""
+  import torch
+  from vizier_amd._src.ops import dispatch
+  ext = dispatch.require_ext()
+  from vizier_amd._src.gp import acquisitions as acq_lib
+  from vizier_amd._src.gp import gp_model
+  from vizier_amd._src.algorithms.optimizers.eagle import (
+      EagleStrategyConfig, VectorizedEagleStrategy)
+  from vizier_amd._src.ops import dispatch as ops
+  g = torch.Generator().manual_seed(0)
+  x = torch.rand(300, 12, generator=g).cuda()
+  y = (-((x - 0.4) ** 2).sum(-1) + 0.01 * torch.randn(300, generator=g).cuda())
+  post = gp_model.train_gp(x, y, num_restarts=2, max_iters=15)
+  onehot = torch.zeros(12, dtype=torch.bool, device='cuda')
+  tr = acq_lib.TrustRegion(post.x, onehot)
+  scoring = acq_lib.ScoringFunction(post, acq_lib.UCB(coefficient=1.8), tr)
+
+  def score_fn(batch):
+    return scoring(batch.continuous[:, 0, :])
+
+  def make(seed=7):
+    s = VectorizedEagleStrategy(
+        n_continuous=12, categorical_sizes=[], batch_size=25,
+        config=EagleStrategyConfig(), n_parallel=1, seed=seed,
+        device='cuda')
+    st = s.init_state()
+    nb = s.pool_size // s.batch_size
+    for _ in range(nb + 2):
+      b = s.suggest(st)
+      r = score_fn(b).detach()
+      s.update(st, b, r)
+    return s, st
+
+  # Eager reference for ONE more iteration (keep copies BEFORE).
+  sA, stA = make()
+  batch = sA.suggest(stA)
+  xq = batch.continuous[:, 0, :].clone()
+  rewards_eager = score_fn(batch).detach().clone()
+  sA.update(stA, batch, rewards_eager)
+
+  # Megakernel ONE iteration from the same warm state.
+  sB, stB = make()
+  n, bsz = 300, 25
+  inv_ls = (1.0 / post.params.lengthscales).contiguous()
+  k_ws = torch.empty(bsz, n, device='cuda')
+  mu_ws = torch.empty(bsz, device='cuda')
+  dist_ws = torch.empty(bsz, device='cuda')
+  var_ws = torch.empty(bsz, 10, device='cuda')
+  bar = torch.zeros(2, dtype=torch.int32, device='cuda')
+  cfg = sB.config
+  ext.eagle_sweep(
+      stB.continuous, stB.rewards, stB.perturbations,
+      stB.best_reward.reshape(1), sB._iter_t, bar, post.x, inv_ls,
+      post.alpha, post.K_inv, sB._out_cont, k_ws, mu_ws, dist_ws,
+      var_ws, sB.pool_size // bsz, bsz, sB.pool_size,
+      stB.iterations, 1, cfg.visibility, cfg.gravity,
+      cfg.negative_gravity, cfg.normalization_scale,
+      cfg.penalize_factor, cfg.perturbation_lower_bound,
+      cfg.perturbation, sB._seed, sB._seed ^ 0xABCDEF,
+      scoring._amp * scoring._amp, scoring._mean_c,
+      ops.ACQ_CODES['ucb'], scoring._coef, scoring._best,
+      scoring._tr_radius)
+  torch.cuda.synchronize()
+  out_mega = sB._out_cont[:, 0, :] if sB._out_cont.dim() == 3 else       sB._out_cont
+  print('suggest equal:', torch.equal(out_mega, xq),
+        'maxdiff', float((out_mega - xq).abs().max()))
+  # Scores: recompute finalize from the megakernel partials.
+  var = var_ws.sum(-1)
+  amp2 = scoring._amp * scoring._amp
+  sd = (amp2 - var).clamp_min(1e-12).sqrt()
+  mu = mu_ws + scoring._mean_c
+  sc = mu + 1.8 * sd
+  pen = dist_ws > scoring._tr_radius
+  sc = torch.where(pen, -1e4 - dist_ws, sc)
+  print('scores equal:', torch.equal(sc, rewards_eager),
+        'maxdiff', float((sc - rewards_eager).abs().max()))
+  print('pool equal:', torch.equal(stA.continuous, stB.continuous),
+        'rewards equal:', torch.equal(stA.rewards, stB.rewards),
+        'rdiff', float((stA.rewards - stB.rewards).abs().max()))
 
 
 if __name__ == '__main__':
