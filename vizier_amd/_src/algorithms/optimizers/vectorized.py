@@ -238,7 +238,9 @@ class VectorizedOptimizer:
       graph = torch.cuda.CUDAGraph()
       with torch.cuda.graph(graph):
         one_iter()
-      remaining -= 1
+      # Stream capture DEFERS execution to replay — the captured
+      # iteration has not run yet, so it must not consume a slot
+      # (verified bitwise against the eager and megakernel paths).
       for _ in range(remaining):
         graph.replay()
 

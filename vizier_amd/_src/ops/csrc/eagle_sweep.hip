@@ -60,29 +60,44 @@ __device__ __forceinline__ void cstore(float* p, float v) {
 }
 
 // Sense-reversing grid barrier on agent-scope atomics. bar[0] = arrival
-// count, bar[1] = generation. Every thread first drains its own
-// outstanding global ops (s_waitcnt 0), so all relaxed data stores are
-// at the coherence point before the generation flips.
+// count, bar[1] = generation.
+//
+// Deliberately RELAXED atomics: acq_rel orderings compile to
+// buffer_wbl2 / buffer_inv (full per-XCD L2 writeback+invalidate),
+// which costs tens of microseconds per barrier AND evicts the
+// L2-resident K^-1 that phase B depends on. Ordering is instead
+// established by hand:
+//   - every thread drains its own outstanding global ops
+//     (s_waitcnt 0) before the block barrier, so all sc1 data stores
+//     have reached the coherence point before tid 0 arrives;
+//   - all cross-workgroup data moves through sc1 accesses (cload /
+//     cstore), which always read/write the coherence point, so no
+//     cache invalidation is needed on the acquire side;
+//   - waves issue memory ops in order, so the spin-exit load ordering
+//     is sufficient (compiler reordering is fenced with asm).
 __device__ __forceinline__ void grid_sync(unsigned int* bar) {
   __builtin_amdgcn_s_waitcnt(0);
   __syncthreads();
   if (threadIdx.x == 0) {
     const unsigned int gen =
-        __hip_atomic_load(bar + 1, __ATOMIC_ACQUIRE,
+        __hip_atomic_load(bar + 1, __ATOMIC_RELAXED,
                           __HIP_MEMORY_SCOPE_AGENT);
+    asm volatile("" ::: "memory");
     const unsigned int arrived = __hip_atomic_fetch_add(
-        bar, 1u, __ATOMIC_ACQ_REL, __HIP_MEMORY_SCOPE_AGENT);
+        bar, 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
     if (arrived == gridDim.x - 1) {
       __hip_atomic_store(bar, 0u, __ATOMIC_RELAXED,
                          __HIP_MEMORY_SCOPE_AGENT);
-      __hip_atomic_fetch_add(bar + 1, 1u, __ATOMIC_RELEASE,
+      __builtin_amdgcn_s_waitcnt(0);  // count reset lands before gen++
+      __hip_atomic_fetch_add(bar + 1, 1u, __ATOMIC_RELAXED,
                              __HIP_MEMORY_SCOPE_AGENT);
     } else {
-      while (__hip_atomic_load(bar + 1, __ATOMIC_ACQUIRE,
+      while (__hip_atomic_load(bar + 1, __ATOMIC_RELAXED,
                                __HIP_MEMORY_SCOPE_AGENT) == gen) {
         __builtin_amdgcn_s_sleep(8);
       }
     }
+    asm volatile("" ::: "memory");
   }
   __syncthreads();
 }
