@@ -18,9 +18,12 @@ if __name__ == '__main__':
   if len(sys.argv) > 1:
     case(int(sys.argv[1]), int(sys.argv[2]), sys.argv[3])
   else:
-    for backend in ('default', 'cusolver', 'magma'):
-      for (r, n) in [(1, 300), (3, 300), (3, 100), (3, 25), (5, 1000),
-                     (3, 513), (24, 300), (3, 1000)]:
+    import itertools
+    ns = [128, 160, 200, 255, 256, 257, 288, 300, 320, 384, 448,
+          511, 512, 513, 576, 640, 768, 900, 1000, 1024, 1500]
+    for backend in ('default',):
+      for (r, n) in itertools.chain(((4, n) for n in ns),
+                                    ((2, n) for n in (300, 400, 500))):
         p = subprocess.run(['timeout', '120', sys.executable, __file__,
                             str(r), str(n), backend],
                            capture_output=True, text=True)
