@@ -122,9 +122,10 @@ class VectorizedOptimizer:
     q == 1, and a fused-able ScoringFunction (same conditions as the
     chunked HIP scorer fast path)."""
     import os
-    if os.environ.get('VIZIER_AMD_MEGAKERNEL', '0') != '1':
-      # Experimental until it beats the hipGraph path (see DESIGN.md);
-      # opt in with VIZIER_AMD_MEGAKERNEL=1.
+    if os.environ.get('VIZIER_AMD_MEGAKERNEL', '1') != '1':
+      # Default ON: 68 us/iteration vs 85 for the hipGraph replay
+      # (profiles/megatime3.log), bit-identical results (GPU test).
+      # Set VIZIER_AMD_MEGAKERNEL=0 to force the hipGraph path.
       return False
     strategy = self.strategy
     scoring = getattr(score_fn, 'scoring', None)

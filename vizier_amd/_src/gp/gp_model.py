@@ -27,16 +27,31 @@ from vizier_amd._src.gp import lbfgs
 from vizier_amd._src.gp.matern import gram_matern52
 from vizier_amd._src.ops import dispatch as ops
 
-# ROCm 7.2 / gfx950: MAGMA's batched Cholesky hits hipErrorLaunchFailure
-# for some (batch, N) shapes (e.g. 3x300x300 — see
-# profiles/cholprobe.log); torch's default heuristic routes certain
-# batched sizes there. hipSOLVER handles every shape we probed, so pin
-# the linalg backend once at import.
-try:
-  if torch.version.hip is not None:
-    torch.backends.cuda.preferred_linalg_library('cusolver')
-except Exception:  # pragma: no cover - older torch without the knob
-  pass
+# ROCm 7.2 / gfx950: MAGMA's batched Cholesky (the default backend for
+# batched inputs) hits hipErrorLaunchFailure for batch > 1 when
+# 257 <= N <= ~320 (mapped in profiles/cholprobe2.log; 256 and 384+ are
+# fine). Pinning hipSOLVER instead costs 4x on every warm ARD fit
+# (profiles/fitprobe.log: 784 vs 201 ms), so we keep the fast default
+# backend and identity-pad batched factorizations through the crash
+# window — block-diagonal padding is exact: chol([[K,0],[0,I]]) =
+# [[chol(K),0],[0,I]], logdet unchanged, and we slice the result back.
+_MAGMA_BAD_LO, _MAGMA_BAD_HI = 257, 511  # generous upper margin
+_MAGMA_PAD_N = 512
+
+
+def safe_cholesky_ex(K: torch.Tensor):
+  """torch.linalg.cholesky_ex that avoids MAGMA's batched crash window."""
+  n = K.shape[-1]
+  if (K.is_cuda and K.dim() > 2 and K.shape[0] > 1 and
+      _MAGMA_BAD_LO <= n <= _MAGMA_BAD_HI):
+    shape = K.shape[:-2] + (_MAGMA_PAD_N, _MAGMA_PAD_N)
+    Kp = torch.zeros(shape, dtype=K.dtype, device=K.device)
+    Kp[..., :n, :n] = K
+    idx = torch.arange(n, _MAGMA_PAD_N, device=K.device)
+    Kp[..., idx, idx] = 1.0
+    L, info = torch.linalg.cholesky_ex(Kp)
+    return L[..., :n, :n].contiguous(), info
+  return torch.linalg.cholesky_ex(K)
 
 # SoftClip-style bounds (log-space), mirroring tuned_gp_models.py:147-199.
 _LOG_AMP_BOUNDS = (math.log(1e-3), math.log(10.0))
@@ -101,7 +116,7 @@ def cholesky_with_jitter(K: torch.Tensor, amplitude2: torch.Tensor,
   jitter = 1e-6
   eye = torch.eye(K.shape[-1], dtype=K.dtype, device=K.device)
   for _ in range(max_tries):
-    L, info = torch.linalg.cholesky_ex(K)
+    L, info = safe_cholesky_ex(K)
     if bool((info == 0).all()):
       return L
     scale = amplitude2.reshape(amplitude2.shape + (1, 1)) \
@@ -110,7 +125,7 @@ def cholesky_with_jitter(K: torch.Tensor, amplitude2: torch.Tensor,
     K = K + bad * jitter * scale * eye
     jitter *= 10.0
   # Last resort: return whatever factorization we can get.
-  return torch.linalg.cholesky_ex(K)[0]
+  return safe_cholesky_ex(K)[0]
 
 
 def negative_log_marginal_likelihood(raw: torch.Tensor, x: torch.Tensor,
@@ -122,7 +137,7 @@ def negative_log_marginal_likelihood(raw: torch.Tensor, x: torch.Tensor,
                     params.amplitude)
   noise = params.noise.reshape(-1, 1, 1)
   K = K + noise * torch.eye(n, dtype=x.dtype, device=x.device)
-  L, info = torch.linalg.cholesky_ex(K)
+  L, info = safe_cholesky_ex(K)
   resid = (y.unsqueeze(0) - params.mean.unsqueeze(-1)).unsqueeze(-1)
   # NOTE: torch.cholesky_solve hits a hipErrorLaunchFailure on this ROCm
   # build (see profiles/notes); two triangular solves are equivalent.

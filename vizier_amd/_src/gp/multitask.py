@@ -113,7 +113,7 @@ def negative_log_marginal_likelihood(raw: torch.Tensor, x: torch.Tensor,
   m = y.shape[1]
   params = MultitaskParams.from_raw(raw, d, m, kind)
   k = _joint_cov(params, x)
-  L, info = torch.linalg.cholesky_ex(k)
+  L, info = gp_model.safe_cholesky_ex(k)
   resid = (y.unsqueeze(0) - params.means.unsqueeze(-2)).reshape(
       raw.shape[0], n * m, 1)
   alpha = gp_model._chol_solve(L, resid)
