@@ -229,3 +229,64 @@ class TestVizierAlias:
     assert _clients.Study is not None
     assert _pythia.Policy is not None
     assert _benchmarks.BenchmarkRunner is not None
+
+
+class TestFaultInjection:
+  """Failing designers exercise the service's error-capture path
+  (Pythia exceptions land in op.error)."""
+
+  def _servicer_with_failing_policy(self):
+    from vizier_amd._src.algorithms.policies import designer_policy as dp
+    from vizier_amd._src.algorithms.testing import failing
+    from vizier_amd._src.service import pythia_service, vizier_service
+    from vizier_amd._src.service.policy_factory import PolicyFactory
+
+    class FailingFactory(PolicyFactory):
+
+      def __call__(self, problem_statement, algorithm, policy_supporter,
+                   study_name):
+        return dp.DesignerPolicy(
+            policy_supporter,
+            lambda problem, **kw: failing.FailingDesigner())
+
+    pythia = pythia_service.PythiaServicer(
+        policy_factory=FailingFactory())
+    servicer = vizier_service.VizierServicer(
+        database_url=None, default_pythia_service=pythia)
+    pythia.connect_to_vizier(servicer)
+    return servicer
+
+  def test_suggest_error_captured_in_operation(self):
+    from vizier_amd import pyvizier as vz
+    from vizier_amd._src.service.proto import (
+        study_pb2,
+        vizier_service_pb2,
+    )
+    servicer = self._servicer_with_failing_policy()
+    config = vz.StudyConfig(algorithm='RANDOM_SEARCH')
+    config.search_space.root.add_float_param('x', 0.0, 1.0)
+    config.metric_information.append(vz.MetricInformation(
+        name='obj', goal=vz.ObjectiveMetricGoal.MAXIMIZE))
+    study = study_pb2.Study(display_name='s')
+    study.study_spec.CopyFrom(config.to_proto())
+    created = servicer.CreateStudy(
+        vizier_service_pb2.CreateStudyRequest(parent='owners/o',
+                                              study=study))
+    op = servicer.SuggestTrials(vizier_service_pb2.SuggestTrialsRequest(
+        parent=created.name, suggestion_count=1, client_id='w'))
+    assert op.done
+    assert op.error.message, 'expected the designer failure in op.error'
+    assert 'FailedSuggestError' in op.error.message or \
+        op.error.code != 0
+
+  def test_alternate_failing_designer_alternates(self):
+    from vizier_amd import pyvizier as vz
+    from vizier_amd._src.algorithms.testing import failing
+    space = vz.SearchSpace()
+    space.root.add_float_param('x', 0.0, 1.0)
+    d = failing.AlternateFailingDesigner(space)
+    first = d.suggest(1)
+    assert len(first) == 1
+    import pytest as _pytest
+    with _pytest.raises(failing.FailedSuggestError):
+      d.suggest(1)

@@ -240,7 +240,11 @@ class VizierServicer:
         decision_proto = pythia_service.Suggest(request_proto)
       except grpc.RpcError as e:
         logger.exception('Pythia failed to suggest trials')
-        return _finish_with_error(str(e))
+        # Prefer gRPC details (carries the exception type for in-process
+        # LocalRpcError, whose str() may be empty).
+        details = e.details() if callable(getattr(e, 'details', None)) \
+            else None
+        return _finish_with_error(details or str(e) or type(e).__name__)
 
       decision = pythia_converters.SuggestConverter.from_decision_proto(
           decision_proto)
