@@ -59,7 +59,26 @@ class TestPosteriorScoreKernel:
                       mean=post.params.mean.cuda())
     return GPPosterior(x=post.x.cuda(), params=params, L=post.L.cuda(),
                        alpha=post.alpha.cuda(),
-                       K_inv=post.K_inv.cuda(), nll=post.nll)
+                       K_inv=post.K_inv.cuda(), nll=post.nll,
+                       noise_eff=post.noise_eff)
+
+  def _oracle_mean_std(self, post, xq):
+    """float64 posterior oracle with the cache's exact noise floor.
+    The fp32 K_inv quadform (kernel AND torch fast path) carries a
+    documented cancellation error ~6e-5*amp^2 in variance near the
+    floor (see train_gp), so fp32-vs-fp32 comparisons are circular —
+    compare both against this instead."""
+    from vizier_amd._src.gp.matern import gram_matern52
+    x = post.x.double()
+    ls = post.params.lengthscales.double()
+    amp = post.params.amplitude.double()
+    K = gram_matern52(x, None, ls, amp)
+    K = K + post.noise_eff * torch.eye(x.shape[0], dtype=torch.float64)
+    k = gram_matern52(xq.double(), x, ls, amp)
+    sol = torch.linalg.solve(K, k.T)
+    mean = post.params.mean.double() + k @ post.alpha.double()
+    var = (amp * amp - (k * sol.T).sum(-1)).clamp_min(1e-12)
+    return mean.float(), var.sqrt().float()
 
   @pytest.mark.parametrize('acq,code', [('ucb', 0), ('lcb', 1), ('ei', 2),
                                         ('pi', 3)])
@@ -68,7 +87,7 @@ class TestPosteriorScoreKernel:
     gpost = self._to_cuda(post)
     g = torch.Generator().manual_seed(2)
     xq = torch.rand(64, 10, generator=g)
-    mean, stddev = post.predict(xq)
+    mean, stddev = self._oracle_mean_std(post, xq)
     if acq == 'ucb':
       want = mean + 1.8 * stddev
     elif acq == 'lcb':
@@ -85,8 +104,13 @@ class TestPosteriorScoreKernel:
         xq.cuda(), gpost.x, gpost.params.lengthscales,
         float(gpost.params.amplitude), float(gpost.params.mean),
         gpost.alpha, gpost.K_inv, onehot, code, 1.8, 0.5, 0.0)
+    # fp32 quadform bound: ~6e-5*amp^2 variance error -> ~1e-3*amp
+    # stddev error away from degeneracy; PI/EI amplify via z.
+    amp = float(post.params.amplitude)
+    tol = {'ucb': 0.02 * amp, 'lcb': 0.02 * amp,
+           'ei': 0.02 * amp, 'pi': 0.05}[acq]
     err = (got.cpu() - want).abs().max()
-    assert float(err) < 5e-3, f'{acq}: max err {err}'
+    assert float(err) < tol, f'{acq}: max err {err}'
 
   def test_trust_region_penalty_matches(self, ext):
     from vizier_amd._src.gp import acquisitions as acq_lib
@@ -113,8 +137,10 @@ class TestPosteriorScoreKernel:
     xq = torch.rand(32, 10).cuda()
     assert scoring._can_fuse(xq)
     got = scoring(xq)
-    mean, stddev = post.predict(xq.cpu())
-    assert torch.allclose(got.cpu(), mean + 1.8 * stddev, atol=5e-3)
+    mean, stddev = self._oracle_mean_std(post, xq.cpu())
+    amp = float(post.params.amplitude)
+    assert torch.allclose(got.cpu(), mean + 1.8 * stddev,
+                          atol=0.02 * amp)
 
 
 class TestEagleKernels:

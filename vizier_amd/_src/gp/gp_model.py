@@ -164,6 +164,7 @@ class GPPosterior:
   K_inv: Optional[torch.Tensor]  # (N, N), for the GEMM variance path
   nll: float                 # training loss of the selected restart
   raw: Optional[torch.Tensor] = None  # unconstrained params (warm starts)
+  noise_eff: float = 0.0     # cache noise floor actually used (see train_gp)
 
   def predict(self, xq: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
     """Returns (mean, stddev) of the latent f at query points (Q, D)."""
@@ -253,4 +254,5 @@ def train_gp(x: torch.Tensor, y: torch.Tensor, *,
     K_inv = (z.T @ z).to(x.dtype)
   return GPPosterior(x=x, params=params, L=L64.to(x.dtype),
                      alpha=alpha.to(x.dtype), K_inv=K_inv,
-                     nll=float(best_f[idx]), raw=raw.detach())
+                     nll=float(best_f[idx]), raw=raw.detach(),
+                     noise_eff=float(noise_eff))
