@@ -1,0 +1,31 @@
+"""Public experimenters API (parity with vizier/benchmarks/experimenters)."""
+
+from vizier_amd._src.benchmarks.experimenters import experimenter_factory
+from vizier_amd._src.benchmarks.experimenters.combo import (
+    CentroidExperimenter,
+    ContaminationExperimenter,
+    IsingExperimenter,
+    MAXSATExperimenter,
+    PestControlExperimenter,
+)
+from vizier_amd._src.benchmarks.experimenters.experimenter import (
+    Experimenter,
+)
+from vizier_amd._src.benchmarks.experimenters.experimenter_factory import (
+    BBOBExperimenterFactory,
+    SingleObjectiveExperimenterFactory,
+)
+from vizier_amd._src.benchmarks.experimenters.numpy_experimenter import (
+    NumpyExperimenter,
+)
+from vizier_amd._src.benchmarks.experimenters.wrappers import (
+    DiscretizingExperimenter,
+    InfeasibleExperimenter,
+    MultiObjectiveExperimenter,
+    NoisyExperimenter,
+    NormalizingExperimenter,
+    ShiftingExperimenter,
+    SignFlipExperimenter,
+    SwitchExperimenter,
+)
+from vizier_amd._src.benchmarks.experimenters.synthetic import bbob
