@@ -47,7 +47,11 @@ class TestPosteriorScoreKernel:
     from vizier_amd._src.gp import gp_model
     g = torch.Generator().manual_seed(seed)
     x = torch.rand(n, d, generator=g)
-    y = torch.sin(3 * x[:, 0]) + x[:, 1]
+    # Real observation noise keeps the fitted noise (and cond(K)) in a
+    # regime where the fp32 K_inv quadform is accurate, so the kernel
+    # numerics test can be TIGHT. The near-noiseless/degenerate regime
+    # is exercised end-to-end by the designer + regret evidence.
+    y = torch.sin(3 * x[:, 0]) + x[:, 1] +         0.1 * torch.randn(n, generator=g)
     post = gp_model.train_gp(x, y, num_restarts=2, max_iters=20, seed=seed)
     return post
 
@@ -111,12 +115,12 @@ class TestPosteriorScoreKernel:
     amp = float(post.params.amplitude)
     got = got.cpu()
     if acq in ('pi', 'ei'):
-      keep = stddev > 0.1 * amp
+      keep = stddev > 0.05 * amp
       assert int(keep.sum()) > 10, 'test needs non-degenerate points'
       got, want = got[keep], want[keep]
-      tol = 0.05 if acq == 'pi' else 0.05 * amp
+      tol = 0.02 if acq == 'pi' else 0.01 * amp
     else:
-      tol = 0.06 * amp
+      tol = 0.01 * amp
     err = (got - want).abs().max()
     assert float(err) < tol, f'{acq}: max err {err}'
 
@@ -148,7 +152,7 @@ class TestPosteriorScoreKernel:
     mean, stddev = self._oracle_mean_std(post, xq.cpu())
     amp = float(post.params.amplitude)
     assert torch.allclose(got.cpu(), mean + 1.8 * stddev,
-                          atol=0.06 * amp)
+                          atol=0.01 * amp)
 
 
 class TestEagleKernels:
