@@ -139,10 +139,13 @@ def negative_log_marginal_likelihood(raw: torch.Tensor, x: torch.Tensor,
   K = K + noise * torch.eye(n, dtype=x.dtype, device=x.device)
   L, info = safe_cholesky_ex(K)
   resid = (y.unsqueeze(0) - params.mean.unsqueeze(-1)).unsqueeze(-1)
-  # NOTE: torch.cholesky_solve hits a hipErrorLaunchFailure on this ROCm
-  # build (see profiles/notes); two triangular solves are equivalent.
-  alpha = _chol_solve(L, resid)
-  quad = (resid * alpha).sum(dim=(-1, -2))
+  # quad = r^T K^-1 r = ||L^-1 r||^2: ONE triangular solve instead of a
+  # _chol_solve pair — the batched vector-RHS solves dispatch as serial
+  # rocBLAS trsv calls (~80 ms/suggest at the bench shape, see
+  # profiles/bench_kernel_stats2.csv), so halving them matters.
+  # (torch.cholesky_solve itself hipErrorLaunchFailures on this build.)
+  z = torch.linalg.solve_triangular(L, resid, upper=False)
+  quad = (z * z).sum(dim=(-1, -2))
   logdet = 2.0 * torch.log(torch.diagonal(L, dim1=-2, dim2=-1)).sum(-1)
   nll = 0.5 * (quad + logdet + n * math.log(2 * math.pi))
   # Mild pull toward the raw-space origin (the reference regularizes via

@@ -107,8 +107,8 @@ def negative_log_posterior(raw: torch.Tensor, x: torch.Tensor,
       n, dtype=x.dtype, device=x.device)
   L, info = gp_model.safe_cholesky_ex(K)
   resid = y.reshape(1, n, 1).expand(raw.shape[0], n, 1)
-  alpha = gp_model._chol_solve(L, resid)
-  quad = (resid * alpha).sum(dim=(-1, -2))
+  z = torch.linalg.solve_triangular(L, resid, upper=False)
+  quad = (z * z).sum(dim=(-1, -2))
   logdet = 2.0 * torch.log(
       torch.diagonal(L, dim1=-2, dim2=-1)).sum(-1)
   nll = 0.5 * (quad + logdet + n * math.log(2 * math.pi))

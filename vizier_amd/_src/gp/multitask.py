@@ -116,8 +116,8 @@ def negative_log_marginal_likelihood(raw: torch.Tensor, x: torch.Tensor,
   L, info = gp_model.safe_cholesky_ex(k)
   resid = (y.unsqueeze(0) - params.means.unsqueeze(-2)).reshape(
       raw.shape[0], n * m, 1)
-  alpha = gp_model._chol_solve(L, resid)
-  quad = (resid * alpha).sum(dim=(-1, -2))
+  z = torch.linalg.solve_triangular(L, resid, upper=False)
+  quad = (z * z).sum(dim=(-1, -2))
   logdet = 2.0 * torch.log(torch.diagonal(L, dim1=-2, dim2=-1)).sum(-1)
   nll = 0.5 * (quad + logdet + n * m * math.log(2 * math.pi))
   nll = nll + 0.01 * (raw * raw).sum(-1)
