@@ -49,9 +49,11 @@ def _forward_cholesky_ex(K: torch.Tensor):
   if (K.is_cuda and K.dim() == 3 and K.shape[0] > 1 and
       not K.requires_grad and K.shape[-1] <= _CHOL_KERNEL_MAX_N and
       ops.extension_available()):
-    ext = ops.require_ext()
-    L, info = ext.batched_cholesky(K.contiguous())
-    return L, info
+    try:
+      ext = ops.require_ext()
+      return ext.batched_cholesky(K.contiguous())
+    except RuntimeError:  # launch refusal -> library fallback
+      pass
   return safe_cholesky_ex(K)
 
 

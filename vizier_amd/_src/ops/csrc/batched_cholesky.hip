@@ -147,9 +147,13 @@ extern "C" int launch_batched_cholesky(float* a, int* info, int batch,
   const size_t shmem = (size_t)n * PST * sizeof(float);
   static int attr_set = 0;
   if (!attr_set) {
+    // May be unsupported for this symbol form on some ROCm builds;
+    // harmless either way, but it leaves a sticky error code that a
+    // later hipGetLastError would misattribute to the launch — clear.
     (void)hipFuncSetAttribute(
         (const void*)batched_cholesky_kernel,
         hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+    (void)hipGetLastError();
     attr_set = 1;
   }
   hipLaunchKernelGGL(batched_cholesky_kernel, dim3(batch),
