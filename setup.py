@@ -20,7 +20,6 @@ _SRC = [
     'vizier_amd/_src/ops/csrc/posterior_score.hip',
     'vizier_amd/_src/ops/csrc/eagle_step.hip',
     'vizier_amd/_src/ops/csrc/eagle_sweep.hip',
-    'vizier_amd/_src/ops/csrc/batched_cholesky.hip',
 ]
 
 setup(

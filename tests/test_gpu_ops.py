@@ -458,39 +458,3 @@ class TestGramFp8MFMA:
     # to fp32 summation-order differences (measured 3e-3).
     assert torch.allclose(torch.diagonal(K),
                           torch.full((64,), 4.0).cuda(), atol=1e-2)
-
-
-class TestBatchedCholesky:
-
-  @pytest.mark.parametrize('b,n', [(4, 64), (12, 333), (3, 1000),
-                                   (25, 1241)])
-  def test_matches_torch_cholesky(self, ext, b, n):
-    g = torch.Generator().manual_seed(0)
-    a = torch.randn(b, n, 40 + n // 4, generator=g)
-    K = (a @ a.mT + n * torch.eye(n)).cuda()
-    want = torch.linalg.cholesky(K)
-    got, info = ext.batched_cholesky(K.clone())
-    assert int(info.abs().sum()) == 0
-    err = (got - want).abs().max() / want.abs().max()
-    assert float(err) < 1e-4, f'rel err {err}'
-
-  def test_flags_non_spd(self, ext):
-    K = torch.randn(2, 50, 50).cuda()
-    K = K + K.mT  # symmetric but indefinite
-    _, info = ext.batched_cholesky(K.clone().contiguous())
-    assert int(info[0]) > 0 and int(info[1]) > 0
-
-  def test_nll_ladder_uses_kernel_and_matches_torch(self, ext):
-    """The no-grad NLL (line-search path) through the HIP Cholesky is
-    numerically equivalent to the torch path."""
-    from vizier_amd._src.gp import gp_model
-    g = torch.Generator().manual_seed(1)
-    x = torch.rand(500, 12, generator=g).cuda()
-    y = torch.randn(500, generator=g).cuda()
-    raw = (torch.randn(8, 15, generator=g) * 0.5).cuda()
-    with torch.no_grad():
-      got = gp_model.negative_log_marginal_likelihood(raw, x, y)
-    want = gp_model.negative_log_marginal_likelihood(
-        raw.requires_grad_(True), x, y)
-    assert torch.allclose(got, want.detach(), rtol=1e-4, atol=1e-2), \
-        (got, want)

@@ -39,24 +39,6 @@ _MAGMA_BAD_LO, _MAGMA_BAD_HI = 257, 511  # generous upper margin
 _MAGMA_PAD_N = 512
 
 
-_CHOL_KERNEL_MAX_N = 1241  # LDS panel bound of the HIP kernel
-
-
-def _forward_cholesky_ex(K: torch.Tensor):
-  """Value-only batched Cholesky: the hand-written HIP kernel when no
-  gradient is required (the L-BFGS line-search ladder — MAGMA's batched
-  POTRF is launch-overhead-bound there), torch otherwise."""
-  if (K.is_cuda and K.dim() == 3 and K.shape[0] > 1 and
-      not K.requires_grad and K.shape[-1] <= _CHOL_KERNEL_MAX_N and
-      ops.extension_available()):
-    try:
-      ext = ops.require_ext()
-      return ext.batched_cholesky(K.contiguous())
-    except RuntimeError:  # launch refusal -> library fallback
-      pass
-  return safe_cholesky_ex(K)
-
-
 def safe_cholesky_ex(K: torch.Tensor):
   """torch.linalg.cholesky_ex that avoids MAGMA's batched crash window."""
   n = K.shape[-1]
@@ -155,7 +137,7 @@ def negative_log_marginal_likelihood(raw: torch.Tensor, x: torch.Tensor,
                     params.amplitude)
   noise = params.noise.reshape(-1, 1, 1)
   K = K + noise * torch.eye(n, dtype=x.dtype, device=x.device)
-  L, info = _forward_cholesky_ex(K)
+  L, info = safe_cholesky_ex(K)
   resid = (y.unsqueeze(0) - params.mean.unsqueeze(-1)).unsqueeze(-1)
   # quad = r^T K^-1 r = ||L^-1 r||^2: ONE triangular solve instead of a
   # _chol_solve pair — the batched vector-RHS solves dispatch as serial

@@ -66,10 +66,6 @@ extern "C" int launch_eagle_sweep(
     float amp2, float mean_c, int acq, float coef, float best_value,
     float tr_radius, hipStream_t stream);
 
-extern "C" int launch_batched_cholesky(float* a, int* info, int batch,
-                                       int n, hipStream_t stream);
-extern "C" int batched_cholesky_max_n(void);
-
 extern "C" void launch_eagle_update(
     float* pool_cont, long* pool_cat, float* rewards, float* perturbations,
     const float* batch_cont, const long* batch_cat,
@@ -387,19 +383,6 @@ int64_t eagle_sweep(
   return ret;
 }
 
-std::vector<torch::Tensor> batched_cholesky(torch::Tensor a) {
-  a = check_f32(a, "a");
-  TORCH_CHECK(a.dim() == 3 && a.size(1) == a.size(2),
-              "expected (B, N, N)");
-  const int b = a.size(0), n = a.size(1);
-  TORCH_CHECK(n <= batched_cholesky_max_n(),
-              "batched_cholesky supports N <= ", batched_cholesky_max_n());
-  auto info = torch::zeros({b}, a.options().dtype(torch::kInt32));
-  const int err = launch_batched_cholesky(
-      a.data_ptr<float>(), info.data_ptr<int>(), b, n, current_stream());
-  TORCH_CHECK(err == 0, "batched_cholesky launch failed: ", err);
-  return {a, info};
-}
 
 }  // namespace
 
@@ -421,8 +404,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("eagle_suggest", &eagle_suggest,
         "Fused Eagle suggest step (gfx950)");
   m.def("eagle_update", &eagle_update, "Fused Eagle update step (gfx950)");
-  m.def("batched_cholesky", &batched_cholesky,
-        "In-place batched lower Cholesky, line-search path (gfx950)");
   m.def("eagle_sweep", &eagle_sweep,
         "Persistent cooperative Eagle sweep megakernel (gfx950)");
 }
