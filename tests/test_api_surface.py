@@ -1,0 +1,233 @@
+"""Public API surface parity + functional checks for the new pieces."""
+
+import ast
+import importlib
+import os
+
+import numpy as np
+import pytest
+
+from vizier_amd import pyvizier as vz
+
+REFERENCE = '/root/reference'
+
+_MODULES = {
+    'vizier/pyvizier/__init__.py': 'vizier_amd.pyvizier',
+    'vizier/algorithms/__init__.py': 'vizier_amd.algorithms',
+    'vizier/benchmarks/__init__.py': 'vizier_amd.benchmarks',
+    'vizier/benchmarks/experimenters/__init__.py':
+        'vizier_amd.benchmarks.experimenters',
+    'vizier/benchmarks/analyzers.py': 'vizier_amd.benchmarks.analyzers',
+    'vizier/service/__init__.py': 'vizier_amd.service',
+}
+
+
+@pytest.mark.skipif(not os.path.isdir(REFERENCE),
+                    reason='reference not mounted')
+@pytest.mark.parametrize('ref_path,module', sorted(_MODULES.items()))
+def test_public_exports_cover_reference(ref_path, module):
+  """Every name the reference's public __init__ exports must exist."""
+  tree = ast.parse(open(os.path.join(REFERENCE, ref_path)).read())
+  names = [a.asname or a.name for node in ast.walk(tree)
+           if isinstance(node, ast.ImportFrom) for a in node.names]
+  mine = importlib.import_module(module)
+  missing = [n for n in names if n != 'annotations' and
+             not hasattr(mine, n)]
+  assert not missing, f'{module} missing {missing}'
+
+
+class TestClassicExperimenters:
+
+  def test_branin_known_minimum(self):
+    from vizier_amd._src.benchmarks.experimenters.synthetic import classic
+    t = vz.Trial({'x1': -np.pi, 'x2': 12.275})  # a global minimizer
+    classic.Branin2DExperimenter().evaluate([t])
+    assert t.final_measurement.metrics['value'].value == \
+        pytest.approx(0.397887, abs=1e-4)
+
+  def test_hartmann6_known_minimum(self):
+    from vizier_amd._src.benchmarks.experimenters.synthetic import classic
+    x = [0.20169, 0.150011, 0.476874, 0.275332, 0.311652, 0.6573]
+    t = vz.Trial({f'x{i+1}': v for i, v in enumerate(x)})
+    classic.HartmannExperimenter.from_6d().evaluate([t])
+    assert t.final_measurement.metrics['value'].value == \
+        pytest.approx(-3.32237, abs=1e-3)
+
+  def test_dh1_two_objectives(self):
+    from vizier_amd._src.benchmarks.experimenters.synthetic import classic
+    exp = classic.DHExperimenter.DH1(num_dimensions=3)
+    t = vz.Trial({'x0': 0.5, 'x1': 0.2, 'x2': -0.1})
+    exp.evaluate([t])
+    m = t.final_measurement.metrics
+    assert m['f0'].value == pytest.approx(0.5)
+    assert np.isfinite(m['f1'].value)
+    assert len(exp.problem_statement().metric_information) == 2
+
+  def test_multiarm(self):
+    from vizier_amd._src.benchmarks.experimenters.synthetic import classic
+    fixed = classic.FixedMultiArmExperimenter({'a': 0.3, 'b': 0.9})
+    t = vz.Trial({'arm': 'b'})
+    fixed.evaluate([t])
+    assert t.final_measurement.metrics['reward'].value == 0.9
+    bern = classic.BernoulliMultiArmExperimenter({'a': 1.0}, seed=0)
+    t2 = vz.Trial({'arm': 'a'})
+    bern.evaluate([t2])
+    assert t2.final_measurement.metrics['reward'].value == 1.0
+
+
+class TestExtraExperimenters:
+
+  def _sphere(self):
+    from vizier_amd._src.benchmarks.experimenters.numpy_experimenter import (
+        NumpyExperimenter)
+    problem = vz.ProblemStatement()
+    problem.search_space.root.add_float_param('x0', -1, 1)
+    problem.search_space.root.add_float_param('x1', -1, 1)
+    problem.metric_information.append(vz.MetricInformation(
+        name='obj', goal=vz.ObjectiveMetricGoal.MINIMIZE))
+    return NumpyExperimenter(lambda x: float((x ** 2).sum()), problem)
+
+  def test_l1_categorical(self):
+    from vizier_amd._src.benchmarks.experimenters import extra
+    exp = extra.L1CategorialExperimenter(num_categories=[3, 4],
+                                         optimum=[1, 2])
+    best = exp.optimal_trial
+    assert best.final_measurement.metrics['objective'].value == 0.0
+    t = vz.Trial({'c0': '0', 'c1': '2'})
+    exp.evaluate([t])
+    assert t.final_measurement.metrics['objective'].value == 1.0
+
+  def test_hashing_infeasible_deterministic(self):
+    from vizier_amd._src.benchmarks.experimenters import extra
+    exp = extra.HashingInfeasibleExperimenter(self._sphere(),
+                                              infeasible_prob=0.5, seed=1)
+    outcomes = []
+    for _ in range(2):
+      t = vz.Trial({'x0': 0.25, 'x1': -0.5})
+      exp.evaluate([t])
+      outcomes.append(t.infeasible)
+    assert outcomes[0] == outcomes[1]
+
+  def test_param_region_infeasible(self):
+    from vizier_amd._src.benchmarks.experimenters import extra
+    exp = extra.ParamRegionInfeasibleExperimenter(
+        self._sphere(), 'x0', infeasible_interval=(0.0, 0.3))
+    t_bad = vz.Trial({'x0': -0.9, 'x1': 0.0})   # scaled ~0.05 -> infeasible
+    t_ok = vz.Trial({'x0': 0.9, 'x1': 0.0})     # scaled ~0.95 -> feasible
+    exp.evaluate([t_bad, t_ok])
+    assert t_bad.infeasible and not t_ok.infeasible
+
+  def test_hypercube(self):
+    from vizier_amd._src.benchmarks.experimenters import extra
+    exp = extra.HyperCubeExperimenter(self._sphere())
+    space = exp.problem_statement().search_space
+    assert all(p.bounds == (0.0, 1.0) for p in space.parameters)
+    t = vz.Trial({'h0': 0.5, 'h1': 0.5})        # centre -> x = (0, 0)
+    exp.evaluate([t])
+    assert t.final_measurement.metrics['obj'].value == pytest.approx(0.0)
+
+  def test_multiobjective_numpy(self):
+    from vizier_amd._src.benchmarks.experimenters import extra
+    problem = vz.ProblemStatement()
+    problem.search_space.root.add_float_param('x0', 0, 1)
+    for n in ('f1', 'f2'):
+      problem.metric_information.append(vz.MetricInformation(
+          name=n, goal=vz.ObjectiveMetricGoal.MINIMIZE))
+    exp = extra.MultiObjectiveNumpyExperimenter(
+        lambda x: [float(x[0]), float(1 - x[0])], problem)
+    t = vz.Trial({'x0': 0.25})
+    exp.evaluate([t])
+    assert t.final_measurement.metrics['f1'].value == pytest.approx(0.25)
+    assert t.final_measurement.metrics['f2'].value == pytest.approx(0.75)
+
+
+class TestSequentialParameterBuilder:
+
+  def test_dfs_unlocks_children(self):
+    from vizier_amd._src.pyvizier.parameter_iterators import (
+        SequentialParameterBuilder)
+    space = vz.SearchSpace()
+    root = space.root
+    root.add_categorical_param('model', ['linear', 'dnn'])
+    root.select('model', ['dnn']).add_float_param('lr', 1e-4, 1e-1)
+    root.add_int_param('batch', 1, 128)
+    builder = SequentialParameterBuilder(space)
+    chosen = {'model': 'dnn', 'lr': 0.01, 'batch': 32}
+    seen = []
+    for pc in builder:
+      seen.append(pc.name)
+      builder.choose_value(chosen[pc.name])
+    assert seen == ['model', 'lr', 'batch']
+    assert {k: v.value for k, v in builder.parameters.items()} == chosen
+
+  def test_skip_and_inactive_child(self):
+    from vizier_amd._src.pyvizier.parameter_iterators import (
+        SequentialParameterBuilder)
+    space = vz.SearchSpace()
+    root = space.root
+    root.add_categorical_param('model', ['linear', 'dnn'])
+    root.select('model', ['dnn']).add_float_param('lr', 1e-4, 1e-1)
+    builder = SequentialParameterBuilder(space)
+    for pc in builder:
+      if pc.name == 'model':
+        builder.choose_value('linear')  # lr never becomes active
+      else:
+        builder.skip()
+    assert list(builder.parameters) == ['model']
+
+
+class TestAnalyzerAdditions:
+
+  def _curves(self):
+    from vizier_amd._src.benchmarks.analyzers.convergence_curve import (
+        ConvergenceCurve)
+    g = np.random.default_rng(0)
+    xs = np.arange(1, 21)
+    base = np.sort(g.random((5, 20)), axis=1)
+    comp = np.sort(g.random((5, 20)) + 0.3, axis=1)
+    return (ConvergenceCurve(xs, base), ConvergenceCurve(xs, comp))
+
+  def test_winrate_pair_comparator(self):
+    from vizier_amd._src.benchmarks.analyzers import convergence_curve as cc
+    base, comp = self._curves()
+    comparator = cc.WinRateConvergenceCurveComparatorFactory()(base, comp)
+    assert comparator.score() > 0.3     # comp dominates
+    reverse = cc.WinRateConvergenceCurveComparatorFactory()(comp, base)
+    assert reverse.score() < -0.3
+
+  def test_stateful_converter_distributive(self):
+    from vizier_amd._src.benchmarks.analyzers.convergence_curve import (
+        ConvergenceCurveConverter)
+    mi = vz.MetricInformation(name='m',
+                              goal=vz.ObjectiveMetricGoal.MAXIMIZE)
+    def make(vals, start_id=1):
+      out = []
+      for i, v in enumerate(vals):
+        t = vz.Trial({'x': 0.5}, id=start_id + i)
+        t.complete(vz.Measurement(metrics={'m': v}))
+        out.append(t)
+      return out
+    whole = ConvergenceCurveConverter(mi).convert(make([1, 3, 2, 5]))
+    split = ConvergenceCurveConverter(mi)
+    c1 = split.convert(make([1, 3]))
+    c2 = split.convert(make([2, 5], start_id=3))
+    assert np.allclose(np.concatenate([c1.ys[0], c2.ys[0]]), whole.ys[0])
+    assert np.allclose(np.concatenate([c1.xs, c2.xs]), whole.xs)
+
+  def test_record_analyzer_and_plot(self, tmp_path):
+    from vizier_amd._src.benchmarks.analyzers import state_analyzer as sa
+    from vizier_amd._src.benchmarks.analyzers import plot_utils
+    base, comp = self._curves()
+    records = [
+        sa.BenchmarkRecord('baseline', {'exp': 'sphere'},
+                           {'objective': sa.PlotElement(curve=base)}),
+        sa.BenchmarkRecord('candidate', {'exp': 'sphere'},
+                           {'objective': sa.PlotElement(curve=comp)}),
+    ]
+    out = sa.BenchmarkRecordAnalyzer.add_comparison_metrics(
+        records, 'baseline')
+    assert len(out) == 2
+    assert any('objective:score' in r.plot_elements for r in out)
+    fig, _ = plot_utils.plot_from_records(out)
+    fig.savefig(tmp_path / 'plot.png')
+    assert (tmp_path / 'plot.png').exists()

@@ -9,7 +9,7 @@ from __future__ import annotations
 
 import abc
 import dataclasses
-from typing import List, Optional, Sequence
+from typing import Callable, List, Optional, Sequence, TypeVar
 
 import numpy as np
 
@@ -48,6 +48,11 @@ class Prediction:
   mean: np.ndarray
   stddev: np.ndarray
   metadata: Optional[Metadata] = None
+
+
+_T = TypeVar('_T')
+# Factory protocol: problem statement (+kwargs) -> designer instance.
+DesignerFactory = Callable[..., _T]
 
 
 class Designer(abc.ABC):

@@ -7,6 +7,7 @@ Capability parity with vizier/_src/algorithms/optimizers/base.py
 from __future__ import annotations
 
 import abc
+import dataclasses
 from typing import Callable, Dict, List, Optional, Sequence
 
 import numpy as np
@@ -27,6 +28,18 @@ class GradientFreeOptimizer(abc.ABC):
                seed_candidates: Sequence[vz.TrialSuggestion] = ()
                ) -> List[vz.TrialSuggestion]:
     ...
+
+
+@dataclasses.dataclass(frozen=True)
+class BranchSelection:
+  """A flat subspace plus the number of suggestions to generate in it.
+
+  N suggestions on a conditional space are decomposed into
+  N_1 + ... + N_k suggestions over flat subspaces (base.py:50).
+  """
+
+  problem: vz.ProblemStatement
+  num_suggestions: int
 
 
 class BranchSelector(abc.ABC):

@@ -139,3 +139,80 @@ def _metadata_from_blob(blob: str) -> Metadata:
     for k, v in items.items():
       view[k] = v
   return md
+
+
+class InRamDesignerPolicy(pythia_policy.Policy):
+  """Keeps ONE designer instance in RAM and feeds it only new trials.
+
+  Parity with designer_policy.py:347: no serialization — efficient for
+  in-process services where the policy object survives across calls.
+  """
+
+  def __init__(self, problem: ProblemStatement, supporter: PolicySupporter,
+               designer_factory: Callable[[ProblemStatement], object]):
+    self._problem = problem
+    self._supporter = supporter
+    self._designer_factory = designer_factory
+    self._designer = None
+    self._max_seen_id = 0
+
+  def suggest(self, request: SuggestRequest) -> SuggestDecision:
+    if self._designer is None:
+      self._designer = self._designer_factory(request.study_config)
+      self._max_seen_id = 0
+    completed = self._supporter.GetTrials(
+        study_guid=request.study_guid, min_trial_id=self._max_seen_id + 1,
+        status_matches=TrialStatus.COMPLETED)
+    active = self._supporter.GetTrials(
+        study_guid=request.study_guid, status_matches=TrialStatus.ACTIVE)
+    self._designer.update(CompletedTrials(completed), ActiveTrials(active))
+    self._max_seen_id = max([t.id for t in completed],
+                            default=self._max_seen_id)
+    return SuggestDecision(self._designer.suggest(request.count))
+
+
+class SerializableDesignerPolicy(pythia_policy.Policy):
+  """Wraps a fully SerializableDesigner: state restored via the class's
+  recover() from study metadata, falling back to a fresh designer +
+  full-history update (designer_policy.py:377)."""
+
+  def __init__(self, problem: ProblemStatement, supporter: PolicySupporter,
+               designer_factory: Callable[[ProblemStatement], object],
+               designer_cls, *, ns_root: str = 'designer'):
+    self._problem = problem
+    self._supporter = supporter
+    self._designer_factory = designer_factory
+    self._designer_cls = designer_cls
+    self._tracker = _StateTracker(ns_root)
+
+  def suggest(self, request: SuggestRequest) -> SuggestDecision:
+    study_md = request.study_config.metadata
+    state_blob, max_seen_id = self._tracker.load(study_md)
+    designer = None
+    incremental = False
+    if state_blob is not None:
+      try:
+        designer = self._designer_cls.recover(
+            _metadata_from_blob(state_blob))
+        incremental = True
+      except Exception:
+        designer = None
+    if designer is None:
+      designer = self._designer_factory(request.study_config)
+    if incremental:
+      completed = self._supporter.GetTrials(
+          study_guid=request.study_guid, min_trial_id=max_seen_id + 1,
+          status_matches=TrialStatus.COMPLETED)
+    else:
+      completed = self._supporter.GetTrials(
+          study_guid=request.study_guid,
+          status_matches=TrialStatus.COMPLETED)
+    active = self._supporter.GetTrials(
+        study_guid=request.study_guid, status_matches=TrialStatus.ACTIVE)
+    designer.update(CompletedTrials(completed), ActiveTrials(active))
+    suggestions = designer.suggest(request.count)
+    new_max = max([t.id for t in completed], default=max_seen_id)
+    delta = self._tracker.updates(_metadata_to_blob(designer.dump()),
+                                  new_max)
+    return SuggestDecision(suggestions, metadata=delta)
+

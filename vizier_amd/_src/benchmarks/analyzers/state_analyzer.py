@@ -11,6 +11,7 @@ from __future__ import annotations
 
 import collections
 import dataclasses
+import json
 from typing import Dict, List, Optional, Sequence
 
 import numpy as np
@@ -26,13 +27,26 @@ from vizier_amd._src.benchmarks.runners.benchmark_state import (
 
 
 @dataclasses.dataclass
+class PlotElement:
+  """One subplot's payload (reference state_analyzer.py:37): either a
+  curve (error-bar) or a raw array (histogram/scatter)."""
+
+  curve: object = None                 # ConvergenceCurve, for error-bar
+  plot_array: object = None            # np.ndarray, for histogram/scatter
+  plot_type: str = 'error-bar'         # 'error-bar'|'histogram'|'scatter'
+  xlabel: str = 'Num Trials'
+  yscale: str = 'linear'
+  percentile_error_bar: tuple = (25, 75)
+
+
+@dataclasses.dataclass
 class BenchmarkRecord:
   """One (algorithm, experimenter) benchmark outcome."""
 
   algorithm: str
   experimenter_metadata: Dict[str, str]
-  plot_elements: Dict[str, ConvergenceCurve] = dataclasses.field(
-      default_factory=dict)
+  plot_elements: Dict[str, object] = dataclasses.field(
+      default_factory=dict)  # name -> PlotElement or bare curve
 
 
 class BenchmarkStateAnalyzer:
@@ -69,8 +83,13 @@ class BenchmarkStateAnalyzer:
     import pandas as pd
     rows = []
     for r in records:
-      final = {name: float(np.nanmedian(curve.ys[:, -1]))
-               for name, curve in r.plot_elements.items()}
+      final = {}
+      for name, el in r.plot_elements.items():
+        curve = getattr(el, 'curve', el)
+        if curve is not None and hasattr(curve, 'ys'):
+          final[name] = float(np.nanmedian(curve.ys[:, -1]))
+        elif getattr(el, 'plot_array', None) is not None:
+          final[name] = float(np.nanmedian(el.plot_array))
       rows.append({'algorithm': r.algorithm,
                    **r.experimenter_metadata, **final})
     return pd.DataFrame(rows)
@@ -95,3 +114,52 @@ def compute_parameter_entropy(trials: Sequence[vz.Trial],
   p = counts / counts.sum()
   p = p[p > 0]
   return float(-(p * np.log(p)).sum())
+
+
+class BenchmarkRecordAnalyzer:
+  """Cross-record comparisons (reference state_analyzer.py:195)."""
+
+  @classmethod
+  def add_comparison_metrics(
+      cls, records: Sequence[BenchmarkRecord], baseline_algo: str, *,
+      compare_metric: str = 'objective',
+      comparator_factory=None) -> List[BenchmarkRecord]:
+    """Scores every record's `compare_metric` curve against the
+    baseline algorithm's (grouped by experimenter metadata)."""
+    from vizier_amd._src.benchmarks.analyzers import convergence_curve as cc
+    if comparator_factory is None:
+      comparator_factory = \
+          cc.LogEfficiencyConvergenceCurveComparatorFactory()
+
+    def curve_of(rec):
+      el = rec.plot_elements.get(compare_metric)
+      curve = getattr(el, 'curve', el)
+      if curve is None:
+        raise ValueError(
+            f'Record {rec.algorithm} has no curve for {compare_metric}')
+      return curve
+
+    def exp_key(rec):
+      return json.dumps(dict(rec.experimenter_metadata), sort_keys=True)
+
+    groups: Dict[str, List[BenchmarkRecord]] = {}
+    for rec in records:
+      groups.setdefault(exp_key(rec), []).append(rec)
+
+    out: List[BenchmarkRecord] = []
+    for group in groups.values():
+      baselines = [r for r in group if r.algorithm == baseline_algo]
+      if not baselines:
+        raise ValueError(f'Baseline {baseline_algo} not found in group.')
+      base_curve = curve_of(baselines[0])
+      for rec in group:
+        comparator = comparator_factory(base_curve, curve_of(rec))
+        new_elements = dict(rec.plot_elements)
+        new_elements[f'{compare_metric}:score'] = PlotElement(
+            plot_array=np.asarray([comparator.score()]),
+            plot_type='histogram')
+        out.append(BenchmarkRecord(
+            algorithm=rec.algorithm,
+            experimenter_metadata=rec.experimenter_metadata,
+            plot_elements=new_elements))
+    return out

@@ -45,6 +45,21 @@ class EvaluateActiveTrials(BenchmarkSubroutine):
 
 
 @dataclasses.dataclass
+class FillActiveTrials(BenchmarkSubroutine):
+  """Tops up suggestions until `num_active_trials_limit` are ACTIVE
+  (benchmark_runner.py:123)."""
+
+  num_active_trials_limit: int = 1
+
+  def run(self, state: BenchmarkState) -> None:
+    active = state.algorithm.supporter.GetTrials(
+        status_matches=vz.TrialStatus.ACTIVE)
+    missing = self.num_active_trials_limit - len(active)
+    if missing > 0:
+      state.algorithm.suggest(missing)
+
+
+@dataclasses.dataclass
 class GenerateAndEvaluate(BenchmarkSubroutine):
   num_suggestions: int = 1
 

@@ -80,3 +80,39 @@ class DesignerBenchmarkStateFactory:
   def __call__(self, seed: Optional[int] = None) -> BenchmarkState:
     return BenchmarkState.from_designer_factory(
         self.designer_factory, self.experimenter, seed)
+
+
+# Protocol: (problem, seed) -> Policy.
+SeededPolicyFactory = Callable[[vz.ProblemStatement, Optional[int]], Policy]
+
+
+@dataclasses.dataclass(frozen=True)
+class ExperimenterDesignerBenchmarkStateFactory:
+  """BenchmarkState from (experimenter factory, designer factory)
+  (benchmark_state.py:110)."""
+
+  experimenter_factory: Callable[[], Experimenter]
+  designer_factory: Callable[[vz.ProblemStatement], Designer]
+
+  def __call__(self, seed: Optional[int] = None) -> BenchmarkState:
+    experimenter = self.experimenter_factory()
+    return DesignerBenchmarkStateFactory(
+        experimenter=experimenter,
+        designer_factory=self.designer_factory)(seed=seed)
+
+
+@dataclasses.dataclass(frozen=True)
+class PolicyBenchmarkStateFactory:
+  """BenchmarkState from a seeded Pythia policy factory
+  (benchmark_state.py:154)."""
+
+  experimenter: Experimenter
+  policy_factory: SeededPolicyFactory
+
+  def __call__(self, seed: Optional[int] = None) -> BenchmarkState:
+    problem = self.experimenter.problem_statement()
+    supporter = InRamPolicySupporter(problem)
+    return BenchmarkState(
+        experimenter=self.experimenter,
+        algorithm=PolicySuggester(self.policy_factory(problem, seed),
+                                  supporter))

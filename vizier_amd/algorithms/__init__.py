@@ -17,3 +17,16 @@ from vizier_amd._src.algorithms.policies.random_policy import RandomPolicy
 from vizier_amd._src.pythia.local_policy_supporters import (
     InRamPolicySupporter,
 )
+
+from vizier_amd._src.algorithms.core.abstractions import DesignerFactory
+from vizier_amd._src.algorithms.optimizers.base import (
+    BatchTrialScoreFunction,
+    BranchSelection,
+    BranchSelector,
+    BranchThenOptimizer,
+    GradientFreeOptimizer,
+)
+from vizier_amd._src.algorithms.policies.designer_policy import (
+    InRamDesignerPolicy,
+    SerializableDesignerPolicy,
+)

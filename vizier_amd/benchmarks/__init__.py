@@ -40,3 +40,13 @@ from vizier_amd._src.benchmarks.analyzers.convergence_curve import (
     PercentageBetterConvergenceCurveComparator,
     WinRateComparator,
 )
+
+from vizier_amd._src.benchmarks.runners.benchmark_runner import (
+    FillActiveTrials,
+)
+from vizier_amd._src.benchmarks.runners.benchmark_state import (
+    BenchmarkStateFactory,
+    ExperimenterDesignerBenchmarkStateFactory,
+    PolicyBenchmarkStateFactory,
+    SeededPolicyFactory,
+)

@@ -17,3 +17,24 @@ from vizier_amd._src.benchmarks.analyzers.state_analyzer import (
     BenchmarkStateAnalyzer,
     compute_parameter_entropy,
 )
+
+from vizier_amd._src.benchmarks.analyzers.convergence_curve import (
+    ConvergenceComparator,
+    ConvergenceComparatorFactory,
+    LogEfficiencyConvergenceCurveComparatorFactory,
+    MultiMetricCurveConverter,
+    PercentageBetterConvergenceCurveComparatorFactory,
+    RestartingCurveConverter,
+    StatefulCurveConverter,
+    WinRateConvergenceCurveComparator,
+    WinRateConvergenceCurveComparatorFactory,
+)
+from vizier_amd._src.benchmarks.analyzers.plot_utils import (
+    plot_from_records,
+    plot_mean_convergence,
+    plot_median_convergence,
+)
+from vizier_amd._src.benchmarks.analyzers.state_analyzer import (
+    BenchmarkRecordAnalyzer,
+    PlotElement,
+)

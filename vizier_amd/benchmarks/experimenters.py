@@ -29,3 +29,29 @@ from vizier_amd._src.benchmarks.experimenters.wrappers import (
     SwitchExperimenter,
 )
 from vizier_amd._src.benchmarks.experimenters.synthetic import bbob
+
+from vizier_amd._src.benchmarks.experimenters.extra import (
+    CombinedExperimenterFactory,
+    ExperimenterFactory,
+    HashingInfeasibleExperimenter,
+    HyperCubeExperimenter,
+    L1CategorialExperimenter,
+    MultiObjectiveNumpyExperimenter,
+    ParamRegionInfeasibleExperimenter,
+    PredictorExperimenter,
+    SerializableExperimenterFactory,
+)
+from vizier_amd._src.benchmarks.experimenters.synthetic.classic import (
+    BernoulliMultiArmExperimenter,
+    Branin2DExperimenter,
+    DHExperimenter,
+    FixedMultiArmExperimenter,
+    HartmannExperimenter,
+)
+from vizier_amd._src.benchmarks.experimenters.synthetic.simplekd import (
+    SimpleKDExperimenter,
+)
+from vizier_amd._src.benchmarks.experimenters.wrappers import (
+    PermutingExperimenter,
+    SparseExperimenter,
+)

@@ -62,3 +62,19 @@ class StudyStateInfo:
   def __init__(self, state: StudyState, details: str = ''):
     self.state = StudyState(state)
     self.details = details
+
+# Reference-compat additions (vizier/pyvizier/__init__.py parity).
+from typing import Sequence as _Sequence, Union as _Union
+
+from vizier_amd._src.pyvizier.parameter_iterators import (
+    SequentialParameterBuilder,
+)
+from vizier_amd._src.pyvizier.trial import (
+    CompletedTrialWithMeasurements,
+    PendingTrialWithMeasurements,
+)
+from vizier_amd._src.pythia.policy import StudyDescriptor
+
+ParameterValueSequence = _Union[ParameterValueTypes, _Sequence[int],
+                                _Sequence[float], _Sequence[str],
+                                _Sequence[bool]]

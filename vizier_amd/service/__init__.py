@@ -16,3 +16,10 @@ from vizier_amd._src.service.vizier_server import (
     DistributedPythiaVizierServer,
 )
 from vizier_amd._src.service.vizier_service import VizierServicer
+
+from vizier_amd._src.service.constants import (
+    NO_ENDPOINT,
+    SQL_LOCAL_URL,
+    SQL_MEMORY_URL,
+    VIZIER_DB_PATH,
+)
