@@ -19,6 +19,10 @@ _MODULES = {
         'vizier_amd.benchmarks.experimenters',
     'vizier/benchmarks/analyzers.py': 'vizier_amd.benchmarks.analyzers',
     'vizier/service/__init__.py': 'vizier_amd.service',
+    'vizier/service/clients/__init__.py': 'vizier_amd.service.clients',
+    'vizier/service/servers/__init__.py': 'vizier_amd.service.servers',
+    'vizier/service/pyvizier/__init__.py': 'vizier_amd.service.pyvizier',
+    'vizier/raytune/__init__.py': 'vizier_amd.raytune',
 }
 
 

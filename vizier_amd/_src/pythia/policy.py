@@ -8,6 +8,7 @@ from __future__ import annotations
 
 import abc
 import dataclasses
+import enum
 from typing import FrozenSet, Iterable, List, Optional, Sequence, Union
 
 from vizier_amd._src.pyvizier.base_study_config import ProblemStatement
@@ -18,6 +19,22 @@ from vizier_amd._src.pyvizier.trial import (
     MetadataDelta,
     TrialSuggestion,
 )
+
+
+class StudyState(enum.Enum):
+  """Cross-platform study lifecycle state (pythia/study.py:25)."""
+  ACTIVE = 'ACTIVE'
+  ABORTED = 'ABORTED'
+  COMPLETED = 'COMPLETED'
+
+
+@dataclasses.dataclass
+class StudyStateInfo:
+  state: StudyState
+  explanation: str = ''
+
+  def __post_init__(self):
+    self.state = StudyState(self.state)
 
 
 @dataclasses.dataclass(frozen=True)

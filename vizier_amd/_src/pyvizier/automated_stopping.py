@@ -29,3 +29,8 @@ class AutomatedStoppingConfig:
     if not isinstance(other, AutomatedStoppingConfig):
       return NotImplemented
     return self._proto == other._proto
+
+
+from vizier_amd._src.service.proto import study_pb2 as _study_pb2
+
+AutomatedStoppingConfigProto = _study_pb2.StudySpec.DefaultEarlyStoppingSpec

@@ -437,3 +437,44 @@ class TrialSuggestionConverter:
 
 # Alias matching the reference's naming (proto_converters.py:869).
 SuggestConverter = TrialSuggestionConverter
+
+
+# Parameter sequence alias (oss/proto_converters.py:41).
+from vizier_amd._src.pyvizier.parameter_config import (  # noqa: E402
+    MonotypeParameterSequence,
+)
+
+
+class StudyStateConverter:
+  """pythia StudyState <-> study_pb2.Study.State (oss converters :44)."""
+
+  _TO_PROTO = {'ACTIVE': 'ACTIVE', 'ABORTED': 'INACTIVE',
+               'COMPLETED': 'COMPLETED'}
+
+  @classmethod
+  def to_proto(cls, state):
+    from vizier_amd._src.service.proto import study_pb2
+    return study_pb2.Study.State.Value(cls._TO_PROTO[state.value])
+
+  @classmethod
+  def from_proto(cls, proto_state):
+    from vizier_amd._src.pythia.policy import StudyState
+    from vizier_amd._src.service.proto import study_pb2
+    name = study_pb2.Study.State.Name(proto_state)
+    for k, v in cls._TO_PROTO.items():
+      if v == name:
+        return StudyState(k)
+    return StudyState.ACTIVE
+
+
+class MetadataDeltaConverter:
+  """MetadataDelta <-> UnitMetadataUpdate protos (oss converters :765)."""
+
+  @classmethod
+  def to_protos(cls, delta):
+    from vizier_amd._src.pyvizier import metadata_util
+    updates = metadata_util.study_metadata_to_update_list(delta.on_study)
+    updates.extend(
+        metadata_util.trial_metadata_to_update_list(delta.on_trials))
+    return updates
+

@@ -10,6 +10,8 @@ from typing import Any, Callable, Iterator, List, Mapping, Optional, Type
 
 from vizier_amd import pyvizier as vz
 from vizier_amd.client import client_abc
+from vizier_amd.client.client_abc import ResourceNotFoundError
+from vizier_amd._src.service.constants import UNUSED_CLIENT_ID
 from vizier_amd._src.pyvizier import proto_converters as pc
 from vizier_amd._src.service import resources
 from vizier_amd._src.service import vizier_client
