@@ -122,7 +122,10 @@ class VizierGPUCBPEBandit(Designer):
     suggestions: List[vz.TrialSuggestion] = []
     for i in range(count):
       use_ucb = self._choose_ucb(has_new and i == 0 and not self._active)
-      dense = self._optimize_one(use_ucb, torch.cat(x_all, dim=0))
+      # Single-element cat would COPY, breaking the trust-region anchor
+      # (posterior.x identity) that unlocks the fused scorer path.
+      x_cat = x_all[0] if len(x_all) == 1 else torch.cat(x_all, dim=0)
+      dense = self._optimize_one(use_ucb, x_cat)
       x_all.append(dense.reshape(1, -1))
       params = self._converter.to_parameters(
           dense.detach().cpu().numpy())[0]
@@ -228,10 +231,17 @@ class VizierGPUCBPEBandit(Designer):
 
       def score_fn(batch: CandidateBatch) -> torch.Tensor:
         return scoring(self._codec.decode(batch)[:, 0, :])
-      # The trust region anchors at completed+pending points (not the
-      # GP train set), so scoring uses rocBLAS — not hipGraph-capturable
-      # on this ROCm build.
-      score_fn.graph_safe = False
+      if scoring._tr_anchored and scoring._acq_name is not None:
+        # No pending/hallucinated points: the trust region anchors at
+        # the GP train set, so the pure-HIP chunked scorer applies —
+        # hipGraph-capturable and megakernel-eligible (the common
+        # sequential-suggest case; same fast path as GP-Bandit).
+        score_fn.scoring = scoring
+        score_fn.codec_identity = self._codec.identity
+      else:
+        # Pending points anchor the trust region elsewhere: composed
+        # rocBLAS path, not capturable on this ROCm build.
+        score_fn.graph_safe = False
     else:
       # Promising-region threshold: predicted mean at the observed point
       # with the highest UCB (gp_ucb_pe.py:175-205).
