@@ -69,6 +69,7 @@ class GPBanditConfig:
   ard_warm_restarts: int = 2  # random restarts kept on warm refits
   use_trust_region: bool = True
   num_scalarizations: int = 1000  # multi-objective
+  scorer_gram_dtype: str = 'fp32'  # 'fp32'|'bf16'|'fp8' candidate grams
   data_parallel: bool = False     # shard the sweep across dist ranks
   device: Optional[str] = None
   dtype: torch.dtype = torch.float32
@@ -251,12 +252,27 @@ class VizierGPBandit(Designer, Predictor):
           cfg.num_scalarizations, len(self._posteriors), seed=self._seed,
           reference_point=acq_lib.get_reference_point(self._warped_labels))
 
+      def _predict(post, flat):
+        if cfg.scorer_gram_dtype == 'fp32':
+          return post.predict(flat)
+        # bf16 / fp8 MFMA candidate gram (config 5), fp32 GEMMs after.
+        from vizier_amd._src.ops import dispatch as ops
+        ext = ops.require_ext()
+        gram = (ext.gram_matern52_fp8 if cfg.scorer_gram_dtype == 'fp8'
+                else ext.gram_matern52_bf16)
+        k = gram(flat, post.x, post.params.lengthscales,
+                 float(post.params.amplitude))
+        mean = post.params.mean + k @ post.alpha
+        amp2 = post.params.amplitude ** 2
+        var = (amp2 - (k * (k @ post.K_inv)).sum(-1)).clamp_min(1e-12)
+        return mean, var.sqrt()
+
       def score_fn(batch: CandidateBatch) -> torch.Tensor:
         dense = self._codec.decode(batch)
         flat = dense.reshape(-1, dense.shape[-1])
         per_metric = []
         for post in self._posteriors:
-          mean, stddev = post.predict(flat)
+          mean, stddev = _predict(post, flat)
           per_metric.append(mean + cfg.ucb_coefficient * stddev)
         ys = torch.stack(per_metric, dim=-1)       # (B*q, M)
         scores = scalarizer(ys).mean(dim=0)        # (B*q,)
@@ -333,7 +349,8 @@ class VizierGPBandit(Designer, Predictor):
       acquisition = acq_lib.Sample(seed=self._seed)
     else:
       acquisition = acq_lib.UCB(coefficient=cfg.ucb_coefficient)
-    scoring = acq_lib.ScoringFunction(posterior, acquisition, trust_region)
+    scoring = acq_lib.ScoringFunction(posterior, acquisition, trust_region,
+                                      gram_dtype=cfg.scorer_gram_dtype)
 
     def score_fn(batch: CandidateBatch) -> torch.Tensor:
       dense = self._codec.decode(batch)[:, 0, :]

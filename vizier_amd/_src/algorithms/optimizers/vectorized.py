@@ -136,6 +136,7 @@ class VectorizedOptimizer:
       return False
     post = scoring.posterior
     return (scoring._acq_name is not None and scoring._tr_anchored and
+            getattr(scoring, 'gram_dtype', 'fp32') == 'fp32' and
             post.K_inv is not None and post.x.is_cuda and
             strategy.pool_size <= 128 and post.x.shape[0] <= 8192)
 

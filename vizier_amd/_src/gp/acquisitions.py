@@ -278,10 +278,16 @@ class ScoringFunction:
 
   def __init__(self, posterior: GPPosterior,
                acquisition: AcquisitionFunction,
-               trust_region: Optional[TrustRegion] = None):
+               trust_region: Optional[TrustRegion] = None,
+               gram_dtype: str = 'fp32'):
     self.posterior = posterior
     self.acquisition = acquisition
     self.trust_region = trust_region
+    # 'fp32' (default) | 'bf16' | 'fp8': compute dtype for the
+    # candidate k-vector Gram in the COMPOSED path (bf16/fp8 MFMA
+    # kernels; configs 2/5 of BASELINE.json). The fused chunked kernel
+    # stays fp32, so a non-fp32 gram_dtype forces the composed path.
+    self.gram_dtype = gram_dtype
     # Pre-extract scalars once so the hot loop never syncs the device.
     self._amp = float(posterior.params.amplitude)
     self._mean_c = float(posterior.params.mean)
@@ -308,7 +314,8 @@ class ScoringFunction:
     if self._can_fuse(xs):
       post = self.posterior
       from vizier_amd._src.ops import dispatch as ops
-      if self._acq_name is not None and self._tr_anchored:
+      if (self._acq_name is not None and self._tr_anchored and
+          self.gram_dtype == 'fp32'):
         # Primary GPU path: the 3-kernel chunked HIP scorer — one
         # Python op; k-vectors + chip-filling K^-1 quadform chunks +
         # acquisition/trust-region finalize (posterior_score.hip).
@@ -325,9 +332,17 @@ class ScoringFunction:
             self._tr_radius if self.trust_region is not None else 0.0)
       # Composed path: hand-written HIP gram kernel for the k-vectors +
       # rocBLAS for the plain K^-1 quadform GEMM (used for exotic
-      # acquisitions / unanchored trust regions).
-      k = ops.gram_matern52(xs, post.x, post.params.lengthscales,
-                            post.params.amplitude)
+      # acquisitions / unanchored trust regions / bf16+fp8 grams).
+      ext = ops.require_ext()
+      if self.gram_dtype == 'bf16':
+        k = ext.gram_matern52_bf16(xs, post.x, post.params.lengthscales,
+                                   float(post.params.amplitude))
+      elif self.gram_dtype == 'fp8':
+        k = ext.gram_matern52_fp8(xs, post.x, post.params.lengthscales,
+                                  float(post.params.amplitude))
+      else:
+        k = ops.gram_matern52(xs, post.x, post.params.lengthscales,
+                              post.params.amplitude)
       amp2 = self._amp * self._amp
       var = (amp2 - (k * (k @ post.K_inv)).sum(-1)).clamp_min(1e-12)
       mean = self._mean_c + k @ post.alpha
