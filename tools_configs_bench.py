@@ -72,11 +72,15 @@ def timed_suggest(designer, count=1, reps=3):
   return float(np.mean(times) * 1e3)
 
 
-def main():
+def run_one(which: str) -> None:
+  main(only=which)
+
+
+def main(only=None):
   out = {}
 
   # Config 2: 8D, N=500, bf16 grams.
-  for dtype in ('fp32', 'bf16'):
+  for dtype in (('fp32', 'bf16') if only in (None, '2') else ()):
     d = VizierGPBandit(make_problem(8), GPBanditConfig(
         max_evaluations=75000, device='cuda',
         scorer_gram_dtype=dtype), seed=0)
@@ -87,6 +91,27 @@ def main():
           flush=True)
 
   # Config 3: 20D, N=2000, qEI count=8.
+  if only in (None, '3'):
+    _config3(out)
+
+  # Config 4: 50D, N=10000, 1.25M evals (per-GPU slice of 10M DP=8).
+  if only in (None, '4'):
+    _config4(out)
+
+  # Config 5: 30D MO + trust region, fp8 grams.
+  if only in (None, '5'):
+    _config5(out)
+
+  print(json.dumps(out))
+  import os
+  path = 'gpurun_out/configs_bench.json'
+  old = json.load(open(path)) if os.path.exists(path) else {}
+  old.update(out)
+  with open(path, 'w') as f:
+    json.dump(old, f, indent=2)
+
+
+def _config3(out):
   d = VizierGPBandit(make_problem(20), GPBanditConfig(
       max_evaluations=75000, acquisition='qei', device='cuda'), seed=0)
   preload(d, 20, 2000)
@@ -95,7 +120,8 @@ def main():
   print(f'config 3 (20D N=2000 qEI q=8): {ms:.1f} ms/suggest(8)',
         flush=True)
 
-  # Config 4: 50D, N=10000, 1.25M evals (per-GPU slice of 10M DP=8).
+
+def _config4(out):
   d = VizierGPBandit(make_problem(50), GPBanditConfig(
       max_evaluations=1_250_000, device='cuda'), seed=0)
   preload(d, 50, 10000)
@@ -104,7 +130,8 @@ def main():
   print(f'config 4 (50D N=10000, 1.25M evals): {ms:.1f} ms/suggest',
         flush=True)
 
-  # Config 5: 30D MO + trust region, fp8 grams.
+
+def _config5(out):
   for dtype in ('fp32', 'fp8'):
     d = VizierGPBandit(make_problem(30, metrics=('f1', 'f2')),
                        GPBanditConfig(max_evaluations=75000,
@@ -115,11 +142,10 @@ def main():
     out[f'config5_30d_mo_{dtype}'] = ms
     print(f'config 5 (30D MO N=1000, {dtype}): {ms:.1f} ms/suggest',
           flush=True)
-
-  print(json.dumps(out))
-  with open('gpurun_out/configs_bench.json', 'w') as f:
-    json.dump(out, f, indent=2)
+    del d
+    torch.cuda.empty_cache()
 
 
 if __name__ == '__main__':
-  main()
+  import sys as _sys
+  main(only=_sys.argv[1] if len(_sys.argv) > 1 else None)

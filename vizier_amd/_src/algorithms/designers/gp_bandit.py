@@ -351,6 +351,14 @@ class VizierGPBandit(Designer, Predictor):
       acquisition = acq_lib.UCB(coefficient=cfg.ucb_coefficient)
     scoring = acq_lib.ScoringFunction(posterior, acquisition, trust_region,
                                       gram_dtype=cfg.scorer_gram_dtype)
+    if cfg.scorer_gram_dtype != 'fp32':
+      # bf16/fp8 grams run the composed path (host-side range scaling
+      # inside the binding), which is not capture-safe.
+      def score_fn(batch: CandidateBatch) -> torch.Tensor:
+        dense = self._codec.decode(batch)[:, 0, :]
+        return scoring(dense)
+      score_fn.graph_safe = False
+      return score_fn, 1
 
     def score_fn(batch: CandidateBatch) -> torch.Tensor:
       dense = self._codec.decode(batch)[:, 0, :]
