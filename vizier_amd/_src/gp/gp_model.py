@@ -220,6 +220,13 @@ def train_gp(x: torch.Tensor, y: torch.Tensor, *,
     raw0 = torch.cat([warm_start_raw.reshape(1, -1).to(raw0), raw0], 0)
 
   def loss_fn(raw: torch.Tensor) -> torch.Tensor:
+    if n >= 8000 and raw.requires_grad:
+      # hipBLAS's batched trsm in the Cholesky backward requests a
+      # workspace it cannot allocate at R x 10000^2 (ALLOC_FAILED,
+      # ROCm 7.2). Per-row graphs keep each backward trsm at (N, N).
+      return torch.cat([
+          negative_log_marginal_likelihood(raw[i:i + 1], x, y)
+          for i in range(raw.shape[0])])
     return negative_log_marginal_likelihood(raw, x, y)
 
   best_raw, best_f = lbfgs.minimize_batched(loss_fn, raw0,
