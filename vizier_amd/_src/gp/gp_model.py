@@ -36,6 +36,7 @@ from vizier_amd._src.ops import dispatch as ops
 # window — block-diagonal padding is exact: chol([[K,0],[0,I]]) =
 # [[chol(K),0],[0,I]], logdet unchanged, and we slice the result back.
 _MAGMA_BAD_LO, _MAGMA_BAD_HI = 257, 511  # generous upper margin
+_NO_GRAD_FIT_N = 8000  # above this, hyperparameters freeze (see train_gp)
 _MAGMA_PAD_N = 512
 
 
@@ -220,18 +221,21 @@ def train_gp(x: torch.Tensor, y: torch.Tensor, *,
     raw0 = torch.cat([warm_start_raw.reshape(1, -1).to(raw0), raw0], 0)
 
   def loss_fn(raw: torch.Tensor) -> torch.Tensor:
-    if n >= 8000 and raw.requires_grad:
-      # hipBLAS's batched trsm in the Cholesky backward requests a
-      # workspace it cannot allocate at R x 10000^2 (ALLOC_FAILED,
-      # ROCm 7.2). Per-row graphs keep each backward trsm at (N, N).
-      return torch.cat([
-          negative_log_marginal_likelihood(raw[i:i + 1], x, y)
-          for i in range(raw.shape[0])])
     return negative_log_marginal_likelihood(raw, x, y)
 
-  best_raw, best_f = lbfgs.minimize_batched(loss_fn, raw0,
-                                            max_iters=max_iters,
-                                            check_every=5)
+  if max_iters <= 0 or n >= _NO_GRAD_FIT_N:
+    # Gradient-free selection: hipBLAS's trsm BACKWARD fails with
+    # ALLOC_FAILED at N=10000 on ROCm 7.2 (even unbatched), so huge
+    # studies cannot autograd the NLL. By that size the warm-started
+    # hyperparameters are stable; pick the best candidate row by
+    # no-grad NLL (forward factorizations work fine).
+    with torch.no_grad():
+      f0 = loss_fn(raw0)
+    best_raw, best_f = raw0, f0
+  else:
+    best_raw, best_f = lbfgs.minimize_batched(loss_fn, raw0,
+                                              max_iters=max_iters,
+                                              check_every=5)
   idx = int(torch.argmin(best_f))
   raw = best_raw[idx]
   params = GPParams.from_raw(raw)
@@ -260,7 +264,17 @@ def train_gp(x: torch.Tensor, y: torch.Tensor, *,
   K_inv = None
   if precompute_inverse:
     eye = torch.eye(n, dtype=x64.dtype, device=x.device)
-    z = torch.linalg.solve_triangular(L64, eye, upper=False)
+    if n >= _NO_GRAD_FIT_N:
+      # Column-blocked solves: one huge trsm can hit hipBLAS's
+      # ALLOC_FAILED workspace failure at this scale (ROCm 7.2).
+      cols = []
+      step = 2048
+      for c0 in range(0, n, step):
+        cols.append(torch.linalg.solve_triangular(
+            L64, eye[:, c0:c0 + step], upper=False))
+      z = torch.cat(cols, dim=1)
+    else:
+      z = torch.linalg.solve_triangular(L64, eye, upper=False)
     K_inv = (z.T @ z).to(x.dtype)
   return GPPosterior(x=x, params=params, L=L64.to(x.dtype),
                      alpha=alpha.to(x.dtype), K_inv=K_inv,
