@@ -105,6 +105,24 @@ def _init_raw(num_restarts: int, dim: int, generator: torch.Generator,
   return raw
 
 
+def _blocked_solve_lower(L: torch.Tensor, B: torch.Tensor,
+                         nb: int = 2048) -> torch.Tensor:
+  """Solves L Z = B (L lower-triangular) by blocked forward
+  substitution: (nb, nb) diagonal trsm solves + GEMM updates. hipBLAS's
+  trsm fails with ALLOC_FAILED whenever lda is ~10000 on ROCm 7.2
+  (any rhs width > 1), so huge solves must never hand it the full L."""
+  n = L.shape[-1]
+  Z = torch.empty_like(B)
+  for i0 in range(0, n, nb):
+    i1 = min(i0 + nb, n)
+    rhs = B[i0:i1].clone()
+    if i0 > 0:
+      rhs -= L[i0:i1, :i0] @ Z[:i0]
+    Z[i0:i1] = torch.linalg.solve_triangular(
+        L[i0:i1, i0:i1].contiguous(), rhs, upper=False)
+  return Z
+
+
 def _chol_solve(L: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
   """(L L^T)^-1 b via two triangular solves (batched)."""
   z = torch.linalg.solve_triangular(L, b, upper=False)
@@ -265,14 +283,7 @@ def train_gp(x: torch.Tensor, y: torch.Tensor, *,
   if precompute_inverse:
     eye = torch.eye(n, dtype=x64.dtype, device=x.device)
     if n >= _NO_GRAD_FIT_N:
-      # Column-blocked solves: one huge trsm can hit hipBLAS's
-      # ALLOC_FAILED workspace failure at this scale (ROCm 7.2).
-      cols = []
-      step = 2048
-      for c0 in range(0, n, step):
-        cols.append(torch.linalg.solve_triangular(
-            L64, eye[:, c0:c0 + step], upper=False))
-      z = torch.cat(cols, dim=1)
+      z = _blocked_solve_lower(L64, eye)
     else:
       z = torch.linalg.solve_triangular(L64, eye, upper=False)
     K_inv = (z.T @ z).to(x.dtype)
