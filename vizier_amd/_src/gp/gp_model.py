@@ -112,14 +112,19 @@ def _blocked_solve_lower(L: torch.Tensor, B: torch.Tensor,
   trsm fails with ALLOC_FAILED whenever lda is ~10000 on ROCm 7.2
   (any rhs width > 1), so huge solves must never hand it the full L."""
   n = L.shape[-1]
+  ncols = B.shape[-1]
   Z = torch.empty_like(B)
-  for i0 in range(0, n, nb):
-    i1 = min(i0 + nb, n)
-    rhs = B[i0:i1].clone()
-    if i0 > 0:
-      rhs -= L[i0:i1, :i0] @ Z[:i0]
-    Z[i0:i1] = torch.linalg.solve_triangular(
-        L[i0:i1, i0:i1].contiguous(), rhs, upper=False)
+  # Panel BOTH dimensions: hipBLAS trsm ALLOC_FAILs whenever either
+  # extent is ~10^4, so every call here is at most (nb, nb).
+  for c0 in range(0, ncols, nb):
+    c1 = min(c0 + nb, ncols)
+    for i0 in range(0, n, nb):
+      i1 = min(i0 + nb, n)
+      rhs = B[i0:i1, c0:c1].clone()
+      if i0 > 0:
+        rhs -= L[i0:i1, :i0] @ Z[:i0, c0:c1]
+      Z[i0:i1, c0:c1] = torch.linalg.solve_triangular(
+          L[i0:i1, i0:i1].contiguous(), rhs, upper=False)
   return Z
 
 
