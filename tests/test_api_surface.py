@@ -235,3 +235,53 @@ class TestAnalyzerAdditions:
     fig, _ = plot_utils.plot_from_records(out)
     fig.savefig(tmp_path / 'plot.png')
     assert (tmp_path / 'plot.png').exists()
+
+
+class TestBenchmarkRunnerIntegration:
+  """End-to-end: runner subroutines over the new experimenters."""
+
+  def test_runner_over_branin_with_random_designer(self):
+    from vizier_amd._src.algorithms.designers.random import RandomDesigner
+    from vizier_amd._src.benchmarks.experimenters.synthetic import classic
+    from vizier_amd._src.benchmarks.runners.benchmark_runner import (
+        BenchmarkRunner,
+        FillActiveTrials,
+        EvaluateActiveTrials,
+    )
+    from vizier_amd._src.benchmarks.runners.benchmark_state import (
+        BenchmarkState,
+    )
+    exp = classic.Branin2DExperimenter()
+    state = BenchmarkState.from_designer_factory(
+        lambda p: RandomDesigner(p.search_space, seed=0), exp)
+    runner = BenchmarkRunner(
+        benchmark_subroutines=[FillActiveTrials(3),
+                               EvaluateActiveTrials()],
+        num_repeats=5)
+    runner.run(state)
+    trials = state.algorithm.supporter.GetTrials()
+    assert len(trials) == 15
+    assert all(t.final_measurement is not None for t in trials)
+
+  def test_policy_state_factory(self):
+    from vizier_amd._src.algorithms.policies.random_policy import (
+        RandomPolicy,
+    )
+    from vizier_amd._src.benchmarks.experimenters.synthetic import classic
+    from vizier_amd._src.benchmarks.runners.benchmark_state import (
+        PolicyBenchmarkStateFactory,
+    )
+    exp = classic.Branin2DExperimenter()
+
+    def policy_factory(problem, seed):
+      from vizier_amd._src.pythia.local_policy_supporters import (
+          InRamPolicySupporter,
+      )
+      del seed
+      return RandomPolicy(InRamPolicySupporter(problem))
+
+    state = PolicyBenchmarkStateFactory(
+        experimenter=exp, policy_factory=policy_factory)(seed=1)
+    state.algorithm.suggest(4)
+    trials = state.algorithm.supporter.GetTrials()
+    assert len(trials) == 4
