@@ -290,3 +290,33 @@ class TestFaultInjection:
     import pytest as _pytest
     with _pytest.raises(failing.FailedSuggestError):
       d.suggest(1)
+
+
+class TestAttrsUtils:
+
+  def test_validators(self):
+    import numpy as np
+    import pytest as _pytest
+    from vizier_amd.utils import attrs_utils as au
+
+    au.assert_not_empty(None, 'f', [1])
+    with _pytest.raises(ValueError):
+      au.assert_not_empty(None, 'f', [])
+    au.assert_not_negative(None, 'f', 0)
+    with _pytest.raises(ValueError):
+      au.assert_not_negative(None, 'f', -1)
+    with _pytest.raises(ValueError):
+      au.assert_not_none(None, 'f', None)
+    au.assert_between(0, 1)(None, 'f', 0.5)
+    with _pytest.raises(ValueError):
+      au.assert_between(0, 1)(None, 'f', 2.0)
+    au.assert_re_fullmatch(r'[a-z]+')(None, 'f', 'abc')
+    with _pytest.raises(ValueError):
+      au.assert_re_fullmatch(r'[a-z]+')(None, 'f', 'ABC')
+
+    class Holder:
+      n = 3
+    v = au.shape_equals(lambda inst: (inst.n, None))
+    v(Holder(), 'f', np.zeros((3, 7)))
+    with _pytest.raises(ValueError):
+      v(Holder(), 'f', np.zeros((4, 7)))
