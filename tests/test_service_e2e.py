@@ -305,3 +305,23 @@ class TestMultiClientStress:
     import numpy as _np
     print(f'suggest+complete latency: p50={_np.median(latencies)*1e3:.1f}'
           f'ms p95={_np.percentile(latencies, 95)*1e3:.1f}ms')
+
+
+class TestClientAbcConformance(
+    __import__('vizier_amd.client.client_abc_testing',
+               fromlist=['TestCaseMixin']).TestCaseMixin):
+  """Runs the client ABC conformance mixin against clients.Study
+  (reference client_abc_testing.py:36-48 pattern)."""
+
+  def create_study(self, problem, study_id: str):
+    from vizier_amd._src.service import clients as service_clients
+    from vizier_amd._src.service import constants
+    config = vz.StudyConfig(algorithm='RANDOM_SEARCH')
+    for pc in problem.search_space.parameters:
+      config.search_space.add(pc)
+    for mi in problem.metric_information:
+      config.metric_information.append(mi)
+    service_clients.environment_variables.server_endpoint = \
+        constants.NO_ENDPOINT
+    return service_clients.Study.from_study_config(
+        config, owner='conformance', study_id=study_id)
