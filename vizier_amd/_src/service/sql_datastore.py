@@ -163,6 +163,12 @@ class SQLDataStore(DataStore):
 
   def max_trial_id(self, study_name: str) -> int:
     with self._engine.begin() as conn:
+      exists = conn.execute(sqla.select(self._studies.c.study_name).where(
+          self._studies.c.study_name == study_name)).first()
+      if exists is None:
+        # Reference parity (sql_datastore.py:335): missing study raises.
+        raise custom_errors.NotFoundError(
+            f'Study {study_name} does not exist.')
       row = conn.execute(sqla.select(
           sqla.func.max(self._trials.c.trial_id)).where(
               self._trials.c.study_name == study_name)).first()
