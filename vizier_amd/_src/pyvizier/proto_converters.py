@@ -63,7 +63,7 @@ def _timestamp_to_datetime(ts) -> datetime.datetime:
 def _datetime_to_timestamp(dt: datetime.datetime, ts) -> None:
   secs = dt.timestamp()
   ts.seconds = int(secs)
-  ts.nanos = int(1e9 * (secs - int(secs)))
+  ts.nanos = round(1e9 * (secs - int(secs)))
 
 
 class ParameterConfigConverter:
@@ -224,7 +224,8 @@ class MeasurementConverter:
     proto.step_count = int(measurement.steps)
     secs = float(measurement.elapsed_secs)
     proto.elapsed_duration.seconds = int(secs)
-    proto.elapsed_duration.nanos = int(1e9 * (secs - int(secs)))
+    # round, not truncate: int() of e.g. 0.2s*1e9 = 199999999 nanos.
+    proto.elapsed_duration.nanos = round(1e9 * (secs - int(secs)))
     return proto
 
   @classmethod
