@@ -149,3 +149,44 @@ class TestParetoProperties:
       dominates = ((pts >= pts[i]).all(axis=1) &
                    (pts > pts[i]).any(axis=1))
       assert not dominates.any()
+
+
+class TestStudyConfigRoundtrip:
+
+  @_SETTINGS
+  @given(
+      n_float=st.integers(0, 3), n_cat=st.integers(0, 2),
+      child_under=st.integers(0, 1), algo=st.sampled_from(
+          ['RANDOM_SEARCH', 'QUASI_RANDOM_SEARCH', 'NSGA2']),
+      lo=st.floats(-10, 0, allow_nan=False),
+      width=st.floats(0.1, 10, allow_nan=False))
+  def test_conditional_study_config_proto_roundtrip(
+      self, n_float, n_cat, child_under, algo, lo, width):
+    config = vz.StudyConfig(algorithm=algo)
+    root = config.search_space.root
+    for i in range(n_float):
+      root.add_float_param(f'f{i}', lo, lo + width)
+    for i in range(n_cat):
+      root.add_categorical_param(f'c{i}', ['a', 'b', 'c'])
+    if n_cat and child_under:
+      root.select('c0', ['b']).add_int_param('child', 1, 5)
+    config.metric_information.append(vz.MetricInformation(
+        name='obj', goal=vz.ObjectiveMetricGoal.MAXIMIZE))
+
+    proto = config.to_proto()
+    back = vz.StudyConfig.from_proto(proto)
+    assert back.algorithm == config.algorithm
+    space_a = config.search_space
+    space_b = back.search_space
+    assert space_a.num_parameters() == space_b.num_parameters()
+    for pa in space_a.parameters:
+      pb = space_b.get(pa.name)
+      assert pb.type == pa.type
+      if pa.type == vz.ParameterType.DOUBLE:
+        assert pb.bounds == pytest.approx(pa.bounds)
+      else:
+        assert list(pb.feasible_values) == list(pa.feasible_values)
+      assert [c.name for c in pb.child_parameter_configs] == \
+          [c.name for c in pa.child_parameter_configs]
+    # Round trip again: proto -> config -> proto must be stable.
+    assert back.to_proto() == proto
