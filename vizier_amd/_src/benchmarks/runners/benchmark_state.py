@@ -44,8 +44,17 @@ class PolicySuggester:
       designer_factory: Callable[[vz.ProblemStatement], Designer],
       seed: Optional[int] = None) -> 'PolicySuggester':
     supporter = InRamPolicySupporter(problem)
-    policy = DesignerPolicy(supporter, designer_factory)
-    del seed
+    if seed is not None:
+      # Forward the repeat seed when the factory accepts one (the
+      # reference's benchmark repeats vary designers by seed).
+      def seeded_factory(p, _factory=designer_factory, _seed=seed):
+        try:
+          return _factory(p, seed=_seed)
+        except TypeError:
+          return _factory(p)
+      policy = DesignerPolicy(supporter, seeded_factory)
+    else:
+      policy = DesignerPolicy(supporter, designer_factory)
     return cls(policy, supporter)
 
 
