@@ -96,6 +96,14 @@ def apply(store, op):
     if kind == 'get_es_op':
       return store.get_early_stopping_operation(
           op[1]).SerializeToString(), None
+    if kind == 'list_sugg_ops':
+      return sorted(
+          o.name for o in store.list_suggestion_operations(
+              op[1], op[2])), None
+    if kind == 'update_metadata':
+      kv = study_pb2.KeyValue(key=op[2], ns=':fuzz', value=op[3])
+      store.update_metadata(op[1], [kv], [])
+      return store.load_study(op[1]).SerializeToString(), None
     raise AssertionError(op)
   except custom_errors.NotFoundError:
     return None, 'NotFoundError'
@@ -110,7 +118,8 @@ def random_op(rng, owners, studies, trial_ids):
       'create_study', 'load_study', 'list_studies', 'create_trial',
       'get_trial', 'list_trials', 'max_trial_id', 'delete_trial',
       'delete_study', 'create_sugg_op', 'get_sugg_op',
-      'max_sugg_number', 'create_es_op', 'get_es_op'])
+      'max_sugg_number', 'create_es_op', 'get_es_op',
+      'list_sugg_ops', 'update_metadata'])
   owner = rng.choice(owners)
   study = rng.choice(studies)
   study_name = f'owners/{owner}/studies/{study}'
@@ -148,6 +157,11 @@ def random_op(rng, owners, studies, trial_ids):
     return ('get_es_op',
             f'owners/{owner}/operations/earlystopping/{study}/'
             f'{rng.randint(1, 3)}')
+  if choice == 'list_sugg_ops':
+    return ('list_sugg_ops', study_name, f'cl{rng.randint(0, 1)}')
+  if choice == 'update_metadata':
+    return ('update_metadata', study_name, f'k{rng.randint(0, 2)}',
+            f'v{rng.randint(0, 9)}')
   return ('max_trial_id', study_name)
 
 
