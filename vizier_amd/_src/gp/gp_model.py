@@ -36,7 +36,7 @@ from vizier_amd._src.ops import dispatch as ops
 # window — block-diagonal padding is exact: chol([[K,0],[0,I]]) =
 # [[chol(K),0],[0,I]], logdet unchanged, and we slice the result back.
 _MAGMA_BAD_LO, _MAGMA_BAD_HI = 257, 511  # generous upper margin
-_NO_GRAD_FIT_N = 8000  # above this, hyperparameters freeze (see train_gp)
+_NO_GRAD_FIT_N = 8000  # above this, analytic-gradient fit (see train_gp)
 _MAGMA_PAD_N = 512
 
 
@@ -180,6 +180,93 @@ def negative_log_marginal_likelihood(raw: torch.Tensor, x: torch.Tensor,
   return nll
 
 
+def nll_value_and_grad(raw: torch.Tensor, x: torch.Tensor,
+                       y: torch.Tensor
+                       ) -> Tuple[torch.Tensor, torch.Tensor]:
+  """Batched NLL and its ANALYTIC gradient (no autograd).
+
+  Purpose: torch's Cholesky/solve backward needs hipBLAS trsm calls
+  that ALLOC_FAIL at N~10^4 on ROCm 7.2 (profiles/config4*.log), so
+  huge studies cannot autograd the NLL. This path uses only forward
+  factorizations + the classic GP gradient identities:
+
+    dNLL/dtheta = 1/2 tr((K^-1 - alpha alpha^T) dK/dtheta)
+                  (+ mean-parameter term  -sum(alpha)),
+
+  with the per-lengthscale traces reduced to GEMV forms
+  (sum_ij A_ij (z_id - z_jd)^2 = 2 z_d^2 . rowsum(A) - 2 z_d^T A z_d
+  for symmetric A). Verified against autograd in tests/test_gp_core.py.
+
+  raw (R, D+3) -> (nll (R,), grad (R, D+3)).
+  """
+  r_batch, p = raw.shape
+  n, d = x.shape
+  params = GPParams.from_raw(raw)
+  amp2 = (params.amplitude ** 2).reshape(-1, 1, 1)          # (R,1,1)
+  ls = params.lengthscales                                   # (R, D)
+  z = x.unsqueeze(0) / ls.unsqueeze(1)                       # (R, N, D)
+  d2 = ((z * z).sum(-1, keepdim=True) +
+        (z * z).sum(-1).unsqueeze(-2) -
+        2.0 * z @ z.transpose(-1, -2)).clamp_min(1e-18)
+  r = d2.sqrt()
+  sr = math.sqrt(5.0) * r
+  e = torch.exp(-sr)
+  k0 = (1.0 + sr + sr * sr / 3.0) * e                        # unit-amp
+  K = amp2 * k0 + params.noise.reshape(-1, 1, 1) * torch.eye(
+      n, dtype=x.dtype, device=x.device)
+  L, info = safe_cholesky_ex(K)
+  resid = (y.unsqueeze(0) - params.mean.unsqueeze(-1)).unsqueeze(-1)
+  zsol = torch.linalg.solve_triangular(L, resid, upper=False)
+  quad = (zsol * zsol).sum(dim=(-1, -2))
+  logdet = 2.0 * torch.log(torch.diagonal(L, dim1=-2, dim2=-1)).sum(-1)
+  nll = 0.5 * (quad + logdet + n * math.log(2 * math.pi))
+  nll = nll + 0.01 * (raw * raw).sum(-1)
+
+  # alpha and K^-1 per restart via blocked forward solves (safe at any
+  # N; hipBLAS never sees a ~10^4-wide trsm).
+  alpha = torch.linalg.solve_triangular(L.mT, zsol, upper=True)  # (R,N,1)
+  eye = torch.eye(n, dtype=x.dtype, device=x.device)
+  W = torch.empty_like(K)
+  for i in range(r_batch):
+    if n >= _NO_GRAD_FIT_N:
+      zi = _blocked_solve_lower(L[i], eye)
+    else:
+      zi = torch.linalg.solve_triangular(L[i], eye, upper=False)
+    W[i] = zi.T @ zi
+  M = W - alpha @ alpha.mT                                   # (R, N, N)
+
+  grad = torch.zeros_like(raw)
+  # d(bounded log-param)/d(raw) for the sigmoid reparameterization.
+  sig = torch.sigmoid(raw)
+  dbound = sig * (1 - sig)
+
+  # amplitude (raw col 0): dK/dv = 2 * amp^2 * k0, v = log amp.
+  g_amp = 0.5 * (M * (2.0 * amp2 * k0)).sum(dim=(-1, -2))
+  grad[:, 0] = g_amp * (_LOG_AMP_BOUNDS[1] - _LOG_AMP_BOUNDS[0]) *       dbound[:, 0]
+  # noise (col 1): dK/dv = noise * I.
+  g_noise = 0.5 * params.noise * torch.diagonal(
+      M, dim1=-2, dim2=-1).sum(-1)
+  grad[:, 1] = g_noise * (_LOG_NOISE_BOUNDS[1] -
+                          _LOG_NOISE_BOUNDS[0]) * dbound[:, 1]
+  # mean (col 2): dNLL/dm = -sum(alpha).
+  g_mean = -alpha.sum(dim=(-1, -2))
+  grad[:, 2] = g_mean * (_MEAN_BOUNDS[1] - _MEAN_BOUNDS[0]) *       dbound[:, 2]
+  # lengthscales (cols 3:): A = 1/2 M * G with
+  # G = amp^2 * (5/3)(1 + sqrt5 r) e^{-sqrt5 r}; symmetric.
+  A = 0.5 * M * (amp2 * (5.0 / 3.0) * (1.0 + sr) * e)
+  arow = A.sum(-1)                                           # (R, N)
+  for dd in range(d):
+    zd = z[:, :, dd]                                          # (R, N)
+    t = 2.0 * (zd * zd * arow).sum(-1) -         2.0 * torch.einsum('ri,rij,rj->r', zd, A, zd)
+    grad[:, 3 + dd] = t * (_LOG_LS_BOUNDS[1] - _LOG_LS_BOUNDS[0]) *         dbound[:, 3 + dd]
+
+  grad = grad + 0.02 * raw
+  bad = info != 0
+  nll = torch.where(bad, torch.full_like(nll, float('inf')), nll)
+  grad = torch.where(bad.unsqueeze(-1), torch.zeros_like(grad), grad)
+  return nll, grad
+
+
 @dataclasses.dataclass
 class GPPosterior:
   """Cached posterior state for fast repeated prediction."""
@@ -246,15 +333,18 @@ def train_gp(x: torch.Tensor, y: torch.Tensor, *,
   def loss_fn(raw: torch.Tensor) -> torch.Tensor:
     return negative_log_marginal_likelihood(raw, x, y)
 
-  if max_iters <= 0 or n >= _NO_GRAD_FIT_N:
-    # Gradient-free selection: hipBLAS's trsm BACKWARD fails with
-    # ALLOC_FAILED at N=10000 on ROCm 7.2 (even unbatched), so huge
-    # studies cannot autograd the NLL. By that size the warm-started
-    # hyperparameters are stable; pick the best candidate row by
-    # no-grad NLL (forward factorizations work fine).
+  if max_iters <= 0:
     with torch.no_grad():
       f0 = loss_fn(raw0)
     best_raw, best_f = raw0, f0
+  elif n >= _NO_GRAD_FIT_N:
+    # hipBLAS's trsm BACKWARD fails with ALLOC_FAILED at N~10^4 on
+    # ROCm 7.2 (even unbatched), so huge studies cannot autograd the
+    # NLL. Fit with ANALYTIC gradients instead (forward-only solves,
+    # verified against autograd to 1e-9 in tests/test_gp_core.py).
+    best_raw, best_f = lbfgs.minimize_batched(
+        loss_fn, raw0, max_iters=max_iters, check_every=5,
+        value_and_grad_fn=lambda raw: nll_value_and_grad(raw, x, y))
   else:
     best_raw, best_f = lbfgs.minimize_batched(loss_fn, raw0,
                                               max_iters=max_iters,

@@ -13,7 +13,7 @@ handled upstream by a sigmoid reparameterization (see gp_model.py).
 
 from __future__ import annotations
 
-from typing import Callable, Tuple
+from typing import Callable, Optional, Tuple
 
 import torch
 
@@ -27,6 +27,9 @@ def minimize_batched(
     grad_tol: float = 1e-7,
     ls_steps: Tuple[float, ...] = (1.0, 0.3, 0.08, 0.02),
     check_every: int = 10,
+    value_and_grad_fn: Optional[Callable[[torch.Tensor],
+                                         Tuple[torch.Tensor,
+                                               torch.Tensor]]] = None,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
   """Minimizes loss_fn over a batch of R independent parameter vectors.
 
@@ -52,9 +55,14 @@ def minimize_batched(
 
   def value_and_grad(params: torch.Tensor
                      ) -> Tuple[torch.Tensor, torch.Tensor]:
-    params = params.detach().requires_grad_(True)
-    loss = loss_fn(params)
-    grad, = torch.autograd.grad(loss.sum(), params)
+    if value_and_grad_fn is not None:
+      # Analytic gradients (e.g. gp_model.nll_value_and_grad) — used
+      # where autograd is unavailable (huge-N trsm-backward failures).
+      loss, grad = value_and_grad_fn(params.detach())
+    else:
+      params = params.detach().requires_grad_(True)
+      loss = loss_fn(params)
+      grad, = torch.autograd.grad(loss.sum(), params)
     bad = ~torch.isfinite(loss)
     loss = torch.where(bad, inf, loss)
     grad = torch.where(bad.unsqueeze(-1) | ~torch.isfinite(grad),
