@@ -110,6 +110,25 @@ class TestScheduled:
     assert captured[0] == 10.0
     assert captured[1] == pytest.approx(5.0)
 
+  def test_scheduled_gp_factories(self):
+    from vizier_amd._src.algorithms.designers.scheduled_designer import (
+        scheduled_gp_bandit, scheduled_gp_ucb_pe)
+    problem = vz.ProblemStatement()
+    problem.search_space.root.add_float_param('x', 0.0, 1.0)
+    problem.metric_information.append(vz.MetricInformation(name='m',
+                                                           goal=1))
+    for make in (scheduled_gp_bandit, scheduled_gp_ucb_pe):
+      d = make(problem, expected_total_num_trials=20)
+      suggestions = d.suggest(2)
+      assert len(suggestions) == 2
+      trials = []
+      for i, s in enumerate(suggestions):
+        t = s.to_trial(i + 1)
+        t.complete(vz.Measurement(metrics={'m': float(i)}))
+        trials.append(t)
+      d.update(CompletedTrials(trials), ActiveTrials())
+      assert len(d.suggest(1)) == 1
+
 
 class TestEnsembles:
 

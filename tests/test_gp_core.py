@@ -183,6 +183,45 @@ class TestOutputWarpers:
     out = output_warpers.ZScoreLabels().warp(labels)
     assert abs(out.mean()) < 1e-9
 
+  def test_detect_outliers_marks_only_outliers(self):
+    rng = np.random.default_rng(0)
+    labels = np.concatenate([rng.normal(5.0, 1.0, 40),
+                             [-1e7, -1e76]])[:, None]
+    out = output_warpers.DetectOutliers().warp(labels.copy())
+    assert np.isnan(out[-1, 0]) and np.isnan(out[-2, 0])
+    assert np.isfinite(out[:40]).all()
+    np.testing.assert_allclose(out[:40], labels[:40])
+
+  def test_warp_outliers_pipeline_finite(self):
+    rng = np.random.default_rng(1)
+    labels = np.concatenate([rng.normal(0.0, 2.0, 30),
+                             [np.nan, -1e76]])[:, None]
+    out = output_warpers.create_warp_outliers_warper().warp(
+        labels.copy())
+    assert out.shape == labels.shape
+    assert np.isfinite(out).all()
+
+  def test_transform_to_gaussian_monotone(self):
+    labels = np.array([[0.1], [3.0], [1.5], [7.0]])
+    out = output_warpers.TransformToGaussian().warp(labels.copy())
+    order_in = np.argsort(labels[:, 0])
+    order_out = np.argsort(out[:, 0])
+    np.testing.assert_array_equal(order_in, order_out)
+    # With use_rank the spacing is rank-based but the order holds too.
+    out_r = output_warpers.TransformToGaussian(use_rank=True).warp(
+        labels.copy())
+    np.testing.assert_array_equal(np.argsort(out_r[:, 0]), order_in)
+
+  def test_linear_output_warper_roundtrip(self):
+    y = np.array([[1.0, 10.0], [3.0, 20.0], [2.0, 12.0]])
+    w = output_warpers.LinearOutputWarper.from_obs(y)
+    z = w.warp(y)
+    assert z.min() >= -2.0 - 1e-12 and z.max() <= 2.0 + 1e-12
+    np.testing.assert_allclose(w.unwarp(z), y, rtol=1e-12)
+    yt = torch.tensor(y)
+    wt = output_warpers.LinearOutputWarper.from_obs(yt)
+    assert torch.allclose(wt.unwarp(wt.warp(yt)), yt)
+
 
 class TestTransferLearning:
 

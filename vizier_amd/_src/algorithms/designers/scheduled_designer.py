@@ -110,3 +110,39 @@ def scheduled_gp_bandit(problem: vz.ProblemStatement, *,
       {'ucb_coefficient': ExponentialScheduledParam(
           init_ucb_coefficient, final_ucb_coefficient, decay_rate)},
       expected_total_num_trials=expected_total_num_trials)
+
+
+def scheduled_gp_ucb_pe(problem: vz.ProblemStatement, *,
+                        expected_total_num_trials: int,
+                        init_ucb_coefficient: float = 4.0,
+                        final_ucb_coefficient: float = 1.0,
+                        decay_ucb_coefficient: float = 1.2,
+                        init_explore_region_ucb_coefficient: float = 1.0,
+                        final_explore_region_ucb_coefficient: float = 0.5,
+                        decay_explore_region_ucb_coefficient: float = 1.2,
+                        **pe_kwargs) -> ScheduledDesigner:
+  """GP-UCB-PE with exponentially decaying UCB and explore-region
+  coefficients (parity with scheduled_gp_ucb_pe.py:29
+  ScheduledGPUCBPEFactory)."""
+  from vizier_amd._src.algorithms.designers.gp_ucb_pe import (
+      UCBPEConfig,
+      VizierGPUCBPEBandit,
+  )
+
+  def factory(p, ucb_coefficient, explore_region_ucb_coefficient):
+    cfg = UCBPEConfig(
+        ucb_coefficient=ucb_coefficient,
+        explore_region_ucb_coefficient=explore_region_ucb_coefficient,
+        **pe_kwargs)
+    return VizierGPUCBPEBandit(p, cfg)
+
+  return ScheduledDesigner(
+      problem, factory,
+      {'ucb_coefficient': ExponentialScheduledParam(
+          init_ucb_coefficient, final_ucb_coefficient,
+          decay_ucb_coefficient),
+       'explore_region_ucb_coefficient': ExponentialScheduledParam(
+           init_explore_region_ucb_coefficient,
+           final_explore_region_ucb_coefficient,
+           decay_explore_region_ucb_coefficient)},
+      expected_total_num_trials=expected_total_num_trials)

@@ -212,3 +212,152 @@ def create_default_warper(*, half_rank_warp: bool = True,
   if infeasible_warp:
     warpers.append(InfeasibleWarperComponent())
   return OutputWarperPipeline(warpers)
+
+
+class DetectOutliers(OutputWarper):
+  """Marks unreasonably-bad finite labels as NaN (infeasible).
+
+  Parity with output_warpers.py:578 (DetectOutliers). The variance of
+  the good half is estimated from (max - median) with the sample-size
+  dependent divisors of Hozo et al. 2005 (the reference's source);
+  labels below median - min_zscore * std become NaN so the infeasible
+  warper can handle them.
+  """
+
+  def __init__(self, *, min_zscore: float = 6.0,
+               max_zscore: Optional[float] = None):
+    self.min_zscore = min_zscore
+    self.max_zscore = max_zscore
+
+  def _estimate_variance(self, labels_arr: np.ndarray) -> float:
+    num_points = labels_arr.shape[0]
+    labels_median = float(np.nanmedian(labels_arr))
+    labels_max = float(np.nanmax(labels_arr))
+    if not np.isfinite(labels_max):
+      raise ValueError('The max label value should be finite.')
+    if not np.isfinite(labels_median):
+      raise ValueError('The median label value should be finite.')
+    if self.max_zscore:
+      return (labels_max - labels_median) / self.min_zscore
+    if num_points >= 70:
+      return (labels_max - labels_median) / 3
+    if num_points >= 15:
+      return (labels_max - labels_median) / 2
+    # Small-sample range-based estimator (eq. 12 of the paper above),
+    # hallucinating the min at median - max as the reference does.
+    labels_min = labels_median - np.max(labels_arr)
+    if labels_min < 0:
+      labels_min = 0.0
+    a, m, b, n = labels_min, labels_median, labels_max, num_points
+    out = a ** 2 + m ** 2 + b ** 2
+    out += ((n - 3) / 2) * ((a + m) ** 2 + (b + m) ** 2) / 4
+    out -= n * ((a + 2 * m + b) / 4 + (a - 2 * m + b) / (4 * n)) ** 2
+    return out / (n - 1)
+
+  def warp(self, labels_arr: np.ndarray) -> np.ndarray:
+    labels_arr = _validate(labels_arr)
+    finite = np.isfinite(labels_arr)
+    vals = labels_arr[finite]
+    median = np.median(vals)
+    std = np.sqrt(self._estimate_variance(vals))
+    threshold = median - self.min_zscore * std
+    vals[vals < threshold] = np.nan
+    labels_arr[finite] = vals
+    return labels_arr
+
+
+def _softclip(x: np.ndarray, low: float, high: float,
+              softness: float) -> np.ndarray:
+  """Smooth clip of x into (low, high): softplus-based, identity for
+  interior values when softness is small (TFP SoftClip semantics)."""
+  sp = lambda t: np.logaddexp(0.0, t)  # softplus, overflow-safe
+  return low + softness * sp((x - low) / softness) \
+      - softness * sp((x - high) / softness)
+
+
+class TransformToGaussian(OutputWarper):
+  """Quantile-transforms labels toward a standard Gaussian.
+
+  Parity with output_warpers.py:666. Labels (or their ranks when
+  use_rank) are min-max normalized, soft-clipped into (0, 1), and
+  mapped through the normal quantile function.
+  """
+
+  def __init__(self, *, softclip_low: float = 1e-10,
+               softclip_high: float = 1 - 1e-10,
+               softclip_hinge_softness: float = 0.01,
+               use_rank: bool = False):
+    self.softclip_low = softclip_low
+    self.softclip_high = softclip_high
+    self.softclip_hinge_softness = softclip_hinge_softness
+    self.use_rank = use_rank
+
+  def warp(self, labels_arr: np.ndarray) -> np.ndarray:
+    labels_arr = _validate(labels_arr)
+    flat = labels_arr.flatten()
+    base = np.argsort(flat).astype(np.float64) if self.use_rank else flat
+    span = np.max(base) - np.min(base)
+    normalized = (base - np.min(base)) / span if span > 0 \
+        else np.zeros_like(base)
+    clipped = _softclip(normalized, self.softclip_low,
+                        self.softclip_high,
+                        self.softclip_hinge_softness)
+    from scipy import special
+    return special.ndtri(clipped).reshape(labels_arr.shape)
+
+
+class LinearOutputWarper:
+  """Invertible affine map of each metric column to [low, high].
+
+  Parity with output_warpers.py:729 (equinox LinearOutputWarper):
+  per-metric min/max from observations, slope floored by min_range.
+  Works on numpy arrays and torch tensors alike (pure arithmetic).
+  """
+
+  def __init__(self, *, low_bound, high_bound, min_value, max_value,
+               min_range):
+    self.low_bound = low_bound
+    self.high_bound = high_bound
+    self.min_value = min_value
+    self.max_value = max_value
+    self.min_range = min_range
+
+  @classmethod
+  def from_obs(cls, y_obs, low_bound: float = -2.0,
+               high_bound: float = 2.0,
+               min_range: float = 1e-20) -> 'LinearOutputWarper':
+    min_value = y_obs.min(axis=0) if isinstance(y_obs, np.ndarray) \
+        else y_obs.min(dim=0).values
+    max_value = y_obs.max(axis=0) if isinstance(y_obs, np.ndarray) \
+        else y_obs.max(dim=0).values
+    return cls(low_bound=low_bound, high_bound=high_bound,
+               min_value=min_value, max_value=max_value,
+               min_range=min_range)
+
+  def _slope(self):
+    rng = self.max_value - self.min_value
+    if isinstance(rng, np.ndarray):
+      rng = np.maximum(rng, self.min_range)
+    else:
+      rng = rng.clamp_min(self.min_range)
+    return (self.high_bound - self.low_bound) / rng
+
+  def warp(self, y):
+    return (y - self.min_value) * self._slope() + self.low_bound
+
+  def unwarp(self, y):
+    return (y - self.low_bound) / self._slope() + self.min_value
+
+
+def create_warp_outliers_warper(
+    *, warp_outliers: bool = True, infeasible_warp: bool = True,
+    transform_gaussian: bool = True) -> OutputWarperPipeline:
+  """Outlier-robust pipeline (output_warpers.py:216)."""
+  warpers: List[OutputWarper] = []
+  if warp_outliers:
+    warpers.append(DetectOutliers())
+  if infeasible_warp:
+    warpers.append(InfeasibleWarperComponent())
+  if transform_gaussian:
+    warpers.append(TransformToGaussian())
+  return OutputWarperPipeline(warpers)
