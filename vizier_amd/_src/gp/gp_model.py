@@ -19,6 +19,7 @@ from __future__ import annotations
 
 import dataclasses
 import math
+import os
 from typing import Optional, Tuple
 
 import torch
@@ -333,11 +334,18 @@ def train_gp(x: torch.Tensor, y: torch.Tensor, *,
   def loss_fn(raw: torch.Tensor) -> torch.Tensor:
     return negative_log_marginal_likelihood(raw, x, y)
 
+  # VIZIER_AMD_ANALYTIC_NLL: 'auto' (default — analytic gradients only
+  # at N >= _NO_GRAD_FIT_N where trsm-backward is broken), 'always', or
+  # 'never'. 'always' is a round-2 A/B candidate: the analytic path
+  # does one K^-1 build instead of autograd's taped solve chain.
+  analytic_mode = os.environ.get('VIZIER_AMD_ANALYTIC_NLL', 'auto')
+  use_analytic = (analytic_mode == 'always' or
+                  (analytic_mode != 'never' and n >= _NO_GRAD_FIT_N))
   if max_iters <= 0:
     with torch.no_grad():
       f0 = loss_fn(raw0)
     best_raw, best_f = raw0, f0
-  elif n >= _NO_GRAD_FIT_N:
+  elif use_analytic:
     # hipBLAS's trsm BACKWARD fails with ALLOC_FAILED at N~10^4 on
     # ROCm 7.2 (even unbatched), so huge studies cannot autograd the
     # NLL. Fit with ANALYTIC gradients instead (forward-only solves,
