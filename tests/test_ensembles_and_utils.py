@@ -236,6 +236,14 @@ class TestUtils:
       pass
     assert 'scope_b' not in profiler.get_latencies_dict(events)
 
+  def test_profiler_scopes_appear_in_torch_traces(self):
+    import torch
+    from torch.profiler import ProfilerActivity, profile
+    with profile(activities=[ProfilerActivity.CPU]) as prof:
+      with profiler.timeit('vz_scope'):
+        torch.ones(4).sum()
+    assert any('vz_scope' in e.key for e in prof.key_averages())
+
   def test_json_numpy_roundtrip(self):
     obj = {'a': np.arange(6, dtype=np.int32).reshape(2, 3),
            'b': [np.float64(1.5)]}

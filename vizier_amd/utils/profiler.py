@@ -72,7 +72,20 @@ def collect_events() -> Iterator[List[ProfileEvent]]:
 @contextlib.contextmanager
 def timeit(name: str, also_log: bool = False,
            sync_gpu: bool = False) -> Iterator[None]:
-  """Times a scope; optionally synchronizes the GPU at both edges."""
+  """Times a scope; optionally synchronizes the GPU at both edges.
+
+  Scopes are also emitted as torch.profiler record_function ranges, so
+  they appear as named regions in torch.profiler / rocTracer traces
+  (`rocprofv3 --hip-trace` + torch profiler export) alongside the HIP
+  kernels they launch.
+  """
+  range_cm = None
+  try:
+    import torch
+    range_cm = torch.profiler.record_function(name)
+    range_cm.__enter__()
+  except ImportError:
+    pass
   if sync_gpu:
     try:
       import torch
@@ -91,6 +104,8 @@ def timeit(name: str, also_log: bool = False,
           torch.cuda.synchronize()
       except ImportError:
         pass
+    if range_cm is not None:
+      range_cm.__exit__(None, None, None)
     duration = datetime.timedelta(seconds=time.monotonic() - start)
     _storage.record(ProfileEvent(name, duration, time.time()))
     if also_log:
