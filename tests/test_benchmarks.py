@@ -202,3 +202,38 @@ class TestStatisticalGates:
     exptr.evaluate([t])
     assert t.final_measurement.metrics['value'].value == pytest.approx(
         exptr.optimal_value)
+
+
+class TestExplorationScores:
+
+  def _problem(self):
+    p = vz.ProblemStatement()
+    p.search_space.root.add_float_param('x', 0.0, 1.0)
+    p.search_space.root.add_categorical_param('c', ['a', 'b', 'c'])
+    p.metric_information.append(vz.MetricInformation(name='m', goal=1))
+    return p
+
+  def test_parameter_entropy_uniform_vs_constant(self):
+    import numpy as np
+    from vizier_amd._src.benchmarks.analyzers.state_analyzer import (
+        compute_average_marginal_parameter_entropy,
+        compute_parameter_entropy)
+    p = self._problem()
+    cfgs = {c.name: c for c in p.search_space.parameters}
+    rng = np.random.default_rng(0)
+    spread = [vz.Trial({'x': float(rng.random()),
+                        'c': rng.choice(['a', 'b', 'c'])}, id=k + 1)
+              for k in range(90)]
+    const = [vz.Trial({'x': 0.5, 'c': 'a'}, id=k + 1)
+             for k in range(90)]
+    # Exploration scores separate the two regimes on every parameter.
+    assert compute_parameter_entropy(spread, cfgs['c']) == \
+        pytest.approx(np.log(3), abs=0.15)
+    assert compute_parameter_entropy(const, cfgs['c']) == 0.0
+    assert compute_parameter_entropy(spread, cfgs['x']) > \
+        compute_parameter_entropy(const, cfgs['x'])
+    avg_spread = compute_average_marginal_parameter_entropy(
+        [(p, spread)])
+    avg_const = compute_average_marginal_parameter_entropy([(p, const)])
+    assert avg_spread > avg_const
+    assert compute_average_marginal_parameter_entropy([]) == 0.0

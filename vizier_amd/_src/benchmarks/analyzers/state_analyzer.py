@@ -12,7 +12,7 @@ from __future__ import annotations
 import collections
 import dataclasses
 import json
-from typing import Dict, List, Optional, Sequence
+from typing import Dict, List, Optional, Sequence, Tuple
 
 import numpy as np
 
@@ -97,23 +97,48 @@ class BenchmarkStateAnalyzer:
 
 def compute_parameter_entropy(trials: Sequence[vz.Trial],
                               config: vz.ParameterConfig,
-                              num_bins: int = 10) -> float:
-  """Exploration score: entropy of visited values for one parameter."""
+                              num_bins: Optional[int] = None) -> float:
+  """Exploration score: entropy of visited values for one parameter.
+
+  Parity with exploration_score_utils.py:28: categorical/discrete/
+  integer parameters count unique values; continuous parameters
+  histogram with the cube-root bin rule num_bins = 30/100^(1/3) *
+  n^(1/3) (capped at n) unless num_bins is given explicitly.
+  """
   values = [t.parameters.get_value(config.name, None) for t in trials]
   values = [v for v in values if v is not None]
   if not values:
     return 0.0
-  if config.type == vz.ParameterType.CATEGORICAL:
+  if config.type in (vz.ParameterType.CATEGORICAL,
+                     vz.ParameterType.DISCRETE,
+                     vz.ParameterType.INTEGER):
     counts = np.asarray(list(collections.Counter(values).values()),
                         dtype=np.float64)
   else:
     lo, hi = config.bounds
+    n = len(values)
+    if num_bins is None:
+      alpha = 1.0 / 3.0
+      num_bins = min(int(30.0 / (100 ** alpha) * n ** alpha), n)
+      num_bins = max(num_bins, 1)
     hist, _ = np.histogram([float(v) for v in values], bins=num_bins,
                            range=(lo, hi))
     counts = hist.astype(np.float64)
   p = counts / counts.sum()
   p = p[p > 0]
   return float(-(p * np.log(p)).sum())
+
+
+def compute_average_marginal_parameter_entropy(
+    studies: Sequence[Tuple[vz.ProblemStatement, Sequence[vz.Trial]]]
+) -> float:
+  """Mean per-parameter entropy across studies
+  (exploration_score_utils.py:97)."""
+  entropies = []
+  for problem, trials in studies:
+    for config in problem.search_space.parameters:
+      entropies.append(compute_parameter_entropy(trials, config))
+  return float(np.mean(entropies)) if entropies else 0.0
 
 
 class BenchmarkRecordAnalyzer:
