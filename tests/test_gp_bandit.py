@@ -194,3 +194,51 @@ class TestGPBandit:
     trials = run_loop(designer, 8)
     best = max(t.final_measurement.metrics['obj'].value for t in trials)
     assert best > -0.4, f'transfer designer failed to adapt: {best}'
+
+
+class TestEnsembleDesigners:
+
+  def _problem(self):
+    p = vz.ProblemStatement()
+    for i in range(3):
+      p.search_space.root.add_float_param(f'x{i}', -1.0, 1.0)
+    p.metric_information.append(vz.MetricInformation(
+        name='m', goal=vz.ObjectiveMetricGoal.MAXIMIZE))
+    return p
+
+  def _trials(self, n=14):
+    rng = np.random.default_rng(0)
+    trials = []
+    for uid in range(1, n + 1):
+      params = {f'x{i}': float(rng.uniform(-1, 1)) for i in range(3)}
+      t = vz.Trial(params, id=uid)
+      x = np.array(list(params.values()))
+      t.complete(vz.Measurement(metrics={'m': float(-(x ** 2).sum())}))
+      trials.append(t)
+    return trials
+
+  def test_gp_bandit_ensemble_end_to_end(self):
+    from vizier_amd._src.gp.gp_model import EnsembleGPPosterior
+    d = VizierGPBandit(self._problem(), GPBanditConfig(
+        max_evaluations=400, ard_restarts=3, ard_max_iters=8,
+        ensemble_size=3))
+    d.update(CompletedTrials(self._trials()), ActiveTrials())
+    assert len(d.suggest(1)) == 1
+    assert isinstance(d._posteriors[0], EnsembleGPPosterior)
+    # Warm refit keeps the ensemble and still suggests.
+    t = vz.Trial({f'x{i}': 0.1 for i in range(3)}, id=99)
+    t.complete(vz.Measurement(metrics={'m': 0.0}))
+    d.update(CompletedTrials([t]), ActiveTrials())
+    assert len(d.suggest(1)) == 1
+
+  def test_gp_ucb_pe_ensemble_both_phases(self):
+    from vizier_amd._src.algorithms.designers.gp_ucb_pe import (
+        UCBPEConfig, VizierGPUCBPEBandit)
+    from vizier_amd._src.gp.gp_model import EnsembleGPPosterior
+    d = VizierGPUCBPEBandit(self._problem(), UCBPEConfig(
+        max_evaluations=400, ard_restarts=3, ard_max_iters=8,
+        ensemble_size=3))
+    d.update(CompletedTrials(self._trials()), ActiveTrials())
+    # count=3 exercises UCB (first) and PE fill (rest) with the mixture.
+    assert len(d.suggest(3)) == 3
+    assert isinstance(d._posterior, EnsembleGPPosterior)

@@ -374,3 +374,50 @@ class TestAnalyticNLLGradient:
     assert post.nll < gp_model.negative_log_marginal_likelihood(
         gp_model._init_raw(1, 3, torch.Generator().manual_seed(0),
                            'cpu', torch.float32), x, y)[0] - 1.0
+
+
+class TestEnsemblePosterior:
+
+  def _fit(self, ensemble_size):
+    g = torch.Generator().manual_seed(0)
+    x = torch.rand(30, 2, generator=g)
+    y = torch.randn(30, generator=g)
+    return gp_model.train_gp(x, y, num_restarts=4, max_iters=10,
+                             ensemble_size=ensemble_size), x
+
+  def test_default_returns_single_posterior(self):
+    post, _ = self._fit(1)
+    assert isinstance(post, gp_model.GPPosterior)
+
+  def test_ensemble_members_sorted_by_nll(self):
+    post, _ = self._fit(3)
+    assert isinstance(post, gp_model.EnsembleGPPosterior)
+    nlls = [m.nll for m in post.members]
+    assert nlls == sorted(nlls)
+    # Duck-type surface used by the designers.
+    assert post.params is post.members[0].params
+    assert post.K_inv is None
+
+  def test_mixture_moments_match_hand_computation(self):
+    post, x = self._fit(3)
+    g = torch.Generator().manual_seed(1)
+    xq = torch.rand(7, 2, generator=g)
+    mean, stddev = post.predict(xq)
+    ms = torch.stack([m.predict(xq)[0] for m in post.members])
+    ss = torch.stack([m.predict(xq)[1] for m in post.members])
+    want_mean = ms.mean(0)
+    want_var = (ss.square() + ms.square()).mean(0) - want_mean.square()
+    assert torch.allclose(mean, want_mean, atol=1e-6)
+    assert torch.allclose(stddev, want_var.clamp_min(1e-12).sqrt(),
+                          atol=1e-6)
+    # Mixture stddev >= smallest member stddev (spread adds variance).
+    assert (stddev >= ss.min(0).values - 1e-6).all()
+
+  def test_scoring_function_uses_mixture(self):
+    from vizier_amd._src.gp import acquisitions as acq_lib
+    post, _ = self._fit(3)
+    g = torch.Generator().manual_seed(2)
+    xq = torch.rand(5, 2, generator=g)
+    scoring = acq_lib.ScoringFunction(post, acq_lib.UCB(1.8))
+    mean, stddev = post.predict(xq)
+    assert torch.allclose(scoring(xq), mean + 1.8 * stddev, atol=1e-6)

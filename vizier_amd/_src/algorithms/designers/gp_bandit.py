@@ -70,6 +70,7 @@ class GPBanditConfig:
   use_trust_region: bool = True
   num_scalarizations: int = 1000  # multi-objective
   scorer_gram_dtype: str = 'fp32'  # 'fp32'|'bf16'|'fp8' candidate grams
+  ensemble_size: int = 1   # best-N ARD restarts mixed (gp_models.py:201)
   data_parallel: bool = False     # shard the sweep across dist ranks
   device: Optional[str] = None
   dtype: torch.dtype = torch.float32
@@ -218,9 +219,11 @@ class VizierGPBandit(Designer, Predictor):
           else cfg.ard_restarts,
           max_iters=cfg.ard_warm_iters if warm is not None
           else cfg.ard_max_iters,
-          seed=self._seed, warm_start_raw=warm)
+          seed=self._seed, warm_start_raw=warm,
+          ensemble_size=cfg.ensemble_size)
       if cfg.data_parallel:
-        sharded_sweep.broadcast_posterior(post)
+        for member in getattr(post, 'members', [post]):
+          sharded_sweep.broadcast_posterior(member)
       self._posteriors.append(post)
     if transfer:
       self._stacked = transfer_learning.StackedResidualGP(
@@ -309,7 +312,31 @@ class VizierGPBandit(Designer, Predictor):
       score_fn.graph_safe = False  # multi-GP chain uses rocBLAS
       return score_fn, 1
 
+    if isinstance(posterior, gp_model.EnsembleGPPosterior) and not (
+        cfg.acquisition == 'qei' and count > 1):
+      if cfg.acquisition == 'ei':
+        acquisition = acq_lib.EI(best_value=best_value)
+      elif cfg.acquisition == 'pi':
+        acquisition = acq_lib.PI(best_value=best_value)
+      elif cfg.acquisition == 'thompson':
+        acquisition = acq_lib.Sample(seed=self._seed)
+      else:
+        acquisition = acq_lib.UCB(coefficient=cfg.ucb_coefficient)
+      ensemble = posterior
+
+      def score_fn(batch: CandidateBatch) -> torch.Tensor:
+        dense = self._codec.decode(batch)[:, 0, :]
+        mean, stddev = ensemble.predict(dense)
+        scores = acquisition(mean, stddev)
+        if trust_region is not None:
+          scores = trust_region.apply(dense, scores)
+        return scores
+      score_fn.graph_safe = False  # mixture loops members (rocBLAS)
+      return score_fn, 1
+
     if cfg.acquisition == 'qei' and count > 1:
+      if isinstance(posterior, gp_model.EnsembleGPPosterior):
+        posterior = posterior.members[0]
       qei = acq_lib.QEI(best_value=best_value, seed=self._seed)
 
       def score_fn(batch: CandidateBatch) -> torch.Tensor:

@@ -52,6 +52,7 @@ class UCBPEConfig:
   ucb_overwrite_probability: float = 0.25
   pe_overwrite_probability: float = 0.1
   pe_overwrite_probability_in_high_noise: float = 0.7
+  ensemble_size: int = 1   # best-N ARD restarts mixed (gp_ucb_pe.py:651)
   signal_to_noise_threshold: float = 0.7
   max_evaluations: int = 75000
   suggestion_batch_size: int = 25
@@ -170,7 +171,8 @@ class VizierGPUCBPEBandit(Designer):
         num_restarts=cfg.ard_warm_restarts if warm is not None
         else cfg.ard_restarts,
         max_iters=cfg.ard_warm_iters if warm is not None
-        else cfg.ard_max_iters, seed=self._seed, warm_start_raw=warm)
+        else cfg.ard_max_iters, seed=self._seed, warm_start_raw=warm,
+        ensemble_size=cfg.ensemble_size)
     self._warped_labels = y
     self._last_fit_count = len(self._completed)
 
@@ -231,7 +233,9 @@ class VizierGPUCBPEBandit(Designer):
 
       def score_fn(batch: CandidateBatch) -> torch.Tensor:
         return scoring(self._codec.decode(batch)[:, 0, :])
-      if scoring._tr_anchored and scoring._acq_name is not None:
+      single_gp = not isinstance(posterior, gp_model.EnsembleGPPosterior)
+      if single_gp and scoring._tr_anchored and \
+          scoring._acq_name is not None:
         # No pending/hallucinated points: the trust region anchors at
         # the GP train set, so the pure-HIP chunked scorer applies —
         # hipGraph-capturable and megakernel-eligible (the common

@@ -310,16 +310,67 @@ class GPPosterior:
     return mean, cov
 
 
+@dataclasses.dataclass
+class EnsembleGPPosterior:
+  """Equal-weight mixture of the best-N ARD restarts' posteriors.
+
+  Parity with UniformEnsemblePredictive
+  (vizier/_src/jax/stochastic_process_model.py:835) selected by
+  ensemble_size (gp_models.py:201): predictions are a uniform Gaussian
+  mixture over members — mean is the member-mean average, variance is
+  E[m^2 + s^2] - E[m]^2. Members are sorted by NLL, so .params/.raw/
+  .nll expose the best member (warm starts keep working).
+  """
+
+  members: list
+
+  @property
+  def x(self) -> torch.Tensor:
+    return self.members[0].x
+
+  @property
+  def params(self) -> GPParams:
+    return self.members[0].params
+
+  @property
+  def nll(self) -> float:
+    return self.members[0].nll
+
+  @property
+  def raw(self) -> Optional[torch.Tensor]:
+    return self.members[0].raw
+
+  @property
+  def noise_eff(self) -> float:
+    return self.members[0].noise_eff
+
+  @property
+  def K_inv(self):
+    # None: the fused K^-1 quadform scorers apply to a single GP;
+    # ScoringFunction then falls back to predict() (the mixture).
+    return None
+
+  def predict(self, xq: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
+    means, stds = zip(*(m.predict(xq) for m in self.members))
+    means = torch.stack(means)
+    stds = torch.stack(stds)
+    mix_mean = means.mean(0)
+    mix_var = (stds.square() + means.square()).mean(0) - mix_mean.square()
+    return mix_mean, mix_var.clamp_min(1e-12).sqrt()
+
+
 def train_gp(x: torch.Tensor, y: torch.Tensor, *,
              num_restarts: int = 4, max_iters: int = 50,
              seed: int = 0, precompute_inverse: bool = True,
-             warm_start_raw: Optional[torch.Tensor] = None
-             ) -> GPPosterior:
+             warm_start_raw: Optional[torch.Tensor] = None,
+             ensemble_size: int = 1):
   """Fits GP hyperparameters by restarting batched L-BFGS on the NLL.
 
   Mirrors gp_models.train_gp (vizier/_src/algorithms/designers/gp/
   gp_models.py:169-223): restart init -> ARD optimize -> best restart ->
-  posterior precompute.
+  posterior precompute. With ensemble_size > 1 (gp_models.py:201) the
+  best N finite restarts are kept and returned as an
+  EnsembleGPPosterior (uniform mixture), else a single GPPosterior.
   """
   x = x.detach()
   y = y.detach().reshape(-1)
@@ -357,8 +408,30 @@ def train_gp(x: torch.Tensor, y: torch.Tensor, *,
     best_raw, best_f = lbfgs.minimize_batched(loss_fn, raw0,
                                               max_iters=max_iters,
                                               check_every=5)
-  idx = int(torch.argmin(best_f))
-  raw = best_raw[idx]
+  def _build_posterior(raw: torch.Tensor, nll_one: float) -> GPPosterior:
+    return _build_posterior_cache(
+        x, y, raw, nll_one, precompute_inverse=precompute_inverse)
+
+  order = torch.argsort(best_f)
+  if ensemble_size <= 1:
+    i = int(order[0])
+    return _build_posterior(best_raw[i], float(best_f[i]))
+  members = []
+  for i in order[:ensemble_size].tolist():
+    if math.isfinite(float(best_f[i])):
+      members.append(_build_posterior(best_raw[i], float(best_f[i])))
+  if len(members) <= 1:
+    i = int(order[0])
+    return members[0] if members else _build_posterior(
+        best_raw[i], float(best_f[i]))
+  return EnsembleGPPosterior(members=members)
+
+
+def _build_posterior_cache(x: torch.Tensor, y: torch.Tensor,
+                           raw: torch.Tensor, nll_one: float, *,
+                           precompute_inverse: bool) -> GPPosterior:
+  """Builds the fp64 posterior cache for one raw-parameter vector."""
+  n = x.shape[0]
   params = GPParams.from_raw(raw)
 
   # Posterior caches in fp64, stored fp32: once the fit drives noise
@@ -392,5 +465,5 @@ def train_gp(x: torch.Tensor, y: torch.Tensor, *,
     K_inv = (z.T @ z).to(x.dtype)
   return GPPosterior(x=x, params=params, L=L64.to(x.dtype),
                      alpha=alpha.to(x.dtype), K_inv=K_inv,
-                     nll=float(best_f[idx]), raw=raw.detach(),
+                     nll=nll_one, raw=raw.detach(),
                      noise_eff=float(noise_eff))
