@@ -242,3 +242,12 @@ class TestEnsembleDesigners:
     # count=3 exercises UCB (first) and PE fill (rest) with the mixture.
     assert len(d.suggest(3)) == 3
     assert isinstance(d._posterior, EnsembleGPPosterior)
+
+  def test_custom_output_warper_and_ref_scaling(self):
+    from vizier_amd._src.gp import output_warpers as ow
+    d = VizierGPBandit(self._problem(), GPBanditConfig(
+        max_evaluations=200, ard_restarts=2, ard_max_iters=5,
+        output_warper_factory=ow.create_warp_outliers_warper,
+        ref_scaling=0.1))
+    d.update(CompletedTrials(self._trials(10)), ActiveTrials())
+    assert len(d.suggest(1)) == 1

@@ -17,7 +17,7 @@ from __future__ import annotations
 
 import dataclasses
 import time
-from typing import List, Optional, Sequence
+from typing import Callable, List, Optional, Sequence
 
 import numpy as np
 import torch
@@ -71,6 +71,10 @@ class GPBanditConfig:
   num_scalarizations: int = 1000  # multi-objective
   scorer_gram_dtype: str = 'fp32'  # 'fp32'|'bf16'|'fp8' candidate grams
   ensemble_size: int = 1   # best-N ARD restarts mixed (gp_models.py:201)
+  ref_scaling: float = 0.01  # MO reference-point margin (gp_bandit.py:156)
+  # Custom label warper factory (gp_bandit.py:150 _output_warper); None
+  # = the reference default HalfRank+Log+Infeasible pipeline.
+  output_warper_factory: Optional[Callable[[], object]] = None
   data_parallel: bool = False     # shard the sweep across dist ranks
   device: Optional[str] = None
   dtype: torch.dtype = torch.float32
@@ -185,9 +189,11 @@ class VizierGPBandit(Designer, Predictor):
 
   def _prepare_labels(self, raw_labels: np.ndarray) -> np.ndarray:
     """Per-metric default warping -> (N, M) finite labels."""
+    factory = (self._config.output_warper_factory or
+               output_warpers.create_default_warper)
     warped = np.zeros_like(raw_labels)
     for m in range(raw_labels.shape[1]):
-      warper = output_warpers.create_default_warper()
+      warper = factory()
       warped[:, m] = warper.warp(raw_labels[:, m:m + 1]).flatten()
     return warped
 
@@ -253,7 +259,8 @@ class VizierGPBandit(Designer, Predictor):
     if multi_objective:
       scalarizer = acq_lib.create_hv_scalarization(
           cfg.num_scalarizations, len(self._posteriors), seed=self._seed,
-          reference_point=acq_lib.get_reference_point(self._warped_labels))
+          reference_point=acq_lib.get_reference_point(
+              self._warped_labels, scale=cfg.ref_scaling))
 
       def _predict(post, flat):
         if cfg.scorer_gram_dtype == 'fp32':
