@@ -71,6 +71,10 @@ class GPBanditConfig:
   num_scalarizations: int = 1000  # multi-objective
   scorer_gram_dtype: str = 'fp32'  # 'fp32'|'bf16'|'fp8' candidate grams
   ensemble_size: int = 1   # best-N ARD restarts mixed (gp_models.py:201)
+  # Additive continuous-only linear kernel scale (gp_bandit.py:131
+  # _linear_coef; tuned_gp_models.py:204): None disables it. Posteriors
+  # with a linear term score via the composed predict path.
+  linear_coef: Optional[float] = None
   ref_scaling: float = 0.01  # MO reference-point margin (gp_bandit.py:156)
   # Custom label warper factory (gp_bandit.py:150 _output_warper); None
   # = the reference default HalfRank+Log+Infeasible pipeline.
@@ -219,14 +223,20 @@ class VizierGPBandit(Designer, Predictor):
       # Warm refits: the previous optimum is almost always the winner,
       # so keep only a couple of random restarts — the restart batch
       # multiplies the per-iteration Cholesky cost (R x N^3).
-      post = gp_model.train_gp(
-          x, y,
-          num_restarts=cfg.ard_warm_restarts if warm is not None
-          else cfg.ard_restarts,
-          max_iters=cfg.ard_warm_iters if warm is not None
-          else cfg.ard_max_iters,
-          seed=self._seed, warm_start_raw=warm,
-          ensemble_size=cfg.ensemble_size)
+      restarts = cfg.ard_warm_restarts if warm is not None \
+          else cfg.ard_restarts
+      iters = cfg.ard_warm_iters if warm is not None \
+          else cfg.ard_max_iters
+      if cfg.linear_coef is not None:
+        from vizier_amd._src.gp import linear_matern
+        post = linear_matern.train_linear_matern_gp(
+            x, y, linear_coef=cfg.linear_coef, num_restarts=restarts,
+            max_iters=iters, seed=self._seed, warm_start_raw=warm)
+      else:
+        post = gp_model.train_gp(
+            x, y, num_restarts=restarts, max_iters=iters,
+            seed=self._seed, warm_start_raw=warm,
+            ensemble_size=cfg.ensemble_size)
       if cfg.data_parallel:
         for member in getattr(post, 'members', [post]):
           sharded_sweep.broadcast_posterior(member)
@@ -319,8 +329,8 @@ class VizierGPBandit(Designer, Predictor):
       score_fn.graph_safe = False  # multi-GP chain uses rocBLAS
       return score_fn, 1
 
-    if isinstance(posterior, gp_model.EnsembleGPPosterior) and not (
-        cfg.acquisition == 'qei' and count > 1):
+    plain_gp = isinstance(posterior, gp_model.GPPosterior)
+    if not plain_gp and not (cfg.acquisition == 'qei' and count > 1):
       if cfg.acquisition == 'ei':
         acquisition = acq_lib.EI(best_value=best_value)
       elif cfg.acquisition == 'pi':
@@ -329,16 +339,16 @@ class VizierGPBandit(Designer, Predictor):
         acquisition = acq_lib.Sample(seed=self._seed)
       else:
         acquisition = acq_lib.UCB(coefficient=cfg.ucb_coefficient)
-      ensemble = posterior
+      composed = posterior
 
       def score_fn(batch: CandidateBatch) -> torch.Tensor:
         dense = self._codec.decode(batch)[:, 0, :]
-        mean, stddev = ensemble.predict(dense)
+        mean, stddev = composed.predict(dense)
         scores = acquisition(mean, stddev)
         if trust_region is not None:
           scores = trust_region.apply(dense, scores)
         return scores
-      score_fn.graph_safe = False  # mixture loops members (rocBLAS)
+      score_fn.graph_safe = False  # composed predict uses rocBLAS
       return score_fn, 1
 
     if cfg.acquisition == 'qei' and count > 1:
@@ -493,6 +503,19 @@ def posterior_batched_cov(posterior: gp_model.GPPosterior,
   from vizier_amd._src.ops import dispatch as ops
   B, q, D = dense.shape
   flat = dense.reshape(B * q, D)
+  from vizier_amd._src.gp import linear_matern
+  if isinstance(posterior, linear_matern.LinearMaternPosterior):
+    # Combined Matérn+linear kernel: compose k and k_qq explicitly.
+    k = linear_matern._combined_gram(
+        posterior.params, posterior.linear_coef, flat,
+        posterior.x).reshape(B, q, -1)
+    mean = posterior.params.mean + k @ posterior.alpha
+    kqq = linear_matern._combined_gram(
+        posterior.params, posterior.linear_coef, dense, None)
+    v = torch.linalg.solve_triangular(
+        posterior.L, k.reshape(B * q, -1).T, upper=False)
+    v = v.T.reshape(B, q, -1)
+    return mean, kqq - torch.einsum('bqn,brn->bqr', v, v)
   k = ops.gram_matern52(flat, posterior.x, posterior.params.lengthscales,
                         posterior.params.amplitude).reshape(B, q, -1)
   mean = posterior.params.mean + k @ posterior.alpha

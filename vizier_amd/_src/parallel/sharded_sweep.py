@@ -52,12 +52,18 @@ def broadcast_posterior(posterior, src: int = 0) -> None:
       posterior.params.noise.reshape(()),
       posterior.params.mean.reshape(()),
   ])
+  aux = getattr(posterior, 'aux', None)  # e.g. linear slope/shift
+  if aux is not None:
+    tensors.append(aux)
   for t in tensors:
     dist.broadcast(t, src=src)
   dist.broadcast(scalars, src=src)
   posterior.params.amplitude = scalars[0]
   posterior.params.noise = scalars[1]
   posterior.params.mean = scalars[2]
+  if aux is not None and hasattr(posterior.params, 'slope'):
+    posterior.params.slope = aux[0]
+    posterior.params.shift = aux[1]
 
 
 def allgather_topk(features: torch.Tensor, rewards: torch.Tensor,
