@@ -132,3 +132,40 @@ class TestTraining:
     err_solo = float((mean_solo - 0.8 * f).abs().mean())
     # The joint model must be at least comparable (no degradation).
     assert err_joint < err_solo * 1.5, (err_joint, err_solo)
+
+
+class TestMultitaskDesignerIntegration:
+
+  def test_gp_bandit_multitask_type(self):
+    import numpy as np
+    from vizier_amd import pyvizier as vz
+    from vizier_amd._src.algorithms.core.abstractions import (
+        ActiveTrials, CompletedTrials)
+    from vizier_amd._src.algorithms.designers.gp_bandit import (
+        GPBanditConfig, VizierGPBandit)
+    p = vz.ProblemStatement()
+    for i in range(2):
+      p.search_space.root.add_float_param(f'x{i}', -1.0, 1.0)
+    for name in ('m1', 'm2'):
+      p.metric_information.append(vz.MetricInformation(
+          name=name, goal=vz.ObjectiveMetricGoal.MAXIMIZE))
+    rng = np.random.default_rng(0)
+    trials = []
+    for uid in range(1, 14):
+      params = {f'x{i}': float(rng.uniform(-1, 1)) for i in range(2)}
+      t = vz.Trial(params, id=uid)
+      x = np.array(list(params.values()))
+      t.complete(vz.Measurement(metrics={'m1': float(x.sum()),
+                                         'm2': float(-(x**2).sum())}))
+      trials.append(t)
+    for kind, expect_joint in (('separable', True),
+                               ('separable_diag', True),
+                               ('independent', False)):
+      d = VizierGPBandit(p, GPBanditConfig(
+          max_evaluations=300, ard_restarts=2, ard_max_iters=6,
+          num_scalarizations=50, multitask_type=kind))
+      d.update(CompletedTrials(trials), ActiveTrials())
+      assert len(d.suggest(1)) == 1
+      assert (d._mt_posterior is not None) == expect_joint
+      pred = d.predict([vz.TrialSuggestion({'x0': 0.0, 'x1': 0.0})])
+      assert pred.mean.shape[-1] == 2 or len(pred.mean.shape) == 2
