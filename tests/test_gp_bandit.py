@@ -283,3 +283,21 @@ class TestEnsembleDesigners:
         acquisition='qei', linear_coef=1.0))
     d2.update(CompletedTrials(self._trials()), ActiveTrials())
     assert len(d2.suggest(4)) == 4
+
+  def test_custom_scoring_function_factory(self):
+    import torch
+    calls = []
+
+    def factory(posterior, best_value, trust_region):
+      def score(xs: torch.Tensor) -> torch.Tensor:
+        calls.append(xs.shape)
+        mean, stddev = posterior.predict(xs)
+        return mean + 0.5 * stddev  # custom low-explore UCB
+      return score
+
+    d = VizierGPBandit(self._problem(), GPBanditConfig(
+        max_evaluations=200, ard_restarts=2, ard_max_iters=5,
+        scoring_function_factory=factory))
+    d.update(CompletedTrials(self._trials(10)), ActiveTrials())
+    assert len(d.suggest(1)) == 1
+    assert calls, 'custom scorer was never invoked'

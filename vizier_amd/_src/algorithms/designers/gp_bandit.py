@@ -79,6 +79,11 @@ class GPBanditConfig:
   # or 'separable'/'separable_diag' joint task kernels
   # (gp_bandit.py:157 _multitask_type; multitask_tuned_gp_models.py:41).
   multitask_type: str = 'independent'
+  # Custom acquisition hook (gp_bandit.py:132 _scoring_function_factory):
+  # called as factory(posterior, best_value, trust_region) -> callable
+  # mapping xs (B, D) -> scores (B,). Overrides `acquisition`; runs on
+  # the composed path (not fused/megakernel).
+  scoring_function_factory: Optional[Callable] = None
   ref_scaling: float = 0.01  # MO reference-point margin (gp_bandit.py:156)
   # Custom label warper factory (gp_bandit.py:150 _output_warper); None
   # = the reference default HalfRank+Log+Infeasible pipeline.
@@ -344,6 +349,16 @@ class VizierGPBandit(Designer, Predictor):
 
     posterior = self._posteriors[0]
     best_value = float(self._warped_labels[:, 0].max())
+
+    if cfg.scoring_function_factory is not None:
+      custom = cfg.scoring_function_factory(posterior, best_value,
+                                            trust_region)
+
+      def score_fn(batch: CandidateBatch) -> torch.Tensor:
+        dense = self._codec.decode(batch)[:, 0, :]
+        return custom(dense)
+      score_fn.graph_safe = False  # arbitrary user torch code
+      return score_fn, 1
 
     if self._stacked is not None and cfg.acquisition in (
         'ucb', 'ei', 'pi'):
