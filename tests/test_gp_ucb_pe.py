@@ -96,3 +96,41 @@ class TestGPUCBPE:
     first = designer.suggest(2)
     assert len(first) == 2
     assert first[0].parameters.get_value('x0') == pytest.approx(0.5)
+
+
+class TestUCBPELinearKernel:
+
+  def test_mixes_linear_kernel_both_phases(self):
+    import numpy as np
+    from vizier_amd import pyvizier as vz
+    from vizier_amd._src.algorithms.core.abstractions import (
+        ActiveTrials, CompletedTrials)
+    from vizier_amd._src.algorithms.designers.gp_ucb_pe import (
+        UCBPEConfig, VizierGPUCBPEBandit)
+    from vizier_amd._src.gp.linear_matern import LinearMaternPosterior
+    p = vz.ProblemStatement()
+    for i in range(2):
+      p.search_space.root.add_float_param(f'x{i}', -1.0, 1.0)
+    p.metric_information.append(vz.MetricInformation(
+        name='m', goal=vz.ObjectiveMetricGoal.MAXIMIZE))
+    rng = np.random.default_rng(0)
+    trials = []
+    for uid in range(1, 14):
+      params = {f'x{i}': float(rng.uniform(-1, 1)) for i in range(2)}
+      t = vz.Trial(params, id=uid)
+      x = np.array(list(params.values()))
+      t.complete(vz.Measurement(metrics={'m': float(x.sum())}))
+      trials.append(t)
+    d = VizierGPUCBPEBandit(p, UCBPEConfig(
+        max_evaluations=300, ard_restarts=2, ard_max_iters=6,
+        mixes_linear_kernel=True))
+    d.update(CompletedTrials(trials), ActiveTrials())
+    # count=3: UCB phase (combined-kernel scoring) + PE fill (combined-
+    # kernel hallucinated variance posterior).
+    assert len(d.suggest(3)) == 3
+    assert isinstance(d._posterior, LinearMaternPosterior)
+    # Warm refit through the linear raw layout.
+    t2 = vz.Trial({'x0': 0.2, 'x1': 0.2}, id=99)
+    t2.complete(vz.Measurement(metrics={'m': 0.4}))
+    d.update(CompletedTrials([t2]), ActiveTrials())
+    assert len(d.suggest(1)) == 1

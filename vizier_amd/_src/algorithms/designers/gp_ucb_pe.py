@@ -53,6 +53,9 @@ class UCBPEConfig:
   pe_overwrite_probability: float = 0.1
   pe_overwrite_probability_in_high_noise: float = 0.7
   ensemble_size: int = 1   # best-N ARD restarts mixed (gp_ucb_pe.py:651)
+  # Matérn + continuous-only linear kernel (gp_ucb_pe.py:676
+  # _mixes_linear_kernel -> linear_coef=1.0). Composed scoring path.
+  mixes_linear_kernel: bool = False
   signal_to_noise_threshold: float = 0.7
   max_evaluations: int = 75000
   suggestion_batch_size: int = 25
@@ -166,13 +169,20 @@ class VizierGPUCBPEBandit(Designer):
     x = torch.as_tensor(x_np, dtype=cfg.dtype, device=self._device)
     y = torch.as_tensor(y_np, dtype=cfg.dtype, device=self._device)
     warm = self._posterior.raw if self._posterior is not None else None
-    self._posterior = gp_model.train_gp(
-        x, y,
-        num_restarts=cfg.ard_warm_restarts if warm is not None
-        else cfg.ard_restarts,
-        max_iters=cfg.ard_warm_iters if warm is not None
-        else cfg.ard_max_iters, seed=self._seed, warm_start_raw=warm,
-        ensemble_size=cfg.ensemble_size)
+    restarts = cfg.ard_warm_restarts if warm is not None \
+        else cfg.ard_restarts
+    iters = cfg.ard_warm_iters if warm is not None \
+        else cfg.ard_max_iters
+    if cfg.mixes_linear_kernel:
+      from vizier_amd._src.gp import linear_matern
+      self._posterior = linear_matern.train_linear_matern_gp(
+          x, y, linear_coef=1.0, num_restarts=restarts,
+          max_iters=iters, seed=self._seed, warm_start_raw=warm)
+    else:
+      self._posterior = gp_model.train_gp(
+          x, y, num_restarts=restarts, max_iters=iters,
+          seed=self._seed, warm_start_raw=warm,
+          ensemble_size=cfg.ensemble_size)
     self._warped_labels = y
     self._last_fit_count = len(self._completed)
 
@@ -203,6 +213,17 @@ class VizierGPUCBPEBandit(Designer):
     params = self._posterior.params
     from vizier_amd._src.gp.matern import gram_matern52
     n = x_all.shape[0]
+    from vizier_amd._src.gp import linear_matern
+    if isinstance(self._posterior, linear_matern.LinearMaternPosterior):
+      lc = self._posterior.linear_coef
+      K = linear_matern._combined_gram(params, lc, x_all, None)
+      K = K + params.noise * torch.eye(n, dtype=x_all.dtype,
+                                       device=x_all.device)
+      L = gp_model.cholesky_with_jitter(K, params.amplitude ** 2)
+      return linear_matern.LinearMaternPosterior(
+          x=x_all, params=params, linear_coef=lc, L=L,
+          alpha=torch.zeros(n, dtype=x_all.dtype, device=x_all.device),
+          nll=0.0)
     K = gram_matern52(x_all, None, params.lengthscales, params.amplitude)
     K = K + params.noise * torch.eye(n, dtype=x_all.dtype,
                                      device=x_all.device)
@@ -233,7 +254,7 @@ class VizierGPUCBPEBandit(Designer):
 
       def score_fn(batch: CandidateBatch) -> torch.Tensor:
         return scoring(self._codec.decode(batch)[:, 0, :])
-      single_gp = not isinstance(posterior, gp_model.EnsembleGPPosterior)
+      single_gp = isinstance(posterior, gp_model.GPPosterior)
       if single_gp and scoring._tr_anchored and \
           scoring._acq_name is not None:
         # No pending/hallucinated points: the trust region anchors at
