@@ -134,3 +134,51 @@ class TestUCBPELinearKernel:
     t2.complete(vz.Measurement(metrics={'m': 0.4}))
     d.update(CompletedTrials([t2]), ActiveTrials())
     assert len(d.suggest(1)) == 1
+
+
+class TestUCBPEMultimetric:
+
+  def _problem(self):
+    import numpy as np
+    from vizier_amd import pyvizier as vz
+    p = vz.ProblemStatement()
+    for i in range(2):
+      p.search_space.root.add_float_param(f'x{i}', -1.0, 1.0)
+    for name in ('m1', 'm2'):
+      p.metric_information.append(vz.MetricInformation(
+          name=name, goal=vz.ObjectiveMetricGoal.MAXIMIZE))
+    return p
+
+  def test_default_algorithm_serves_mo_studies(self):
+    import numpy as np
+    from vizier_amd import pyvizier as vz
+    from vizier_amd._src.algorithms.core.abstractions import (
+        ActiveTrials, CompletedTrials)
+    from vizier_amd._src.algorithms.designers.gp_ucb_pe import (
+        UCBPEConfig, VizierGPUCBPEBandit)
+    p = self._problem()
+    rng = np.random.default_rng(0)
+    trials = []
+    for uid in range(1, 14):
+      params = {f'x{i}': float(rng.uniform(-1, 1)) for i in range(2)}
+      t = vz.Trial(params, id=uid)
+      x = np.array(list(params.values()))
+      t.complete(vz.Measurement(metrics={'m1': float(x.sum()),
+                                         'm2': float(-(x**2).sum())}))
+      trials.append(t)
+    d = VizierGPUCBPEBandit(p, UCBPEConfig(
+        max_evaluations=300, ard_restarts=2, ard_max_iters=6,
+        num_scalarizations=50))
+    d.update(CompletedTrials(trials), ActiveTrials())
+    s = d.suggest(3)
+    assert len(s) == 3
+    assert d._mo_posteriors is not None and len(d._mo_posteriors) == 2
+    # Both phases appear across a batch (first=UCB, fills=PE by
+    # construction when there are fresh trials and no actives).
+    kinds = {x.metadata.ns('gp_ucb_pe')['acquisition'] for x in s}
+    assert 'pe' in kinds and 'ucb' in kinds
+    # Warm refit across the per-metric fits.
+    t2 = vz.Trial({'x0': 0.2, 'x1': 0.1}, id=99)
+    t2.complete(vz.Measurement(metrics={'m1': 0.3, 'm2': -0.05}))
+    d.update(CompletedTrials([t2]), ActiveTrials())
+    assert len(d.suggest(1)) == 1

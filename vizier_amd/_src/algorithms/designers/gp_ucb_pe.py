@@ -56,6 +56,7 @@ class UCBPEConfig:
   # Matérn + continuous-only linear kernel (gp_ucb_pe.py:676
   # _mixes_linear_kernel -> linear_coef=1.0). Composed scoring path.
   mixes_linear_kernel: bool = False
+  num_scalarizations: int = 1000  # multimetric UCB scalarization dirs
   signal_to_noise_threshold: float = 0.7
   max_evaluations: int = 75000
   suggestion_batch_size: int = 25
@@ -84,6 +85,8 @@ class VizierGPUCBPEBandit(Designer):
     self._codec = EagleFeatureCodec(self._converter)
     self._device = self._config.device or gp_bandit.default_device()
     self._posterior: Optional[gp_model.GPPosterior] = None
+    self._mo_posteriors = None    # per-metric GPs (multimetric studies)
+    self._mo_scalarizer = None
     self._last_fit_count = -1
     self._last_suggest_completed = 0
     from vizier_amd._src.algorithms.designers.quasi_random import (
@@ -164,9 +167,44 @@ class VizierGPUCBPEBandit(Designer):
       return
     cfg = self._config
     x_np, y_np = self._converter.to_xy(self._completed)
+    x = torch.as_tensor(x_np, dtype=cfg.dtype, device=self._device)
+
+    if y_np.shape[1] > 1:
+      # Multimetric (the DEFAULT algorithm must serve MO studies,
+      # gp_ucb_pe.py:66,300): per-metric GPs + hypervolume-scalarized
+      # UCB; the PE phase explores summed posterior stddev.
+      warped = np.zeros_like(y_np)
+      for m in range(y_np.shape[1]):
+        warped[:, m] = output_warpers.create_default_warper().warp(
+            y_np[:, m:m + 1]).flatten()
+      prev = self._mo_posteriors or []
+      posts = []
+      for m in range(y_np.shape[1]):
+        y_m = torch.as_tensor(warped[:, m], dtype=cfg.dtype,
+                              device=self._device)
+        warm = prev[m].raw if m < len(prev) else None
+        posts.append(gp_model.train_gp(
+            x, y_m,
+            num_restarts=cfg.ard_warm_restarts if warm is not None
+            else cfg.ard_restarts,
+            max_iters=cfg.ard_warm_iters if warm is not None
+            else cfg.ard_max_iters,
+            seed=self._seed + m, warm_start_raw=warm,
+            ensemble_size=cfg.ensemble_size))
+      self._mo_posteriors = posts
+      self._posterior = posts[0]
+      labels = torch.as_tensor(warped, dtype=cfg.dtype,
+                               device=self._device)
+      self._warped_labels = labels
+      self._mo_scalarizer = acq_lib.create_hv_scalarization(
+          cfg.num_scalarizations, y_np.shape[1], seed=self._seed,
+          reference_point=acq_lib.get_reference_point(labels))
+      self._last_fit_count = len(self._completed)
+      return
+    self._mo_posteriors = None
+
     warper = output_warpers.create_default_warper()
     y_np = warper.warp(y_np[:, :1]).flatten()
-    x = torch.as_tensor(x_np, dtype=cfg.dtype, device=self._device)
     y = torch.as_tensor(y_np, dtype=cfg.dtype, device=self._device)
     warm = self._posterior.raw if self._posterior is not None else None
     restarts = cfg.ard_warm_restarts if warm is not None \
@@ -204,18 +242,19 @@ class VizierGPUCBPEBandit(Designer):
       return False
     return self._rng.random() < cfg.ucb_overwrite_probability
 
-  def _variance_posterior(self, x_all: torch.Tensor
+  def _variance_posterior(self, x_all: torch.Tensor, posterior=None
                           ) -> gp_model.GPPosterior:
     """GP conditioned on all (completed+hallucinated) features.
 
     Only the predictive variance is used, so alpha is zeros.
     """
-    params = self._posterior.params
+    posterior = posterior if posterior is not None else self._posterior
+    params = posterior.params
     from vizier_amd._src.gp.matern import gram_matern52
     n = x_all.shape[0]
     from vizier_amd._src.gp import linear_matern
-    if isinstance(self._posterior, linear_matern.LinearMaternPosterior):
-      lc = self._posterior.linear_coef
+    if isinstance(posterior, linear_matern.LinearMaternPosterior):
+      lc = posterior.linear_coef
       K = linear_matern._combined_gram(params, lc, x_all, None)
       K = K + params.noise * torch.eye(n, dtype=x_all.dtype,
                                        device=x_all.device)
@@ -248,7 +287,46 @@ class VizierGPUCBPEBandit(Designer):
     trust_region = acq_lib.TrustRegion(x_all, onehot) \
         if cfg.use_trust_region else None
 
-    if use_ucb:
+    if self._mo_posteriors is not None:
+      posts = self._mo_posteriors
+      scalarizer = self._mo_scalarizer
+
+      def _scalarized(xs: torch.Tensor, coef: float) -> torch.Tensor:
+        per = [p_.predict(xs) for p_ in posts]
+        ys = torch.stack([m_ + coef * s_ for m_, s_ in per], dim=-1)
+        return scalarizer(ys).mean(dim=0)
+
+      if use_ucb:
+        def score_fn(batch: CandidateBatch) -> torch.Tensor:
+          xs = self._codec.decode(batch)[:, 0, :]
+          scores = _scalarized(xs, cfg.ucb_coefficient)
+          if trust_region is not None:
+            scores = trust_region.apply(xs, scores)
+          return scores
+      else:
+        # Promising region: scalarized mean at the observed point with
+        # the best scalarized UCB (multimetric analogue of
+        # gp_ucb_pe.py:384 PEScoreFunction).
+        with torch.no_grad():
+          obs_ucb = _scalarized(posterior.x, cfg.ucb_coefficient)
+          obs_mean = _scalarized(posterior.x, 0.0)
+          threshold = obs_mean[int(torch.argmax(obs_ucb))]
+        var_posts = [self._variance_posterior(x_all, p_) for p_ in posts]
+
+        def score_fn(batch: CandidateBatch) -> torch.Tensor:
+          xs = self._codec.decode(batch)[:, 0, :]
+          explore = _scalarized(
+              xs, cfg.explore_region_ucb_coefficient)
+          stddev_sum = sum(vp.predict(xs)[1] for vp in var_posts)
+          penalty = cfg.cb_violation_penalty_coefficient * \
+              torch.minimum(explore - threshold,
+                            torch.zeros_like(explore))
+          scores = stddev_sum + penalty
+          if trust_region is not None:
+            scores = trust_region.apply(xs, scores)
+          return scores
+      score_fn.graph_safe = False  # multi-GP loops use rocBLAS
+    elif use_ucb:
       scoring = acq_lib.ScoringFunction(
           posterior, acq_lib.UCB(cfg.ucb_coefficient), trust_region)
 
