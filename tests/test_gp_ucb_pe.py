@@ -182,3 +182,29 @@ class TestUCBPEMultimetric:
     t2.complete(vz.Measurement(metrics={'m1': 0.3, 'm2': -0.05}))
     d.update(CompletedTrials([t2]), ActiveTrials())
     assert len(d.suggest(1)) == 1
+
+  def test_mo_prior_seeding_with_many_trials(self):
+    # Regression: Eagle prior seeding takes SCALAR rewards; with
+    # multi-metric labels and enough trials to overflow the firefly
+    # pool the dedup comparison used to see (M,)-shaped rewards.
+    import numpy as np
+    from vizier_amd import pyvizier as vz
+    from vizier_amd._src.algorithms.core.abstractions import (
+        ActiveTrials, CompletedTrials)
+    from vizier_amd._src.algorithms.designers.gp_ucb_pe import (
+        UCBPEConfig, VizierGPUCBPEBandit)
+    p = self._problem()
+    rng = np.random.default_rng(1)
+    trials = []
+    for uid in range(1, 41):
+      params = {f'x{i}': float(rng.uniform(-1, 1)) for i in range(2)}
+      t = vz.Trial(params, id=uid)
+      x = np.array(list(params.values()))
+      t.complete(vz.Measurement(metrics={'m1': float(x.sum()),
+                                         'm2': float(-(x**2).sum())}))
+      trials.append(t)
+    d = VizierGPUCBPEBandit(p, UCBPEConfig(
+        max_evaluations=200, ard_restarts=2, ard_max_iters=5,
+        num_scalarizations=30))
+    d.update(CompletedTrials(trials), ActiveTrials())
+    assert len(d.suggest(2)) == 2

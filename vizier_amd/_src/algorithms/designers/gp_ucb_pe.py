@@ -376,7 +376,14 @@ class VizierGPUCBPEBandit(Designer):
         seed=self._seed + len(self._completed) + x_all.shape[0],
         device=self._device, dtype=cfg.dtype)
 
-    rewards = self._warped_labels.cpu().numpy()
+    if self._mo_posteriors is not None:
+      # Eagle prior seeding needs scalar rewards: use the scalarized
+      # warped labels (same scalarizer as the UCB phase).
+      with torch.no_grad():
+        rewards = self._mo_scalarizer(
+            self._warped_labels).mean(dim=0).cpu().numpy()
+    else:
+      rewards = self._warped_labels.cpu().numpy()
     prior_features, prior_rewards = trials_to_sorted_features(
         self._converter, self._codec, self._completed, rewards,
         device=self._device, dtype=cfg.dtype)
