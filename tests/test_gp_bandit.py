@@ -301,3 +301,16 @@ class TestEnsembleDesigners:
     d.update(CompletedTrials(self._trials(10)), ActiveTrials())
     assert len(d.suggest(1)) == 1
     assert calls, 'custom scorer was never invoked'
+
+  def test_default_config_keeps_fused_scorer_attach(self):
+    # The GPU sweep (hipGraph + megakernel) keys off score_fn.scoring /
+    # codec_identity; optional surrogates must not cost the default
+    # path its fast-path eligibility.
+    d = VizierGPBandit(self._problem(), GPBanditConfig(
+        max_evaluations=200, ard_restarts=2, ard_max_iters=5))
+    d.update(CompletedTrials(self._trials(10)), ActiveTrials())
+    d._fit()
+    score_fn, _ = d._score_factory(1)
+    assert hasattr(score_fn, 'scoring')
+    assert hasattr(score_fn, 'codec_identity')
+    assert not getattr(score_fn, 'graph_safe', True) is False
