@@ -208,3 +208,30 @@ class TestUCBPEMultimetric:
         num_scalarizations=30))
     d.update(CompletedTrials(trials), ActiveTrials())
     assert len(d.suggest(2)) == 2
+
+  def test_mo_multitask_surrogate(self):
+    import numpy as np
+    from vizier_amd import pyvizier as vz
+    from vizier_amd._src.algorithms.core.abstractions import (
+        ActiveTrials, CompletedTrials)
+    from vizier_amd._src.algorithms.designers.gp_ucb_pe import (
+        UCBPEConfig, VizierGPUCBPEBandit)
+    p = self._problem()
+    rng = np.random.default_rng(2)
+    trials = []
+    for uid in range(1, 14):
+      params = {f'x{i}': float(rng.uniform(-1, 1)) for i in range(2)}
+      t = vz.Trial(params, id=uid)
+      x = np.array(list(params.values()))
+      t.complete(vz.Measurement(metrics={'m1': float(x.sum()),
+                                         'm2': float(-(x**2).sum())}))
+      trials.append(t)
+    d = VizierGPUCBPEBandit(p, UCBPEConfig(
+        max_evaluations=200, ard_restarts=2, ard_max_iters=5,
+        num_scalarizations=30, multitask_type='separable'))
+    d.update(CompletedTrials(trials), ActiveTrials())
+    assert len(d.suggest(3)) == 3  # UCB + PE through the joint GP
+    t2 = vz.Trial({'x0': 0.1, 'x1': 0.1}, id=99)
+    t2.complete(vz.Measurement(metrics={'m1': 0.2, 'm2': -0.02}))
+    d.update(CompletedTrials([t2]), ActiveTrials())
+    assert len(d.suggest(1)) == 1  # joint warm refit

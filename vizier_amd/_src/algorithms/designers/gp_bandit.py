@@ -565,6 +565,16 @@ class _MultitaskTaskView:
   def raw(self):
     return None  # warm starts go through the joint posterior's raw
 
+  @property
+  def params(self):
+    """Marginal single-task view of the joint hyperparameters."""
+    joint = self._mt.params
+    amp = joint.task_cov[self.task, self.task].clamp_min(1e-12).sqrt()
+    return gp_model.GPParams(
+        amplitude=amp, noise=joint.noise,
+        lengthscales=joint.lengthscales,
+        mean=joint.means[self.task])
+
   def predict(self, xq: torch.Tensor):
     mean, stddev = self._mt.predict(xq)
     return mean[:, self.task], stddev[:, self.task]
