@@ -12,7 +12,9 @@ import random
 import pytest
 
 from vizier_amd._src.service import custom_errors
-from vizier_amd._src.service.proto import study_pb2, vizier_oss_pb2
+from vizier_amd._src.service.proto import (study_pb2,
+                                           vizier_oss_pb2,
+                                           vizier_service_pb2)
 from vizier_amd._src.service.ram_datastore import NestedDictRAMDataStore
 
 
@@ -104,6 +106,14 @@ def apply(store, op):
       kv = study_pb2.KeyValue(key=op[2], ns=':fuzz', value=op[3])
       store.update_metadata(op[1], [kv], [])
       return store.load_study(op[1]).SerializeToString(), None
+    if kind == 'update_trial_metadata':
+      u = vizier_service_pb2.UnitMetadataUpdate(trial_id=str(op[2]))
+      u.metadatum.key = op[3]
+      u.metadatum.ns = ':fuzz'
+      u.metadatum.value = op[4]
+      store.update_metadata(op[1], [], [u])
+      return store.get_trial(
+          f'{op[1]}/trials/{op[2]}').SerializeToString(), None
     raise AssertionError(op)
   except custom_errors.NotFoundError:
     return None, 'NotFoundError'
@@ -111,6 +121,10 @@ def apply(store, op):
     return None, 'AlreadyExistsError'
   except custom_errors.ImmutableStudyError:
     return None, 'ImmutableStudyError'
+  except KeyError:
+    # Trial-metadata update on a missing trial raises KeyError in the
+    # reference's ram_datastore; both stores must match it.
+    return None, 'KeyError'
 
 
 def random_op(rng, owners, studies, trial_ids):
@@ -119,7 +133,7 @@ def random_op(rng, owners, studies, trial_ids):
       'get_trial', 'list_trials', 'max_trial_id', 'delete_trial',
       'delete_study', 'create_sugg_op', 'get_sugg_op',
       'max_sugg_number', 'create_es_op', 'get_es_op',
-      'list_sugg_ops', 'update_metadata'])
+      'list_sugg_ops', 'update_metadata', 'update_trial_metadata'])
   owner = rng.choice(owners)
   study = rng.choice(studies)
   study_name = f'owners/{owner}/studies/{study}'
@@ -162,6 +176,9 @@ def random_op(rng, owners, studies, trial_ids):
   if choice == 'update_metadata':
     return ('update_metadata', study_name, f'k{rng.randint(0, 2)}',
             f'v{rng.randint(0, 9)}')
+  if choice == 'update_trial_metadata':
+    return ('update_trial_metadata', study_name, tid,
+            f'k{rng.randint(0, 2)}', f'v{rng.randint(0, 9)}')
   return ('max_trial_id', study_name)
 
 
