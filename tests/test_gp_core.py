@@ -146,6 +146,52 @@ class TestAcquisitions:
     assert float(out[0]) == 1.0
     assert float(out[1]) < -1e4 + 1
 
+  def test_trust_region_dof_counts_categorical_params_once(self):
+    # Regression for ADVICE r1 (medium): one 20-value categorical must
+    # contribute ONE dof (reference acquisitions.py:752-768), not 20.
+    from vizier_amd import pyvizier as vz
+    from vizier_amd.converters.core import TrialToArrayConverter
+    problem = vz.ProblemStatement()
+    problem.search_space.root.add_float_param('a', 0.0, 1.0)
+    problem.search_space.root.add_float_param('b', 0.0, 1.0)
+    problem.search_space.root.add_categorical_param(
+        'c', [f'v{i}' for i in range(20)])
+    problem.metric_information.append(vz.MetricInformation(
+        name='obj', goal=vz.ObjectiveMetricGoal.MAXIMIZE))
+    conv = TrialToArrayConverter(problem)
+    assert conv.n_features == 22
+    trusted = torch.rand(30, 22)
+    tr = acq_lib.TrustRegion.for_converter(trusted, conv)
+    # dof = 2 continuous + 1 categorical param = 3.
+    expected = 0.2 + 0.3 * 30 / (5 * 4)
+    assert tr.trust_radius == pytest.approx(expected)
+    # One-hot columns excluded from the L-inf distance.
+    xs = trusted[0].clone()
+    xs[2:] = 1.0 - xs[2:]  # flip every one-hot column
+    assert float(tr.min_linf_distance(xs[None])) == pytest.approx(0.0)
+
+  def test_trust_region_excludes_wide_gap_discretes(self):
+    from vizier_amd import pyvizier as vz
+    from vizier_amd.converters.core import TrialToArrayConverter
+    problem = vz.ProblemStatement()
+    problem.search_space.root.add_float_param('a', 0.0, 1.0)
+    # Two feasible values -> scaled gap 1.0 > min_radius: excluded.
+    problem.search_space.root.add_discrete_param('d', [0.0, 100.0])
+    # Eleven evenly spaced values -> gap 0.1 <= 0.2: included.
+    problem.search_space.root.add_discrete_param(
+        'e', [float(v) for v in range(0, 101, 10)])
+    problem.metric_information.append(vz.MetricInformation(
+        name='obj', goal=vz.ObjectiveMetricGoal.MAXIMIZE))
+    conv = TrialToArrayConverter(problem)
+    mask, n_cat = acq_lib.converter_trust_masks(conv)
+    assert n_cat == 0
+    assert mask == [False, True, False]
+    trusted = torch.rand(10, 3)
+    tr = acq_lib.TrustRegion.for_converter(trusted, conv)
+    # dof = 2 (a and e; d excluded).
+    expected = 0.2 + 0.3 * 10 / (5 * 3)
+    assert tr.trust_radius == pytest.approx(expected)
+
   def test_hv_scalarization(self):
     s = acq_lib.create_hv_scalarization(100, 2, seed=0)
     ys = torch.tensor([[1.0, 1.0], [0.1, 0.1]])
@@ -211,6 +257,33 @@ class TestOutputWarpers:
     out_r = output_warpers.TransformToGaussian(use_rank=True).warp(
         labels.copy())
     np.testing.assert_array_equal(np.argsort(out_r[:, 0]), order_in)
+
+  def test_infeasible_warper_keeps_bad_below_feasible(self):
+    # Regression for ADVICE r1 (high): the shift must apply to ALL
+    # entries after NaN substitution, so infeasible trials stay the
+    # WORST labels, not the best (reference output_warpers.py
+    # InfeasibleWarperComponent).
+    w = output_warpers.InfeasibleWarperComponent()
+    labels = np.array([[100.0], [101.0], [102.0], [np.nan]])
+    out = w.warp(labels.copy())
+    # The substituted infeasible entry is strictly below every feasible.
+    assert out[3, 0] < out[:3, 0].min()
+    # Relative order of feasible entries preserved.
+    assert out[0, 0] < out[1, 0] < out[2, 0]
+    # unwarp is the exact inverse on the feasible entries.
+    back = w.unwarp(out)
+    np.testing.assert_allclose(back[:3], labels[:3], rtol=1e-12)
+
+  def test_infeasible_warper_centering(self):
+    # E[warp] over the substituted array is ~0 weighted by p_feasible
+    # construction: mean of output == mean(sub) + shift.
+    w = output_warpers.InfeasibleWarperComponent()
+    labels = np.array([[1.0], [2.0], [np.nan], [3.0]])
+    out = w.warp(labels.copy())
+    rng_ = np.nanmax(labels) - np.nanmin(labels)
+    warped_bad = np.nanmin(labels) - (0.5 * rng_ + 1)
+    sub = np.array([1.0, 2.0, warped_bad, 3.0])
+    np.testing.assert_allclose(out[:, 0], sub + w._shift, rtol=1e-12)
 
   def test_linear_output_warper_roundtrip(self):
     y = np.array([[1.0, 10.0], [3.0, 20.0], [2.0, 12.0]])

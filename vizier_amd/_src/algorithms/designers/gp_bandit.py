@@ -290,12 +290,7 @@ class VizierGPBandit(Designer, Predictor):
   def _make_trust_region(self) -> Optional[acq_lib.TrustRegion]:
     if not self._config.use_trust_region:
       return None
-    onehot = torch.zeros(self._converter.n_features, dtype=torch.bool,
-                         device=self._device)
-    for col in self._converter.output_specs:
-      if col.is_onehot:
-        onehot[col.start:col.start + col.width] = True
-    return acq_lib.TrustRegion(self._x, onehot)
+    return acq_lib.TrustRegion.for_converter(self._x, self._converter)
 
   def _score_factory(self, count: int):
     """Returns (score_fn over CandidateBatch, n_parallel)."""

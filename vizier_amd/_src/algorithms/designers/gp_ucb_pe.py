@@ -305,13 +305,8 @@ class VizierGPUCBPEBandit(Designer):
                     ) -> torch.Tensor:
     cfg = self._config
     posterior = self._posterior
-    onehot = torch.zeros(self._converter.n_features, dtype=torch.bool,
-                         device=self._device)
-    for col in self._converter.output_specs:
-      if col.is_onehot:
-        onehot[col.start:col.start + col.width] = True
-    trust_region = acq_lib.TrustRegion(x_all, onehot) \
-        if cfg.use_trust_region else None
+    trust_region = acq_lib.TrustRegion.for_converter(
+        x_all, self._converter) if cfg.use_trust_region else None
 
     if self._mo_posteriors is not None:
       posts = self._mo_posteriors

@@ -218,19 +218,85 @@ def get_reference_point(labels: torch.Tensor, scale: float = 0.01
 # -- trust region ------------------------------------------------------------
 
 
+def converter_trust_masks(converter) -> tuple:
+  """(exclude_mask list[bool] over feature columns, n_categorical_params).
+
+  Mirrors the reference's TrustRegion.__post_init__ dimension masking
+  (acquisitions.py:736-749): a continuified discrete/integer column is
+  excluded when its feasible values, scaled to [0,1], have a consecutive
+  gap > min_radius (0.2), or when it has a single feasible value.
+  """
+  from vizier_amd import pyvizier as vz
+  min_radius = TrustRegion.MIN_RADIUS
+  exclude = [False] * converter.n_features
+  n_cat = 0
+  for col in converter.output_specs:
+    if col.is_onehot:
+      n_cat += 1
+      for c in range(col.start, col.start + col.width):
+        exclude[c] = True
+      continue
+    cfg = col.config
+    if cfg.type in (vz.ParameterType.DISCRETE, vz.ParameterType.INTEGER):
+      lo, hi = cfg.bounds
+      if hi == lo:
+        exclude[col.start] = True
+        continue
+      if cfg.type == vz.ParameterType.DISCRETE:
+        vals = sorted(float(v) for v in cfg.feasible_values)
+      else:
+        # Integers: enumerate when small; for wide ranges the largest
+        # scaled gap is between the two lowest values under LOG scaling
+        # (linear spacing is uniform), so those two suffice.
+        if hi - lo <= 64:
+          vals = [float(v) for v in range(int(lo), int(hi) + 1)]
+        else:
+          vals = [float(lo), float(lo) + 1.0, float(hi)]
+      if len(vals) < 2:
+        exclude[col.start] = True
+        continue
+      scaled = [_scale_value(cfg, v) for v in vals]
+      max_gap = max(b - a for a, b in zip(scaled[:-1], scaled[1:]))
+      if max_gap > min_radius:
+        exclude[col.start] = True
+  return exclude, n_cat
+
+
+def _scale_value(cfg, value: float) -> float:
+  """[0,1] scaling consistent with converters.core._scale."""
+  import math
+  lo, hi = cfg.bounds
+  if hi == lo:
+    return 0.0
+  st = cfg.scale_type
+  from vizier_amd import pyvizier as vz
+  if st == vz.ScaleType.LOG and lo > 0:
+    return (math.log(value) - math.log(lo)) / (math.log(hi) - math.log(lo))
+  if st == vz.ScaleType.REVERSE_LOG and lo > 0:
+    flipped = hi + lo - value
+    return 1.0 - (math.log(flipped) - math.log(lo)) / (
+        math.log(hi) - math.log(lo))
+  return (value - lo) / (hi - lo)
+
+
 class TrustRegion:
   """Union of L-inf balls around observed points (acquisitions.py:691).
 
   radius = 0.2 + (0.5 - 0.2) * num_obs / (5 * (dof + 1)); a radius > 0.5
-  disables the constraint. One-hot (categorical) feature columns are
-  excluded from the distance but counted in dof.
+  disables the constraint. Matches the reference's dof accounting
+  (acquisitions.py:752-768): dof = number of continuous feature columns
+  participating in the distance + ONE per categorical parameter (not one
+  per one-hot column). Columns flagged in `onehot_column_mask` — one-hot
+  blocks plus wide-gap discretes excluded via `for_converter` — are
+  excluded from the L-inf distance.
   """
 
   MIN_RADIUS = 0.2
   DIMENSION_FACTOR = 5.0
 
   def __init__(self, trusted: torch.Tensor,
-               onehot_column_mask: Optional[torch.Tensor] = None):
+               onehot_column_mask: Optional[torch.Tensor] = None,
+               *, n_categorical_params: Optional[int] = None):
     """trusted: (N, D) observed features in [0,1]."""
     self._trusted = trusted
     d = trusted.shape[-1]
@@ -238,14 +304,37 @@ class TrustRegion:
       onehot_column_mask = torch.zeros(d, dtype=torch.bool,
                                        device=trusted.device)
     self._onehot = onehot_column_mask
+    n_excluded = int(onehot_column_mask.sum())
+    if n_categorical_params is None:
+      # Fallback when built from a bare mask: each contiguous run of
+      # excluded columns is treated as one categorical parameter's
+      # one-hot block.
+      m = onehot_column_mask.to('cpu', torch.int8)
+      n_categorical_params = int(
+          ((m[1:] - m[:-1]) == 1).sum() + (1 if d and m[0] else 0))
     num_obs = trusted.shape[0]
-    dof = d  # continuous dims + categorical columns
+    dof = (d - n_excluded) + n_categorical_params
     trust_level = num_obs / (self.DIMENSION_FACTOR * (dof + 1))
     if num_obs == 0:
       self.trust_radius = 1.0
     else:
       self.trust_radius = self.MIN_RADIUS + (0.5 - self.MIN_RADIUS) * \
           trust_level
+
+  @classmethod
+  def for_converter(cls, trusted: torch.Tensor, converter
+                    ) -> 'TrustRegion':
+    """Builds the reference-faithful region from a TrialToArrayConverter.
+
+    Excludes from the distance: one-hot blocks, single-value parameters,
+    and continuified discrete/integer parameters whose largest gap
+    between consecutive scaled feasible values exceeds MIN_RADIUS
+    (reference acquisitions.py:736-749 `_continuous_dimensions_mask`).
+    """
+    mask, n_cat = converter_trust_masks(converter)
+    return cls(trusted,
+               torch.as_tensor(mask, device=trusted.device),
+               n_categorical_params=n_cat)
 
   def min_linf_distance(self, xs: torch.Tensor) -> torch.Tensor:
     """xs: (..., D) -> (...) L-inf distance to the nearest trusted point."""

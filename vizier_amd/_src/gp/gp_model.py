@@ -450,7 +450,14 @@ def _build_posterior_cache(x: torch.Tensor, y: torch.Tensor,
   # bounds that error to ~6e-5*amp^2 (a ~3% stddev error at the floor)
   # at the cost of a ~0.03*amp stddev floor near training points —
   # negligible against UCB-scale acquisition scores.
-  noise_eff = torch.maximum(p64.noise, 1e-3 * p64.amplitude ** 2)
+  # The floor only matters when the cache feeds the fp32 explicit-inverse
+  # quadform; the triangular-solve (composed/predict) path is fp64-backed
+  # and matches reference variances better with the true fitted noise
+  # (reference lower bound is 1e-10 in float64).
+  if precompute_inverse:
+    noise_eff = torch.maximum(p64.noise, 1e-3 * p64.amplitude ** 2)
+  else:
+    noise_eff = p64.noise
   K = K + noise_eff * torch.eye(n, dtype=x64.dtype, device=x.device)
   L64 = cholesky_with_jitter(K, p64.amplitude ** 2)
   resid = (y.double() - p64.mean).unsqueeze(-1)

@@ -150,9 +150,13 @@ class InfeasibleWarperComponent(OutputWarper):
     p_feasible = (0.5 + num_feasible) / (1 + flat.size)
     self._shift = (-np.nanmean(flat) * p_feasible
                    - warped_bad * (1 - p_feasible))
+    # Match the reference (output_warpers.py InfeasibleWarperComponent):
+    # substitute warped_bad for NaNs FIRST, then shift ALL entries so that
+    # E[warp] = 0 holds over the substituted array (and unwarp() below,
+    # which subtracts the shift from everything, is the exact inverse).
     nan_mask = np.isnan(flat)
     flat[nan_mask] = warped_bad
-    flat[~nan_mask] += self._shift
+    flat += self._shift
     return flat[:, None]
 
   def unwarp(self, labels_arr: np.ndarray) -> np.ndarray:

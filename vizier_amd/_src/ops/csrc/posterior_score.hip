@@ -221,24 +221,37 @@ ps_quadform_kernel(const float* __restrict__ k_in,   // (B, N)
   const float* k = k_in + (long)q * n;
   float acc = 0.0f;
   const int n4 = n / 4;
-  for (int j = j0 + wave; j < j1; j += waves) {
-    const float4* row4 = reinterpret_cast<const float4*>(
-        kinv + (long)j * n);
-    const float4* k4 = reinterpret_cast<const float4*>(k);
-    float t_j = 0.0f;
-    for (int i4 = lane; i4 < n4; i4 += WAVE_SIZE) {
-      const float4 r = row4[i4];
-      const float4 kv = k4[i4];
-      t_j = fmaf(r.x, kv.x, t_j);
-      t_j = fmaf(r.y, kv.y, t_j);
-      t_j = fmaf(r.z, kv.z, t_j);
-      t_j = fmaf(r.w, kv.w, t_j);
+  // float4 loads are only legal when every row base (kinv + j*n, k_in +
+  // q*n) is 16-byte aligned, i.e. n % 4 == 0 (the host pads the
+  // workspace leading dim, but defend here too: misaligned vector
+  // loads are UB on gfx950).
+  if ((n & 3) == 0) {
+    for (int j = j0 + wave; j < j1; j += waves) {
+      const float4* row4 = reinterpret_cast<const float4*>(
+          kinv + (long)j * n);
+      const float4* k4 = reinterpret_cast<const float4*>(k);
+      float t_j = 0.0f;
+      for (int i4 = lane; i4 < n4; i4 += WAVE_SIZE) {
+        const float4 r = row4[i4];
+        const float4 kv = k4[i4];
+        t_j = fmaf(r.x, kv.x, t_j);
+        t_j = fmaf(r.y, kv.y, t_j);
+        t_j = fmaf(r.z, kv.z, t_j);
+        t_j = fmaf(r.w, kv.w, t_j);
+      }
+      t_j = wave_reduce_sum(t_j);
+      if (lane == 0) acc = fmaf(k[j], t_j, acc);
     }
-    for (int i = 4 * n4 + lane; i < n; i += WAVE_SIZE) {
-      t_j = fmaf(kinv[(long)j * n + i], k[i], t_j);
+  } else {
+    for (int j = j0 + wave; j < j1; j += waves) {
+      const float* row = kinv + (long)j * n;
+      float t_j = 0.0f;
+      for (int i = lane; i < n; i += WAVE_SIZE) {
+        t_j = fmaf(row[i], k[i], t_j);
+      }
+      t_j = wave_reduce_sum(t_j);
+      if (lane == 0) acc = fmaf(k[j], t_j, acc);
     }
-    t_j = wave_reduce_sum(t_j);
-    if (lane == 0) acc = fmaf(k[j], t_j, acc);
   }
   // acc lives in lane 0 of each wave; combine across waves.
   auto fsum = [](float a, float c) { return a + c; };
