@@ -67,6 +67,11 @@ class GPBanditConfig:
   ard_max_iters: int = 50
   ard_warm_iters: int = 12   # iters when warm-starting from the last fit
   ard_warm_restarts: int = 2  # random restarts kept on warm refits
+  # False = every refit runs the full cold budget (ard_max_iters x
+  # ard_restarts, no warm start) — the reference's exact ARD schedule
+  # (gp_models.py:200-208). True keeps the regret-validated warm-refit
+  # optimization (profiles/warmiters.log, profiles/regret2.json).
+  warm_refit: bool = True
   use_trust_region: bool = True
   num_scalarizations: int = 1000  # multi-objective
   scorer_gram_dtype: str = 'fp32'  # 'fp32'|'bf16'|'fp8' candidate grams
@@ -231,7 +236,8 @@ class VizierGPBandit(Designer, Predictor):
               if cfg.multitask_type == 'separable_diag'
               else multitask.MultiTaskType.SEPARABLE)
       y_all = torch.as_tensor(y_np, dtype=cfg.dtype, device=self._device)
-      warm = self._mt_posterior.raw if self._mt_posterior is not None \
+      warm = self._mt_posterior.raw if (cfg.warm_refit and
+                                        self._mt_posterior is not None) \
           else None
       self._mt_posterior = multitask.train_multitask_gp(
           x, y_all, multitask_type=kind,
@@ -255,8 +261,8 @@ class VizierGPBandit(Designer, Predictor):
         with torch.no_grad():
           prior_mean, _ = self._prior_stack.predict(x)
         y = y - prior_mean
-      warm = prev[m].raw if m < len(prev) and prev[m].raw is not None \
-          else None
+      warm = prev[m].raw if (cfg.warm_refit and m < len(prev) and
+                             prev[m].raw is not None) else None
       # Warm refits: the previous optimum is almost always the winner,
       # so keep only a couple of random restarts — the restart batch
       # multiplies the per-iteration Cholesky cost (R x N^3).

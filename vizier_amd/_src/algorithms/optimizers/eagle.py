@@ -121,7 +121,9 @@ class VectorizedEagleStrategy:
       self._cat_sizes_t = torch.tensor(self.categorical_sizes,
                                        device=self.device)
     # The fused HIP path replaces the ~30-launch torch step with 2 launches.
-    if self.device.type == 'cuda':
+    # The eagle kernels are fp32; fp64 sweeps (--fp64 parity mode) run
+    # the torch path on rocBLAS DGEMMs instead.
+    if self.device.type == 'cuda' and dtype == torch.float32:
       from vizier_amd._src.ops import dispatch as ops
       self._ext = ops.require_ext()
       self._iter_t = torch.zeros(2, dtype=torch.long, device=self.device)

@@ -396,8 +396,13 @@ class ScoringFunction:
       self._tr_anchored = True
 
   def _can_fuse(self, xs: torch.Tensor) -> bool:
-    """True when the GPU K^-1 quadform path applies (any acquisition)."""
-    return xs.is_cuda and self.posterior.K_inv is not None
+    """True when the GPU K^-1 quadform path applies (any acquisition).
+
+    The HIP scorer kernels are fp32; fp64 parity mode scores through
+    the torch predict path (rocBLAS DGEMM / fp64 matrix cores).
+    """
+    return (xs.is_cuda and xs.dtype == torch.float32 and
+            self.posterior.K_inv is not None)
 
   def __call__(self, xs: torch.Tensor) -> torch.Tensor:
     if self._can_fuse(xs):

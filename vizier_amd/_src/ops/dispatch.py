@@ -51,7 +51,7 @@ def gram_matern52(x1: torch.Tensor, x2: Optional[torch.Tensor],
                   lengthscales: torch.Tensor,
                   amplitude: torch.Tensor) -> torch.Tensor:
   """Matern-5/2 ARD Gram / cross-Gram matrix."""
-  if x1.is_cuda:
+  if x1.is_cuda and x1.dtype == torch.float32:
     ext = require_ext()
     x2t = x1 if x2 is None else x2
     return ext.gram_matern52(x1.contiguous(), x2t.contiguous(),
