@@ -298,6 +298,75 @@ ps_finalize_kernel(const float* __restrict__ mu_in,
   out[q] = score;
 }
 
+// Finalize variant consuming a precomputed (b,) quadform directly.
+//
+// At huge N the per-candidate K^-1 streaming in ps_quadform_kernel
+// re-reads K_inv once per candidate per iteration (b * N^2 * 4 bytes =
+// 10 GB at N=10^4, B=25 — none of it L2/MALL-resident at that size),
+// which measured 4.2 ms per Eagle iteration on config 4. The host
+// instead computes T = k @ K_inv with ONE rocBLAS SGEMM (K_inv read
+// once, candidates reused from registers/LDS by the GEMM tiling) and
+// quad[q] = sum_i k[q,i] T[q,i]; this kernel applies the acquisition +
+// trust region to that.
+extern "C" __global__ void
+ps_finalize_direct_kernel(const float* __restrict__ mu_in,
+                          const float* __restrict__ dist_in,
+                          const float* __restrict__ quad_in,
+                          float* __restrict__ out, int b, float amp2,
+                          float mean_c, int acq, float coef,
+                          float best_value, float tr_radius) {
+  const int q = blockIdx.x * blockDim.x + threadIdx.x;
+  if (q >= b) return;
+  float var = fmaxf(amp2 - quad_in[q], 1e-12f);
+  const float sd = sqrtf(var);
+  const float mu = mu_in[q] + mean_c;
+  float score;
+  switch (acq) {
+    case ACQ_LCB: score = mu - coef * sd; break;
+    case ACQ_EI: {
+      const float z = (mu - best_value) / sd;
+      score = sd * (z * normal_cdf(z) + normal_pdf(z));
+      break;
+    }
+    case ACQ_PI: {
+      const float z = (mu - best_value) / sd;
+      score = normal_cdf(z);
+      break;
+    }
+    case ACQ_MEAN: score = mu; break;
+    case ACQ_STDDEV: score = sd; break;
+    case ACQ_UCB:
+    default: score = mu + coef * sd; break;
+  }
+  const float dist = dist_in[q];
+  if (tr_radius > 0.0f && tr_radius <= 0.5f && dist > tr_radius) {
+    score = -1e4f - dist;
+  }
+  out[q] = score;
+}
+
+extern "C" void launch_ps_kvec(
+    const float* xq, const float* x, const float* inv_ls,
+    const float* alpha, const unsigned char* onehot, float* k_ws,
+    float* mu_ws, float* dist_ws, int b, int n, int d, float amp2,
+    hipStream_t stream) {
+  hipLaunchKernelGGL(ps_kvec_kernel, dim3(b), dim3(BLOCK), 0, stream, xq,
+                     x, inv_ls, alpha, onehot, k_ws, mu_ws, dist_ws, b, n,
+                     d, amp2);
+}
+
+extern "C" void launch_ps_finalize_direct(
+    const float* mu_ws, const float* dist_ws, const float* quad,
+    float* out, int b, float amp2, float mean_c, int acq, float coef,
+    float best_value, float tr_radius, hipStream_t stream) {
+  const int fin_block = 256;
+  hipLaunchKernelGGL(ps_finalize_direct_kernel,
+                     dim3((b + fin_block - 1) / fin_block),
+                     dim3(fin_block), 0, stream, mu_ws, dist_ws, quad,
+                     out, b, amp2, mean_c, acq, coef, best_value,
+                     tr_radius);
+}
+
 extern "C" void launch_posterior_score_chunked(
     const float* xq, const float* x, const float* inv_ls,
     const float* alpha, const float* kinv, const unsigned char* onehot,
