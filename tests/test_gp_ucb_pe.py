@@ -91,6 +91,59 @@ class TestGPUCBPE:
     out = designer.suggest(1)
     assert len(out) == 1
 
+  def test_set_pe_batch_diversity(self):
+    # SetPE (logdet joint acquisition, reference gp_ucb_pe.py:510):
+    # exploration members of a batch must DECORRELATE — pairwise
+    # distances of the jointly-optimized set are bounded away from 0,
+    # and the suggest flow labels them 'set_pe'.
+    designer = VizierGPUCBPEBandit(
+        make_problem(), cfg(optimize_set_acquisition_for_exploration=True,
+                            max_evaluations=1500), seed=5)
+    trials = []
+    rng = np.random.default_rng(0)
+    for uid in range(1, 9):
+      s = vz.TrialSuggestion(
+          {f'x{i}': float(v) for i, v in enumerate(rng.uniform(0, 1, 3))})
+      t = s.to_trial(uid)
+      t.complete(vz.Measurement(metrics={'obj': evaluate(s)}))
+      trials.append(t)
+    designer.update(CompletedTrials(trials), ActiveTrials())
+    batch = designer.suggest(4)
+    assert len(batch) == 4
+    kinds = [s.metadata.abs_ns(('gp_ucb_pe',))['acquisition']
+             for s in batch]
+    # One UCB exploit (new data arrived), rest jointly explored.
+    assert kinds.count('set_pe') == 3
+    pts = np.array([[s.parameters.get_value(f'x{i}') for i in range(3)]
+                    for s in batch if 'set_pe' in
+                    s.metadata.abs_ns(('gp_ucb_pe',))['acquisition']])
+    dists = [np.abs(pts[a] - pts[b]).max()
+             for a in range(len(pts)) for b in range(a + 1, len(pts))]
+    # logdet of the joint covariance collapses to -inf for duplicate
+    # points, so the optimized set must be spread out.
+    assert min(dists) > 1e-3
+
+  def test_set_pe_no_new_trials_all_set(self):
+    designer = VizierGPUCBPEBandit(
+        make_problem(), cfg(optimize_set_acquisition_for_exploration=True,
+                            max_evaluations=800), seed=6)
+    trials = []
+    rng = np.random.default_rng(1)
+    for uid in range(1, 7):
+      s = vz.TrialSuggestion(
+          {f'x{i}': float(v) for i, v in enumerate(rng.uniform(0, 1, 3))})
+      t = s.to_trial(uid)
+      t.complete(vz.Measurement(metrics={'obj': evaluate(s)}))
+      trials.append(t)
+    designer.update(CompletedTrials(trials), ActiveTrials())
+    designer.suggest(1)   # consumes the has-new-trials credit
+    batch = designer.suggest(3)
+    kinds = [s.metadata.abs_ns(('gp_ucb_pe',))['acquisition']
+             for s in batch]
+    # No new completed trials since the last suggest: the whole batch
+    # comes from the set acquisition (gp_ucb_pe.py:1423-1431).
+    assert kinds == ['set_pe'] * 3
+
   def test_seed_phase(self):
     designer = VizierGPUCBPEBandit(make_problem(), cfg(), seed=4)
     first = designer.suggest(2)

@@ -61,6 +61,12 @@ class UCBPEConfig:
   # 'independent' per-metric GPs or 'separable'/'separable_diag' joint.
   multitask_type: str = 'independent'
   signal_to_noise_threshold: float = 0.7
+  # Batch PE via the SET acquisition (gp_ucb_pe.py:118,510): the
+  # exploration members of a batch are optimized JOINTLY, scored by the
+  # log-determinant of their predictive covariance (conditioned on
+  # completed + pending trials) — intra-batch correlation is penalized,
+  # so the batch decorrelates, unlike the sequential summed-stddev fill.
+  optimize_set_acquisition_for_exploration: bool = False
   max_evaluations: int = 75000
   suggestion_batch_size: int = 25
   num_seed_trials: int = 2
@@ -131,6 +137,31 @@ class VizierGPUCBPEBandit(Designer):
           device=self._device))
 
     suggestions: List[vz.TrialSuggestion] = []
+    if (cfg.optimize_set_acquisition_for_exploration and count > 1 and
+        self._mo_posteriors is None):
+      # Set-PE batch (gp_ucb_pe.py:1423-1437): at most ONE UCB exploit
+      # suggestion (only when new completed trials arrived), then the
+      # rest jointly via the logdet set acquisition.
+      if has_new:
+        x_cat = x_all[0] if len(x_all) == 1 else torch.cat(x_all, dim=0)
+        dense = self._optimize_one(True, x_cat)
+        x_all.append(dense.reshape(1, -1))
+        params = self._converter.to_parameters(
+            dense.detach().cpu().numpy())[0]
+        s = vz.TrialSuggestion(params)
+        s.metadata.ns('gp_ucb_pe')['acquisition'] = 'ucb'
+        suggestions.append(s)
+      remaining = count - len(suggestions)
+      if remaining > 0:
+        x_cat = x_all[0] if len(x_all) == 1 else torch.cat(x_all, dim=0)
+        rows = self._optimize_set(x_cat, remaining)
+        for p in self._converter.to_parameters(
+            rows.detach().cpu().numpy()):
+          s = vz.TrialSuggestion(p)
+          s.metadata.ns('gp_ucb_pe')['acquisition'] = 'set_pe'
+          suggestions.append(s)
+      return suggestions
+
     for i in range(count):
       use_ucb = self._choose_ucb(has_new and i == 0 and not self._active)
       # Single-element cat would COPY, breaking the trust-region anchor
@@ -300,6 +331,71 @@ class VizierGPUCBPEBandit(Designer):
         x=x_all, params=params, L=L,
         alpha=torch.zeros(n, dtype=x_all.dtype, device=x_all.device),
         K_inv=K_inv, nll=0.0)
+
+  def _optimize_set(self, x_all: torch.Tensor, q: int) -> torch.Tensor:
+    """Jointly optimizes q exploration points via the SET-PE acquisition.
+
+    Reference parity: SetPEScoreFunction (gp_ucb_pe.py:510-595).
+    score(set) = logdet(cov(set | completed + pending))
+                 + penalty_coef * sum_q min(explore_ucb(x_q) - thresh, 0)
+    with the trust-region set penalty of gp_ucb_pe.py:245-269. Returns
+    (q, D) feature rows.
+    """
+    cfg = self._config
+    posterior = self._posterior
+    if isinstance(posterior, gp_model.EnsembleGPPosterior):
+      posterior = posterior.members[0]
+    trust_region = acq_lib.TrustRegion.for_converter(
+        x_all, self._converter) if cfg.use_trust_region else None
+
+    with torch.no_grad():
+      mean_obs, stddev_obs = posterior.predict(posterior.x)
+      ucb_obs = mean_obs + cfg.ucb_coefficient * stddev_obs
+      threshold = mean_obs[int(torch.argmax(ucb_obs))]
+    var_post = self._variance_posterior(x_all, posterior)
+
+    def score_fn(batch: CandidateBatch) -> torch.Tensor:
+      xs = self._codec.decode(batch)                 # (B, q, D)
+      B = xs.shape[0]
+      flat = xs.reshape(B * q, -1)
+      mean, stddev = posterior.predict(flat)
+      explore_ucb = (mean + cfg.explore_region_ucb_coefficient *
+                     stddev).reshape(B, q)
+      penalty = cfg.cb_violation_penalty_coefficient * torch.minimum(
+          explore_ucb - threshold,
+          torch.zeros_like(explore_ucb)).sum(-1)
+      _, cov = gp_bandit.posterior_batched_cov(var_post, xs)
+      # Jitter scaled to the prior variance: near-duplicate rows (the
+      # optimizer will try them) make cov singular; the reference maps
+      # those to -inf via NaN logdet, cholesky_ex info does the same.
+      amp2 = float(var_post.params.amplitude) ** 2
+      eye = torch.eye(q, dtype=cov.dtype, device=cov.device)
+      L, info = torch.linalg.cholesky_ex(cov + 1e-10 * amp2 * eye)
+      logdet = 2.0 * torch.log(
+          torch.diagonal(L, dim1=-2, dim2=-1).clamp_min(1e-30)).sum(-1)
+      logdet = torch.where(info == 0, logdet,
+                           torch.full_like(logdet, -float('inf')))
+      scores = logdet + penalty
+      if trust_region is not None:
+        dist = trust_region.min_linf_distance(xs)    # (B, q)
+        r = trust_region.trust_radius
+        if r <= 0.5:
+          scores = scores + torch.where(
+              dist > r, -1e4 - dist, torch.zeros_like(dist)).sum(-1)
+      return scores
+    score_fn.graph_safe = False  # batched cholesky uses rocSOLVER
+
+    factory = VectorizedOptimizerFactory(
+        eagle_config=EagleStrategyConfig(),
+        max_evaluations=cfg.max_evaluations,
+        suggestion_batch_size=cfg.suggestion_batch_size)
+    optimizer = factory(
+        n_continuous=self._codec.n_continuous,
+        categorical_sizes=self._codec.categorical_sizes, n_parallel=q,
+        seed=self._seed + len(self._completed) + x_all.shape[0],
+        device=self._device, dtype=cfg.dtype)
+    results = optimizer.optimize(score_fn, count=1)
+    return self._codec.decode(results.features)[0]   # (q, D)
 
   def _optimize_one(self, use_ucb: bool, x_all: torch.Tensor
                     ) -> torch.Tensor:
