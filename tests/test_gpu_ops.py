@@ -154,6 +154,63 @@ class TestPosteriorScoreKernel:
     assert torch.allclose(got.cpu(), mean + 1.8 * stddev,
                           atol=0.01 * amp)
 
+  def test_bf16_cached_scorer_matches_fp32(self, ext):
+    # The cached-operand bf16 fused scorer (posterior_scores_bf16) must
+    # agree with the fp32 fused scorer to bf16-rounding tolerance, and
+    # ScoringFunction(gram_dtype='bf16') must take the fused path
+    # (graph-capturable; config-2 parity).
+    from vizier_amd._src.gp import acquisitions as acq_lib
+    post = self._posterior()
+    gpost = self._to_cuda(post)
+    g = torch.Generator().manual_seed(5)
+    xq = torch.rand(64, 10, generator=g).cuda()
+    s32 = acq_lib.ScoringFunction(gpost, acq_lib.UCB(1.8))
+    s16 = acq_lib.ScoringFunction(gpost, acq_lib.UCB(1.8),
+                                  gram_dtype='bf16')
+    assert s16._bf16_cache is not None
+    got32 = s32(xq)
+    got16 = s16(xq)
+    amp = float(gpost.params.amplitude)
+    # bf16 rounding perturbs d^2 by ~2^-8 relative: scores agree to a
+    # few percent of the amplitude scale.
+    assert torch.allclose(got16, got32, atol=0.05 * amp)
+
+  def test_bf16_cached_scorer_matches_torch_oracle(self, ext):
+    # posterior_scores_bf16 vs a pure-torch recomputation of the SAME
+    # bf16-rounded norm-trick distances (tight tolerance: identical
+    # rounding, only summation order differs).
+    from vizier_amd._src.gp import acquisitions as acq_lib
+    post = self._posterior()
+    gpost = self._to_cuda(post)
+    g = torch.Generator().manual_seed(6)
+    xq = torch.rand(16, 10, generator=g).cuda()
+    s16 = acq_lib.ScoringFunction(gpost, acq_lib.UCB(1.8),
+                                  gram_dtype='bf16')
+    z2b, n2 = s16._bf16_cache
+    args = (xq, gpost.x, z2b, n2, gpost.params.lengthscales,
+            float(gpost.params.amplitude), float(gpost.params.mean),
+            gpost.alpha, gpost.K_inv,
+            torch.zeros(10, dtype=torch.uint8).cuda(), 0, 1.8, 0.0, 0.0)
+    t = ext.posterior_scores_bf16(*args)
+    # Torch reference of the same bf16-rounded quadform:
+    z1 = (xq / gpost.params.lengthscales)
+    d = z1.shape[1]
+    z1b = torch.zeros(xq.shape[0], z2b.shape[1], dtype=torch.bfloat16,
+                      device=xq.device)
+    z1b[:, :d] = z1.to(torch.bfloat16)
+    n1 = (z1b.float() ** 2).sum(-1)
+    dot = z1b.float() @ z2b.float().T
+    d2 = (n1[:, None] + n2[None, :] - 2 * dot).clamp_min(0)
+    r = d2.sqrt()
+    sr = (5.0 ** 0.5) * r
+    amp = float(gpost.params.amplitude)
+    kv = amp * amp * (1 + sr + sr * sr / 3) * torch.exp(-sr)
+    mu = float(gpost.params.mean) + kv @ gpost.alpha
+    var = (amp * amp -
+           (kv * (kv @ gpost.K_inv)).sum(-1)).clamp_min(1e-12)
+    want = mu + 1.8 * var.sqrt()
+    assert torch.allclose(t, want, atol=0.02 * amp)
+
 
 class TestEagleKernels:
 

@@ -450,12 +450,17 @@ class VizierGPBandit(Designer, Predictor):
     scoring = acq_lib.ScoringFunction(posterior, acquisition, trust_region,
                                       gram_dtype=cfg.scorer_gram_dtype)
     if cfg.scorer_gram_dtype != 'fp32':
-      # bf16/fp8 grams run the composed path (host-side range scaling
-      # inside the binding), which is not capture-safe.
       def score_fn(batch: CandidateBatch) -> torch.Tensor:
         dense = self._codec.decode(batch)[:, 0, :]
         return scoring(dense)
-      score_fn.graph_safe = False
+      if scoring._bf16_cache is not None and scoring._tr_anchored:
+        # bf16 with cached training operands runs the fused HIP scorer
+        # (3 launches, no host-side conversion): hipGraph-capturable.
+        score_fn.graph_safe = True
+      else:
+        # fp8 (and unanchored bf16) grams run the composed path with
+        # host-side range scaling inside the binding: not capture-safe.
+        score_fn.graph_safe = False
       return score_fn, 1
 
     def score_fn(batch: CandidateBatch) -> torch.Tensor:

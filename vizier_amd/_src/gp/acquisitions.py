@@ -384,6 +384,23 @@ class ScoringFunction:
     self._coef = getattr(acquisition, 'coefficient', 0.0)
     self._best = getattr(acquisition, 'best_value', 0.0)
     self._onehot_u8 = None
+    # bf16 fused path: cache the TRAINING-side operands once per
+    # suggest (z2b = bf16(x / lengthscales), n2 = rounded row norms);
+    # the HIP kernel then only rounds the 25 candidates per iteration,
+    # so the bf16 scorer is a 3-launch graph-capturable sequence like
+    # fp32 instead of ~15 eager conversion launches.
+    self._bf16_cache = None
+    if (gram_dtype == 'bf16' and posterior.x.is_cuda and
+        posterior.K_inv is not None and self._acq_name is not None):
+      with torch.no_grad():
+        z = posterior.x / posterior.params.lengthscales
+        d = z.shape[1]
+        dp = (d + 31) // 32 * 32
+        z2b = torch.zeros(z.shape[0], dp, dtype=torch.bfloat16,
+                          device=z.device)
+        z2b[:, :d] = z.to(torch.bfloat16)
+        n2 = (z2b.float() ** 2).sum(-1)
+      self._bf16_cache = (z2b, n2)
     if trust_region is not None:
       self._onehot_u8 = trust_region._onehot.to(torch.uint8)
       self._tr_radius = float(trust_region.trust_radius)
@@ -408,6 +425,21 @@ class ScoringFunction:
     if self._can_fuse(xs):
       post = self.posterior
       from vizier_amd._src.ops import dispatch as ops
+      if (self._acq_name is not None and self._tr_anchored and
+          self._bf16_cache is not None):
+        # bf16 fused scorer: cached training operands, 3 launches.
+        onehot = self._onehot_u8
+        if onehot is None:
+          onehot = torch.zeros(xs.shape[-1], dtype=torch.uint8,
+                               device=xs.device)
+          self._onehot_u8 = onehot
+        z2b, n2 = self._bf16_cache
+        ext = ops.require_ext()
+        return ext.posterior_scores_bf16(
+            xs, post.x, z2b, n2, post.params.lengthscales, self._amp,
+            self._mean_c, post.alpha, post.K_inv, onehot,
+            ops.ACQ_CODES[self._acq_name], self._coef, self._best,
+            self._tr_radius if self.trust_region is not None else 0.0)
       if (self._acq_name is not None and self._tr_anchored and
           self.gram_dtype == 'fp32'):
         # Primary GPU path: the 3-kernel chunked HIP scorer — one

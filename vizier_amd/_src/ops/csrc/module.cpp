@@ -54,6 +54,19 @@ extern "C" void launch_ps_finalize_direct(
     float* out, int b, float amp2, float mean_c, int acq, float coef,
     float best_value, float tr_radius, hipStream_t stream);
 
+extern "C" void launch_ps_kvec_bf16(
+    const float* xq, const float* x, const unsigned short* z2b,
+    const float* n2, const float* inv_ls, const float* alpha,
+    const unsigned char* onehot, float* k_ws, float* mu_ws,
+    float* dist_ws, int b, int n, int d, int dp, float amp2,
+    hipStream_t stream);
+
+extern "C" void launch_ps_quadform_finalize(
+    const float* k_ws, const float* kinv, const float* mu_ws,
+    const float* dist_ws, float* var_ws, float* out, int b, int n,
+    float amp2, float mean_c, int acq, float coef, float best_value,
+    float tr_radius, hipStream_t stream);
+
 extern "C" void launch_eagle_suggest(
     const float* pool_cont, const long* pool_cat, const float* rewards,
     const float* perturbations, const long* cat_sizes, float* out_cont,
@@ -307,6 +320,67 @@ torch::Tensor posterior_scores_chunked(
   return out;
 }
 
+torch::Tensor posterior_scores_bf16(
+    torch::Tensor xq, torch::Tensor x, torch::Tensor z2b,
+    torch::Tensor n2, torch::Tensor lengthscales, double amplitude,
+    double mean_c, torch::Tensor alpha, torch::Tensor kinv,
+    torch::Tensor onehot, int64_t acq, double coef, double best_value,
+    double tr_radius) {
+  // bf16 candidate grams against CACHED training operands (z2b =
+  // bf16(x / lengthscales), n2 = rounded row norms — computed once per
+  // suggest by ScoringFunction). Same 3-launch graph-capturable shape
+  // as the fp32 chunked scorer.
+  xq = check_f32(xq, "xq");
+  x = check_f32(x, "x");
+  n2 = check_f32(n2, "n2");
+  lengthscales = check_f32(lengthscales, "lengthscales");
+  alpha = check_f32(alpha, "alpha");
+  kinv = check_f32(kinv, "kinv");
+  TORCH_CHECK(z2b.scalar_type() == torch::kBFloat16 && z2b.is_cuda(),
+              "z2b must be bf16 cuda");
+  z2b = z2b.contiguous();
+  onehot = onehot.contiguous();
+  const int b = xq.size(0), d = xq.size(1), n = x.size(0);
+  const int dp = z2b.size(1);
+  TORCH_CHECK(dp % 32 == 0 && dp <= 512, "dp must be mult of 32, <=512");
+  auto inv_ls = 1.0f / lengthscales;
+  auto k_ws = torch::empty({b, n}, xq.options());
+  auto mu_ws = torch::empty({b}, xq.options());
+  auto dist_ws = torch::empty({b}, xq.options());
+  auto out = torch::empty({b}, xq.options());
+  launch_ps_kvec_bf16(
+      xq.data_ptr<float>(), x.data_ptr<float>(),
+      (const unsigned short*)z2b.data_ptr(), n2.data_ptr<float>(),
+      inv_ls.data_ptr<float>(), alpha.data_ptr<float>(),
+      onehot.data_ptr<unsigned char>(), k_ws.data_ptr<float>(),
+      mu_ws.data_ptr<float>(), dist_ws.data_ptr<float>(), b, n, d, dp,
+      (float)(amplitude * amplitude), current_stream());
+  static const int gemm_n_threshold = []() {
+    const char* s = getenv("VIZIER_AMD_PS_GEMM_N");
+    return s ? atoi(s) : 4096;
+  }();
+  if (n >= gemm_n_threshold) {
+    auto t = at::matmul(k_ws, kinv);
+    auto quad = (k_ws * t).sum(-1);
+    launch_ps_finalize_direct(
+        mu_ws.data_ptr<float>(), dist_ws.data_ptr<float>(),
+        quad.data_ptr<float>(), out.data_ptr<float>(), b,
+        (float)(amplitude * amplitude), (float)mean_c, (int)acq,
+        (float)coef, (float)best_value, (float)tr_radius,
+        current_stream());
+    return out;
+  }
+  auto var_ws = torch::empty({b, 10}, xq.options());
+  launch_ps_quadform_finalize(
+      k_ws.data_ptr<float>(), kinv.data_ptr<float>(),
+      mu_ws.data_ptr<float>(), dist_ws.data_ptr<float>(),
+      var_ws.data_ptr<float>(), out.data_ptr<float>(), b, n,
+      (float)(amplitude * amplitude), (float)mean_c, (int)acq,
+      (float)coef, (float)best_value, (float)tr_radius,
+      current_stream());
+  return out;
+}
+
 std::vector<torch::Tensor> eagle_suggest(
     torch::Tensor pool_cont, torch::Tensor pool_cat, torch::Tensor rewards,
     torch::Tensor perturbations, torch::Tensor cat_sizes,
@@ -438,6 +512,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "128x128 LDS-tiled fp8 e4m3 MFMA Matern-5/2 Gram (gfx950)");
   m.def("posterior_scores", &posterior_scores,
         "Fused GP posterior + acquisition + trust region (gfx950)");
+  m.def("posterior_scores_bf16", &posterior_scores_bf16,
+        "Chunked scorer with cached-bf16 candidate grams (gfx950)");
   m.def("posterior_scores_chunked", &posterior_scores_chunked,
         "3-kernel chunked GP posterior scorer (chip-filling, gfx950)");
   m.def("eagle_suggest", &eagle_suggest,
