@@ -191,3 +191,59 @@ class TestSpatioTemporalConverters:
     assert grid.tolist() == [1.0, 2.0]
     assert y[0].tolist() == [1.0, 2.0]
     assert np.isnan(y[1][0]) and y[1][1] == 5.0
+
+
+class TestPaddingSchedule:
+  """Parity with reference converters/padding.py:28-97 (3-axis buckets)."""
+
+  def _trials(self, n):
+    out = []
+    for uid in range(1, n + 1):
+      t = vz.Trial({'lr': 0.01, 'opt': 'adam', 'mom': 0.5,
+                    'layers': 2, 'act': 'relu'}, id=uid)
+      t.complete(vz.Measurement(metrics={'acc': float(uid)}))
+      out.append(t)
+    return out
+
+  def test_three_axis_buckets(self):
+    from vizier_amd.converters.core import PaddingSchedule, PaddingType
+    s = PaddingSchedule(num_trials=PaddingType.MULTIPLES_OF_10,
+                        num_features=PaddingType.POWERS_OF_2,
+                        num_metrics=PaddingType.MULTIPLES_OF_10)
+    conv = TrialToArrayConverter(mixed_problem(), padding_schedule=s)
+    x, y, mask = conv.to_padded_xy(self._trials(13))
+    assert x.shape[0] == 20 and y.shape[0] == 20
+    assert mask.sum() == 13
+    # 8 raw features (lr + 3 onehot + mom + layers + 2 onehot) -> 8
+    # (already a power of 2); metrics 1 -> 10.
+    assert x.shape[1] == 8
+    assert y.shape[1] == 10
+    assert np.isnan(y[:, 1:]).all()
+    assert np.isnan(y[13:, 0]).all()
+
+  def test_feature_padding_distance_neutral(self):
+    # Zero-padded feature columns are identical across rows, so GP
+    # pairwise distances are unchanged.
+    import torch
+    from vizier_amd.converters.core import PaddingSchedule, PaddingType
+    from vizier_amd._src.gp.matern import gram_matern52
+    s = PaddingSchedule(num_features=PaddingType.MULTIPLES_OF_10)
+    conv_p = TrialToArrayConverter(mixed_problem(), padding_schedule=s)
+    conv = TrialToArrayConverter(mixed_problem())
+    trials = self._trials(5)
+    xp, _, _ = conv_p.to_padded_xy(trials)
+    x, _ = conv.to_xy(trials)
+    assert xp.shape[1] == 10 and x.shape[1] == 8
+    ls_p = torch.ones(10)
+    ls = torch.ones(8)
+    amp = torch.tensor(1.0)
+    Kp = gram_matern52(torch.tensor(xp), None, ls_p, amp)
+    K = gram_matern52(torch.tensor(x), None, ls, amp)
+    assert torch.allclose(Kp, K, atol=1e-6)
+
+  def test_bucket_stability_across_growth(self):
+    from vizier_amd.converters.core import PaddingSchedule, PaddingType
+    s = PaddingSchedule(num_trials=PaddingType.POWERS_OF_2)
+    # 65..128 trials all land in the SAME padded shape: one hipGraph.
+    sizes = {s.padded_size(n) for n in range(65, 129)}
+    assert sizes == {128}

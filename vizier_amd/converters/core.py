@@ -32,16 +32,39 @@ class PaddingType(enum.Enum):
 
 @dataclasses.dataclass(frozen=True)
 class PaddingSchedule:
-  """Pads the trials axis so kernel shapes re-use across suggest calls."""
+  """Pads (trials, features, metrics) so kernel shapes re-use.
+
+  Parity with vizier/pyvizier/converters/padding.py:28-97: each axis
+  buckets independently (NONE / MULTIPLES_OF_10 / POWERS_OF_2). Trials
+  bucketing stabilizes shapes across suggest calls (Gram/Cholesky/
+  hipGraph re-use); feature bucketing stabilizes D across studies in
+  one process (conditional spaces, transfer stacks) — zero-padded
+  feature columns are shared by every row, so they contribute 0 to all
+  pairwise distances and the GP math is unchanged; metric bucketing
+  pads labels with NaN (masked as missing).
+  """
 
   num_trials: PaddingType = PaddingType.NONE
+  num_features: PaddingType = PaddingType.NONE
+  num_metrics: PaddingType = PaddingType.NONE
 
-  def padded_size(self, n: int) -> int:
-    if n == 0 or self.num_trials == PaddingType.NONE:
+  @staticmethod
+  def _bucket(n: int, kind: PaddingType) -> int:
+    if n == 0 or kind == PaddingType.NONE:
       return n
-    if self.num_trials == PaddingType.MULTIPLES_OF_10:
+    if kind == PaddingType.MULTIPLES_OF_10:
       return int(math.ceil(n / 10.0) * 10)
     return 1 << (n - 1).bit_length()
+
+  def padded_size(self, n: int) -> int:
+    """Trials-axis bucket (back-compat name)."""
+    return self._bucket(n, self.num_trials)
+
+  def padded_features(self, d: int) -> int:
+    return self._bucket(d, self.num_features)
+
+  def padded_metrics(self, m: int) -> int:
+    return self._bucket(m, self.num_metrics)
 
 
 def pad_rows(arr: np.ndarray, target: int,
@@ -187,11 +210,25 @@ class TrialToArrayConverter:
 
   def to_padded_xy(self, trials: Sequence[vz.Trial]
                    ) -> Tuple[np.ndarray, np.ndarray, np.ndarray]:
-    """Returns (features, labels, valid_mask) padded per the schedule."""
+    """Returns (features, labels, valid_mask) padded per the schedule.
+
+    All three axes bucket: trials (rows, mask returned), features
+    (zero columns — distance-neutral for the GP kernels), metrics
+    (NaN columns — masked as missing labels).
+    """
     x, y = self.to_xy(trials)
     target = self._padding.padded_size(len(trials))
     x, mask = pad_rows(x, target)
     y, _ = pad_rows(y, target, fill=np.nan)
+    df = self._padding.padded_features(x.shape[1])
+    if df > x.shape[1]:
+      x = np.concatenate(
+          [x, np.zeros((x.shape[0], df - x.shape[1]), x.dtype)], axis=1)
+    dm = self._padding.padded_metrics(y.shape[1])
+    if dm > y.shape[1]:
+      y = np.concatenate(
+          [y, np.full((y.shape[0], dm - y.shape[1]), np.nan, y.dtype)],
+          axis=1)
     return x, y, mask
 
   # -- inverse --------------------------------------------------------------
