@@ -146,6 +146,199 @@ class TestEagleDesigner:
     assert best >= rand_best - 1e-6
 
 
+class TestEagleUtils:
+  """Behavior-level checks mirroring eagle_strategy_utils_test.py."""
+
+  def _utils(self, problem=None, **cfg_kw):
+    from vizier_amd._src.algorithms.designers.eagle_strategy.eagle_utils \
+        import EagleUtils, FireflyAlgorithmConfig
+    problem = problem or mixed_eagle_problem()
+    return EagleUtils(problem, FireflyAlgorithmConfig(**cfg_kw),
+                      np.random.default_rng(0))
+
+  def test_pool_capacity_formula(self):
+    # min(10 + round((df^1.2 + df) * 0.5), 1000) — reference :235.
+    u = self._utils(continuous_problem(4))
+    assert u.pool_capacity() == 10 + round((4 ** 1.2 + 4) * 0.5)
+    u100 = self._utils(continuous_problem(100))
+    expected = min(10 + round((100 ** 1.2 + 100) * 0.5), 1000)
+    assert u100.pool_capacity() == expected
+
+  def test_pull_weights_per_type_visibility(self):
+    u = self._utils()
+    v1 = {'f': 0.2, 'c': 'a', 'd': 0.0, 'i': 0.5}
+    v2 = {'f': 0.8, 'c': 'b', 'd': 1.0, 'i': 0.0}
+    w = u.pull_weights_by_type(v1, v2, other_is_better=True)
+    # Continuous visibility (3.0) decays harder than categorical (0.2)
+    # for the same per-dof distance scale.
+    assert 0 < w[vz.ParameterType.DOUBLE] < 1
+    assert 0 < w[vz.ParameterType.CATEGORICAL] <= 1
+    # Worse fly pushes (negative weight scaled by negative_gravity).
+    w_neg = u.pull_weights_by_type(v1, v2, other_is_better=False)
+    assert w_neg[vz.ParameterType.DOUBLE] < 0
+
+  def test_matching_categories_weaken_pull(self):
+    # Reference counts EQUAL categories into distance^2 (:216).
+    u = self._utils()
+    same = {'f': 0.5, 'c': 'a', 'd': 0.5, 'i': 0.5}
+    diff = {'f': 0.5, 'c': 'b', 'd': 0.5, 'i': 0.5}
+    w_same = u.pull_weights_by_type(same, same, True)
+    w_diff = u.pull_weights_by_type(diff, same, True)
+    assert w_same[vz.ParameterType.CATEGORICAL] < \
+        w_diff[vz.ParameterType.CATEGORICAL]
+
+  def test_combine_categorical_bernoulli(self):
+    u = self._utils()
+    cfg = [c for c in u.parameter_configs if c.name == 'c'][0]
+    # Extreme weights are deterministic.
+    assert u.combine(cfg, 'a', 'b', 1.5) == 'a'
+    assert u.combine(cfg, 'a', 'b', -0.5) == 'b'
+    picks = [u.combine(cfg, 'a', 'b', 0.8) for _ in range(200)]
+    frac_a = sum(p == 'a' for p in picks) / 200
+    assert 0.65 < frac_a < 0.95
+
+  def test_combine_numeric_is_linear_mix_clipped(self):
+    u = self._utils()
+    cfg = [c for c in u.parameter_configs if c.name == 'f'][0]
+    assert u.combine(cfg, 0.8, 0.2, 0.5) == pytest.approx(0.5)
+    # Weights outside [0,1] extrapolate, then clip to the scaled box.
+    assert u.combine(cfg, 1.0, 0.0, 2.0) == 1.0
+
+  def test_perturbation_scales_by_type(self):
+    u = self._utils()
+    scales = u.perturbation_scales()
+    names = [c.name for c in u.parameter_configs]
+    assert scales[names.index('c')] == 25.0      # categorical factor
+    # Discrete: 10 / (n_feasible * base_perturbation).
+    d_cfg = [c for c in u.parameter_configs if c.name == 'd'][0]
+    nfeas = len(d_cfg.feasible_values)
+    assert scales[names.index('d')] == pytest.approx(
+        10.0 / (nfeas * 0.1))
+    assert scales[names.index('f')] == 1.0
+
+  def test_categorical_perturb_is_replacement_probability(self):
+    u = self._utils()
+    cfg = [c for c in u.parameter_configs if c.name == 'c'][0]
+    kept = sum(u.perturb(cfg, 'a', 0.0) == 'a' for _ in range(50))
+    assert kept == 50
+    changed = sum(u.perturb(cfg, 'a', 1.0) != 'a' for _ in range(200))
+    # With prob 1 a uniform category is drawn; 'a' itself has 1/3 mass.
+    assert changed > 100
+
+  def test_pure_categorical_constant_perturbation(self):
+    p = vz.ProblemStatement()
+    p.search_space.root.add_categorical_param('c1', ['a', 'b'])
+    p.search_space.root.add_categorical_param('c2', ['x', 'y'])
+    p.metric_information.append(vz.MetricInformation(
+        name='obj', goal=vz.ObjectiveMetricGoal.MAXIMIZE))
+    u = self._utils(p)
+    assert u.is_pure_categorical()
+    np.testing.assert_allclose(u.create_perturbations(0.3),
+                               [0.1, 0.1])
+
+
+def mixed_eagle_problem():
+  p = vz.ProblemStatement()
+  root = p.search_space.root
+  root.add_float_param('f', 0.0, 10.0)
+  root.add_categorical_param('c', ['a', 'b', 'z'])
+  root.add_discrete_param('d', [1.0, 2.0, 4.0, 8.0])
+  root.add_int_param('i', 0, 10)
+  p.metric_information.append(vz.MetricInformation(
+      name='obj', goal=vz.ObjectiveMetricGoal.MAXIMIZE))
+  return p
+
+
+class TestEagleDesignerBehavior:
+
+  def _completed(self, designer, s, uid, value):
+    t = s.to_trial(uid)
+    t.complete(vz.Measurement(metrics={'obj': value}))
+    designer.update(CompletedTrials([t]), ActiveTrials())
+    return t
+
+  def test_stuck_fly_escalates_perturbation(self):
+    from vizier_amd._src.algorithms.designers.eagle_strategy.eagle_utils \
+        import Firefly
+    problem = continuous_problem(2)
+    designer = EagleStrategyDesigner(problem, seed=0)
+    # Install one fly and report a NON-improving trial with IDENTICAL
+    # parameters: perturbation must x10 (capped at max_perturbation).
+    values = {'x0': 0.5, 'x1': 0.5}
+    designer._firefly_pool._pool[7] = Firefly(
+        id_=7, perturbation=0.1, generation=1, values=dict(values),
+        reward=1.0)
+    s = vz.TrialSuggestion({'x0': 0.5, 'x1': 0.5})
+    s.metadata.ns('eagle')['parent_fly_id'] = '7'
+    self._completed(designer, s, 1, 0.0)   # worse reward
+    assert designer._pool[7].perturbation == pytest.approx(0.5)
+
+  def test_penalize_decay_and_best_fly_survives(self):
+    from vizier_amd._src.algorithms.designers.eagle_strategy.eagle_utils \
+        import Firefly
+    problem = continuous_problem(2)
+    designer = EagleStrategyDesigner(problem, seed=0)
+    pool = designer._firefly_pool
+    # Fill to capacity; fly 0 is the best with perturbation at floor.
+    for i in range(pool.capacity):
+      pool._pool[i] = Firefly(
+          id_=i, perturbation=1e-3 if i == 0 else 0.1, generation=1,
+          values={'x0': i / 100.0, 'x1': 0.5},
+          reward=10.0 if i == 0 else float(i) / 100)
+    s = vz.TrialSuggestion({'x0': 0.9, 'x1': 0.9})
+    s.metadata.ns('eagle')['parent_fly_id'] = '0'
+    self._completed(designer, s, 1, -1.0)  # non-improving
+    # perturbation decayed below the bound, but the BEST fly survives.
+    assert 0 in designer._pool
+    # A non-best fly at the floor gets removed at capacity.
+    designer._pool[1].perturbation = 1e-3
+    s2 = vz.TrialSuggestion({'x0': 0.8, 'x1': 0.8})
+    s2.metadata.ns('eagle')['parent_fly_id'] = '1'
+    self._completed(designer, s2, 2, -1.0)
+    assert 1 not in designer._pool
+
+  def test_foreign_trial_adopted_by_closest_parent_only_if_better(self):
+    from vizier_amd._src.algorithms.designers.eagle_strategy.eagle_utils \
+        import Firefly
+    problem = continuous_problem(2)
+    designer = EagleStrategyDesigner(problem, seed=0)
+    pool = designer._firefly_pool
+    for i in range(pool.capacity):
+      pool._pool[i] = Firefly(
+          id_=i, perturbation=0.1, generation=1,
+          values={'x0': i / pool.capacity, 'x1': 0.0}, reward=0.5)
+    pool._max_fly_id = pool.capacity
+    # A trial with NO eagle metadata lands near fly 0; better reward
+    # replaces the closest fly's values.
+    s = vz.TrialSuggestion({'x0': 0.01, 'x1': 0.02})
+    self._completed(designer, s, 1, 2.0)
+    assert designer._pool[0].reward == pytest.approx(2.0)
+
+  def test_mixed_space_round_trip(self):
+    problem = mixed_eagle_problem()
+    designer = EagleStrategyDesigner(problem, seed=1)
+    trials = RandomMetricsRunner(problem, iters=30, batch_size=1,
+                                 seed=2).run_designer(designer)
+    assert len(trials) == 30
+    for t in trials:
+      assert t.parameters.get_value('d') in (1.0, 2.0, 4.0, 8.0)
+      assert t.parameters.get_value('c') in ('a', 'b', 'z')
+      assert isinstance(t.parameters.get_value('i'), int)
+
+  def test_serialization_roundtrip_preserves_rng_stream(self):
+    problem = continuous_problem(3)
+    d1 = EagleStrategyDesigner(problem, seed=9)
+    trials = RandomMetricsRunner(problem, iters=25, batch_size=1,
+                                 seed=3).run_designer(d1)
+    assert trials
+    state = d1.dump()
+    d2 = EagleStrategyDesigner(problem, seed=123)  # different seed
+    d2.load(state)
+    s1 = [s.parameters.as_dict() for s in d1.suggest(3)]
+    s2 = [s.parameters.as_dict() for s in d2.suggest(3)]
+    assert s1 == s2
+
+
 class TestBOCS:
 
   def test_runs_and_finds_good_bits(self):

@@ -2,75 +2,114 @@
 
 Capability parity with
 vizier/_src/algorithms/designers/eagle_strategy/eagle_strategy.py
-(EagleStrategyDesigner :95, FireflyPool/EagleStrategyUtils
-eagle_strategy_utils.py:103,437, serialization.py): a persistent firefly
-pool evolved one suggestion at a time through the service protocol, with
-full state (pool, perturbations, rewards, ids) serialized into study
-metadata so the designer survives across Pythia calls. This serves the
-EAGLE_STRATEGY algorithm string; the GPU-vectorized variant used inside
-GP-Bandit lives in vizier_amd/_src/algorithms/optimizers/eagle.py.
+(EagleStrategyDesigner :95) at the reference's behavioral depth:
+per-parameter-TYPE pull forces and perturbations (eagle_utils.py —
+reference eagle_strategy_utils.py:103-435), exploration-rate
+accentuation, categorical Bernoulli mixing, stuck-fly detection
+(identical parameters => perturbation x10 capped), capacity-gated
+removal that never evicts the best fly, closest-parent adoption of
+foreign trials, and full pool + RNG serialization into study metadata
+(PartiallySerializableDesigner). Serves the EAGLE_STRATEGY algorithm
+string; the GPU-vectorized variant used inside GP-Bandit lives in
+vizier_amd/_src/algorithms/optimizers/eagle.py.
 """
 
 from __future__ import annotations
 
 import json
 import math
-from typing import Dict, List, Optional, Sequence
+from typing import Optional, Sequence
 
 import numpy as np
 
 from vizier_amd import pyvizier as vz
-from vizier_amd.converters.core import TrialToArrayConverter
 from vizier_amd._src.algorithms.core.abstractions import (
     ActiveTrials,
     CompletedTrials,
     PartiallySerializableDesigner,
 )
-from vizier_amd._src.algorithms.optimizers.eagle import (
-    EagleStrategyConfig,
-    compute_pool_size,
+from vizier_amd._src.algorithms.designers.eagle_strategy.eagle_utils import (
+    EagleUtils,
+    FireflyAlgorithmConfig,
+    FireflyPool,
 )
 
 _NS = 'eagle'
-_FLY_ID_KEY = 'firefly_id'
-
-
-class _Firefly:
-
-  def __init__(self, fly_id: int, features: np.ndarray,
-               reward: float = -math.inf, perturbation: float = 0.16):
-    self.id = fly_id
-    self.features = features            # dense converter features
-    self.reward = reward
-    self.perturbation = perturbation
-
-  def to_json(self) -> dict:
-    return {'id': self.id, 'features': self.features.tolist(),
-            'reward': self.reward, 'perturbation': self.perturbation}
-
-  @classmethod
-  def from_json(cls, d: dict) -> '_Firefly':
-    return cls(d['id'], np.asarray(d['features'], dtype=np.float64),
-               d['reward'], d['perturbation'])
+_PARENT_KEY = 'parent_fly_id'
 
 
 class EagleStrategyDesigner(PartiallySerializableDesigner):
-  """Firefly pool updated incrementally from completed trials."""
+  """Firefly pool evolved one suggestion at a time."""
 
   def __init__(self, problem: vz.ProblemStatement,
-               config: Optional[EagleStrategyConfig] = None, *,
+               config: Optional[FireflyAlgorithmConfig] = None, *,
                seed: Optional[int] = None):
+    if problem.search_space.is_conditional:
+      raise ValueError('EagleStrategyDesigner does not support '
+                       'conditional search spaces.')
+    metrics = list(problem.metric_information)
+    if len(metrics) != 1:
+      raise ValueError('EagleStrategyDesigner is single-objective.')
     self._problem = problem
-    self._config = config or EagleStrategyConfig()
-    self._converter = TrialToArrayConverter(problem)
+    self._metric_name = metrics[0].name
+    self._config = config or FireflyAlgorithmConfig()
     self._rng = np.random.default_rng(seed)
-    d = self._converter.n_features
-    self._pool_capacity = compute_pool_size(d, None, self._config)
-    self._pool: Dict[int, _Firefly] = {}
-    self._next_id = 0
-    self._best_reward = -math.inf
+    self._utils = EagleUtils(problem, self._config, self._rng)
+    self._firefly_pool = FireflyPool(self._utils,
+                                     self._utils.pool_capacity())
+    from vizier_amd._src.algorithms.designers.quasi_random import (
+        QuasiRandomDesigner,
+    )
+    self._initial_designer = QuasiRandomDesigner(problem.search_space,
+                                                 seed=seed)
+
+  # Back-compat attribute used by tests/PARITY checks.
+  @property
+  def _pool(self):
+    return self._firefly_pool.members
 
   # -- designer protocol ----------------------------------------------------
+
+  def suggest(self, count: Optional[int] = None
+              ) -> Sequence[vz.TrialSuggestion]:
+    return [self._suggest_one() for _ in range(count or 1)]
+
+  def _suggest_one(self) -> vz.TrialSuggestion:
+    if self._firefly_pool.size < self._firefly_pool.capacity:
+      # Underpopulated pool: quasi-random init (reference seeds with a
+      # serializable initial designer, eagle_strategy.py:155-158).
+      params = self._initial_designer.suggest(1)[0].parameters
+      parent_id = self._firefly_pool.generate_new_fly_id()
+      suggestion = vz.TrialSuggestion(params)
+    else:
+      moving = self._firefly_pool.get_next_moving_fly_copy()
+      self._mutate_fly(moving)
+      self._perturb_fly(moving)
+      parent_id = moving.id_
+      suggestion = vz.TrialSuggestion(
+          self._utils.values_to_parameters(moving.values))
+    suggestion.metadata.ns(_NS)[_PARENT_KEY] = str(parent_id)
+    return suggestion
+
+  def _mutate_fly(self, moving) -> None:
+    """Applies pulls from every pool member in shuffled order."""
+    for other in self._firefly_pool.get_shuffled_flies(self._rng):
+      other_better = other.reward > moving.reward
+      weights = self._utils.pull_weights_by_type(
+          other.values, moving.values, other_better)
+      for cfg in self._utils.parameter_configs:
+        w = weights[cfg.type]
+        if other.infeasible:
+          w *= self._config.infeasible_force_factor
+        w = self._utils.explore_weight(w)
+        moving.values[cfg.name] = self._utils.combine(
+            cfg, other.values[cfg.name], moving.values[cfg.name], w)
+
+  def _perturb_fly(self, moving) -> None:
+    amounts = self._utils.create_perturbations(moving.perturbation)
+    for i, cfg in enumerate(self._utils.parameter_configs):
+      moving.values[cfg.name] = self._utils.perturb(
+          cfg, moving.values[cfg.name], float(amounts[i]))
 
   def update(self, completed: CompletedTrials, all_active: ActiveTrials
              ) -> None:
@@ -78,110 +117,72 @@ class EagleStrategyDesigner(PartiallySerializableDesigner):
     for trial in completed.trials:
       self._update_one(trial)
 
-  def _trial_reward(self, trial: vz.Trial) -> float:
-    if trial.final_measurement is None or trial.infeasible:
-      return -math.inf
-    y = self._converter.to_labels([trial])[0, 0]
-    return float(y) if np.isfinite(y) else -math.inf
-
   def _update_one(self, trial: vz.Trial) -> None:
-    reward = self._trial_reward(trial)
-    self._best_reward = max(self._best_reward, reward)
-    features = self._converter.to_features([trial])[0].astype(np.float64)
-    fly_id = trial.metadata.abs_ns((_NS,)).get(_FLY_ID_KEY, None)
-    fly = self._pool.get(int(fly_id)) if fly_id is not None else None
-
-    if fly is None:
-      # Trial from elsewhere (or pre-pool): adopt it if there is room or
-      # it beats the closest pool member.
-      if len(self._pool) < self._pool_capacity:
-        self._spawn(features, reward)
-      else:
-        closest = min(self._pool.values(), key=lambda f: float(
-            np.sum((f.features - features) ** 2)))
-        if reward > closest.reward:
-          closest.features, closest.reward = features, reward
-      return
-
-    if reward > fly.reward:
-      fly.features = features
-      fly.reward = reward
+    values = self._utils.trial_to_values(trial)
+    reward = self._utils.trial_reward(trial, self._metric_name)
+    parent_raw = trial.metadata.abs_ns((_NS,)).get(_PARENT_KEY, None)
+    if parent_raw is None:
+      # Foreign trial: assign a fresh id (reference eagle_strategy.py
+      # :365-369).
+      parent_id = self._firefly_pool.generate_new_fly_id()
     else:
-      fly.perturbation *= self._config.penalize_factor
-      if (fly.perturbation < self._config.perturbation_lower_bound and
-          fly.reward != self._best_reward):
-        # Dead fly: replace with a random restart.
-        del self._pool[fly.id]
-        self._spawn(self._random_features(), -math.inf)
+      parent_id = int(parent_raw)
 
-  def _spawn(self, features: np.ndarray, reward: float) -> _Firefly:
-    fly = _Firefly(self._next_id, features, reward,
-                   self._config.perturbation)
-    self._next_id += 1
-    self._pool[fly.id] = fly
-    return fly
+    if trial.infeasible and self._config.infeasible_force_factor > 0:
+      self._firefly_pool.create_or_update_fly(
+          self._firefly_pool.generate_new_fly_id(), values, -math.inf,
+          infeasible=True)
 
-  def _random_features(self) -> np.ndarray:
-    d = self._converter.n_features
-    return self._rng.uniform(0, 1, d)
+    parent = self._firefly_pool.find_parent_fly(parent_id)
+    if parent is None:
+      if trial.infeasible:
+        return
+      if self._firefly_pool.size < self._firefly_pool.capacity:
+        self._firefly_pool.create_or_update_fly(parent_id, values,
+                                                reward, infeasible=False)
+        return
+      # At capacity: adopt via the closest parent, but only when the
+      # trial improves on it (otherwise the parent is not responsible
+      # for the failure — eagle_strategy.py:407-437).
+      closest = self._firefly_pool.find_closest_parent(values)
+      if closest is None or not (reward > closest.reward):
+        return
+      parent = closest
 
-  def suggest(self, count: Optional[int] = None
-              ) -> Sequence[vz.TrialSuggestion]:
-    count = count or 1
-    out = []
-    for _ in range(count):
-      out.append(self._suggest_one())
-    return out
-
-  def _suggest_one(self) -> vz.TrialSuggestion:
-    if len(self._pool) < self._pool_capacity:
-      fly = self._spawn(self._random_features(), -math.inf)
-      features = fly.features
+    if not trial.infeasible and reward > parent.reward:
+      parent.values = values
+      parent.reward = reward
+      parent.generation += 1
     else:
-      # Round-robin through the pool by id.
-      ids = sorted(self._pool)
-      fly = self._pool[ids[self._next_id % len(ids)]]
-      self._next_id += 1
-      features = self._mutate(fly)
-    suggestion = vz.TrialSuggestion(
-        self._converter.to_parameters(features[None, :])[0])
-    suggestion.metadata.abs_ns((_NS,))[_FLY_ID_KEY] = str(fly.id)
-    return suggestion
+      self._penalize_parent(parent, values)
 
-  def _mutate(self, fly: _Firefly) -> np.ndarray:
-    cfg = self._config
-    d = self._converter.n_features
-    others = [f for f in self._pool.values()
-              if f.id != fly.id and math.isfinite(f.reward)]
-    x = fly.features.copy()
-    pulls, pushes = [], []
-    for other in others:
-      d2 = float(np.sum((other.features - x) ** 2))
-      force = math.exp(-cfg.visibility * d2 / max(d, 1) * 10.0)
-      if other.reward >= fly.reward:
-        pulls.append((cfg.gravity * force, other.features))
-      else:
-        pushes.append((-cfg.negative_gravity * force, other.features))
-    move = np.zeros(d)
-    for group in (pulls, pushes):
-      if group:
-        for s, feat in group:
-          move += (cfg.normalization_scale * s / len(group)) * (feat - x)
-    noise = self._rng.laplace(size=d)
-    noise = np.sign(noise) * fly.perturbation
-    return np.clip(x + move + noise, 0.0, 1.0)
+  def _penalize_parent(self, parent, values) -> None:
+    """Reference eagle_strategy.py:438-464."""
+    if values == parent.values:
+      # Identical parameters: the fly is STUCK — escalate exploration.
+      parent.perturbation = min(parent.perturbation * 10.0,
+                                self._config.max_perturbation)
+    else:
+      parent.perturbation *= self._config.penalize_factor
+    if parent.perturbation < self._config.perturbation_lower_bound:
+      # Remove only at capacity and never the best fly (critical for
+      # studies with few feasible trials).
+      if (self._firefly_pool.size == self._firefly_pool.capacity and
+          not self._firefly_pool.is_best_fly(parent)):
+        self._firefly_pool.remove_fly(parent)
 
   # -- serialization --------------------------------------------------------
 
   def dump(self) -> vz.Metadata:
-    state = {
-        'next_id': self._next_id,
-        'best_reward': (self._best_reward
-                        if math.isfinite(self._best_reward) else None),
-        'pool': [f.to_json() for f in self._pool.values()],
-    }
     md = vz.Metadata()
+    state = {
+        'rng': self._rng.bit_generator.state,
+        'firefly_pool': self._firefly_pool.to_json(),
+        'version': 'v2',
+    }
     md.ns(_NS)['state'] = json.dumps(state)
+    md.ns(_NS).ns('initial_designer').attach(
+        self._initial_designer.dump())
     return md
 
   def load(self, metadata: vz.Metadata) -> None:
@@ -189,7 +190,10 @@ class EagleStrategyDesigner(PartiallySerializableDesigner):
     if blob is None:
       raise ValueError('No Eagle state found in metadata.')
     state = json.loads(blob)
-    self._next_id = state['next_id']
-    self._best_reward = (state['best_reward']
-                         if state['best_reward'] is not None else -math.inf)
-    self._pool = {f['id']: _Firefly.from_json(f) for f in state['pool']}
+    self._rng.bit_generator.state = state['rng']
+    self._firefly_pool.load_json(state['firefly_pool'])
+    try:
+      self._initial_designer.load(
+          metadata.ns(_NS).ns('initial_designer'))
+    except Exception:
+      pass  # harmless: the quasi-random stream restarts
