@@ -165,31 +165,98 @@ class TestEnsembles:
 
 class TestMetaLearning:
 
-  def test_hyperparameter_epochs(self):
+  def _problem(self):
     problem = vz.ProblemStatement()
     problem.search_space.root.add_float_param('x', 0.0, 1.0)
     problem.metric_information.append(vz.MetricInformation(name='m',
                                                            goal=1))
+    return problem
+
+  def _meta_space(self):
     meta_space = vz.SearchSpace()
-    meta_space.root.add_float_param('exploration', 0.1, 1.0)
-    seen = []
+    meta_space.root.add_float_param('exploration', 0.1, 1.0,
+                                    default_value=0.4)
+    return meta_space
 
-    def tuned_factory(p, hparams):
-      seen.append(dict(hparams))
-      return RandomDesigner(p.search_space, seed=0)
-
-    designer = MetaLearningDesigner(
-        problem, tuned_factory, meta_space,
-        config=MetaLearningConfig(num_trials_per_update=3))
+  def _run(self, designer, n, value_fn):
     uid = 0
-    for _ in range(7):
+    for _ in range(n):
       s = designer.suggest(1)[0]
       uid += 1
       t = s.to_trial(uid)
-      t.complete(vz.Measurement(metrics={'m': float(uid)}))
+      t.complete(vz.Measurement(metrics={'m': value_fn(s, uid)}))
       designer.update(CompletedTrials([t]), ActiveTrials())
-    assert len(seen) >= 3  # initial + >= 2 epochs
-    assert all(0.1 <= h['exploration'] <= 1.0 for h in seen)
+
+  def test_state_machine_and_continuous_reward(self):
+    from vizier_amd._src.algorithms.designers.meta_learning import (
+        MetaLearningState,
+    )
+    seen = []
+
+    def tuned_factory(p, seed=None, exploration=None):
+      seen.append(exploration)
+      return RandomDesigner(p.search_space, seed=seed)
+
+    designer = MetaLearningDesigner(
+        self._problem(), tuned_factory, self._meta_space(),
+        config=MetaLearningConfig(num_trials_per_tuning=3,
+                                  tuning_min_num_trials=4,
+                                  tuning_max_num_trials=16),
+        seed=0)
+    # INITIALIZE uses the search-space DEFAULT hyperparameter.
+    assert designer.state == MetaLearningState.INITIALIZE
+    assert seen == [0.4]
+    self._run(designer, 3, lambda s, uid: float(uid))
+    assert designer.state == MetaLearningState.INITIALIZE
+    self._run(designer, 4, lambda s, uid: float(uid))
+    # Past tuning_min: TUNE with at least one epoch rotated.
+    assert designer.state == MetaLearningState.TUNE
+    assert len(seen) >= 2
+    assert all(0.1 <= h <= 1.0 for h in seen)
+    # Meta trials carry the epoch's BEST objective as a continuous
+    # score (reference meta_learning_utils.py:82-85).
+    assert designer._meta_trials
+    first_meta = designer._meta_trials[0]
+    score = first_meta.final_measurement.metrics['score'].value
+    # Epoch trials carried values [1,2,3] then [1] (uid restarts per
+    # _run): the epoch closes at the 4th trial with best value 3.0.
+    assert score == 3.0
+    # Run past tuning_max: locks in the best meta hyperparameters.
+    self._run(designer, 12, lambda s, uid: float(uid))
+    assert designer.state == MetaLearningState.USE_BEST_PARAMS
+    best_meta = max(
+        designer._meta_trials,
+        key=lambda t: t.final_measurement.metrics['score'].value)
+    assert designer.current_hyperparameters['exploration'] == \
+        pytest.approx(best_meta.parameters.get_value('exploration'))
+
+  def test_missing_default_raises(self):
+    meta_space = vz.SearchSpace()
+    meta_space.root.add_float_param('lr', 0.1, 1.0)  # no default
+    with pytest.raises(ValueError, match='default'):
+      MetaLearningDesigner(
+          self._problem(),
+          lambda p, seed=None, **kw: RandomDesigner(p.search_space),
+          meta_space)
+
+  def test_meta_eagle_space_and_factory(self):
+    from vizier_amd._src.algorithms.designers.meta_learning import (
+        meta_eagle_designer_factory,
+        meta_eagle_search_space,
+    )
+    space = meta_eagle_search_space()
+    names = {c.name for top in space.parameters for c in top.traverse()}
+    assert {'perturbation', 'gravity', 'visibility',
+            'pool_size_factor', 'negative_gravity'} <= names
+    designer = MetaLearningDesigner(
+        self._problem(), meta_eagle_designer_factory, space, seed=1,
+        config=MetaLearningConfig(num_trials_per_tuning=4,
+                                  tuning_min_num_trials=4,
+                                  tuning_max_num_trials=40))
+    self._run(designer, 10,
+              lambda s, uid: -abs(s.parameters.get_value('x') - 0.3))
+    # The inner Eagle designer received meta-suggested hyperparams.
+    assert designer._curr_designer._config.perturbation > 0
 
 
 class TestSafetyWrapper:
