@@ -10,6 +10,11 @@ from vizier_amd import pyvizier as vz
 from vizier_amd._src.algorithms.classification import SklearnClassifier
 from vizier_amd._src.algorithms.regression import (
     GBMAutoRegressor,
+    GBMTrialHallucinator,
+    HallucinationOptions,
+    TrialData,
+    WindowedAutoRegressor,
+    sort_dedupe_measurements,
     trials_to_curves,
 )
 
@@ -33,11 +38,99 @@ class TestRegression:
       start = rng.uniform(1.0, 3.0)
       steps = np.arange(20)
       curves.append(start * 0.9 ** steps)
-    model = GBMAutoRegressor(window=3, seed=0).fit(curves)
+    model = WindowedAutoRegressor(window=3, seed=0).fit(curves)
     prefix = 2.0 * 0.9 ** np.arange(8)
     pred = model.predict_final(prefix, total_steps=20)
     true = 2.0 * 0.9 ** 19
     assert abs(pred - true) < 0.1
+
+
+class TestTargetStepRegression:
+  """Reference trial_regression_utils.py behavior checks."""
+
+  def _curve_trial(self, uid, lr, start, decay, n_steps, complete=True):
+    t = vz.Trial({'learning_rate': lr}, id=uid)
+    values = [start * decay ** k for k in range(n_steps)]
+    t.measurements = [
+        vz.Measurement(metrics={'loss': v}, steps=k + 1,
+                       elapsed_secs=float(k + 1))
+        for k, v in enumerate(values)]
+    if complete:
+      t.complete(vz.Measurement(metrics={'loss': values[-1]},
+                                steps=n_steps))
+    return t
+
+  def test_trial_data_extraction_and_extrapolation(self):
+    t = self._curve_trial(1, 0.1, 2.0, 0.9, 5)
+    td = TrialData.from_trial(t, learning_rate_param_name='learning_rate',
+                              metric_name='loss')
+    assert td.learning_rate == pytest.approx(0.1)
+    assert td.steps == [1, 2, 3, 4, 5]
+    td.extrapolate_trial_objective_value(10)
+    assert td.steps[-1] == 10
+    assert td.objective_values[-1] == td.objective_values[-2]
+
+  def test_sort_dedupe(self):
+    s, v = sort_dedupe_measurements([3, 1, 3, 2], [30.0, 10.0, 31.0, 20.0])
+    assert s == [1, 2, 3]
+    assert v == [10.0, 20.0, 31.0]  # last value per step wins
+
+  def test_gbm_predicts_target_step_value(self):
+    rng = np.random.default_rng(0)
+    trials = []
+    for uid in range(1, 41):
+      start = float(rng.uniform(1.0, 3.0))
+      trials.append(self._curve_trial(uid, 0.1, start, 0.9, 20))
+    model = GBMAutoRegressor(target_step=20, min_points=3,
+                             metric_name='loss', random_state=0)
+    model.train(trials)
+    assert model.is_trained
+    assert model.best_params is not None
+    # A stopped trial with an 8-step prefix: predict its value at 20.
+    probe = self._curve_trial(99, 0.1, 2.0, 0.9, 8, complete=False)
+    pred = model.predict(probe)
+    true = 2.0 * 0.9 ** 19
+    assert abs(pred - true) < 0.15
+
+  def test_gbm_returns_none_for_short_prefix(self):
+    trials = [self._curve_trial(uid, 0.1, 2.0, 0.9, 20)
+              for uid in range(1, 31)]
+    model = GBMAutoRegressor(target_step=20, min_points=5,
+                             metric_name='loss', random_state=0)
+    model.train(trials)
+    probe = self._curve_trial(99, 0.1, 2.0, 0.9, 2, complete=False)
+    assert model.predict(probe) is None
+
+  def test_hallucinator_completes_stopped_trials(self):
+    rng = np.random.default_rng(1)
+    problem = vz.ProblemStatement()
+    problem.search_space.root.add_float_param('learning_rate', 0.0, 1.0)
+    problem.metric_information.append(vz.MetricInformation(
+        name='loss', goal=vz.ObjectiveMetricGoal.MINIMIZE))
+    trials = [self._curve_trial(uid, 0.1, float(rng.uniform(1, 3)),
+                                0.9, 20) for uid in range(1, 41)]
+    h = GBMTrialHallucinator(problem, HallucinationOptions(
+        autoregressive_order=3, min_steps=3, max_steps=20,
+        random_state=0))
+    h.train(trials)
+    assert h.is_trained
+    stopped = self._curve_trial(99, 0.1, 2.0, 0.9, 9, complete=False)
+    out = h.update_stopped_trials([stopped])
+    assert out[0].final_measurement is not None
+    hallucinated = out[0].final_measurement.metrics['loss'].value
+    assert abs(hallucinated - 2.0 * 0.9 ** 19) < 0.2
+    assert out[0].final_measurement.steps == 20
+
+  def test_hallucinator_needs_enough_trials(self):
+    problem = vz.ProblemStatement()
+    problem.search_space.root.add_float_param('learning_rate', 0.0, 1.0)
+    problem.metric_information.append(vz.MetricInformation(
+        name='loss', goal=vz.ObjectiveMetricGoal.MINIMIZE))
+    h = GBMTrialHallucinator(problem)
+    h.train([self._curve_trial(1, 0.1, 2.0, 0.9, 10)])
+    assert not h.is_trained
+    stopped = self._curve_trial(9, 0.1, 2.0, 0.9, 8, complete=False)
+    assert h.update_stopped_trials([stopped])[0].final_measurement is None
 
 
 class TestClassification:
