@@ -84,12 +84,22 @@ class BOCSDesigner(Designer):
 
   def __init__(self, problem: vz.ProblemStatement, *,
                order: int = 2, num_init_samples: int = 10,
-               sa_iters: int = 500, seed: Optional[int] = None):
+               sa_iters: int = 500, seed: Optional[int] = None,
+               acquisition_optimizer: str = 'sdp',
+               lamda: float = 1e-4, sdp_repeats: int = 100,
+               sdp_bm_iters: int = 300):
+    if acquisition_optimizer not in ('sdp', 'sa'):
+      raise ValueError("acquisition_optimizer must be 'sdp' or 'sa'")
     self._configs = _bool_configs(problem)
     self._n = len(self._configs)
     self._order = order
     self._num_init = num_init_samples
     self._sa_iters = sa_iters
+    # SDP acquisition (the reference default, bocs.py:537-539).
+    self._acquisition_optimizer = acquisition_optimizer
+    self._lamda = lamda
+    self._sdp_repeats = sdp_repeats
+    self._sdp_bm_iters = sdp_bm_iters
     self._rng = np.random.default_rng(seed)
     self._xs: List[np.ndarray] = []
     self._ys: List[float] = []
@@ -136,6 +146,60 @@ class BOCSDesigner(Designer):
           best_x, best_val = x.copy(), val
     return best_x
 
+  def _sdp_rounding(self, weights: np.ndarray) -> np.ndarray:
+    """SDP-relaxation acquisition (reference bocs.py:448-525).
+
+    Minimizes x^T A x + b^T x over {0,1}^n via the +-1 homogenized SDP
+    min tr(At X), X PSD, diag(X)=1, solved with a Burer-Monteiro
+    low-rank factorization X = V^T V on the product of unit spheres
+    (projected gradient descent — no external SDP solver needed
+    offline; same relaxation cvxpy solves in the reference), followed
+    by Goemans-Williamson random-hyperplane rounding.
+    """
+    n = self._n
+    lamda = self._lamda
+    # Internal convention maximizes; the SDP formulation minimizes.
+    alpha = -weights
+    b = alpha[1:n + 1] + lamda
+    a = alpha[n + 1:]
+    A = np.zeros((n, n))
+    for k, (i, j) in enumerate(self._pairs):
+      A[i, j] = a[k] / 2.0
+      A[j, i] = a[k] / 2.0
+    bt = b / 2.0 + A @ np.ones(n) / 2.0
+    At = np.zeros((n + 1, n + 1))
+    At[:n, :n] = A / 4.0
+    At[:n, n] = bt / 2.0
+    At[n, :n] = bt / 2.0
+    At[n, n] = 2.0
+    # Burer-Monteiro: V (k, n+1) with unit columns; grad tr(At V^T V)
+    # = 2 V At; Riemannian step = project out the radial component.
+    k_rank = max(2, int(np.ceil(np.sqrt(2.0 * (n + 1)))))
+    V = self._rng.standard_normal((k_rank, n + 1))
+    V /= np.linalg.norm(V, axis=0, keepdims=True)
+    scale = max(np.abs(At).max(), 1e-12)
+    lr = 0.2 / scale
+    for it in range(self._sdp_bm_iters):
+      grad = 2.0 * V @ At
+      grad -= V * (V * grad).sum(axis=0, keepdims=True)
+      V -= lr * grad
+      V /= np.linalg.norm(V, axis=0, keepdims=True)
+      if it == self._sdp_bm_iters // 2:
+        lr *= 0.3
+    # Random-hyperplane rounding; keep the best of y and -y per cut.
+    best_x, best_obj = None, np.inf
+    for _ in range(self._sdp_repeats):
+      r = self._rng.standard_normal(k_rank)
+      r /= max(np.linalg.norm(r), 1e-12)
+      y = np.sign(r @ V)
+      y[y == 0] = 1.0
+      for cand in (y, -y):
+        x01 = (cand[:n] + 1.0) / 2.0
+        obj = float(x01 @ A @ x01 + b @ x01)
+        if obj < best_obj:
+          best_obj, best_x = obj, x01
+    return best_x
+
   def suggest(self, count: Optional[int] = None
               ) -> Sequence[vz.TrialSuggestion]:
     count = count or 1
@@ -149,6 +213,8 @@ class BOCSDesigner(Designer):
     for _ in range(count):
       if model is None:
         bits = self._rng.integers(0, 2, self._n)
+      elif self._acquisition_optimizer == 'sdp' and self._pairs:
+        bits = self._sdp_rounding(model.sample_weights())
       else:
         bits = self._simulated_annealing(model.sample_weights())
       params = {pc.name: ('true' if bits[i] > 0.5 else 'false')
