@@ -1,5 +1,7 @@
 """Tests for BBOB, experimenters, runners, analyzers and test harnesses."""
 
+import os
+
 import numpy as np
 import pytest
 
@@ -237,3 +239,160 @@ class TestExplorationScores:
     avg_const = compute_average_marginal_parameter_entropy([(p, const)])
     assert avg_spread > avg_const
     assert compute_average_marginal_parameter_entropy([]) == 0.0
+
+
+class TestHPOB:
+  """HPO-B handler + experimenter over the bundled tiny fixture."""
+
+  ROOT = os.path.join(os.path.dirname(__file__), 'data', 'hpob')
+
+  def test_availability_check(self):
+    from vizier_amd._src.benchmarks.experimenters.hpob import HPOBHandler
+    assert HPOBHandler.is_available(self.ROOT)
+    assert not HPOBHandler.is_available('/nonexistent')
+    with pytest.raises(FileNotFoundError, match='HPO-B'):
+      HPOBHandler('/nonexistent')
+
+  def test_tabular_evaluate_loop(self):
+    from vizier_amd._src.benchmarks.experimenters.hpob import HPOBHandler
+    handler = HPOBHandler(self.ROOT)
+
+    class GreedyMethod:
+      def observe_and_suggest(self, x_obs, y_obs, x_pending):
+        # Pick the pending row closest to the best observed row.
+        best = x_obs[np.argmax(y_obs)]
+        return int(np.argmin(((x_pending - best) ** 2).sum(1)))
+
+    history = handler.evaluate(GreedyMethod(), search_space_id='5970',
+                               dataset_id='3561', seed='test0',
+                               n_trials=6)
+    assert len(history) == 7
+    assert all(b >= a for a, b in zip(history, history[1:]))
+    assert 0.0 <= history[-1] <= 1.0
+
+  def test_surrogate_experimenter_end_to_end(self):
+    from vizier_amd._src.benchmarks.experimenters.hpob import (
+        HPOBExperimenter,
+    )
+    exp = HPOBExperimenter(self.ROOT, '5970', '3561')
+    problem = exp.problem_statement()
+    n_params = sum(1 for top in problem.search_space.parameters
+                   for _ in top.traverse())
+    assert n_params == 3
+    t_good = vz.Trial({'x0': 0.6, 'x1': 0.6, 'x2': 0.6}, id=1)
+    t_bad = vz.Trial({'x0': 0.0, 'x1': 0.0, 'x2': 1.0}, id=2)
+    exp.evaluate([t_good, t_bad])
+    g = t_good.final_measurement.metrics['accuracy'].value
+    b = t_bad.final_measurement.metrics['accuracy'].value
+    assert g > b  # surrogate learned the quadratic's shape
+
+
+class TestNASBench:
+
+  def test_nb101_synthetic_end_to_end(self):
+    from vizier_amd._src.benchmarks.experimenters.nasbench import (
+        NASBench101Experimenter,
+        SyntheticNASBench101,
+    )
+    exp = NASBench101Experimenter(SyntheticNASBench101())
+    problem = exp.problem_statement()
+    names = [c.name for top in problem.search_space.parameters
+             for c in top.traverse()]
+    assert len(names) == 21 + 5  # DAG edges + op slots
+    # A sparse valid architecture.
+    params = {n: 'false' for n in names if '_' in n and 'ops' not in n}
+    params.update({'0_1': 'true', '1_6': 'true'})
+    for i in range(5):
+      params[f'ops_{i}'] = 'conv3x3-bn-relu'
+    t = vz.Trial(params, id=1)
+    exp.evaluate([t])
+    assert not t.infeasible
+    acc = t.final_measurement.metrics['validation_accuracy'].value
+    assert 0.4 <= acc <= 0.9
+    # Determinism: same architecture, same metrics.
+    t2 = vz.Trial(params, id=2)
+    exp.evaluate([t2])
+    assert t2.final_measurement.metrics['validation_accuracy'].value \
+        == acc
+    # A dense architecture (> 9 edges) is infeasible.
+    dense = dict(params)
+    for n in names:
+      if 'ops' not in n:
+        dense[n] = 'true'
+    t3 = vz.Trial(dense, id=3)
+    exp.evaluate([t3])
+    assert t3.infeasible
+
+  def test_nb201_topology_string_and_eval(self):
+    from vizier_amd._src.benchmarks.experimenters.nasbench import (
+        NASBench201Experimenter,
+        SyntheticNASBench201,
+        model_tss_spec,
+    )
+    s = model_tss_spec(['a', 'b', 'c', 'd', 'e', 'f'], 4)
+    assert s == '|a~0|+|b~0|c~1|+|d~0|e~1|f~2|'
+    exp = NASBench201Experimenter(SyntheticNASBench201())
+    t = vz.Trial({f'op_{i}': 'nor_conv_3x3' for i in range(6)}, id=1)
+    exp.evaluate([t])
+    assert 40.0 <= \
+        t.final_measurement.metrics['valid-accuracy'].value <= 90.0
+
+  def test_real_loaders_gated(self):
+    from vizier_amd._src.benchmarks.experimenters.nasbench import (
+        load_nasbench101,
+        load_nasbench201,
+    )
+    with pytest.raises(ImportError, match='nasbench'):
+      load_nasbench101('/nonexistent')
+    with pytest.raises(ImportError, match='nats_bench'):
+      load_nasbench201()
+
+
+class TestAtari100k:
+
+  def test_search_space_parity(self):
+    from vizier_amd._src.benchmarks.experimenters.atari100k import (
+        Atari100kExperimenter,
+    )
+    exp = Atari100kExperimenter()
+    problem = exp.problem_statement()
+    names = {c.name for top in problem.search_space.parameters
+             for c in top.traverse()}
+    assert 'JaxDQNAgent.gamma' in names
+    assert 'create_optimizer.learning_rate' in names
+    assert len(names) == 14
+
+  def test_evaluate_gated_without_backend(self):
+    from vizier_amd._src.benchmarks.experimenters.atari100k import (
+        Atari100kExperimenter,
+    )
+    exp = Atari100kExperimenter()
+    t = vz.Trial({}, id=1)
+    with pytest.raises(ImportError, match='Dopamine'):
+      exp.evaluate([t])
+
+  def test_injected_runner_completes_with_curve(self):
+    from vizier_amd._src.benchmarks.experimenters.atari100k import (
+        Atari100kExperimenter,
+    )
+
+    class FakeRunner:
+      def __init__(self, game, agent, bindings):
+        self.bindings = bindings
+      def run_trial(self):
+        return {'eval_average_return': [1.0, 2.0, 5.0]}
+
+    exp = Atari100kExperimenter(
+        runner_factory=lambda g, a, b: FakeRunner(g, a, b))
+    t = vz.Trial({'JaxDQNAgent.gamma': 0.9}, id=1)
+    exp.evaluate([t])
+    assert len(t.measurements) == 2  # intermediate epochs
+    assert t.final_measurement.metrics[
+        'eval_average_return'].value == 5.0
+
+  def test_invalid_agent_rejected(self):
+    from vizier_amd._src.benchmarks.experimenters.atari100k import (
+        Atari100kExperimenter,
+    )
+    with pytest.raises(ValueError):
+      Atari100kExperimenter(agent_name='NotAnAgent')
