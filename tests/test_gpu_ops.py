@@ -171,9 +171,13 @@ class TestPosteriorScoreKernel:
     got32 = s32(xq)
     got16 = s16(xq)
     amp = float(gpost.params.amplitude)
-    # bf16 rounding perturbs d^2 by ~2^-8 relative: scores agree to a
-    # few percent of the amplitude scale.
-    assert torch.allclose(got16, got32, atol=0.05 * amp)
+    # bf16 rounds z = x/ls to 8 mantissa bits; the d^2 error passes
+    # through the kernel and is AMPLIFIED by the variance cancellation
+    # amp^2 - k^T K_inv k near training points, so score deviations
+    # reach ~7% of amp at this shape (measured 0.026 on amp 0.38).
+    # The tight rounding-exactness check against the torch bf16 oracle
+    # is the next test; here we bound the drift vs full fp32.
+    assert torch.allclose(got16, got32, atol=0.12 * amp)
 
   def test_bf16_cached_scorer_matches_torch_oracle(self, ext):
     # posterior_scores_bf16 vs a pure-torch recomputation of the SAME
