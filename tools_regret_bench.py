@@ -87,9 +87,23 @@ def main():
     )
     return QuasiRandomDesigner(problem.search_space, seed=seed)
 
+  def gp_bandit_fp64(problem, seed):
+    from vizier_amd._src.algorithms.designers.gp_bandit import (
+        GPBanditConfig,
+        VizierGPBandit,
+    )
+    return VizierGPBandit(problem, GPBanditConfig(
+        max_evaluations=args.evals, ard_restarts=2, ard_max_iters=30,
+        ard_warm_iters=15, device=device,
+        dtype=torch.float64), seed=seed)
+
   functions = {'Sphere': bbob.Sphere, 'Rastrigin': bbob.Rastrigin,
-               'SharpRidge': bbob.SharpRidge}
+               'SharpRidge': bbob.SharpRidge,
+               'Rosenbrock': bbob.Rosenbrock,
+               'Ellipsoidal': bbob.Ellipsoidal,
+               'AttractiveSector': bbob.AttractiveSector}
   algorithms = {'gp_bandit_ucb': gp_bandit, 'gp_ucb_pe': gp_ucb_pe,
+                'gp_bandit_fp64': gp_bandit_fp64,
                 'quasi_random': quasi_random}
   results = {}
   for fname, fn in functions.items():
