@@ -127,7 +127,7 @@ class Feedback:
   def skip(self, reason: Optional[str] = None) -> None:
     self._trial_client.complete(
         vz.Measurement(),
-        infeasibility_reason=reason or 'skipped by feedback')
+        infeasible_reason=reason or 'skipped by feedback')
 
   def should_stop_early(self) -> bool:
     return self._trial_client.check_early_stopping()

@@ -424,6 +424,8 @@ class TrialFilter:
     self.ids = frozenset(ids) if ids is not None else None
     self.min_id = min_id
     self.max_id = max_id
+    if isinstance(status, (TrialStatus, str)):
+      status = [status]  # a lone status is a common call-site shape
     self.status = frozenset(TrialStatus(s) if not isinstance(s, TrialStatus)
                             else s for s in status) if status else None
 
