@@ -1,7 +1,7 @@
 // Python bindings for the vizier_amd gfx950 kernels (torch extension).
 
 #include <ATen/cuda/CUDAContext.h>
-#include <torch/extension.h>
+#include <torch/extension.h>\n#include <string>
 
 #include <hip/hip_runtime.h>
 
@@ -48,6 +48,10 @@ extern "C" void launch_ps_kvec(
     const float* alpha, const unsigned char* onehot, float* k_ws,
     float* mu_ws, float* dist_ws, int b, int n, int d, float amp2,
     hipStream_t stream);
+
+extern "C" void launch_ps_quadform_big(
+    const float* k_ws, const float* kinv, float* part, float* quad,
+    int b, int n, hipStream_t stream);
 
 extern "C" void launch_ps_finalize_direct(
     const float* mu_ws, const float* dist_ws, const float* quad,
@@ -261,6 +265,38 @@ torch::Tensor gram_matern52_fp8_tiled(torch::Tensor x1, torch::Tensor x2,
   return gram_matern52_fp8_impl(x1, x2, lengthscales, amplitude, true);
 }
 
+namespace {
+
+// Large-N variance quadform: either ONE rocBLAS SGEMM or the custom
+// split-K column-owner kernel (ps_quadform_big). The custom kernel is
+// the default for b <= 32 (measured faster than the skinny SGEMM);
+// VIZIER_AMD_QUADFORM=gemm forces the library path.
+torch::Tensor quadform_large_n(const torch::Tensor& k_ws,
+                               const torch::Tensor& kinv, int b, int n,
+                               hipStream_t stream) {
+  static const bool force_gemm = []() {
+    const char* s = getenv("VIZIER_AMD_QUADFORM");
+    return s && std::string(s) == "gemm";
+  }();
+  if (!force_gemm && b <= 32) {
+    const int jchunks = (n + 255) / 256;
+    int ichunks = (512 + jchunks - 1) / jchunks;
+    if (ichunks < 1) ichunks = 1;
+    auto part = torch::empty({b, (long)jchunks * ichunks},
+                             k_ws.options());
+    auto quad = torch::empty({b}, k_ws.options());
+    launch_ps_quadform_big(k_ws.data_ptr<float>(),
+                           kinv.data_ptr<float>(),
+                           part.data_ptr<float>(),
+                           quad.data_ptr<float>(), b, n, stream);
+    return quad;
+  }
+  auto t = at::matmul(k_ws, kinv);
+  return (k_ws * t).sum(-1);
+}
+
+}  // namespace
+
 torch::Tensor posterior_scores_chunked(
     torch::Tensor xq, torch::Tensor x, torch::Tensor lengthscales,
     double amplitude, double mean_c, torch::Tensor alpha,
@@ -297,8 +333,7 @@ torch::Tensor posterior_scores_chunked(
         onehot.data_ptr<unsigned char>(), k_ws.data_ptr<float>(),
         mu_ws.data_ptr<float>(), dist_ws.data_ptr<float>(), b, n, d,
         (float)(amplitude * amplitude), current_stream());
-    auto t = at::matmul(k_ws, kinv);      // (b, n) SGEMM
-    auto quad = (k_ws * t).sum(-1);       // (b,)
+    auto quad = quadform_large_n(k_ws, kinv, b, n, current_stream());
     launch_ps_finalize_direct(
         mu_ws.data_ptr<float>(), dist_ws.data_ptr<float>(),
         quad.data_ptr<float>(), out.data_ptr<float>(), b,
@@ -360,8 +395,7 @@ torch::Tensor posterior_scores_bf16(
     return s ? atoi(s) : 4096;
   }();
   if (n >= gemm_n_threshold) {
-    auto t = at::matmul(k_ws, kinv);
-    auto quad = (k_ws * t).sum(-1);
+    auto quad = quadform_large_n(k_ws, kinv, b, n, current_stream());
     launch_ps_finalize_direct(
         mu_ws.data_ptr<float>(), dist_ws.data_ptr<float>(),
         quad.data_ptr<float>(), out.data_ptr<float>(), b,

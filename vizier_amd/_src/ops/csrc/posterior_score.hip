@@ -399,6 +399,125 @@ extern "C" void launch_ps_kvec_bf16(
 
 // Chunked quadform + finalize reuse the fp32 kernels below.
 
+// -- Large-N quadform: hand-written split-K kernel ----------------------
+//
+// quad[q] = k_q^T Kinv k_q for b <= 32 candidates at N ~ 10^4
+// (BASELINE config 4). rocBLAS's skinny SGEMM (25 x N x N) realizes
+// ~1.2 TB/s effective here (0.34 ms/iter measured); the HBM floor is
+// one full Kinv read = N^2 * 4 B ~ 50 us. This kernel is shaped for
+// that floor on the CDNA4 VECTOR ALU (gfx950 has no fp32 MFMA — guide
+// "assuming fp32 MFMA exists" pitfall):
+//
+//  - each THREAD owns one Kinv column j; for each row i the 256
+//    threads of a workgroup read kinv[i][j0..j0+255] -> fully
+//    coalesced row segments, Kinv read EXACTLY once grid-wide;
+//  - the 32 candidate values k[q][i] come from LDS at the same
+//    address for every lane -> conflict-free broadcast reads;
+//  - acc[32] registers/thread accumulate t_j[q] partials; the final
+//    k[q][j] weighting + block reduction writes one partial per
+//    (q, workgroup), summed by ps_finalize_kernel-style pass.
+//
+// Work split: grid = (JCHUNKS, ICHUNKS); JCHUNKS*ICHUNKS workgroups
+// fill the 256 CUs (XCD-aware linearization not needed: each block's
+// stream is long and bandwidth-shaped).
+
+#define QF_QMAX 32
+#define QF_ITILE 512
+
+extern "C" __global__ __launch_bounds__(BLOCK) void
+ps_quadform_big_kernel(const float* __restrict__ k_in,   // (B, N)
+                       const float* __restrict__ kinv,   // (N, N)
+                       float* __restrict__ part,  // (B, n_wgs)
+                       int b, int n, int ichunks) {
+  __shared__ float k_lds[QF_QMAX][QF_ITILE];
+  __shared__ float red_lds[QF_QMAX][BLOCK / WAVE_SIZE];
+  const int tid = threadIdx.x;
+  const int jc = blockIdx.x;          // j-chunk of 256 columns
+  const int ic = blockIdx.y;          // i-panel
+  const int n_wgs = gridDim.x * gridDim.y;
+  const int wg = jc * ichunks + ic;
+  const int j = jc * BLOCK + tid;     // this thread's column
+  const bool j_ok = j < n;
+  const long i0 = (long)ic * n / ichunks;
+  const long i1 = (long)(ic + 1) * n / ichunks;
+
+  float acc[QF_QMAX];
+#pragma unroll
+  for (int q = 0; q < QF_QMAX; ++q) acc[q] = 0.0f;
+
+  for (long t0 = i0; t0 < i1; t0 += QF_ITILE) {
+    const int tlen = (int)min((long)QF_ITILE, i1 - t0);
+    // Cooperative k staging: rows q < b from global, zeros above so
+    // the unrolled q-loop below needs no bound check.
+    for (int e = tid; e < QF_QMAX * QF_ITILE; e += BLOCK) {
+      const int q = e / QF_ITILE;
+      const int i = e % QF_ITILE;
+      k_lds[q][i] = (q < b && i < tlen)
+          ? k_in[(long)q * n + t0 + i] : 0.0f;
+    }
+    __syncthreads();
+    if (j_ok) {
+      for (int i = 0; i < tlen; ++i) {
+        const float kv = kinv[(t0 + i) * n + j];  // coalesced row seg
+#pragma unroll
+        for (int q = 0; q < QF_QMAX; ++q) {
+          acc[q] = fmaf(kv, k_lds[q][i], acc[q]); // LDS broadcast
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  // Weight by k[q][j] and reduce across the workgroup per q.
+  const int wave = tid / WAVE_SIZE;
+  const int lane = tid % WAVE_SIZE;
+#pragma unroll
+  for (int q = 0; q < QF_QMAX; ++q) {
+    float v = 0.0f;
+    if (j_ok && q < b) v = acc[q] * k_in[(long)q * n + j];
+    v = wave_reduce_sum(v);
+    if (lane == 0) red_lds[q][wave] = v;
+  }
+  __syncthreads();
+  if (tid < QF_QMAX) {
+    float s = 0.0f;
+#pragma unroll
+    for (int w = 0; w < BLOCK / WAVE_SIZE; ++w) s += red_lds[tid][w];
+    if (tid < b) part[(long)tid * n_wgs + wg] = s;
+  }
+}
+
+extern "C" __global__ void
+ps_reduce_parts_kernel(const float* __restrict__ part,  // (B, n_wgs)
+                       float* __restrict__ quad,        // (B,)
+                       int b, int n_wgs) {
+  const int q = blockIdx.x;
+  if (q >= b) return;
+  const int tid = threadIdx.x;
+  __shared__ float red[8];
+  float s = 0.0f;
+  for (int w = tid; w < n_wgs; w += blockDim.x) {
+    s += part[(long)q * n_wgs + w];
+  }
+  auto fsum = [](float a, float c) { return a + c; };
+  float total = block_reduce(s, red, fsum, 0.0f);
+  if (tid == 0) quad[q] = total;
+}
+
+extern "C" void launch_ps_quadform_big(
+    const float* k_ws, const float* kinv, float* part, float* quad,
+    int b, int n, hipStream_t stream) {
+  const int jchunks = (n + BLOCK - 1) / BLOCK;
+  // Fill the chip: >= 512 workgroups, each with a long i-stream.
+  int ichunks = (512 + jchunks - 1) / jchunks;
+  if (ichunks < 1) ichunks = 1;
+  hipLaunchKernelGGL(ps_quadform_big_kernel, dim3(jchunks, ichunks),
+                     dim3(BLOCK), 0, stream, k_ws, kinv, part, b, n,
+                     ichunks);
+  hipLaunchKernelGGL(ps_reduce_parts_kernel, dim3(b), dim3(256), 0,
+                     stream, part, quad, b, jchunks * ichunks);
+}
+
 // Finalize variant consuming a precomputed (b,) quadform directly.
 //
 // At huge N the per-candidate K^-1 streaming in ps_quadform_kernel
