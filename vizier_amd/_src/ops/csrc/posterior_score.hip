@@ -422,7 +422,10 @@ extern "C" void launch_ps_kvec_bf16(
 // stream is long and bandwidth-shaped).
 
 #define QF_QMAX 32
-#define QF_ITILE 512
+// 256-float tiles keep k_lds at 32 KB -> 4 workgroups/CU resident
+// (64 KB tiles measured 2 waves/SIMD and 440 GB/s; see
+// profiles/quadform_ab_r2.json).
+#define QF_ITILE 256
 
 extern "C" __global__ __launch_bounds__(BLOCK) void
 ps_quadform_big_kernel(const float* __restrict__ k_in,   // (B, N)
@@ -457,11 +460,32 @@ ps_quadform_big_kernel(const float* __restrict__ k_in,   // (B, N)
     }
     __syncthreads();
     if (j_ok) {
-      for (int i = 0; i < tlen; ++i) {
-        const float kv = kinv[(t0 + i) * n + j];  // coalesced row seg
+      // 4 rows per step: 4 independent global loads in flight per
+      // thread + ds_read_b128 broadcasts of the candidate block.
+      int i = 0;
+      for (; i + 4 <= tlen; i += 4) {
+        const long base = (t0 + i) * (long)n + j;
+        const float kv0 = kinv[base];
+        const float kv1 = kinv[base + n];
+        const float kv2 = kinv[base + 2L * n];
+        const float kv3 = kinv[base + 3L * n];
 #pragma unroll
         for (int q = 0; q < QF_QMAX; ++q) {
-          acc[q] = fmaf(kv, k_lds[q][i], acc[q]); // LDS broadcast
+          const float4 kq =
+              *reinterpret_cast<const float4*>(&k_lds[q][i]);
+          float a = acc[q];
+          a = fmaf(kv0, kq.x, a);
+          a = fmaf(kv1, kq.y, a);
+          a = fmaf(kv2, kq.z, a);
+          a = fmaf(kv3, kq.w, a);
+          acc[q] = a;
+        }
+      }
+      for (; i < tlen; ++i) {
+        const float kv = kinv[(t0 + i) * (long)n + j];
+#pragma unroll
+        for (int q = 0; q < QF_QMAX; ++q) {
+          acc[q] = fmaf(kv, k_lds[q][i], acc[q]);
         }
       }
     }
