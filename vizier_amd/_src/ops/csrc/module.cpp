@@ -1,7 +1,7 @@
 // Python bindings for the vizier_amd gfx950 kernels (torch extension).
 
 #include <ATen/cuda/CUDAContext.h>
-#include <torch/extension.h>\n#include <string>
+#include <torch/extension.h>\n#include <algorithm>\n#include <string>
 
 #include <hip/hip_runtime.h>
 
@@ -48,6 +48,12 @@ extern "C" void launch_ps_kvec(
     const float* alpha, const unsigned char* onehot, float* k_ws,
     float* mu_ws, float* dist_ws, int b, int n, int d, float amp2,
     hipStream_t stream);
+
+extern "C" void launch_ps_kvec_split(
+    const float* xq, const float* x, const float* inv_ls,
+    const float* alpha, const unsigned char* onehot, float* k_ws,
+    float* mu_part, float* dist_part, float* mu_ws, float* dist_ws,
+    int b, int n, int d, float amp2, int rchunks, hipStream_t stream);
 
 extern "C" void launch_ps_quadform_big(
     const float* k_ws, const float* kinv, float* part, float* quad,
@@ -330,12 +336,16 @@ torch::Tensor posterior_scores_chunked(
     return s ? atoi(s) : 4096;
   }();
   if (n >= gemm_n_threshold) {
-    launch_ps_kvec(
+    const int rchunks = std::max(1, 512 / std::max(b, 1));
+    auto mu_part = torch::empty({b, rchunks}, xq.options());
+    auto dist_part = torch::empty({b, rchunks}, xq.options());
+    launch_ps_kvec_split(
         xq.data_ptr<float>(), x.data_ptr<float>(),
         inv_ls.data_ptr<float>(), alpha.data_ptr<float>(),
         onehot.data_ptr<unsigned char>(), k_ws.data_ptr<float>(),
+        mu_part.data_ptr<float>(), dist_part.data_ptr<float>(),
         mu_ws.data_ptr<float>(), dist_ws.data_ptr<float>(), b, n, d,
-        (float)(amplitude * amplitude), current_stream());
+        (float)(amplitude * amplitude), rchunks, current_stream());
     auto quad = quadform_large_n(k_ws, kinv, b, n, current_stream());
     launch_ps_finalize_direct(
         mu_ws.data_ptr<float>(), dist_ws.data_ptr<float>(),

@@ -399,6 +399,96 @@ extern "C" void launch_ps_kvec_bf16(
 
 // Chunked quadform + finalize reuse the fp32 kernels below.
 
+// -- Large-N k-vector: row-split variant --------------------------------
+//
+// ps_kvec_kernel launches one workgroup per candidate (b=25 -> 10% of
+// the chip). At N ~ 10^4 that leaves the k/mu/dist pass latency-bound;
+// this variant splits the row range over gridDim.y chunks (b x RC
+// workgroups) writing per-chunk mu partial sums / dist partial mins,
+// reduced by ps_mu_dist_reduce_kernel.
+
+extern "C" __global__ __launch_bounds__(BLOCK) void
+ps_kvec_split_kernel(const float* __restrict__ xq,
+                     const float* __restrict__ x,
+                     const float* __restrict__ inv_ls,
+                     const float* __restrict__ alpha,
+                     const unsigned char* __restrict__ onehot,
+                     float* __restrict__ k_out,      // (B, N)
+                     float* __restrict__ mu_part,    // (B, RC)
+                     float* __restrict__ dist_part,  // (B, RC)
+                     int b, int n, int d, float amp2, int rchunks) {
+  __shared__ float red[8];
+  __shared__ float xq_lds[512];
+  const int q = blockIdx.x;
+  const int rc = blockIdx.y;
+  if (q >= b) return;
+  const int tid = threadIdx.x;
+  for (int j = tid; j < d; j += BLOCK) xq_lds[j] = xq[(long)q * d + j];
+  __syncthreads();
+  const int r0 = (int)((long)rc * n / rchunks);
+  const int r1 = (int)((long)(rc + 1) * n / rchunks);
+  float mu_acc = 0.0f;
+  float min_linf = INFINITY;
+  for (int row = r0 + tid; row < r1; row += BLOCK) {
+    const float* xr = x + (long)row * d;
+    float d2 = 0.0f, linf = 0.0f;
+    for (int j = 0; j < d; ++j) {
+      const float diff = xq_lds[j] - xr[j];
+      const float z = diff * inv_ls[j];
+      d2 = fmaf(z, z, d2);
+      if (!onehot[j]) linf = fmaxf(linf, fabsf(diff));
+    }
+    const float kv = amp2 * matern52_of_d2(d2);
+    k_out[(long)q * n + row] = kv;
+    mu_acc = fmaf(kv, alpha[row], mu_acc);
+    min_linf = fminf(min_linf, linf);
+  }
+  auto fsum = [](float a, float c) { return a + c; };
+  auto fmin_ = [](float a, float c) { return fminf(a, c); };
+  float mu = block_reduce(mu_acc, red, fsum, 0.0f);
+  if (tid == 0) mu_part[(long)q * rchunks + rc] = mu;
+  __syncthreads();
+  float dist = block_reduce(min_linf, red, fmin_, INFINITY);
+  if (tid == 0) dist_part[(long)q * rchunks + rc] = dist;
+}
+
+extern "C" __global__ void
+ps_mu_dist_reduce_kernel(const float* __restrict__ mu_part,
+                         const float* __restrict__ dist_part,
+                         float* __restrict__ mu_out,
+                         float* __restrict__ dist_out,
+                         int b, int rchunks) {
+  const int q = blockIdx.x;
+  if (q >= b) return;
+  const int tid = threadIdx.x;
+  __shared__ float red[8];
+  float s = 0.0f, m = INFINITY;
+  for (int c = tid; c < rchunks; c += blockDim.x) {
+    s += mu_part[(long)q * rchunks + c];
+    m = fminf(m, dist_part[(long)q * rchunks + c]);
+  }
+  auto fsum = [](float a, float c) { return a + c; };
+  auto fmin_ = [](float a, float c) { return fminf(a, c); };
+  float mu = block_reduce(s, red, fsum, 0.0f);
+  if (tid == 0) mu_out[q] = mu;
+  __syncthreads();
+  float dist = block_reduce(m, red, fmin_, INFINITY);
+  if (tid == 0) dist_out[q] = dist;
+}
+
+extern "C" void launch_ps_kvec_split(
+    const float* xq, const float* x, const float* inv_ls,
+    const float* alpha, const unsigned char* onehot, float* k_ws,
+    float* mu_part, float* dist_part, float* mu_ws, float* dist_ws,
+    int b, int n, int d, float amp2, int rchunks, hipStream_t stream) {
+  hipLaunchKernelGGL(ps_kvec_split_kernel, dim3(b, rchunks), dim3(BLOCK),
+                     0, stream, xq, x, inv_ls, alpha, onehot, k_ws,
+                     mu_part, dist_part, b, n, d, amp2, rchunks);
+  hipLaunchKernelGGL(ps_mu_dist_reduce_kernel, dim3(b), dim3(256), 0,
+                     stream, mu_part, dist_part, mu_ws, dist_ws, b,
+                     rchunks);
+}
+
 // -- Large-N quadform: hand-written split-K kernel ----------------------
 //
 // quad[q] = k_q^T Kinv k_q for b <= 32 candidates at N ~ 10^4
