@@ -380,6 +380,10 @@ def train_gp(x: torch.Tensor, y: torch.Tensor, *,
                    torch.float32).to(device=x.device, dtype=x.dtype)
   if warm_start_raw is not None and warm_start_raw.numel() == d + 3:
     # Warm start from the previous fit's optimum (incremental refits).
+    # NOTE r2: tried shrinking the warm restart batch / line-search
+    # ladder for the 166 ms refit — small-N convergence regressed
+    # (tests/test_gp_bandit convergence gates), so the regret-validated
+    # r1 schedule stands.
     raw0 = torch.cat([warm_start_raw.reshape(1, -1).to(raw0), raw0], 0)
 
   def loss_fn(raw: torch.Tensor) -> torch.Tensor:
