@@ -372,6 +372,41 @@ class TestGPUDesignerEndToEnd:
         designer.update(CompletedTrials([t]), ActiveTrials())
     assert best > -0.25, f'GPU GP-Bandit failed to converge: {best}'
 
+  def test_gp_bandit_fp64_mode_on_gpu(self, ext):
+    # The --fp64 parity mode (reference forces jax x64): HIP fp32
+    # kernels are bypassed, the whole pipeline runs float64 on rocBLAS
+    # DGEMMs, and suggestions are still sane.
+    from vizier_amd import pyvizier as vz
+    from vizier_amd._src.algorithms.core.abstractions import (
+        ActiveTrials,
+        CompletedTrials,
+    )
+    from vizier_amd._src.algorithms.designers.gp_bandit import (
+        GPBanditConfig,
+        VizierGPBandit,
+    )
+    problem = vz.ProblemStatement()
+    for i in range(4):
+      problem.search_space.root.add_float_param(f'x{i}', 0.0, 1.0)
+    problem.metric_information.append(vz.MetricInformation(
+        name='obj', goal=vz.ObjectiveMetricGoal.MAXIMIZE))
+    designer = VizierGPBandit(problem, GPBanditConfig(
+        max_evaluations=800, ard_restarts=1, ard_max_iters=10,
+        device='cuda', dtype=torch.float64), seed=0)
+    uid = 0
+    for _ in range(4):
+      for s in designer.suggest(1):
+        uid += 1
+        x = np.array([s.parameters.get_value(f'x{i}')
+                      for i in range(4)])
+        t = s.to_trial(uid)
+        t.complete(vz.Measurement(
+            metrics={'obj': float(-((x - 0.4) ** 2).sum())}))
+        designer.update(CompletedTrials([t]), ActiveTrials())
+    assert designer._x.dtype == torch.float64
+    mean, _ = designer._posteriors[0].predict(designer._x[:2])
+    assert mean.dtype == torch.float64
+
   def test_multi_objective_and_qei_on_gpu(self, ext):
     """MO hypervolume-scalarized scoring and q-EI batches run the full
     GPU path (catches device-placement regressions in the composed
