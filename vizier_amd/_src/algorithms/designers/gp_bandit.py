@@ -345,7 +345,10 @@ class VizierGPBandit(Designer, Predictor):
           scores = trust_region.apply(flat.reshape(dense.shape)[:, 0, :],
                                       scores)
         return scores
-      score_fn.graph_safe = False  # multi-posterior path uses rocBLAS
+      # Pure tensor ops (HIP gram + rocBLAS GEMMs + HV scalarizer):
+      # hipGraph-capturable; optimize() auto-falls-back to eager if
+      # capture fails on a given ROCm build.
+      score_fn.graph_safe = True
       return score_fn, 1
 
     posterior = self._posteriors[0]
@@ -380,7 +383,7 @@ class VizierGPBandit(Designer, Predictor):
         if trust_region is not None:
           scores = trust_region.apply(dense, scores)
         return scores
-      score_fn.graph_safe = False  # multi-GP chain uses rocBLAS
+      score_fn.graph_safe = True   # tensor-only stacked-GP chain
       return score_fn, 1
 
     plain_gp = isinstance(posterior, gp_model.GPPosterior)
@@ -402,7 +405,7 @@ class VizierGPBandit(Designer, Predictor):
         if trust_region is not None:
           scores = trust_region.apply(dense, scores)
         return scores
-      score_fn.graph_safe = False  # composed predict uses rocBLAS
+      score_fn.graph_safe = True   # tensor-only composed predict
       return score_fn, 1
 
     if cfg.acquisition == 'qei' and count > 1:
