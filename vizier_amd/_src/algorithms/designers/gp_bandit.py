@@ -345,10 +345,11 @@ class VizierGPBandit(Designer, Predictor):
           scores = trust_region.apply(flat.reshape(dense.shape)[:, 0, :],
                                       scores)
         return scores
-      # Pure tensor ops (HIP gram + rocBLAS GEMMs + HV scalarizer):
-      # hipGraph-capturable; optimize() auto-falls-back to eager if
-      # capture fails on a given ROCm build.
-      score_fn.graph_safe = True
+      # Tried capture in r2: hipErrorStreamCaptureUnsupported on this
+      # ROCm build (gpurun_out/c5_graph.log) — some op in the MO chain
+      # is not capture-safe, so stay eager instead of paying an aborted
+      # capture attempt every suggest.
+      score_fn.graph_safe = False
       return score_fn, 1
 
     posterior = self._posteriors[0]
@@ -383,7 +384,7 @@ class VizierGPBandit(Designer, Predictor):
         if trust_region is not None:
           scores = trust_region.apply(dense, scores)
         return scores
-      score_fn.graph_safe = True   # tensor-only stacked-GP chain
+      score_fn.graph_safe = False  # capture unsupported (see MO note)
       return score_fn, 1
 
     plain_gp = isinstance(posterior, gp_model.GPPosterior)
@@ -405,7 +406,7 @@ class VizierGPBandit(Designer, Predictor):
         if trust_region is not None:
           scores = trust_region.apply(dense, scores)
         return scores
-      score_fn.graph_safe = True   # tensor-only composed predict
+      score_fn.graph_safe = False  # capture unsupported (see MO note)
       return score_fn, 1
 
     if cfg.acquisition == 'qei' and count > 1:

@@ -383,7 +383,7 @@ class VizierGPUCBPEBandit(Designer):
           scores = scores + torch.where(
               dist > r, -1e4 - dist, torch.zeros_like(dist)).sum(-1)
       return scores
-    score_fn.graph_safe = True   # tensor-only (batched cholesky_ex)
+    score_fn.graph_safe = False  # rocSOLVER cholesky: no capture
 
     factory = VectorizedOptimizerFactory(
         eagle_config=EagleStrategyConfig(),
@@ -442,7 +442,7 @@ class VizierGPUCBPEBandit(Designer):
           if trust_region is not None:
             scores = trust_region.apply(xs, scores)
           return scores
-      score_fn.graph_safe = True   # tensor-only multi-GP loop
+      score_fn.graph_safe = False  # capture unsupported on ROCm 7.2
     elif use_ucb:
       scoring = acq_lib.ScoringFunction(
           posterior, acq_lib.UCB(cfg.ucb_coefficient), trust_region)
@@ -461,7 +461,7 @@ class VizierGPUCBPEBandit(Designer):
       else:
         # Pending points anchor the trust region elsewhere: composed
         # rocBLAS path, not capturable on this ROCm build.
-        score_fn.graph_safe = True   # tensor-only composed path
+        score_fn.graph_safe = False  # capture unsupported on ROCm 7.2
     else:
       # Promising-region threshold: predicted mean at the observed point
       # with the highest UCB (gp_ucb_pe.py:175-205).
@@ -481,7 +481,7 @@ class VizierGPUCBPEBandit(Designer):
         if trust_region is not None:
           scores = trust_region.apply(xs, scores)
         return scores
-      score_fn.graph_safe = True   # tensor-only two-posterior path
+      score_fn.graph_safe = False  # capture unsupported on ROCm 7.2
 
     factory = VectorizedOptimizerFactory(
         eagle_config=EagleStrategyConfig(),
