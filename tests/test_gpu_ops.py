@@ -407,6 +407,73 @@ class TestGPUDesignerEndToEnd:
     mean, _ = designer._posteriors[0].predict(designer._x[:2])
     assert mean.dtype == torch.float64
 
+  def test_fused_mo_scorer_matches_eager_chain(self, ext):
+    """The fused MO path (posterior_mean_std + hv_scalarize_tr) must
+    reproduce the eager python scalarization chain."""
+    from vizier_amd import pyvizier as vz
+    from vizier_amd._src.algorithms.core.abstractions import (
+        ActiveTrials,
+        CompletedTrials,
+    )
+    from vizier_amd._src.algorithms.designers.gp_bandit import (
+        GPBanditConfig,
+        VizierGPBandit,
+    )
+    from vizier_amd._src.algorithms.optimizers.eagle import CandidateBatch
+    rng = np.random.default_rng(3)
+    problem = vz.ProblemStatement()
+    for i in range(4):
+      problem.search_space.root.add_float_param(f'x{i}', 0.0, 1.0)
+    problem.metric_information = vz.MetricsConfig([
+        vz.MetricInformation(name=n, goal=vz.ObjectiveMetricGoal.MAXIMIZE)
+        for n in ('f1', 'f2')])
+    designer = VizierGPBandit(problem, GPBanditConfig(
+        max_evaluations=500, ard_restarts=1, ard_max_iters=10,
+        device='cuda'), seed=0)
+    trials = []
+    for uid in range(1, 13):
+      x = rng.uniform(0, 1, 4)
+      t = vz.Trial({f'x{i}': float(x[i]) for i in range(4)}, id=uid)
+      t.complete(vz.Measurement(metrics={
+          'f1': float(-((x - 0.2) ** 2).sum()),
+          'f2': float(-((x - 0.8) ** 2).sum())}))
+      trials.append(t)
+    designer.update(CompletedTrials(trials), ActiveTrials())
+    designer._fit()
+    fused_fn, _ = designer._score_factory(1)
+    assert getattr(fused_fn, 'graph_safe', False), \
+        'fused MO path did not engage'
+    # Build the eager chain by disabling the fused conditions.
+    designer._config.scorer_gram_dtype = 'fp32'
+    mt_save = designer._mt_posterior
+
+    class _Blocker:
+      pass
+    designer._mt_posterior = None
+    kinv_save = designer._posteriors[0].K_inv
+    xs = torch.rand(32, 1, 4, generator=torch.Generator().manual_seed(1)
+                    ).cuda()
+    batch = CandidateBatch(xs, torch.zeros(32, 1, 0, dtype=torch.long,
+                                           device='cuda'))
+    got = fused_fn(batch)
+    # Eager oracle: per-metric predict + python scalarizer + TR.
+    import vizier_amd._src.gp.acquisitions as acq_lib
+    tr = designer._make_trust_region()
+    scal = acq_lib.create_hv_scalarization(
+        designer._config.num_scalarizations, 2, seed=designer._seed,
+        reference_point=acq_lib.get_reference_point(
+            designer._warped_labels,
+            scale=designer._config.ref_scaling))
+    flat = xs[:, 0, :]
+    ys = torch.stack(
+        [m + 1.8 * s for m, s in
+         (p.predict(flat) for p in designer._posteriors)], dim=-1)
+    want = scal(ys).mean(dim=0)
+    want = tr.apply(flat, want)
+    designer._mt_posterior = mt_save
+    assert torch.allclose(got, want, atol=2e-3), \
+        float((got - want).abs().max())
+
   def test_multi_objective_and_qei_on_gpu(self, ext):
     """MO hypervolume-scalarized scoring and q-EI batches run the full
     GPU path (catches device-placement regressions in the composed

@@ -59,6 +59,25 @@ extern "C" void launch_ps_quadform_big(
     const float* k_ws, const float* kinv, float* part, float* quad,
     int b, int n, hipStream_t stream);
 
+extern "C" void launch_ps_finalize_meanstd(
+    const float* mu_ws, const float* var_ws, float* mean_out,
+    float* sd_out, int b, float amp2, float mean_c, int nchunk,
+    hipStream_t stream);
+
+extern "C" void launch_ps_finalize_meanstd_direct(
+    const float* mu_ws, const float* quad, float* mean_out,
+    float* sd_out, int b, float amp2, float mean_c,
+    hipStream_t stream);
+
+extern "C" void launch_hv_scalarize_tr(
+    const float* means, const float* sds, const float* weights,
+    const float* ref, const float* dist, float* out, int b, int m,
+    int s, float coef, float tr_radius, hipStream_t stream);
+
+extern "C" void launch_ps_quadform_kernel_only(
+    const float* k_ws, const float* kinv, float* var_ws, int b, int n,
+    hipStream_t stream);
+
 extern "C" void launch_ps_finalize_direct(
     const float* mu_ws, const float* dist_ws, const float* quad,
     float* out, int b, float amp2, float mean_c, int acq, float coef,
@@ -428,6 +447,96 @@ torch::Tensor posterior_scores_bf16(
   return out;
 }
 
+std::vector<torch::Tensor> posterior_mean_std(
+    torch::Tensor xq, torch::Tensor x, torch::Tensor lengthscales,
+    double amplitude, double mean_c, torch::Tensor alpha,
+    torch::Tensor kinv, torch::Tensor onehot) {
+  // (mean, sd, min-Linf dist) for one GP over a candidate batch in
+  // 3-4 launches — the per-metric piece of the fused MO scorer.
+  xq = check_f32(xq, "xq");
+  x = check_f32(x, "x");
+  lengthscales = check_f32(lengthscales, "lengthscales");
+  alpha = check_f32(alpha, "alpha");
+  kinv = check_f32(kinv, "kinv");
+  onehot = onehot.contiguous();
+  const int b = xq.size(0), d = xq.size(1), n = x.size(0);
+  TORCH_CHECK(d <= 512, "posterior_mean_std supports D <= 512");
+  auto inv_ls = 1.0f / lengthscales;
+  auto k_ws = torch::empty({b, n}, xq.options());
+  auto mu_ws = torch::empty({b}, xq.options());
+  auto dist_ws = torch::empty({b}, xq.options());
+  auto mean = torch::empty({b}, xq.options());
+  auto sd = torch::empty({b}, xq.options());
+  const float amp2 = (float)(amplitude * amplitude);
+  static const int gemm_n_threshold = []() {
+    const char* s = getenv("VIZIER_AMD_PS_GEMM_N");
+    return s ? atoi(s) : 4096;
+  }();
+  if (n >= gemm_n_threshold) {
+    const int rchunks = std::max(1, 512 / std::max(b, 1));
+    auto mu_part = torch::empty({b, rchunks}, xq.options());
+    auto dist_part = torch::empty({b, rchunks}, xq.options());
+    launch_ps_kvec_split(
+        xq.data_ptr<float>(), x.data_ptr<float>(),
+        inv_ls.data_ptr<float>(), alpha.data_ptr<float>(),
+        onehot.data_ptr<unsigned char>(), k_ws.data_ptr<float>(),
+        mu_part.data_ptr<float>(), dist_part.data_ptr<float>(),
+        mu_ws.data_ptr<float>(), dist_ws.data_ptr<float>(), b, n, d,
+        amp2, rchunks, current_stream());
+    auto quad = quadform_large_n(k_ws, kinv, b, n, current_stream());
+    launch_ps_finalize_meanstd_direct(
+        mu_ws.data_ptr<float>(), quad.data_ptr<float>(),
+        mean.data_ptr<float>(), sd.data_ptr<float>(), b, amp2,
+        (float)mean_c, current_stream());
+    return {mean, sd, dist_ws};
+  }
+  auto var_ws = torch::empty({b, 10}, xq.options());
+  launch_ps_kvec(
+      xq.data_ptr<float>(), x.data_ptr<float>(),
+      inv_ls.data_ptr<float>(), alpha.data_ptr<float>(),
+      onehot.data_ptr<unsigned char>(), k_ws.data_ptr<float>(),
+      mu_ws.data_ptr<float>(), dist_ws.data_ptr<float>(), b, n, d,
+      amp2, current_stream());
+  launch_ps_quadform_kernel_only(
+      k_ws.data_ptr<float>(), kinv.data_ptr<float>(),
+      var_ws.data_ptr<float>(), b, n, current_stream());
+  launch_ps_finalize_meanstd(
+      mu_ws.data_ptr<float>(), var_ws.data_ptr<float>(),
+      mean.data_ptr<float>(), sd.data_ptr<float>(), b, amp2,
+      (float)mean_c, 10, current_stream());
+  return {mean, sd, dist_ws};
+}
+
+torch::Tensor hv_scalarize_tr(
+    torch::Tensor means, torch::Tensor sds, torch::Tensor weights,
+    c10::optional<torch::Tensor> ref, c10::optional<torch::Tensor> dist,
+    double coef, double tr_radius) {
+  means = check_f32(means, "means");    // (M, B)
+  sds = check_f32(sds, "sds");
+  weights = check_f32(weights, "weights");  // (S, M)
+  const int m = means.size(0), b = means.size(1);
+  const int s = weights.size(0);
+  TORCH_CHECK(weights.size(1) == m, "weights/means metric mismatch");
+  torch::Tensor ref_t, dist_t;
+  const float* ref_p = nullptr;
+  const float* dist_p = nullptr;
+  if (ref.has_value()) {
+    ref_t = check_f32(*ref, "ref");
+    ref_p = ref_t.data_ptr<float>();
+  }
+  if (dist.has_value()) {
+    dist_t = check_f32(*dist, "dist");
+    dist_p = dist_t.data_ptr<float>();
+  }
+  auto out = torch::empty({b}, means.options());
+  launch_hv_scalarize_tr(
+      means.data_ptr<float>(), sds.data_ptr<float>(),
+      weights.data_ptr<float>(), ref_p, dist_p,
+      out.data_ptr<float>(), b, m, s, (float)coef, (float)tr_radius,
+      current_stream());
+  return out;
+}
+
 std::vector<torch::Tensor> eagle_suggest(
     torch::Tensor pool_cont, torch::Tensor pool_cat, torch::Tensor rewards,
     torch::Tensor perturbations, torch::Tensor cat_sizes,
@@ -559,6 +668,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "128x128 LDS-tiled fp8 e4m3 MFMA Matern-5/2 Gram (gfx950)");
   m.def("posterior_scores", &posterior_scores,
         "Fused GP posterior + acquisition + trust region (gfx950)");
+  m.def("posterior_mean_std", &posterior_mean_std,
+        "Fused (mean, sd, dist) for one GP over candidates (gfx950)");
+  m.def("hv_scalarize_tr", &hv_scalarize_tr,
+        "Hypervolume scalarization + trust region in one launch");
   m.def("posterior_scores_bf16", &posterior_scores_bf16,
         "Chunked scorer with cached-bf16 candidate grams (gfx950)");
   m.def("posterior_scores_chunked", &posterior_scores_chunked,

@@ -489,6 +489,109 @@ extern "C" void launch_ps_kvec_split(
                      rchunks);
 }
 
+// -- Multi-objective fused helpers --------------------------------------
+//
+// The MO (config 5) scorer was ~25 eager launches per Eagle iteration
+// (per-metric predicts + the HV-scalarization chain) and the op mix is
+// not stream-capturable on ROCm 7.2. These two kernels collapse it to
+// 3 launches per metric (k-vec + quadform + mean/std finalize) plus
+// ONE scalarize+trust-region kernel.
+
+extern "C" __global__ void
+ps_finalize_meanstd_kernel(const float* __restrict__ mu_in,
+                           const float* __restrict__ var_part,
+                           float* __restrict__ mean_out,
+                           float* __restrict__ sd_out, int b,
+                           float amp2, float mean_c, int nchunk) {
+  const int q = blockIdx.x * blockDim.x + threadIdx.x;
+  if (q >= b) return;
+  float var = 0.0f;
+  for (int c = 0; c < nchunk; ++c) var += var_part[q * nchunk + c];
+  var = fmaxf(amp2 - var, 1e-12f);
+  mean_out[q] = mu_in[q] + mean_c;
+  sd_out[q] = sqrtf(var);
+}
+
+extern "C" __global__ void
+ps_finalize_meanstd_direct_kernel(const float* __restrict__ mu_in,
+                                  const float* __restrict__ quad_in,
+                                  float* __restrict__ mean_out,
+                                  float* __restrict__ sd_out, int b,
+                                  float amp2, float mean_c) {
+  const int q = blockIdx.x * blockDim.x + threadIdx.x;
+  if (q >= b) return;
+  mean_out[q] = mu_in[q] + mean_c;
+  sd_out[q] = sqrtf(fmaxf(amp2 - quad_in[q], 1e-12f));
+}
+
+// score[q] = mean_s min_m (ucb[m,q] - ref[m]) / w[s,m], then the
+// trust-region cutoff using the precomputed min-L-inf distance.
+// means/sds are (M, B) row-major; weights (S, M).
+extern "C" __global__ __launch_bounds__(BLOCK) void
+hv_scalarize_tr_kernel(const float* __restrict__ means,
+                       const float* __restrict__ sds,
+                       const float* __restrict__ weights,
+                       const float* __restrict__ ref,   // (M,) or NULL
+                       const float* __restrict__ dist,  // (B,) or NULL
+                       float* __restrict__ out, int b, int m, int s,
+                       float coef, float tr_radius) {
+  __shared__ float red[8];
+  const int q = blockIdx.x;
+  if (q >= b) return;
+  const int tid = threadIdx.x;
+  float acc = 0.0f;
+  for (int si = tid; si < s; si += BLOCK) {
+    float v = INFINITY;
+    for (int mi = 0; mi < m; ++mi) {
+      float y = means[mi * b + q] + coef * sds[mi * b + q];
+      if (ref) y -= ref[mi];
+      v = fminf(v, y / weights[si * m + mi]);
+    }
+    acc += v;
+  }
+  auto fsum = [](float a, float c) { return a + c; };
+  float total = block_reduce(acc, red, fsum, 0.0f);
+  if (tid == 0) {
+    float score = total / (float)s;
+    if (dist && tr_radius > 0.0f && tr_radius <= 0.5f &&
+        dist[q] > tr_radius) {
+      score = -1e4f - dist[q];
+    }
+    out[q] = score;
+  }
+}
+
+extern "C" void launch_ps_finalize_meanstd(
+    const float* mu_ws, const float* var_ws, float* mean_out,
+    float* sd_out, int b, float amp2, float mean_c, int nchunk,
+    hipStream_t stream) {
+  const int fin_block = 256;
+  hipLaunchKernelGGL(ps_finalize_meanstd_kernel,
+                     dim3((b + fin_block - 1) / fin_block),
+                     dim3(fin_block), 0, stream, mu_ws, var_ws,
+                     mean_out, sd_out, b, amp2, mean_c, nchunk);
+}
+
+extern "C" void launch_ps_finalize_meanstd_direct(
+    const float* mu_ws, const float* quad, float* mean_out,
+    float* sd_out, int b, float amp2, float mean_c,
+    hipStream_t stream) {
+  const int fin_block = 256;
+  hipLaunchKernelGGL(ps_finalize_meanstd_direct_kernel,
+                     dim3((b + fin_block - 1) / fin_block),
+                     dim3(fin_block), 0, stream, mu_ws, quad,
+                     mean_out, sd_out, b, amp2, mean_c);
+}
+
+extern "C" void launch_hv_scalarize_tr(
+    const float* means, const float* sds, const float* weights,
+    const float* ref, const float* dist, float* out, int b, int m,
+    int s, float coef, float tr_radius, hipStream_t stream) {
+  hipLaunchKernelGGL(hv_scalarize_tr_kernel, dim3(b), dim3(BLOCK), 0,
+                     stream, means, sds, weights, ref, dist, out, b, m,
+                     s, coef, tr_radius);
+}
+
 // -- Large-N quadform: hand-written split-K kernel ----------------------
 //
 // quad[q] = k_q^T Kinv k_q for b <= 32 candidates at N ~ 10^4
@@ -699,6 +802,13 @@ extern "C" void launch_ps_finalize_direct(
                      dim3(fin_block), 0, stream, mu_ws, dist_ws, quad,
                      out, b, amp2, mean_c, acq, coef, best_value,
                      tr_radius);
+}
+
+extern "C" void launch_ps_quadform_kernel_only(
+    const float* k_ws, const float* kinv, float* var_ws, int b, int n,
+    hipStream_t stream) {
+  hipLaunchKernelGGL(ps_quadform_kernel, dim3(b, NCHUNK), dim3(BLOCK), 0,
+                     stream, k_ws, kinv, var_ws, b, n);
 }
 
 extern "C" void launch_ps_quadform_finalize(
