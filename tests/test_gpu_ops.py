@@ -613,6 +613,23 @@ class TestGramFp8MFMA:
     # e4m3 has ~2 significant digits; the Gram is O(1).
     assert float(err) < 0.15, f'max err {err}'
 
+  def test_fp8_cached_gram_matches_per_call(self, ext):
+    # Fp8GramCache (operand cache, deterministic unit-box scale, no
+    # per-call host sync) vs the per-call conversion path vs fp32 —
+    # all three agree to e4m3 tolerance.
+    from vizier_amd._src.gp import acquisitions as acq_lib
+    g = torch.Generator().manual_seed(9)
+    x = torch.rand(200, 12, generator=g).cuda()
+    xq = torch.rand(25, 12, generator=g).cuda()
+    ls = (torch.rand(12, generator=g) * 0.8 + 0.4).cuda()
+    cache = acq_lib.Fp8GramCache(x, ls, amplitude=1.3)
+    got = cache.gram(xq)
+    percall = ext.gram_matern52_fp8(xq, x, ls, 1.3)
+    fp32 = ext.gram_matern52(xq, x, ls, 1.3)
+    assert float((got - fp32).abs().max()) < 0.15
+    assert float((percall - fp32).abs().max()) < 0.15
+    assert float((got - percall).abs().max()) < 0.2
+
   def test_fp8_diag_near_exact(self, ext):
     x = torch.rand(64, 16).cuda()
     ls = torch.full((16,), 0.5).cuda()

@@ -325,6 +325,35 @@ torch::Tensor quadform_large_n(const torch::Tensor& k_ws,
 
 }  // namespace
 
+torch::Tensor gram_matern52_fp8_pre(
+    torch::Tensor z1q, torch::Tensor z2q, torch::Tensor n1,
+    torch::Tensor n2, double amplitude, double scale) {
+  // Cross-gram from PRE-QUANTIZED fp8 operands: the per-call
+  // host-side range scan (.item() sync) and conversions of the
+  // training side are hoisted out by the caller (once per suggest),
+  // so this is one launch and stream-capture-safe.
+  TORCH_CHECK(z1q.scalar_type() == torch::kFloat8_e4m3fn &&
+              z2q.scalar_type() == torch::kFloat8_e4m3fn,
+              "z1q/z2q must be float8_e4m3fn");
+  z1q = z1q.contiguous();
+  z2q = z2q.contiguous();
+  n1 = check_f32(n1, "n1");
+  n2 = check_f32(n2, "n2");
+  const int n = z1q.size(0), m = z2q.size(0), dp = z1q.size(1);
+  TORCH_CHECK(z2q.size(1) == dp && dp % 32 == 0, "dp mismatch");
+  auto out = torch::empty({n, m},
+                          n1.options().dtype(torch::kFloat32));
+  const bool tiled = n >= 512 && m >= 512;
+  auto launch = tiled ? launch_gram_matern52_fp8_tiled
+                      : launch_gram_matern52_fp8;
+  launch((const unsigned char*)z1q.data_ptr(),
+         (const unsigned char*)z2q.data_ptr(), n1.data_ptr<float>(),
+         n2.data_ptr<float>(), out.data_ptr<float>(), n, m, dp,
+         (float)(amplitude * amplitude), (float)(scale * scale),
+         current_stream());
+  return out;
+}
+
 torch::Tensor posterior_scores_chunked(
     torch::Tensor xq, torch::Tensor x, torch::Tensor lengthscales,
     double amplitude, double mean_c, torch::Tensor alpha,
@@ -668,6 +697,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "128x128 LDS-tiled fp8 e4m3 MFMA Matern-5/2 Gram (gfx950)");
   m.def("posterior_scores", &posterior_scores,
         "Fused GP posterior + acquisition + trust region (gfx950)");
+  m.def("gram_matern52_fp8_pre", &gram_matern52_fp8_pre,
+        "fp8 cross-gram from pre-quantized operands (gfx950)");
   m.def("posterior_mean_std", &posterior_mean_std,
         "Fused (mean, sd, dist) for one GP over candidates (gfx950)");
   m.def("hv_scalarize_tr", &hv_scalarize_tr,
