@@ -138,7 +138,8 @@ class VectorizedOptimizer:
     return (scoring._acq_name is not None and scoring._tr_anchored and
             getattr(scoring, 'gram_dtype', 'fp32') == 'fp32' and
             post.K_inv is not None and post.x.is_cuda and
-            strategy.pool_size <= 128 and post.x.shape[0] <= 8192)
+            strategy.pool_size <= 128 and strategy.batch_size <= 32
+            and post.x.shape[0] <= 8192)
 
   def _optimize_megakernel(self, score_fn: ScoreFn, count: int, state,
                            iterations: int) -> VectorizedStrategyResults:
@@ -173,7 +174,11 @@ class VectorizedOptimizer:
       k_ws = _torch.empty(b, n, dtype=_torch.float32, device=dev)
       mu_ws = _torch.empty(b, dtype=_torch.float32, device=dev)
       dist_ws = _torch.empty(b, dtype=_torch.float32, device=dev)
-      var_ws = _torch.empty(b, 10, dtype=_torch.float32, device=dev)
+      tiles_n = (n + 63) // 64
+      # (B, T+1): T tile partials + the reduced quadform per candidate
+      # (the megakernel's phase B/B2 layout; see eagle_sweep.hip).
+      var_ws = _torch.empty(b, tiles_n * tiles_n + 1,
+                            dtype=_torch.float32, device=dev)
       barrier_buf = _torch.zeros(2, dtype=_torch.int32, device=dev)
       from vizier_amd._src.ops import dispatch as ops
       amp2 = scoring._amp * scoring._amp

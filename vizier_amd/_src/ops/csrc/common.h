@@ -91,3 +91,61 @@ __device__ __forceinline__ float vz_normal_cdf(float z) {
 __device__ __forceinline__ float vz_normal_pdf(float z) {
   return 0.3989422804014327f * __expf(-0.5f * z * z);
 }
+
+// ---- Shared 64x64-tile K^-1 quadform --------------------------------------
+//
+// partial[q] contribution of tile (ti,tj) to quad[q] = k_q^T Kinv k_q,
+// computed for ALL candidates at once so Kinv is read ONCE per scorer
+// call. The per-candidate streaming variant reads Kinv b times per
+// iteration (100 MB at N=1000, B=25) and measured 32.6 us of a 58 us
+// Eagle iteration (profiles/sweep_kernels_r2.txt). Requires b <= 32
+// and blockDim.x == 256. Used VERBATIM (same float-op order) by both
+// the standalone chunked scorer and the persistent sweep megakernel so
+// the two paths stay bit-identical.
+#define VZ_QF_TILE 64
+#define VZ_QF_QMAX 32
+
+template <typename LoadK, typename StoreP>
+__device__ __forceinline__ void vz_quadform_tile(
+    const float* __restrict__ kinv, int b, int n, int tile, int tiles_n,
+    float* k_i_lds, float* k_j_lds, LoadK loadk, StoreP store) {
+  const int tid = threadIdx.x;
+  const int ti = tile / tiles_n, tj = tile % tiles_n;
+  const int i0 = ti * VZ_QF_TILE, j0 = tj * VZ_QF_TILE;
+  const int ilen = min(VZ_QF_TILE, n - i0);
+  const int jlen = min(VZ_QF_TILE, n - j0);
+  for (int e = tid; e < VZ_QF_QMAX * VZ_QF_TILE; e += 256) {
+    const int q = e / VZ_QF_TILE, c = e % VZ_QF_TILE;
+    k_i_lds[e] = (q < b && c < ilen) ? loadk((long)q * n + i0 + c)
+                                     : 0.0f;
+    k_j_lds[e] = (q < b && c < jlen) ? loadk((long)q * n + j0 + c)
+                                     : 0.0f;
+  }
+  __syncthreads();
+  // Wave w owns candidates [8w, 8w+8); lane = tile column j. Kinv rows
+  // are read coalesced across lanes (the 4 waves share each row
+  // segment through L2).
+  const int wave = tid / WAVE_SIZE;
+  const int lane = tid % WAVE_SIZE;
+  float acc[8];
+#pragma unroll
+  for (int qq = 0; qq < 8; ++qq) acc[qq] = 0.0f;
+  if (lane < jlen) {
+    for (int i = 0; i < ilen; ++i) {
+      const float kv = kinv[(long)(i0 + i) * n + j0 + lane];
+#pragma unroll
+      for (int qq = 0; qq < 8; ++qq) {
+        acc[qq] = fmaf(k_i_lds[(wave * 8 + qq) * VZ_QF_TILE + i], kv,
+                       acc[qq]);
+      }
+    }
+  }
+#pragma unroll
+  for (int qq = 0; qq < 8; ++qq) {
+    const int q = wave * 8 + qq;
+    float v = acc[qq] * k_j_lds[q * VZ_QF_TILE + lane];
+    v = wave_reduce_sum(v);
+    if (lane == 0 && q < b) store(q, v);
+  }
+  __syncthreads();
+}

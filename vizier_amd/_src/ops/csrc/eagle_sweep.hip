@@ -271,46 +271,43 @@ eagle_sweep_kernel(
     }
     grid_sync(barrier_buf);
 
-    // ---- Phase B: K^-1 quadform (wave-per-row, NCHUNK partition) ----
+    // ---- Phase B: K^-1 quadform (64x64 tiles, Kinv read ONCE) ----
+    // Shared template with the standalone chunked scorer
+    // (vz_quadform_tile in common.h): identical float-op order keeps
+    // the megakernel bit-identical to the hipGraph path. var_ws layout
+    // is (B, T+1): T tile partials + the reduced quadform.
     {
-      const int wave = tid / WAVE_SIZE;
-      const int lane = tid % WAVE_SIZE;
-      const int waves = BLOCK / WAVE_SIZE;
-      const int n4 = n / 4;
-      for (int pair = blockIdx.x; pair < batch_size * SWEEP_NCHUNK;
-           pair += G) {
-        const int q = pair % batch_size;
-        const int chunk = pair / batch_size;
-        const int j0 = (int)((long)chunk * n / SWEEP_NCHUNK);
-        const int j1 = (int)((long)(chunk + 1) * n / SWEEP_NCHUNK);
-        // Stage candidate q's k-vector into LDS (coherent loads once;
-        // reuses the pool staging buffer — requires n <= POOL_LDS_CAP,
-        // enforced by the binding).
-        for (int i = tid; i < n; i += BLOCK) {
-          pool_lds[i] = cload(k_ws + (long)q * n + i);
+      const int tiles_n = (n + VZ_QF_TILE - 1) / VZ_QF_TILE;
+      const int total_tiles = tiles_n * tiles_n;
+      float* k_i_lds = pool_lds;                       // 2048 floats
+      float* k_j_lds = pool_lds + VZ_QF_QMAX * VZ_QF_TILE;
+      for (int tile = blockIdx.x; tile < total_tiles; tile += G) {
+        const int t = tile;
+        vz_quadform_tile(
+            kinv, batch_size, n, t, tiles_n, k_i_lds, k_j_lds,
+            [&](long idx) { return cload(k_ws + idx); },
+            [&](int q, float v) {
+              cstore(var_ws + (long)q * (total_tiles + 1) + t, v);
+            });
+      }
+    }
+    grid_sync(barrier_buf);
+
+    // ---- Phase B2: reduce tile partials (order == the standalone
+    // ps_reduce_parts_kernel) ----
+    {
+      const int tiles_n = (n + VZ_QF_TILE - 1) / VZ_QF_TILE;
+      const int total_tiles = tiles_n * tiles_n;
+      for (int q = blockIdx.x; q < batch_size; q += G) {
+        float sacc = 0.0f;
+        for (int w = tid; w < total_tiles; w += BLOCK) {
+          sacc += cload(var_ws + (long)q * (total_tiles + 1) + w);
         }
-        __syncthreads();
-        const float* k = pool_lds;
-        float acc = 0.0f;
-        for (int j = j0 + wave; j < j1; j += waves) {
-          const float4* row4 =
-              reinterpret_cast<const float4*>(kinv + (long)j * n);
-          float t_j = 0.0f;
-          for (int i4 = lane; i4 < n4; i4 += WAVE_SIZE) {
-            const float4 r = row4[i4];
-            t_j = fmaf(r.x, k[4 * i4], t_j);
-            t_j = fmaf(r.y, k[4 * i4 + 1], t_j);
-            t_j = fmaf(r.z, k[4 * i4 + 2], t_j);
-            t_j = fmaf(r.w, k[4 * i4 + 3], t_j);
-          }
-          for (int i = 4 * n4 + lane; i < n; i += WAVE_SIZE) {
-            t_j = fmaf(kinv[(long)j * n + i], k[i], t_j);
-          }
-          t_j = wave_reduce_sum(t_j);
-          if (lane == 0) acc = fmaf(k[j], t_j, acc);
+        float total = block_reduce(sacc, red, fsum, 0.0f);
+        if (tid == 0) {
+          cstore(var_ws + (long)q * (total_tiles + 1) + total_tiles,
+                 total);
         }
-        float v = block_reduce(acc, red, fsum, 0.0f);
-        if (tid == 0) cstore(var_ws + q * SWEEP_NCHUNK + chunk, v);
         __syncthreads();
       }
     }
@@ -322,11 +319,12 @@ eagle_sweep_kernel(
       // partials (cheap, deterministic, removes a barrier).
       float local_max = -INFINITY;
       float my_score = -INFINITY;
+      const int tiles_n_c = (n + VZ_QF_TILE - 1) / VZ_QF_TILE;
+      const int total_tiles_c = tiles_n_c * tiles_n_c;
       for (int t = tid; t < batch_size; t += BLOCK) {
-        float var = 0.0f;
-        for (int c = 0; c < SWEEP_NCHUNK; ++c)
-          var += cload(var_ws + t * SWEEP_NCHUNK + c);
-        var = fmaxf(amp2 - var, 1e-12f);
+        const float quad =
+            cload(var_ws + (long)t * (total_tiles_c + 1) + total_tiles_c);
+        float var = fmaxf(amp2 - quad, 1e-12f);
         const float sd = sqrtf(var);
         const float mu = cload(mu_ws + t) + mean_c;
         float score = acq_score(acq, mu, sd, coef, best_value);
@@ -415,7 +413,8 @@ extern "C" int launch_eagle_sweep(
   hipError_t err = hipOccupancyMaxActiveBlocksPerMultiprocessor(
       &blocks_per_cu, (const void*)eagle_sweep_kernel, BLOCK, 0);
   if (err != hipSuccess || blocks_per_cu < 1) return 0;
-  int grid = batch_size * SWEEP_NCHUNK;           // fills phase B
+  const int tiles_n = (n + VZ_QF_TILE - 1) / VZ_QF_TILE;
+  int grid = tiles_n * tiles_n;                   // fills phase B
   const int max_grid = blocks_per_cu * num_cu;
   if (grid > max_grid) grid = max_grid;
   if (grid < 1) grid = 1;

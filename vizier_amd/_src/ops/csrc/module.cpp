@@ -55,6 +55,10 @@ extern "C" void launch_ps_kvec_split(
     float* mu_part, float* dist_part, float* mu_ws, float* dist_ws,
     int b, int n, int d, float amp2, int rchunks, hipStream_t stream);
 
+extern "C" void launch_ps_quadform_tile(
+    const float* k_ws, const float* kinv, float* var_part, float* quad,
+    int b, int n, hipStream_t stream);
+
 extern "C" void launch_ps_quadform_big(
     const float* k_ws, const float* kinv, float* part, float* quad,
     int b, int n, hipStream_t stream);
@@ -299,6 +303,24 @@ namespace {
 // 0.263 ms at (b=25, N=10^4) — the library's tiling wins this skinny
 // shape (profiles/quadform_ab_r2.json); both are ~5x off the 50 us
 // HBM floor, so the kernel stays for future split-K work.
+// Small/medium-N quadform: the 64x64-tile kernel reads Kinv ONCE per
+// call (vs once per candidate in the legacy chunked kernel — 32.6 us
+// of a 58 us Eagle iteration at N=1000, profiles/sweep_kernels_r2.txt).
+// b <= 32 only; returns the (b,) quadform.
+torch::Tensor quadform_tile_small_n(const torch::Tensor& k_ws,
+                                    const torch::Tensor& kinv, int b,
+                                    int n, hipStream_t stream) {
+  const int tiles_n = (n + 63) / 64;
+  const long total = (long)tiles_n * tiles_n;
+  auto var_part = torch::empty({b, total}, k_ws.options());
+  auto quad = torch::empty({b}, k_ws.options());
+  launch_ps_quadform_tile(k_ws.data_ptr<float>(),
+                          kinv.data_ptr<float>(),
+                          var_part.data_ptr<float>(),
+                          quad.data_ptr<float>(), b, n, stream);
+  return quad;
+}
+
 torch::Tensor quadform_large_n(const torch::Tensor& k_ws,
                                const torch::Tensor& kinv, int b, int n,
                                hipStream_t stream) {
@@ -403,6 +425,23 @@ torch::Tensor posterior_scores_chunked(
         current_stream());
     return out;
   }
+  if (b <= 32) {
+    launch_ps_kvec(
+        xq.data_ptr<float>(), x.data_ptr<float>(),
+        inv_ls.data_ptr<float>(), alpha.data_ptr<float>(),
+        onehot.data_ptr<unsigned char>(), k_ws.data_ptr<float>(),
+        mu_ws.data_ptr<float>(), dist_ws.data_ptr<float>(), b, n, d,
+        (float)(amplitude * amplitude), current_stream());
+    auto quad = quadform_tile_small_n(k_ws, kinv, b, n,
+                                      current_stream());
+    launch_ps_finalize_direct(
+        mu_ws.data_ptr<float>(), dist_ws.data_ptr<float>(),
+        quad.data_ptr<float>(), out.data_ptr<float>(), b,
+        (float)(amplitude * amplitude), (float)mean_c, (int)acq,
+        (float)coef, (float)best_value, (float)tr_radius,
+        current_stream());
+    return out;
+  }
   auto var_ws = torch::empty({b, 10}, xq.options());
   launch_posterior_score_chunked(
       xq.data_ptr<float>(), x.data_ptr<float>(), inv_ls.data_ptr<float>(),
@@ -465,6 +504,17 @@ torch::Tensor posterior_scores_bf16(
         current_stream());
     return out;
   }
+  if (b <= 32) {
+    auto quad = quadform_tile_small_n(k_ws, kinv, b, n,
+                                      current_stream());
+    launch_ps_finalize_direct(
+        mu_ws.data_ptr<float>(), dist_ws.data_ptr<float>(),
+        quad.data_ptr<float>(), out.data_ptr<float>(), b,
+        (float)(amplitude * amplitude), (float)mean_c, (int)acq,
+        (float)coef, (float)best_value, (float)tr_radius,
+        current_stream());
+    return out;
+  }
   auto var_ws = torch::empty({b, 10}, xq.options());
   launch_ps_quadform_finalize(
       k_ws.data_ptr<float>(), kinv.data_ptr<float>(),
@@ -519,13 +569,22 @@ std::vector<torch::Tensor> posterior_mean_std(
         (float)mean_c, current_stream());
     return {mean, sd, dist_ws};
   }
-  auto var_ws = torch::empty({b, 10}, xq.options());
   launch_ps_kvec(
       xq.data_ptr<float>(), x.data_ptr<float>(),
       inv_ls.data_ptr<float>(), alpha.data_ptr<float>(),
       onehot.data_ptr<unsigned char>(), k_ws.data_ptr<float>(),
       mu_ws.data_ptr<float>(), dist_ws.data_ptr<float>(), b, n, d,
       amp2, current_stream());
+  if (b <= 32) {
+    auto quad = quadform_tile_small_n(k_ws, kinv, b, n,
+                                      current_stream());
+    launch_ps_finalize_meanstd_direct(
+        mu_ws.data_ptr<float>(), quad.data_ptr<float>(),
+        mean.data_ptr<float>(), sd.data_ptr<float>(), b, amp2,
+        (float)mean_c, current_stream());
+    return {mean, sd, dist_ws};
+  }
+  auto var_ws = torch::empty({b, 10}, xq.options());
   launch_ps_quadform_kernel_only(
       k_ws.data_ptr<float>(), kinv.data_ptr<float>(),
       var_ws.data_ptr<float>(), b, n, current_stream());
@@ -657,7 +716,12 @@ int64_t eagle_sweep(
   TORCH_CHECK(pool_cont.numel() == pool_size * dc,
               "pool/feature shape mismatch (continuous-only, q=1)");
   TORCH_CHECK(pool_size <= 128, "eagle_sweep supports pool <= 128");
-  TORCH_CHECK(n <= 8192, "eagle_sweep stages k in LDS: N <= 8192");
+  TORCH_CHECK(n <= 8192, "eagle_sweep supports N <= 8192");
+  TORCH_CHECK(batch_size <= 32,
+              "eagle_sweep tile quadform supports batch <= 32");
+  const long tiles_n = (n + 63) / 64;
+  TORCH_CHECK(var_ws.numel() >= batch_size * (tiles_n * tiles_n + 1),
+              "var_ws must be (B, tiles^2 + 1)");
   const int ret = launch_eagle_sweep(
       pool_cont.data_ptr<float>(), rewards.data_ptr<float>(),
       perturbations.data_ptr<float>(), best_reward.data_ptr<float>(),

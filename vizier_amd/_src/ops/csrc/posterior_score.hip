@@ -489,6 +489,43 @@ extern "C" void launch_ps_kvec_split(
                      rchunks);
 }
 
+// -- Tile-based small/medium-N quadform ---------------------------------
+
+extern "C" __global__ __launch_bounds__(BLOCK) void
+ps_quadform_tile_kernel(const float* __restrict__ k_in,   // (B, N)
+                        const float* __restrict__ kinv,   // (N, N)
+                        float* __restrict__ var_part,     // (B, T)
+                        int b, int n, int tiles_n) {
+  __shared__ float k_i[VZ_QF_QMAX * VZ_QF_TILE];
+  __shared__ float k_j[VZ_QF_QMAX * VZ_QF_TILE];
+  const int total = tiles_n * tiles_n;
+  for (int tile = blockIdx.x; tile < total; tile += gridDim.x) {
+    const int t = tile;
+    vz_quadform_tile(
+        kinv, b, n, t, tiles_n, k_i, k_j,
+        [&](long idx) { return k_in[idx]; },
+        [&](int q, float v) {
+          var_part[(long)q * total + t] = v;
+        });
+  }
+}
+
+extern "C" __global__ void
+ps_reduce_parts_kernel(const float* __restrict__ part,
+                       float* __restrict__ quad, int b, int n_wgs);
+
+extern "C" void launch_ps_quadform_tile(
+    const float* k_ws, const float* kinv, float* var_part, float* quad,
+    int b, int n, hipStream_t stream) {
+  const int tiles_n = (n + VZ_QF_TILE - 1) / VZ_QF_TILE;
+  const int total = tiles_n * tiles_n;
+  int grid = total < 2048 ? total : 2048;
+  hipLaunchKernelGGL(ps_quadform_tile_kernel, dim3(grid), dim3(BLOCK),
+                     0, stream, k_ws, kinv, var_part, b, n, tiles_n);
+  hipLaunchKernelGGL(ps_reduce_parts_kernel, dim3(b), dim3(256), 0,
+                     stream, var_part, quad, b, total);
+}
+
 // -- Multi-objective fused helpers --------------------------------------
 //
 // The MO (config 5) scorer was ~25 eager launches per Eagle iteration
