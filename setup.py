@@ -18,6 +18,7 @@ _SRC = [
     'vizier_amd/_src/ops/csrc/gram_matern52_fp8.hip',
     'vizier_amd/_src/ops/csrc/gram_matern52_fp8_tiled.hip',
     'vizier_amd/_src/ops/csrc/posterior_score.hip',
+    'vizier_amd/_src/ops/csrc/batched_chol.hip',
     'vizier_amd/_src/ops/csrc/eagle_step.hip',
     'vizier_amd/_src/ops/csrc/eagle_sweep.hip',
 ]

@@ -153,6 +153,22 @@ def cholesky_with_jitter(K: torch.Tensor, amplitude2: torch.Tensor,
   return safe_cholesky_ex(K)[0]
 
 
+def _use_custom_chol(K: torch.Tensor) -> bool:
+  """Custom batched potrf/trsv apply on the no-grad GPU path.
+
+  The line-search NLL evaluations (torch.no_grad, batch S*R ~ 16) are
+  the fit's hot spot: MAGMA's batched spotf2 panel launches cost ~54 ms
+  per suggest and the (R, N, 1) solves dispatch as SERIAL rocblas trsv
+  (~170 us each) — profiles/sweep_kernels_r2.txt. The one-workgroup-
+  per-matrix kernels in batched_chol.hip replace both. Autograd paths
+  keep torch's factorization (its backward needs the taped solve
+  chain).
+  """
+  return (K.is_cuda and K.dtype == torch.float32 and
+          not K.requires_grad and K.shape[0] > 1 and
+          K.shape[-1] <= 2048 and ops.extension_available())
+
+
 def negative_log_marginal_likelihood(raw: torch.Tensor, x: torch.Tensor,
                                      y: torch.Tensor) -> torch.Tensor:
   """Batched NLL over restarts. raw (R, D+3); x (N, D); y (N,)."""
@@ -162,15 +178,21 @@ def negative_log_marginal_likelihood(raw: torch.Tensor, x: torch.Tensor,
                     params.amplitude)
   noise = params.noise.reshape(-1, 1, 1)
   K = K + noise * torch.eye(n, dtype=x.dtype, device=x.device)
-  L, info = safe_cholesky_ex(K)
   resid = (y.unsqueeze(0) - params.mean.unsqueeze(-1)).unsqueeze(-1)
-  # quad = r^T K^-1 r = ||L^-1 r||^2: ONE triangular solve instead of a
-  # _chol_solve pair — the batched vector-RHS solves dispatch as serial
-  # rocBLAS trsv calls (~80 ms/suggest at the bench shape, see
-  # profiles/bench_kernel_stats2.csv), so halving them matters.
-  # (torch.cholesky_solve itself hipErrorLaunchFailures on this build.)
-  z = torch.linalg.solve_triangular(L, resid, upper=False)
-  quad = (z * z).sum(dim=(-1, -2))
+  if _use_custom_chol(K):
+    ext = ops.require_ext()
+    L, info = ext.batched_potrf(K.contiguous())
+    z = ext.batched_trsv_lower(
+        L, resid.squeeze(-1).expand(K.shape[0], n).contiguous())
+    quad = (z * z).sum(dim=-1)
+  else:
+    L, info = safe_cholesky_ex(K)
+    # quad = r^T K^-1 r = ||L^-1 r||^2: ONE triangular solve instead
+    # of a _chol_solve pair — the batched vector-RHS solves dispatch
+    # as serial rocBLAS trsv calls, so halving them matters.
+    # (torch.cholesky_solve itself hipErrorLaunchFailures here.)
+    z = torch.linalg.solve_triangular(L, resid, upper=False)
+    quad = (z * z).sum(dim=(-1, -2))
   logdet = 2.0 * torch.log(torch.diagonal(L, dim1=-2, dim2=-1)).sum(-1)
   nll = 0.5 * (quad + logdet + n * math.log(2 * math.pi))
   # Mild pull toward the raw-space origin (the reference regularizes via

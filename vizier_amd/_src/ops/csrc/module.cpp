@@ -100,6 +100,12 @@ extern "C" void launch_ps_quadform_finalize(
     float amp2, float mean_c, int acq, float coef, float best_value,
     float tr_radius, hipStream_t stream);
 
+extern "C" void launch_batched_potrf(float* A, int* info, int r, int n,
+                                     hipStream_t stream);
+extern "C" void launch_batched_trsv_lower(const float* L, float* b,
+                                          int r, int n,
+                                          hipStream_t stream);
+
 extern "C" void launch_eagle_suggest(
     const float* pool_cont, const long* pool_cat, const float* rewards,
     const float* perturbations, const long* cat_sizes, float* out_cont,
@@ -625,6 +631,32 @@ torch::Tensor hv_scalarize_tr(
   return out;
 }
 
+std::vector<torch::Tensor> batched_potrf(torch::Tensor K) {
+  // (R, N, N) -> (L lower in a fresh tensor, info (R,) int32).
+  K = check_f32(K, "K");
+  TORCH_CHECK(K.dim() == 3 && K.size(1) == K.size(2),
+              "K must be (R, N, N)");
+  const int r = K.size(0), n = K.size(1);
+  auto L = K.clone();
+  auto info = torch::zeros({r}, K.options().dtype(torch::kInt32));
+  launch_batched_potrf(L.data_ptr<float>(), info.data_ptr<int>(), r, n,
+                       current_stream());
+  return {L, info};
+}
+
+torch::Tensor batched_trsv_lower(torch::Tensor L, torch::Tensor b) {
+  L = check_f32(L, "L");
+  b = check_f32(b, "b");
+  TORCH_CHECK(L.dim() == 3 && L.size(1) == L.size(2) &&
+              b.dim() == 2 && b.size(0) == L.size(0) &&
+              b.size(1) == L.size(1), "shape mismatch");
+  auto z = b.clone();
+  launch_batched_trsv_lower(L.data_ptr<float>(), z.data_ptr<float>(),
+                            (int)L.size(0), (int)L.size(1),
+                            current_stream());
+  return z;
+}
+
 std::vector<torch::Tensor> eagle_suggest(
     torch::Tensor pool_cont, torch::Tensor pool_cat, torch::Tensor rewards,
     torch::Tensor perturbations, torch::Tensor cat_sizes,
@@ -761,6 +793,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "128x128 LDS-tiled fp8 e4m3 MFMA Matern-5/2 Gram (gfx950)");
   m.def("posterior_scores", &posterior_scores,
         "Fused GP posterior + acquisition + trust region (gfx950)");
+  m.def("batched_potrf", &batched_potrf,
+        "Batched lower Cholesky, one workgroup per matrix (gfx950)");
+  m.def("batched_trsv_lower", &batched_trsv_lower,
+        "Batched forward substitution L z = b (gfx950)");
   m.def("gram_matern52_fp8_pre", &gram_matern52_fp8_pre,
         "fp8 cross-gram from pre-quantized operands (gfx950)");
   m.def("posterior_mean_std", &posterior_mean_std,
