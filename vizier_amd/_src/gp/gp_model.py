@@ -154,16 +154,21 @@ def cholesky_with_jitter(K: torch.Tensor, amplitude2: torch.Tensor,
 
 
 def _use_custom_chol(K: torch.Tensor) -> bool:
-  """Custom batched potrf/trsv apply on the no-grad GPU path.
+  """Opt-in custom batched potrf/trsv on the no-grad GPU path.
 
-  The line-search NLL evaluations (torch.no_grad, batch S*R ~ 16) are
-  the fit's hot spot: MAGMA's batched spotf2 panel launches cost ~54 ms
-  per suggest and the (R, N, 1) solves dispatch as SERIAL rocblas trsv
-  (~170 us each) — profiles/sweep_kernels_r2.txt. The one-workgroup-
-  per-matrix kernels in batched_chol.hip replace both. Autograd paths
-  keep torch's factorization (its backward needs the taped solve
-  chain).
+  Motivation: MAGMA's batched spotf2 panels cost ~54 ms/suggest in
+  ~5000 launches and the (R,N,1) solves dispatch as serial rocblas
+  trsv (~170 us each, profiles/sweep_kernels_r2.txt). The
+  one-workgroup-per-matrix kernels in batched_chol.hip are numerically
+  verified (tests/test_gpu_ops.py TestBatchedCholesky) but MEASURED
+  SLOWER in context (fit 161 -> 251 ms at the headline shape): with
+  batch 16 only 16 workgroups run, and the right-looking panels are
+  latency-chained, while MAGMA's many small launches spread panel work
+  across far more CUs. Kept behind VIZIER_AMD_CUSTOM_CHOL=1; making it
+  competitive needs multi-workgroup-per-matrix trailing updates.
   """
+  if os.environ.get('VIZIER_AMD_CUSTOM_CHOL', '0') != '1':
+    return False
   return (K.is_cuda and K.dtype == torch.float32 and
           not K.requires_grad and K.shape[0] > 1 and
           K.shape[-1] <= 2048 and ops.extension_available())
