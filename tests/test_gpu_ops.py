@@ -249,7 +249,8 @@ class TestBatchedCholesky:
       err = (got - want).abs().max()
       assert float(err) < 5e-4, f'n={n}: {err}'
 
-  def test_nll_custom_path_matches_torch(self, ext):
+  def test_nll_custom_path_matches_torch(self, ext, monkeypatch):
+    monkeypatch.setenv('VIZIER_AMD_CUSTOM_CHOL', '1')
     from vizier_amd._src.gp import gp_model
     g = torch.Generator().manual_seed(13)
     x = torch.rand(300, 6, generator=g).cuda()
