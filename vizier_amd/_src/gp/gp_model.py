@@ -167,7 +167,18 @@ def _use_custom_chol(K: torch.Tensor) -> bool:
   across far more CUs. Kept behind VIZIER_AMD_CUSTOM_CHOL=1; making it
   competitive needs multi-workgroup-per-matrix trailing updates.
   """
-  if os.environ.get('VIZIER_AMD_CUSTOM_CHOL', '0') != '1':
+  mode = os.environ.get('VIZIER_AMD_CUSTOM_CHOL', '0')
+  if mode not in ('1', 'both'):
+    return False
+  return (K.is_cuda and K.dtype == torch.float32 and
+          not K.requires_grad and K.shape[0] > 1 and
+          K.shape[-1] <= 2048 and ops.extension_available())
+
+
+def _use_custom_trsv(K: torch.Tensor) -> bool:
+  """trsv-only opt-in (VIZIER_AMD_CUSTOM_CHOL=trsv): keep MAGMA's
+  factorization, replace only the serial rocblas trsv dispatch."""
+  if os.environ.get('VIZIER_AMD_CUSTOM_CHOL', '0') != 'trsv':
     return False
   return (K.is_cuda and K.dtype == torch.float32 and
           not K.requires_grad and K.shape[0] > 1 and
@@ -192,12 +203,19 @@ def negative_log_marginal_likelihood(raw: torch.Tensor, x: torch.Tensor,
     quad = (z * z).sum(dim=-1)
   else:
     L, info = safe_cholesky_ex(K)
-    # quad = r^T K^-1 r = ||L^-1 r||^2: ONE triangular solve instead
-    # of a _chol_solve pair — the batched vector-RHS solves dispatch
-    # as serial rocBLAS trsv calls, so halving them matters.
-    # (torch.cholesky_solve itself hipErrorLaunchFailures here.)
-    z = torch.linalg.solve_triangular(L, resid, upper=False)
-    quad = (z * z).sum(dim=(-1, -2))
+    if _use_custom_trsv(K):
+      ext = ops.require_ext()
+      z = ext.batched_trsv_lower(
+          L.contiguous(),
+          resid.squeeze(-1).expand(K.shape[0], n).contiguous())
+      quad = (z * z).sum(dim=-1)
+    else:
+      # quad = r^T K^-1 r = ||L^-1 r||^2: ONE triangular solve instead
+      # of a _chol_solve pair — the batched vector-RHS solves dispatch
+      # as serial rocBLAS trsv calls, so halving them matters.
+      # (torch.cholesky_solve itself hipErrorLaunchFailures here.)
+      z = torch.linalg.solve_triangular(L, resid, upper=False)
+      quad = (z * z).sum(dim=(-1, -2))
   logdet = 2.0 * torch.log(torch.diagonal(L, dim1=-2, dim2=-1)).sum(-1)
   nll = 0.5 * (quad + logdet + n * math.log(2 * math.pi))
   # Mild pull toward the raw-space origin (the reference regularizes via
