@@ -397,6 +397,25 @@ class VizierGPUCBPEBandit(Designer):
     results = optimizer.optimize(score_fn, count=1)
     return self._codec.decode(results.features)[0]   # (q, D)
 
+  @staticmethod
+  def _fusable(post, x_all: Optional[torch.Tensor] = None) -> bool:
+    """Plain fp32 GPPosterior with K_inv on GPU (the HIP fused-scorer
+    contract); when x_all is given, the posterior must be anchored on
+    it so the kernel's trust-region distance is the right one."""
+    return (isinstance(post, gp_model.GPPosterior) and
+            post.K_inv is not None and post.x.is_cuda and
+            post.x.dtype == torch.float32 and
+            (x_all is None or post.x.data_ptr() == x_all.data_ptr()))
+
+  def _tr_kernel_args(self, trust_region, device):
+    onehot = (trust_region._onehot.to(torch.uint8)
+              if trust_region is not None else torch.zeros(
+                  self._converter.n_features, dtype=torch.uint8,
+                  device=device))
+    radius = (float(trust_region.trust_radius)
+              if trust_region is not None else 0.0)
+    return onehot, radius
+
   def _optimize_one(self, use_ucb: bool, x_all: torch.Tensor
                     ) -> torch.Tensor:
     cfg = self._config
@@ -413,13 +432,56 @@ class VizierGPUCBPEBandit(Designer):
         ys = torch.stack([m_ + coef * s_ for m_, s_ in per], dim=-1)
         return scalarizer(ys).mean(dim=0)
 
+      fused_mo = (cfg.dtype == torch.float32 and
+                  all(self._fusable(p) for p in posts))
+      if fused_mo:
+        from vizier_amd._src.ops import dispatch as ops
+        ext = ops.require_ext()
+        weights = scalarizer.weights.to(x_all.device,
+                                        torch.float32).contiguous()
+        ref = scalarizer.reference_point
+        ref = (ref.to(x_all.device, torch.float32).contiguous()
+               if ref is not None else None)
+        onehot, tr_radius = self._tr_kernel_args(trust_region,
+                                                 x_all.device)
+
+        def _fused_scalarized(xs, coef, dist_src=None):
+          means, sds, dist = [], [], None
+          for p in posts:
+            m_, s_, d_ = ext.posterior_mean_std(
+                xs, p.x, p.params.lengthscales,
+                float(p.params.amplitude), float(p.params.mean),
+                p.alpha, p.K_inv, onehot)
+            means.append(m_)
+            sds.append(s_)
+            dist = d_
+          if dist_src is not None:
+            dist = dist_src
+          return means, sds, dist
+
       if use_ucb:
-        def score_fn(batch: CandidateBatch) -> torch.Tensor:
-          xs = self._codec.decode(batch)[:, 0, :]
-          scores = _scalarized(xs, cfg.ucb_coefficient)
-          if trust_region is not None:
-            scores = trust_region.apply(xs, scores)
-          return scores
+        if fused_mo and self._fusable(posts[0], x_all):
+          # Fused MO UCB (same shape as GP-Bandit's config-5 path):
+          # per-metric (mean, sd, dist) kernels + ONE scalarize+TR
+          # launch; the posteriors anchor on x_all (no pending), so
+          # the kernel trust-region distance is exact.
+          def score_fn(batch: CandidateBatch) -> torch.Tensor:
+            xs = self._codec.decode(batch)[:, 0, :]
+            means, sds, dist = _fused_scalarized(
+                xs, cfg.ucb_coefficient)
+            return ext.hv_scalarize_tr(
+                torch.stack(means), torch.stack(sds), weights, ref,
+                dist if trust_region is not None else None,
+                cfg.ucb_coefficient, tr_radius)
+          score_fn.graph_safe = True
+        else:
+          def score_fn(batch: CandidateBatch) -> torch.Tensor:
+            xs = self._codec.decode(batch)[:, 0, :]
+            scores = _scalarized(xs, cfg.ucb_coefficient)
+            if trust_region is not None:
+              scores = trust_region.apply(xs, scores)
+            return scores
+          score_fn.graph_safe = False  # capture unsupported: ROCm 7.2
       else:
         # Promising region: scalarized mean at the observed point with
         # the best scalarized UCB (multimetric analogue of
@@ -430,19 +492,49 @@ class VizierGPUCBPEBandit(Designer):
           threshold = obs_mean[int(torch.argmax(obs_ucb))]
         var_posts = [self._variance_posterior(x_all, p_) for p_ in posts]
 
-        def score_fn(batch: CandidateBatch) -> torch.Tensor:
-          xs = self._codec.decode(batch)[:, 0, :]
-          explore = _scalarized(
-              xs, cfg.explore_region_ucb_coefficient)
-          stddev_sum = sum(vp.predict(xs)[1] for vp in var_posts)
-          penalty = cfg.cb_violation_penalty_coefficient * \
-              torch.minimum(explore - threshold,
-                            torch.zeros_like(explore))
-          scores = stddev_sum + penalty
-          if trust_region is not None:
-            scores = trust_region.apply(xs, scores)
-          return scores
-      score_fn.graph_safe = False  # capture unsupported on ROCm 7.2
+        if fused_mo and all(self._fusable(vp, x_all)
+                            for vp in var_posts):
+          thr = float(threshold)
+          pen_coef = cfg.cb_violation_penalty_coefficient
+
+          def score_fn(batch: CandidateBatch) -> torch.Tensor:
+            xs = self._codec.decode(batch)[:, 0, :]
+            means, sds, _ = _fused_scalarized(
+                xs, cfg.explore_region_ucb_coefficient)
+            explore = ext.hv_scalarize_tr(
+                torch.stack(means), torch.stack(sds), weights, ref,
+                None, cfg.explore_region_ucb_coefficient, 0.0)
+            stddev_sum, dist = None, None
+            for vp in var_posts:
+              _, s_all, d_ = ext.posterior_mean_std(
+                  xs, vp.x, vp.params.lengthscales,
+                  float(vp.params.amplitude), 0.0, vp.alpha, vp.K_inv,
+                  onehot)
+              stddev_sum = s_all if stddev_sum is None \
+                  else stddev_sum + s_all
+              dist = d_
+            penalty = pen_coef * torch.minimum(
+                explore - thr, torch.zeros_like(explore))
+            scores = stddev_sum + penalty
+            if trust_region is not None and tr_radius <= 0.5:
+              scores = torch.where(dist <= tr_radius, scores,
+                                   -1e4 - dist)
+            return scores
+          score_fn.graph_safe = True
+        else:
+          def score_fn(batch: CandidateBatch) -> torch.Tensor:
+            xs = self._codec.decode(batch)[:, 0, :]
+            explore = _scalarized(
+                xs, cfg.explore_region_ucb_coefficient)
+            stddev_sum = sum(vp.predict(xs)[1] for vp in var_posts)
+            penalty = cfg.cb_violation_penalty_coefficient * \
+                torch.minimum(explore - threshold,
+                              torch.zeros_like(explore))
+            scores = stddev_sum + penalty
+            if trust_region is not None:
+              scores = trust_region.apply(xs, scores)
+            return scores
+          score_fn.graph_safe = False  # capture unsupported: ROCm 7.2
     elif use_ucb:
       scoring = acq_lib.ScoringFunction(
           posterior, acq_lib.UCB(cfg.ucb_coefficient), trust_region)
@@ -470,18 +562,56 @@ class VizierGPUCBPEBandit(Designer):
       threshold = mean_obs[int(torch.argmax(ucb_obs))]
       var_post = self._variance_posterior(x_all)
 
-      def score_fn(batch: CandidateBatch) -> torch.Tensor:
-        xs = self._codec.decode(batch)[:, 0, :]
-        mean, stddev = posterior.predict(xs)
-        explore_ucb = mean + cfg.explore_region_ucb_coefficient * stddev
-        _, stddev_all = var_post.predict(xs)
-        penalty = cfg.cb_violation_penalty_coefficient * torch.minimum(
-            explore_ucb - threshold, torch.zeros_like(explore_ucb))
-        scores = stddev_all + penalty
-        if trust_region is not None:
-          scores = trust_region.apply(xs, scores)
-        return scores
-      score_fn.graph_safe = False  # capture unsupported on ROCm 7.2
+      if (cfg.dtype == torch.float32 and self._fusable(posterior) and
+          self._fusable(var_post, x_all)):
+        # Fused PE: two (mean, sd, dist) kernel calls + an elementwise
+        # tail — hipGraph-capturable (the DEFAULT algorithm's batch
+        # exploration phase was previously the ~20-launch eager chain).
+        # The trust-region distance comes from the VAR posterior's
+        # call: it anchors on x_all, exactly the region's trusted set.
+        from vizier_amd._src.ops import dispatch as ops
+        ext = ops.require_ext()
+        onehot, tr_radius = self._tr_kernel_args(trust_region,
+                                                 x_all.device)
+        thr = float(threshold)
+        pen_coef = cfg.cb_violation_penalty_coefficient
+        exp_coef = cfg.explore_region_ucb_coefficient
+        post = posterior
+
+        def score_fn(batch: CandidateBatch) -> torch.Tensor:
+          xs = self._codec.decode(batch)[:, 0, :]
+          mean, stddev, _ = ext.posterior_mean_std(
+              xs, post.x, post.params.lengthscales,
+              float(post.params.amplitude), float(post.params.mean),
+              post.alpha, post.K_inv, onehot)
+          _, stddev_all, dist = ext.posterior_mean_std(
+              xs, var_post.x, var_post.params.lengthscales,
+              float(var_post.params.amplitude), 0.0, var_post.alpha,
+              var_post.K_inv, onehot)
+          explore_ucb = mean + exp_coef * stddev
+          penalty = pen_coef * torch.minimum(
+              explore_ucb - thr, torch.zeros_like(explore_ucb))
+          scores = stddev_all + penalty
+          if trust_region is not None and tr_radius <= 0.5:
+            scores = torch.where(dist <= tr_radius, scores,
+                                 -1e4 - dist)
+          return scores
+        score_fn.graph_safe = True
+      else:
+        def score_fn(batch: CandidateBatch) -> torch.Tensor:
+          xs = self._codec.decode(batch)[:, 0, :]
+          mean, stddev = posterior.predict(xs)
+          explore_ucb = mean + \
+              cfg.explore_region_ucb_coefficient * stddev
+          _, stddev_all = var_post.predict(xs)
+          penalty = cfg.cb_violation_penalty_coefficient * \
+              torch.minimum(explore_ucb - threshold,
+                            torch.zeros_like(explore_ucb))
+          scores = stddev_all + penalty
+          if trust_region is not None:
+            scores = trust_region.apply(xs, scores)
+          return scores
+        score_fn.graph_safe = False  # capture unsupported: ROCm 7.2
 
     factory = VectorizedOptimizerFactory(
         eagle_config=EagleStrategyConfig(),
