@@ -445,13 +445,18 @@ class VizierGPUCBPEBandit(Designer):
         onehot, tr_radius = self._tr_kernel_args(trust_region,
                                                  x_all.device)
 
+        # Scalars hoisted OUT of the closures: float(tensor) syncs the
+        # device, which aborts hipGraph capture (measured: the stream-
+        # capture failures on ROCm 7.2 were exactly these).
+        post_amps = [float(p.params.amplitude) for p in posts]
+        post_means = [float(p.params.mean) for p in posts]
+
         def _fused_scalarized(xs, coef, dist_src=None):
           means, sds, dist = [], [], None
-          for p in posts:
+          for i, p in enumerate(posts):
             m_, s_, d_ = ext.posterior_mean_std(
-                xs, p.x, p.params.lengthscales,
-                float(p.params.amplitude), float(p.params.mean),
-                p.alpha, p.K_inv, onehot)
+                xs, p.x, p.params.lengthscales, post_amps[i],
+                post_means[i], p.alpha, p.K_inv, onehot)
             means.append(m_)
             sds.append(s_)
             dist = d_
@@ -496,6 +501,7 @@ class VizierGPUCBPEBandit(Designer):
                             for vp in var_posts):
           thr = float(threshold)
           pen_coef = cfg.cb_violation_penalty_coefficient
+          vp_amps = [float(vp.params.amplitude) for vp in var_posts]
 
           def score_fn(batch: CandidateBatch) -> torch.Tensor:
             xs = self._codec.decode(batch)[:, 0, :]
@@ -505,11 +511,10 @@ class VizierGPUCBPEBandit(Designer):
                 torch.stack(means), torch.stack(sds), weights, ref,
                 None, cfg.explore_region_ucb_coefficient, 0.0)
             stddev_sum, dist = None, None
-            for vp in var_posts:
+            for vi, vp in enumerate(var_posts):
               _, s_all, d_ = ext.posterior_mean_std(
-                  xs, vp.x, vp.params.lengthscales,
-                  float(vp.params.amplitude), 0.0, vp.alpha, vp.K_inv,
-                  onehot)
+                  xs, vp.x, vp.params.lengthscales, vp_amps[vi], 0.0,
+                  vp.alpha, vp.K_inv, onehot)
               stddev_sum = s_all if stddev_sum is None \
                   else stddev_sum + s_all
               dist = d_
@@ -577,17 +582,20 @@ class VizierGPUCBPEBandit(Designer):
         pen_coef = cfg.cb_violation_penalty_coefficient
         exp_coef = cfg.explore_region_ucb_coefficient
         post = posterior
+        # Hoisted: float(tensor) inside the closure syncs and aborts
+        # stream capture.
+        post_amp = float(post.params.amplitude)
+        post_mean = float(post.params.mean)
+        vp_amp = float(var_post.params.amplitude)
 
         def score_fn(batch: CandidateBatch) -> torch.Tensor:
           xs = self._codec.decode(batch)[:, 0, :]
           mean, stddev, _ = ext.posterior_mean_std(
-              xs, post.x, post.params.lengthscales,
-              float(post.params.amplitude), float(post.params.mean),
-              post.alpha, post.K_inv, onehot)
+              xs, post.x, post.params.lengthscales, post_amp,
+              post_mean, post.alpha, post.K_inv, onehot)
           _, stddev_all, dist = ext.posterior_mean_std(
-              xs, var_post.x, var_post.params.lengthscales,
-              float(var_post.params.amplitude), 0.0, var_post.alpha,
-              var_post.K_inv, onehot)
+              xs, var_post.x, var_post.params.lengthscales, vp_amp,
+              0.0, var_post.alpha, var_post.K_inv, onehot)
           explore_ucb = mean + exp_coef * stddev
           penalty = pen_coef * torch.minimum(
               explore_ucb - thr, torch.zeros_like(explore_ucb))
