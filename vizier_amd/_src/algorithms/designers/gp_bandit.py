@@ -412,10 +412,10 @@ class VizierGPBandit(Designer, Predictor):
           scores = trust_region.apply(flat.reshape(dense.shape)[:, 0, :],
                                       scores)
         return scores
-      # Tried capture in r2: hipErrorStreamCaptureUnsupported on this
-      # ROCm build (gpurun_out/c5_graph.log) — some op in the MO chain
-      # is not capture-safe, so stay eager instead of paying an aborted
-      # capture attempt every suggest.
+      # Not capture-safe: posterior.predict's dispatch converts
+      # amplitude via float(tensor) per call — a device sync, which
+      # aborts stream capture (diagnosed in r2: the fused paths hoist
+      # every scalar and capture cleanly).
       score_fn.graph_safe = False
       return score_fn, 1
 
@@ -451,7 +451,7 @@ class VizierGPBandit(Designer, Predictor):
         if trust_region is not None:
           scores = trust_region.apply(dense, scores)
         return scores
-      score_fn.graph_safe = False  # capture unsupported (see MO note)
+      score_fn.graph_safe = False  # float(tensor) syncs (MO note)
       return score_fn, 1
 
     plain_gp = isinstance(posterior, gp_model.GPPosterior)
@@ -473,7 +473,7 @@ class VizierGPBandit(Designer, Predictor):
         if trust_region is not None:
           scores = trust_region.apply(dense, scores)
         return scores
-      score_fn.graph_safe = False  # capture unsupported (see MO note)
+      score_fn.graph_safe = False  # float(tensor) syncs (MO note)
       return score_fn, 1
 
     if cfg.acquisition == 'qei' and count > 1:

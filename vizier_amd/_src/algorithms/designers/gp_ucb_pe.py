@@ -486,7 +486,7 @@ class VizierGPUCBPEBandit(Designer):
             if trust_region is not None:
               scores = trust_region.apply(xs, scores)
             return scores
-          score_fn.graph_safe = False  # capture unsupported: ROCm 7.2
+          score_fn.graph_safe = False  # float(tensor) syncs in predict
       else:
         # Promising region: scalarized mean at the observed point with
         # the best scalarized UCB (multimetric analogue of
@@ -539,7 +539,7 @@ class VizierGPUCBPEBandit(Designer):
             if trust_region is not None:
               scores = trust_region.apply(xs, scores)
             return scores
-          score_fn.graph_safe = False  # capture unsupported: ROCm 7.2
+          score_fn.graph_safe = False  # float(tensor) syncs in predict
     elif use_ucb:
       scoring = acq_lib.ScoringFunction(
           posterior, acq_lib.UCB(cfg.ucb_coefficient), trust_region)
@@ -558,7 +558,7 @@ class VizierGPUCBPEBandit(Designer):
       else:
         # Pending points anchor the trust region elsewhere: composed
         # rocBLAS path, not capturable on this ROCm build.
-        score_fn.graph_safe = False  # capture unsupported on ROCm 7.2
+        score_fn.graph_safe = False  # float(tensor) syncs in predict
     else:
       # Promising-region threshold: predicted mean at the observed point
       # with the highest UCB (gp_ucb_pe.py:175-205).
@@ -619,7 +619,7 @@ class VizierGPUCBPEBandit(Designer):
           if trust_region is not None:
             scores = trust_region.apply(xs, scores)
           return scores
-        score_fn.graph_safe = False  # capture unsupported: ROCm 7.2
+        score_fn.graph_safe = False  # float(tensor) syncs in predict
 
     factory = VectorizedOptimizerFactory(
         eagle_config=EagleStrategyConfig(),
