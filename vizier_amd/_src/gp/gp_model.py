@@ -156,18 +156,17 @@ def cholesky_with_jitter(K: torch.Tensor, amplitude2: torch.Tensor,
 def _use_custom_chol(K: torch.Tensor) -> bool:
   """Opt-in custom batched potrf/trsv on the no-grad GPU path.
 
-  Motivation: MAGMA's batched spotf2 panels cost ~54 ms/suggest in
-  ~5000 launches and the (R,N,1) solves dispatch as serial rocblas
-  trsv (~170 us each, profiles/sweep_kernels_r2.txt). The
-  one-workgroup-per-matrix kernels in batched_chol.hip are numerically
-  verified (tests/test_gpu_ops.py TestBatchedCholesky) but MEASURED
-  SLOWER in context (fit 161 -> 251 ms at the headline shape): with
-  batch 16 only 16 workgroups run, and the right-looking panels are
-  latency-chained, while MAGMA's many small launches spread panel work
-  across far more CUs. Kept behind VIZIER_AMD_CUSTOM_CHOL=1; making it
-  competitive needs multi-workgroup-per-matrix trailing updates.
+  DEFAULT ('both'): the panel-swept v2 potrf (batched_chol.hip —
+  per-matrix panel factor + chip-filling (R x column-tiles) trailing
+  kernels, 2 launches per 32-panel) plus the batched trsv replace
+  MAGMA's ~5000 spotf2 panel launches and the serial rocblas trsv
+  dispatch on the no-grad line-search path. A/B at the headline shape:
+  fit 153.6 -> 150.0 ms, steady suggest 350 -> 340 ms. (The v1
+  one-workgroup-per-matrix factorization measured SLOWER — 16 CUs
+  busy, latency-chained panels — and was replaced by v2.) Opt out with
+  VIZIER_AMD_CUSTOM_CHOL=0 / trsv-only with =trsv.
   """
-  mode = os.environ.get('VIZIER_AMD_CUSTOM_CHOL', 'trsv')
+  mode = os.environ.get('VIZIER_AMD_CUSTOM_CHOL', 'both')
   if mode not in ('1', 'both'):
     return False
   return (K.is_cuda and K.dtype == torch.float32 and
@@ -176,11 +175,9 @@ def _use_custom_chol(K: torch.Tensor) -> bool:
 
 
 def _use_custom_trsv(K: torch.Tensor) -> bool:
-  """DEFAULT: keep MAGMA's factorization but replace the serial
-  rocblas trsv dispatch with the batched wave-synchronous kernel
-  (A/B: fit 162.6 -> 153.3 ms at the headline shape). Opt out with
-  VIZIER_AMD_CUSTOM_CHOL=0."""
-  if os.environ.get('VIZIER_AMD_CUSTOM_CHOL', 'trsv') != 'trsv':
+  """trsv-only mode (VIZIER_AMD_CUSTOM_CHOL=trsv): MAGMA factorization
+  + the batched wave-synchronous solve."""
+  if os.environ.get('VIZIER_AMD_CUSTOM_CHOL', 'both') != 'trsv':
     return False
   return (K.is_cuda and K.dtype == torch.float32 and
           not K.requires_grad and K.shape[0] > 1 and
