@@ -480,6 +480,14 @@ class VizierGPBandit(Designer, Predictor):
       if isinstance(posterior, gp_model.EnsembleGPPosterior):
         posterior = posterior.members[0]
       qei = acq_lib.QEI(best_value=best_value, seed=self._seed)
+      # Common random numbers across the whole sweep (identical eps in
+      # every iteration — the qEI estimate is a deterministic function
+      # of the candidates, which also stabilizes the Eagle search).
+      # Hoisted OUT of the closure: regenerating on the host each call
+      # was a per-iteration H2D copy that also blocked graph capture.
+      g = torch.Generator(device='cpu').manual_seed(self._seed)
+      eps_dev = torch.randn(128, 1, count, generator=g).to(
+          self._device, cfg.dtype)
 
       def score_fn(batch: CandidateBatch) -> torch.Tensor:
         dense = self._codec.decode(batch)          # (B, q, D)
@@ -488,26 +496,25 @@ class VizierGPBandit(Designer, Predictor):
         eye = torch.eye(q, dtype=cov.dtype, device=cov.device)
         # Near-duplicate candidate rows make cov singular; jitter
         # relative to the diagonal scale, then fall back to a diagonal
-        # factor for any batch element that still fails.
+        # factor for any batch element that still fails. Branchless:
+        # a host `if bad.any()` would sync (and abort graph capture).
         diag = cov.diagonal(dim1=-2, dim2=-1)
         jitter = 1e-4 * diag.mean(-1, keepdim=True).clamp_min(1e-10)
         cov = cov + jitter.unsqueeze(-1) * eye
         L, info = torch.linalg.cholesky_ex(cov)    # (B, q, q)
         bad = (info > 0)
-        if bool(bad.any()):
-          L_diag = torch.diag_embed(diag.clamp_min(1e-12).sqrt())
-          L = torch.where(bad.view(-1, 1, 1), L_diag, L)
-        g = torch.Generator(device='cpu').manual_seed(self._seed)
-        eps = torch.randn(128, 1, q, generator=g).to(dense.device,
-                                                     dense.dtype)
+        L_diag = torch.diag_embed(diag.clamp_min(1e-12).sqrt())
+        L = torch.where(bad.view(-1, 1, 1), L_diag, L)
         samples = mean.unsqueeze(0) + torch.einsum(
-            'sbq,bqr->sbr', eps.expand(128, dense.shape[0], q), L)
+            'sbq,bqr->sbr', eps_dev.expand(128, dense.shape[0], q), L)
         scores = (samples - best_value).clamp_min(0).amax(-1).mean(0)
         if trust_region is not None:
           flat_scores = trust_region.apply(dense[:, 0, :], scores)
           scores = flat_scores
         return scores
-      score_fn.graph_safe = False  # host RNG + rocBLAS inside
+      # Capture-eligible (the optimizer auto-falls-back if the batched
+      # cholesky refuses capture on this build).
+      score_fn.graph_safe = True
       return score_fn, count
 
     if cfg.acquisition == 'ei':
