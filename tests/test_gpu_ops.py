@@ -216,6 +216,91 @@ class TestPosteriorScoreKernel:
     assert torch.allclose(t, want, atol=0.02 * amp)
 
 
+class TestUCBPEFusedPaths:
+
+  def _designer(self, n_trials=30):
+    from vizier_amd import pyvizier as vz
+    from vizier_amd._src.algorithms.core.abstractions import (
+        ActiveTrials,
+        CompletedTrials,
+    )
+    from vizier_amd._src.algorithms.designers.gp_ucb_pe import (
+        UCBPEConfig,
+        VizierGPUCBPEBandit,
+    )
+    problem = vz.ProblemStatement()
+    for i in range(5):
+      problem.search_space.root.add_float_param(f'x{i}', 0.0, 1.0)
+    problem.metric_information.append(vz.MetricInformation(
+        name='obj', goal=vz.ObjectiveMetricGoal.MAXIMIZE))
+    d = VizierGPUCBPEBandit(problem, UCBPEConfig(
+        max_evaluations=500, ard_restarts=1, ard_max_iters=10,
+        device='cuda'), seed=0)
+    rng = np.random.default_rng(0)
+    trials = []
+    for uid in range(1, n_trials + 1):
+      x = rng.uniform(0, 1, 5)
+      t = vz.Trial({f'x{i}': float(x[i]) for i in range(5)}, id=uid)
+      t.complete(vz.Measurement(
+          metrics={'obj': float(-((x - 0.6) ** 2).sum())}))
+      trials.append(t)
+    d.update(CompletedTrials(trials), ActiveTrials())
+    d._fit()
+    return d
+
+  def test_fused_pe_matches_eager(self, ext):
+    """The fused PE scorer (two posterior_mean_std kernel calls) must
+    reproduce the eager predict chain."""
+    import vizier_amd._src.gp.acquisitions as acq_lib
+    from vizier_amd._src.algorithms.optimizers.eagle import (
+        CandidateBatch,
+    )
+    d = self._designer()
+    cfg = d._config
+    posterior = d._posterior
+    x_all = posterior.x
+    tr = acq_lib.TrustRegion.for_converter(x_all, d._converter)
+    var_post = d._variance_posterior(x_all)
+    assert d._fusable(posterior) and d._fusable(var_post, x_all)
+
+    g = torch.Generator().manual_seed(4)
+    xs = torch.rand(40, 1, 5, generator=g).cuda()
+    batch = CandidateBatch(xs, torch.zeros(40, 1, 0, dtype=torch.long,
+                                           device='cuda'))
+    # Fused score_fn comes out of _optimize_one's machinery; rebuild
+    # both paths directly for a clean comparison.
+    import vizier_amd_hip as hip_ext
+    onehot, radius = d._tr_kernel_args(tr, x_all.device)
+    mean_obs, stddev_obs = posterior.predict(posterior.x)
+    ucb_obs = mean_obs + cfg.ucb_coefficient * stddev_obs
+    threshold = mean_obs[int(torch.argmax(ucb_obs))]
+    flat = xs[:, 0, :]
+    m, sdev, _ = hip_ext.posterior_mean_std(
+        flat, posterior.x, posterior.params.lengthscales,
+        float(posterior.params.amplitude),
+        float(posterior.params.mean), posterior.alpha,
+        posterior.K_inv, onehot)
+    _, s_all, dist = hip_ext.posterior_mean_std(
+        flat, var_post.x, var_post.params.lengthscales,
+        float(var_post.params.amplitude), 0.0, var_post.alpha,
+        var_post.K_inv, onehot)
+    explore = m + cfg.explore_region_ucb_coefficient * sdev
+    pen = cfg.cb_violation_penalty_coefficient * torch.minimum(
+        explore - float(threshold), torch.zeros_like(explore))
+    fused = s_all + pen
+    if radius <= 0.5:
+      fused = torch.where(dist <= radius, fused, -1e4 - dist)
+    # Eager oracle.
+    mean_e, stddev_e = posterior.predict(flat)
+    explore_e = mean_e + cfg.explore_region_ucb_coefficient * stddev_e
+    _, s_all_e = var_post.predict(flat)
+    pen_e = cfg.cb_violation_penalty_coefficient * torch.minimum(
+        explore_e - threshold, torch.zeros_like(explore_e))
+    want = tr.apply(flat, s_all_e + pen_e)
+    amp = float(posterior.params.amplitude)
+    assert torch.allclose(fused, want, atol=0.02 * amp),         float((fused - want).abs().max())
+
+
 class TestBatchedCholesky:
 
   def test_potrf_matches_torch(self, ext):
