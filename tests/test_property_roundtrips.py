@@ -192,3 +192,81 @@ class TestStudyConfigRoundtrip:
           [c.name for c in pa.child_parameter_configs]
     # Round trip again: proto -> config -> proto must be stable.
     assert back.to_proto() == proto
+
+
+class TestEagleUtilsProperties:
+  """Invariants of the per-type Eagle operations (round-2 depth)."""
+
+  def _utils(self, seed=0):
+    from vizier_amd._src.algorithms.designers.eagle_strategy.eagle_utils \
+        import EagleUtils, FireflyAlgorithmConfig
+    import vizier_amd.pyvizier as vz
+    p = vz.ProblemStatement()
+    root = p.search_space.root
+    root.add_float_param('f', -3.0, 7.0)
+    root.add_categorical_param('c', ['a', 'b', 'z'])
+    root.add_discrete_param('d', [1.0, 2.0, 8.0])
+    root.add_int_param('i', -2, 9)
+    p.metric_information.append(vz.MetricInformation(
+        name='obj', goal=vz.ObjectiveMetricGoal.MAXIMIZE))
+    return EagleUtils(p, FireflyAlgorithmConfig(),
+                      np.random.default_rng(seed))
+
+  @given(w=st.floats(-5.0, 5.0, allow_nan=False),
+         a=st.floats(0.0, 1.0), b=st.floats(0.0, 1.0),
+         seed=st.integers(0, 50))
+  @settings(max_examples=60, deadline=None)
+  def test_combine_numeric_stays_in_unit_box(self, w, a, b, seed):
+    u = self._utils(seed)
+    cfg = [c for c in u.parameter_configs if c.name == 'f'][0]
+    out = u.combine(cfg, a, b, w)
+    assert 0.0 <= out <= 1.0
+
+  @given(amount=st.floats(-3.0, 3.0, allow_nan=False),
+         v=st.floats(0.0, 1.0), seed=st.integers(0, 50))
+  @settings(max_examples=60, deadline=None)
+  def test_perturb_numeric_stays_in_unit_box(self, amount, v, seed):
+    u = self._utils(seed)
+    cfg = [c for c in u.parameter_configs if c.name == 'f'][0]
+    out = u.perturb(cfg, v, amount)
+    assert 0.0 <= out <= 1.0
+
+  @given(fv=st.floats(0.0, 1.0), cv=st.sampled_from(['a', 'b', 'z']),
+         dv=st.floats(0.0, 1.0), iv=st.floats(0.0, 1.0),
+         seed=st.integers(0, 20))
+  @settings(max_examples=60, deadline=None)
+  def test_values_to_parameters_always_feasible(self, fv, cv, dv, iv,
+                                                seed):
+    u = self._utils(seed)
+    params = u.values_to_parameters(
+        {'f': fv, 'c': cv, 'd': dv, 'i': iv})
+    assert -3.0 <= params['f'].value <= 7.0
+    assert params['c'].value in ('a', 'b', 'z')
+    assert params['d'].value in (1.0, 2.0, 8.0)
+    assert isinstance(params['i'].value, int)
+    assert -2 <= params['i'].value <= 9
+    # Round trip through trial_to_values stays consistent.
+    import vizier_amd.pyvizier as vz
+    back = u.trial_to_values(vz.TrialSuggestion(params))
+    assert back['c'] == params['c'].value
+    assert 0.0 <= back['f'] <= 1.0
+
+
+class TestPaddingProperties:
+
+  @given(n=st.integers(0, 200), trials_kind=st.sampled_from(
+      ['NONE', 'MULTIPLES_OF_10', 'POWERS_OF_2']))
+  @settings(max_examples=80, deadline=None)
+  def test_padded_size_monotone_and_bounding(self, n, trials_kind):
+    from vizier_amd.converters.core import PaddingSchedule, PaddingType
+    s = PaddingSchedule(num_trials=PaddingType[trials_kind])
+    p = s.padded_size(n)
+    assert p >= n
+    if trials_kind == 'NONE':
+      assert p == n
+    elif n > 0:
+      # Bounded waste: at most 9 extra rows / less than 2x.
+      assert (p - n <= 9) if trials_kind == 'MULTIPLES_OF_10' \
+          else (p < 2 * max(n, 1))
+    # Monotone in n.
+    assert s.padded_size(n + 1) >= p
