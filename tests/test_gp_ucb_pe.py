@@ -288,3 +288,48 @@ class TestUCBPEMultimetric:
     t2.complete(vz.Measurement(metrics={'m1': 0.2, 'm2': -0.02}))
     d.update(CompletedTrials([t2]), ActiveTrials())
     assert len(d.suggest(1)) == 1  # joint warm refit
+
+
+class TestFusedPathGuards:
+  """The HIP fused scorers must NOT engage on CPU (torch fallback)."""
+
+  def test_pe_phase_runs_on_cpu_with_pending(self):
+    designer = VizierGPUCBPEBandit(make_problem(), cfg(), seed=8)
+    trials = []
+    rng = np.random.default_rng(2)
+    for uid in range(1, 7):
+      s = vz.TrialSuggestion(
+          {f'x{i}': float(v) for i, v in enumerate(rng.uniform(0, 1, 3))})
+      t = s.to_trial(uid)
+      t.complete(vz.Measurement(metrics={'obj': evaluate(s)}))
+      trials.append(t)
+    active = [vz.TrialSuggestion(
+        {f'x{i}': 0.9 for i in range(3)}).to_trial(99)]
+    designer.update(CompletedTrials(trials), ActiveTrials(active))
+    out = designer.suggest(2)
+    assert len(out) == 2
+    # _fusable is False on CPU tensors.
+    assert not designer._fusable(designer._posterior)
+
+  def test_set_pe_with_pending_trials(self):
+    designer = VizierGPUCBPEBandit(
+        make_problem(), cfg(optimize_set_acquisition_for_exploration=True,
+                            max_evaluations=600), seed=9)
+    trials = []
+    rng = np.random.default_rng(3)
+    for uid in range(1, 7):
+      s = vz.TrialSuggestion(
+          {f'x{i}': float(v) for i, v in enumerate(rng.uniform(0, 1, 3))})
+      t = s.to_trial(uid)
+      t.complete(vz.Measurement(metrics={'obj': evaluate(s)}))
+      trials.append(t)
+    active = [vz.TrialSuggestion(
+        {f'x{i}': 0.8 for i in range(3)}).to_trial(50)]
+    designer.update(CompletedTrials(trials), ActiveTrials(active))
+    batch = designer.suggest(3)
+    kinds = [s.metadata.abs_ns(('gp_ucb_pe',))['acquisition']
+             for s in batch]
+    # Pending trials condition the set covariance; batch still mixes
+    # one UCB (new data) + jointly-optimized set members.
+    assert kinds[0] == 'ucb'
+    assert kinds[1:] == ['set_pe', 'set_pe']
