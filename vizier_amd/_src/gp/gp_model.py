@@ -169,15 +169,20 @@ def _use_custom_chol(K: torch.Tensor) -> bool:
   mode = os.environ.get('VIZIER_AMD_CUSTOM_CHOL', 'both')
   if mode not in ('1', 'both'):
     return False
+  # N <= 1024 only: at N=2000 the 63 launch-serial panel rounds lose
+  # to MAGMA (measured 11 ms/call vs ~4, profiles/q3_kernels_r2.txt);
+  # at N=1000 v2 wins (fit 153.6 -> 150 ms).
   return (K.is_cuda and K.dtype == torch.float32 and
           not K.requires_grad and K.shape[0] > 1 and
-          K.shape[-1] <= 2048 and ops.extension_available())
+          K.shape[-1] <= 1024 and ops.extension_available())
 
 
 def _use_custom_trsv(K: torch.Tensor) -> bool:
-  """trsv-only mode (VIZIER_AMD_CUSTOM_CHOL=trsv): MAGMA factorization
-  + the batched wave-synchronous solve."""
-  if os.environ.get('VIZIER_AMD_CUSTOM_CHOL', 'both') != 'trsv':
+  """MAGMA factorization + the batched wave-synchronous solve: applies
+  whenever the custom solve is enabled (any non-'0' mode) but the full
+  custom factorization is not (e.g. 1024 < N <= 2048)."""
+  if os.environ.get('VIZIER_AMD_CUSTOM_CHOL', 'both') not in (
+      'trsv', 'both', '1'):
     return False
   return (K.is_cuda and K.dtype == torch.float32 and
           not K.requires_grad and K.shape[0] > 1 and
