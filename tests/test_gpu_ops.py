@@ -765,6 +765,43 @@ class TestGramFp8MFMA:
     assert float((percall - fp32).abs().max()) < 0.15
     assert float((got - percall).abs().max()) < 0.2
 
+  def test_fp8_fused_mean_std_matches_cached_gram(self, ext):
+    # posterior_mean_std_fp8 (kernel-side e4m3fn decode) must agree
+    # with the torch Fp8GramCache composed math on the same operands.
+    from vizier_amd._src.gp import acquisitions as acq_lib
+    from vizier_amd._src.gp import gp_model
+    g = torch.Generator().manual_seed(21)
+    x = torch.rand(150, 8, generator=g)
+    y = torch.sin(3 * x[:, 0]) + 0.1 * torch.randn(150, generator=g)
+    post = gp_model.train_gp(x, y, num_restarts=1, max_iters=15,
+                             seed=3)
+    from vizier_amd._src.gp.gp_model import GPParams, GPPosterior
+    params = GPParams(amplitude=post.params.amplitude.cuda(),
+                      noise=post.params.noise.cuda(),
+                      lengthscales=post.params.lengthscales.cuda(),
+                      mean=post.params.mean.cuda())
+    gp = GPPosterior(x=post.x.cuda(), params=params, L=post.L.cuda(),
+                     alpha=post.alpha.cuda(), K_inv=post.K_inv.cuda(),
+                     nll=post.nll)
+    cache = acq_lib.Fp8GramCache(gp.x, gp.params.lengthscales,
+                                 float(gp.params.amplitude))
+    xq = torch.rand(25, 8, generator=g).cuda()
+    onehot = torch.zeros(8, dtype=torch.uint8).cuda()
+    mean, sd, dist = ext.posterior_mean_std_fp8(
+        xq, gp.x, cache.z2q, cache.n2, cache.scale,
+        gp.params.lengthscales, float(gp.params.amplitude),
+        float(gp.params.mean), gp.alpha, gp.K_inv, onehot)
+    # Composed oracle with the SAME cached gram.
+    k = cache.gram(xq)
+    amp2 = float(gp.params.amplitude) ** 2
+    want_mean = float(gp.params.mean) + k @ gp.alpha
+    want_var = (amp2 - (k * (k @ gp.K_inv)).sum(-1)).clamp_min(1e-12)
+    assert torch.allclose(mean, want_mean, atol=1e-3)
+    assert torch.allclose(sd, want_var.sqrt(), atol=5e-3)
+    # Trust-region distance: min L-inf to the training rows.
+    want_dist = (xq.unsqueeze(1) - gp.x).abs().amax(-1).amin(-1)
+    assert torch.allclose(dist, want_dist, atol=1e-5)
+
   def test_fp8_diag_near_exact(self, ext):
     x = torch.rand(64, 16).cuda()
     ls = torch.full((16,), 0.5).cuda()

@@ -316,7 +316,7 @@ class VizierGPBandit(Designer, Predictor):
       # is hipGraph-capturable (the eager chain is not on ROCm 7.2).
       fused_mo = (
           self._mt_posterior is None and
-          cfg.scorer_gram_dtype == 'fp32' and
+          cfg.scorer_gram_dtype in ('fp32', 'fp8') and
           self._x.is_cuda and self._x.dtype == torch.float32 and
           all(isinstance(p, gp_model.GPPosterior) and
               p.K_inv is not None for p in self._posteriors) and
@@ -340,14 +340,29 @@ class VizierGPBandit(Designer, Predictor):
                      if trust_region is not None else 0.0)
         amps = [float(p.params.amplitude) for p in posts]
         mean_cs = [float(p.params.mean) for p in posts]
+        # fp8 mode: quantized training operands cached per posterior
+        # (config 5's named dtype) — the per-metric kernel decodes
+        # them in place of the fp32 k-vector pass.
+        fp8_caches_fused = None
+        if cfg.scorer_gram_dtype == 'fp8':
+          fp8_caches_fused = [acq_lib.Fp8GramCache(
+              p.x, p.params.lengthscales, amps[i])
+              for i, p in enumerate(posts)]
 
         def score_fn(batch: CandidateBatch) -> torch.Tensor:
           xs = self._codec.decode(batch)[:, 0, :]
           means, sds, dist = [], [], None
           for i, post in enumerate(posts):
-            m_, s_, d_ = ext.posterior_mean_std(
-                xs, post.x, post.params.lengthscales, amps[i],
-                mean_cs[i], post.alpha, post.K_inv, onehot)
+            if fp8_caches_fused is not None:
+              cache = fp8_caches_fused[i]
+              m_, s_, d_ = ext.posterior_mean_std_fp8(
+                  xs, post.x, cache.z2q, cache.n2, cache.scale,
+                  post.params.lengthscales, amps[i], mean_cs[i],
+                  post.alpha, post.K_inv, onehot)
+            else:
+              m_, s_, d_ = ext.posterior_mean_std(
+                  xs, post.x, post.params.lengthscales, amps[i],
+                  mean_cs[i], post.alpha, post.K_inv, onehot)
             means.append(m_)
             sds.append(s_)
             dist = d_

@@ -63,6 +63,13 @@ extern "C" void launch_ps_quadform_big(
     const float* k_ws, const float* kinv, float* part, float* quad,
     int b, int n, hipStream_t stream);
 
+extern "C" void launch_ps_kvec_fp8(
+    const float* z1f, const float* n1, const float* xq, const float* x,
+    const unsigned char* z2q, const float* n2, const float* alpha,
+    const unsigned char* onehot, float* k_ws, float* mu_ws,
+    float* dist_ws, int b, int n, int d, int dp, float amp2,
+    float scale, hipStream_t stream);
+
 extern "C" void launch_ps_finalize_meanstd(
     const float* mu_ws, const float* var_ws, float* mean_out,
     float* sd_out, int b, float amp2, float mean_c, int nchunk,
@@ -601,6 +608,64 @@ std::vector<torch::Tensor> posterior_mean_std(
   return {mean, sd, dist_ws};
 }
 
+std::vector<torch::Tensor> posterior_mean_std_fp8(
+    torch::Tensor xq, torch::Tensor x, torch::Tensor z2q,
+    torch::Tensor n2, double scale, torch::Tensor lengthscales,
+    double amplitude, double mean_c, torch::Tensor alpha,
+    torch::Tensor kinv, torch::Tensor onehot) {
+  // (mean, sd, dist) with fp8 candidate grams against cached training
+  // operands (Fp8GramCache layout) — the per-metric piece of the
+  // fused MO fp8 scorer (config 5).
+  xq = check_f32(xq, "xq");
+  x = check_f32(x, "x");
+  n2 = check_f32(n2, "n2");
+  lengthscales = check_f32(lengthscales, "lengthscales");
+  alpha = check_f32(alpha, "alpha");
+  kinv = check_f32(kinv, "kinv");
+  TORCH_CHECK(z2q.scalar_type() == torch::kFloat8_e4m3fn,
+              "z2q must be float8_e4m3fn");
+  z2q = z2q.contiguous();
+  onehot = onehot.contiguous();
+  const int b = xq.size(0), d = xq.size(1), n = x.size(0);
+  const int dp = z2q.size(1);
+  TORCH_CHECK(dp % 32 == 0 && dp <= 512, "dp must be mult of 32, <=512");
+  // Candidate side: quantize + dequantize with torch (tiny, no sync).
+  auto z1 = xq / lengthscales;
+  auto z1q = torch::zeros({b, dp},
+                          xq.options().dtype(torch::kFloat8_e4m3fn));
+  z1q.index_put_({torch::indexing::Slice(),
+                  torch::indexing::Slice(0, d)},
+                 (z1 / scale).to(torch::kFloat8_e4m3fn));
+  auto z1f = z1q.to(torch::kFloat32) * scale;
+  auto n1 = (z1f * z1f).sum(-1);
+  auto k_ws = torch::empty({b, n}, xq.options());
+  auto mu_ws = torch::empty({b}, xq.options());
+  auto dist_ws = torch::empty({b}, xq.options());
+  auto mean = torch::empty({b}, xq.options());
+  auto sd = torch::empty({b}, xq.options());
+  const float amp2 = (float)(amplitude * amplitude);
+  launch_ps_kvec_fp8(
+      z1f.contiguous().data_ptr<float>(), n1.contiguous().data_ptr<float>(),
+      xq.data_ptr<float>(), x.data_ptr<float>(),
+      (const unsigned char*)z2q.data_ptr(), n2.data_ptr<float>(),
+      alpha.data_ptr<float>(), onehot.data_ptr<unsigned char>(),
+      k_ws.data_ptr<float>(), mu_ws.data_ptr<float>(),
+      dist_ws.data_ptr<float>(), b, n, d, dp, amp2, (float)scale,
+      current_stream());
+  static const int gemm_n_threshold = []() {
+    const char* sE = getenv("VIZIER_AMD_PS_GEMM_N");
+    return sE ? atoi(sE) : 4096;
+  }();
+  auto quad = (n < gemm_n_threshold && b <= 32)
+      ? quadform_tile_small_n(k_ws, kinv, b, n, current_stream())
+      : quadform_large_n(k_ws, kinv, b, n, current_stream());
+  launch_ps_finalize_meanstd_direct(
+      mu_ws.data_ptr<float>(), quad.data_ptr<float>(),
+      mean.data_ptr<float>(), sd.data_ptr<float>(), b, amp2,
+      (float)mean_c, current_stream());
+  return {mean, sd, dist_ws};
+}
+
 torch::Tensor hv_scalarize_tr(
     torch::Tensor means, torch::Tensor sds, torch::Tensor weights,
     c10::optional<torch::Tensor> ref, c10::optional<torch::Tensor> dist,
@@ -799,6 +864,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Batched forward substitution L z = b (gfx950)");
   m.def("gram_matern52_fp8_pre", &gram_matern52_fp8_pre,
         "fp8 cross-gram from pre-quantized operands (gfx950)");
+  m.def("posterior_mean_std_fp8", &posterior_mean_std_fp8,
+        "Fused (mean, sd, dist) with fp8 cached-operand grams");
   m.def("posterior_mean_std", &posterior_mean_std,
         "Fused (mean, sd, dist) for one GP over candidates (gfx950)");
   m.def("hv_scalarize_tr", &hv_scalarize_tr,

@@ -397,6 +397,92 @@ extern "C" void launch_ps_kvec_bf16(
                      dist_ws, b, n, d, dp, amp2);
 }
 
+// -- fp8 (e4m3fn) k-vector with cached training operands ----------------
+//
+// Same shape as the bf16 variant: the training side is quantized ONCE
+// per suggest (Fp8GramCache: z2q = fp8(x/ls/s), n2 = dequantized row
+// norms, s = unit-box bound so no per-call range scan). The CANDIDATE
+// side arrives pre-dequantized from the binding (4 tiny torch ops,
+// capture-safe); this kernel software-decodes the cached e4m3fn bytes
+// and computes d^2 = n1 + n2 - 2 s * z1f . z2 exactly as the fp8 MFMA
+// gram does (gram_matern52_fp8.hip).
+
+__device__ __forceinline__ float vz_fp8_to_f32(unsigned char v) {
+  const int sign = v >> 7;
+  const int e = (v >> 3) & 0xF;
+  const int m = v & 7;
+  float f;
+  if (e == 0) {
+    f = ldexpf((float)m, -9);              // subnormal: m * 2^-9
+  } else {
+    f = ldexpf(1.0f + m * 0.125f, e - 7);  // bias 7
+  }
+  return sign ? -f : f;
+}
+
+extern "C" __global__ __launch_bounds__(BLOCK) void
+ps_kvec_fp8_kernel(const float* __restrict__ z1f,   // (B, Dp) dequant
+                   const float* __restrict__ n1,    // (B,)
+                   const float* __restrict__ xq,    // (B, D) raw
+                   const float* __restrict__ x,     // (N, D) raw
+                   const unsigned char* __restrict__ z2q,  // (N, Dp)
+                   const float* __restrict__ n2,    // (N,)
+                   const float* __restrict__ alpha, // (N,)
+                   const unsigned char* __restrict__ onehot,
+                   float* __restrict__ k_out, float* __restrict__ mu_out,
+                   float* __restrict__ dist_out, int b, int n, int d,
+                   int dp, float amp2, float scale) {
+  __shared__ float red[8];
+  __shared__ float xq_lds[512];
+  __shared__ float z1_lds[512];
+  const int q = blockIdx.x;
+  if (q >= b) return;
+  const int tid = threadIdx.x;
+  for (int j = tid; j < dp; j += BLOCK) {
+    z1_lds[j] = z1f[(long)q * dp + j];
+    if (j < d) xq_lds[j] = xq[(long)q * d + j];
+  }
+  __syncthreads();
+  const float n1q = n1[q];
+  float mu_acc = 0.0f;
+  float min_linf = INFINITY;
+  for (int row = tid; row < n; row += BLOCK) {
+    const unsigned char* zr = z2q + (long)row * dp;
+    float dot = 0.0f;
+    for (int j = 0; j < dp; ++j) {
+      dot = fmaf(z1_lds[j], vz_fp8_to_f32(zr[j]), dot);
+    }
+    const float d2 = fmaxf(n1q + n2[row] - 2.0f * scale * dot, 0.0f);
+    const float kv = amp2 * matern52_of_d2(d2);
+    k_out[(long)q * n + row] = kv;
+    mu_acc = fmaf(kv, alpha[row], mu_acc);
+    const float* xr = x + (long)row * d;
+    float linf = 0.0f;
+    for (int j = 0; j < d; ++j) {
+      if (!onehot[j]) linf = fmaxf(linf, fabsf(xq_lds[j] - xr[j]));
+    }
+    min_linf = fminf(min_linf, linf);
+  }
+  auto fsum = [](float a, float c) { return a + c; };
+  auto fmin_ = [](float a, float c) { return fminf(a, c); };
+  float mu = block_reduce(mu_acc, red, fsum, 0.0f);
+  if (tid == 0) mu_out[q] = mu;
+  __syncthreads();
+  float dist = block_reduce(min_linf, red, fmin_, INFINITY);
+  if (tid == 0) dist_out[q] = dist;
+}
+
+extern "C" void launch_ps_kvec_fp8(
+    const float* z1f, const float* n1, const float* xq, const float* x,
+    const unsigned char* z2q, const float* n2, const float* alpha,
+    const unsigned char* onehot, float* k_ws, float* mu_ws,
+    float* dist_ws, int b, int n, int d, int dp, float amp2,
+    float scale, hipStream_t stream) {
+  hipLaunchKernelGGL(ps_kvec_fp8_kernel, dim3(b), dim3(BLOCK), 0,
+                     stream, z1f, n1, xq, x, z2q, n2, alpha, onehot,
+                     k_ws, mu_ws, dist_ws, b, n, d, dp, amp2, scale);
+}
+
 // Chunked quadform + finalize reuse the fp32 kernels below.
 
 // -- Large-N k-vector: row-split variant --------------------------------
