@@ -791,13 +791,30 @@ class TestGramFp8MFMA:
         xq, gp.x, cache.z2q, cache.n2, cache.scale,
         gp.params.lengthscales, float(gp.params.amplitude),
         float(gp.params.mean), gp.alpha, gp.K_inv, onehot)
-    # Composed oracle with the SAME cached gram.
-    k = cache.gram(xq)
+    # Oracle with torch's EXACT e4m3 dequantization (the kernel's
+    # software decode must match it bitwise; the MFMA gram kernel is a
+    # different fp8 dot implementation and only agrees to fp8
+    # tolerance).
+    d = 8
+    dp = cache.dp
+    z1 = xq / gp.params.lengthscales
+    z1q = torch.zeros(25, dp, dtype=torch.float8_e4m3fn, device='cuda')
+    z1q[:, :d] = (z1 / cache.scale).to(torch.float8_e4m3fn)
+    z1f = z1q.to(torch.float32) * cache.scale
+    n1 = (z1f * z1f).sum(-1)
+    z2f = cache.z2q.to(torch.float32) * cache.scale
+    d2 = (n1[:, None] + cache.n2[None, :] - 2 * z1f @ z2f.T).clamp_min(0)
+    r = d2.sqrt()
+    sr = (5.0 ** 0.5) * r
     amp2 = float(gp.params.amplitude) ** 2
+    k = amp2 * (1 + sr + sr * sr / 3) * torch.exp(-sr)
     want_mean = float(gp.params.mean) + k @ gp.alpha
     want_var = (amp2 - (k * (k @ gp.K_inv)).sum(-1)).clamp_min(1e-12)
-    assert torch.allclose(mean, want_mean, atol=1e-3)
+    assert torch.allclose(mean, want_mean, atol=1e-3),         float((mean - want_mean).abs().max())
     assert torch.allclose(sd, want_var.sqrt(), atol=5e-3)
+    # And the MFMA cached-gram path agrees to fp8-level tolerance.
+    k_mfma = cache.gram(xq)
+    assert float((k_mfma - k).abs().max()) < 0.2
     # Trust-region distance: min L-inf to the training rows.
     want_dist = (xq.unsqueeze(1) - gp.x).abs().amax(-1).amin(-1)
     assert torch.allclose(dist, want_dist, atol=1e-5)
