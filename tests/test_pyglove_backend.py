@@ -219,3 +219,86 @@ class TestBackendEndToEnd:
         backend1._study.resource_name
     f2 = backend2.next()
     assert f2.id != f1.id
+
+
+class TestBackendExtras:
+
+  def _spec(self):
+    from vizier_amd._src.pyglove import converters as cv
+    return cv.to_dna_spec(_mixed_problem().search_space)
+
+  def test_early_stopping_policy_flow(self):
+    """An injected pg.tuning.EarlyStoppingPolicy drives the service's
+    early-stop decisions through TunerPolicy.early_stop."""
+    from vizier_amd._src.pyglove import oss_vizier
+
+    class StopAll(pg.tuning.EarlyStoppingPolicy):
+      def __init__(self):
+        self.seen = []
+      def should_stop_early(self, trial):
+        self.seen.append(trial.id)
+        return True
+
+    stopper = StopAll()
+    oss_vizier.init()
+    backend = oss_vizier.OSSVizierBackend(
+        name='pgstop', group=0, dna_spec=self._spec(),
+        algorithm=RandomGenerator(seed=7),
+        metrics_to_optimize=['reward'], study_owner='tester',
+        early_stopping_policy=stopper)
+    feedback = backend.next()
+    # The feedback surface routes through the service's early-stopping
+    # operation machinery and back into the pyglove policy.
+    assert feedback.should_stop_early() is True
+    assert feedback.id in stopper.seen
+
+  def test_prior_study_warmup_feeds_algorithm(self):
+    """prior_study_ids warm the algorithm via recover() without adding
+    trials to the new study (backend.py:365-372)."""
+    from vizier_amd._src.pyglove import oss_vizier
+    oss_vizier.init()
+    # Build + complete a prior study.
+    algo1 = RandomGenerator(seed=8)
+    b1 = oss_vizier.OSSVizierBackend(
+        name='pgprior_src', group=0, dna_spec=self._spec(),
+        algorithm=algo1, metrics_to_optimize=['reward'],
+        study_owner='tester', num_examples=3)
+    while True:
+      try:
+        fb = b1.next()
+      except StopIteration:
+        break
+      fb.add_measurement(2.5, done=True)
+    prior_name = b1._study.resource_name
+
+    class RecordingGenerator(RandomGenerator):
+      def __init__(self):
+        super().__init__(seed=9)
+        self.recovered = []
+      def recover(self, history):
+        for dna, reward in history:
+          self.recovered.append((dict(dna.decisions), reward))
+
+    algo2 = RecordingGenerator()
+    b2 = oss_vizier.OSSVizierBackend(
+        name='pgprior_dst', group=0, dna_spec=self._spec(),
+        algorithm=algo2, metrics_to_optimize=['reward'],
+        study_owner='tester', prior_study_ids=[prior_name])
+    # recover() saw the prior study's completed trials with rewards.
+    assert len(algo2.recovered) >= 3
+    assert all(r == 2.5 for _, r in algo2.recovered)
+    # The NEW study has no prior trials (add_prior_trials=False).
+    assert len(list(b2._study.trials())) <= 1
+
+  def test_trial_metadata_roundtrip_through_dna(self):
+    from vizier_amd._src.pyglove import constants, converters as cv
+    converter = cv.VizierConverter.from_problem(_mixed_problem())
+    dna = pg.DNA({'lr': 0.02, 'layers': 3, 'units': 128.0,
+                  'opt': 'adam'}, metadata={'gen': 4})
+    dna.use_spec(converter.dna_spec)
+    trial = converter.to_trial(dna, fallback='raise_error')
+    stored = trial.metadata.ns(constants.METADATA_NAMESPACE)[
+        constants.TRIAL_METADATA_KEY_DNA_METADATA]
+    assert pg.from_json_str(stored) == {'gen': 4}
+    back = converter.to_dna(trial)
+    assert back.metadata == {'gen': 4}
