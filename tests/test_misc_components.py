@@ -147,6 +147,36 @@ class TestClassification:
     assert probs[0] > 0.9 and probs[1] < 0.1
 
 
+class TestClassificationValidation:
+
+  def test_rejects_bad_shapes_and_labels(self):
+    from sklearn.linear_model import LogisticRegression
+    x = np.zeros((10, 2))
+    with pytest.raises(ValueError, match='binary'):
+      SklearnClassifier(LogisticRegression(), features_train=x,
+                        labels_train=np.arange(10),
+                        features_test=x)()
+    with pytest.raises(ValueError, match='dims differ'):
+      SklearnClassifier(LogisticRegression(), features_train=x,
+                        labels_train=np.zeros(10),
+                        features_test=np.zeros((4, 3)))()
+    with pytest.raises(ValueError, match='row counts'):
+      SklearnClassifier(LogisticRegression(), features_train=x,
+                        labels_train=np.zeros(7),
+                        features_test=x)()
+
+  def test_decision_metric(self):
+    from sklearn.svm import LinearSVC
+    rng = np.random.default_rng(3)
+    x = rng.standard_normal((60, 2))
+    y = (x[:, 0] > 0).astype(int)
+    scores = SklearnClassifier(
+        LinearSVC(), features_train=x, labels_train=y,
+        features_test=np.array([[2.0, 0.0], [-2.0, 0.0]]),
+        eval_metric='decision')()
+    assert scores[0] > 0 > scores[1]
+
+
 class TestDemos:
 
   def test_client_demo_runs_in_process(self):
