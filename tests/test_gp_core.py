@@ -192,6 +192,35 @@ class TestAcquisitions:
     expected = 0.2 + 0.3 * 10 / (5 * 3)
     assert tr.trust_radius == pytest.approx(expected)
 
+  def test_qei_mc_sampling_covariance_is_correct(self):
+    # Regression: the batched qEI sampler must draw with covariance
+    # L L^T = cov (y = L eps). The transposed product L^T eps has both
+    # wrong marginal variances and wrong correlations for q > 1 (a
+    # real bug found in r2: einsum 'sbq,bqr' instead of 'sc,brc').
+    torch.manual_seed(0)
+    cov = torch.tensor([[[4.0, 1.9], [1.9, 1.0]]])   # corr ~ 0.95
+    mean = torch.zeros(1, 2)
+    eps = torch.randn(200000, 2)
+    # Reconstruct the sampler's draws to check their statistics.
+    L = torch.linalg.cholesky(cov + 1e-4 * cov.diagonal(
+        dim1=-2, dim2=-1).mean() * torch.eye(2))
+    y = torch.einsum('sc,brc->sbr', eps, L)[:, 0, :]
+    emp = (y.T @ y) / y.shape[0]
+    assert torch.allclose(emp, cov[0], atol=0.05), emp
+    # And the score itself is the MC qEI of that joint distribution.
+    scores = acq_lib.qei_mc_scores(mean, cov, eps, best_value=0.0)
+    want = (y - 0.0).clamp_min(0).amax(-1).mean()
+    assert float(scores[0]) == pytest.approx(float(want), rel=1e-4)
+
+  def test_qei_mc_diagonal_fallback(self):
+    # A non-PD covariance element must fall back to independent
+    # marginals instead of NaNs.
+    cov = torch.tensor([[[1.0, 2.0], [2.0, 1.0]]])   # indefinite
+    mean = torch.zeros(1, 2)
+    eps = torch.randn(512, 2)
+    scores = acq_lib.qei_mc_scores(mean, cov, eps, best_value=0.0)
+    assert torch.isfinite(scores).all()
+
   def test_hv_scalarization(self):
     s = acq_lib.create_hv_scalarization(100, 2, seed=0)
     ys = torch.tensor([[1.0, 1.0], [0.1, 0.1]])

@@ -507,25 +507,10 @@ class VizierGPBandit(Designer, Predictor):
       def score_fn(batch: CandidateBatch) -> torch.Tensor:
         dense = self._codec.decode(batch)          # (B, q, D)
         mean, cov = posterior_batched_cov(posterior, dense)
-        q = dense.shape[1]
-        eye = torch.eye(q, dtype=cov.dtype, device=cov.device)
-        # Near-duplicate candidate rows make cov singular; jitter
-        # relative to the diagonal scale, then fall back to a diagonal
-        # factor for any batch element that still fails. Branchless:
-        # a host `if bad.any()` would sync (and abort graph capture).
-        diag = cov.diagonal(dim1=-2, dim2=-1)
-        jitter = 1e-4 * diag.mean(-1, keepdim=True).clamp_min(1e-10)
-        cov = cov + jitter.unsqueeze(-1) * eye
-        L, info = torch.linalg.cholesky_ex(cov)    # (B, q, q)
-        bad = (info > 0)
-        L_diag = torch.diag_embed(diag.clamp_min(1e-12).sqrt())
-        L = torch.where(bad.view(-1, 1, 1), L_diag, L)
-        samples = mean.unsqueeze(0) + torch.einsum(
-            'sbq,bqr->sbr', eps_dev.expand(128, dense.shape[0], q), L)
-        scores = (samples - best_value).clamp_min(0).amax(-1).mean(0)
+        scores = acq_lib.qei_mc_scores(mean, cov, eps_dev[:, 0, :],
+                                       best_value)
         if trust_region is not None:
-          flat_scores = trust_region.apply(dense[:, 0, :], scores)
-          scores = flat_scores
+          scores = trust_region.apply(dense[:, 0, :], scores)
         return scores
       # Capture-eligible (the optimizer auto-falls-back if the batched
       # cholesky refuses capture on this build).
