@@ -181,32 +181,44 @@ batched_potrf_panel_kernel(float* __restrict__ A, int* __restrict__ info,
   float* M = A + (long)r * n * n;
   const int tid = threadIdx.x;
   const int nb = min(NB, n - k0);
-  for (int e = tid; e < nb * nb; e += CB) {
-    diag[e / nb][e % nb] = M[(long)(k0 + e / nb) * n + k0 + e % nb];
-  }
-  __syncthreads();
-  for (int j = 0; j < nb; ++j) {
-    if (tid == 0) {
-      const float v = diag[j][j];
-      if (v > 0.0f) {
-        diag[j][j] = sqrtf(v);
+  // Wave-synchronous 32x32 diagonal factor: lane l owns row l in
+  // registers; pivots broadcast with shfl — ZERO block syncs (the
+  // LDS/block-sync version cost ~25 of the panel kernel's 50 us).
+  if (tid < WAVE_SIZE) {
+    const int lane = tid;
+    float row[NB];
+#pragma unroll
+    for (int c = 0; c < NB; ++c) {
+      row[c] = (lane < nb && c < nb)
+          ? M[(long)(k0 + lane) * n + k0 + c] : 0.0f;
+    }
+#pragma unroll
+    for (int j = 0; j < NB; ++j) {
+      if (j >= nb) break;
+      const float piv = __shfl(row[j], j, WAVE_SIZE);
+      float d;
+      if (piv > 0.0f) {
+        d = sqrtf(piv);
       } else {
-        diag[j][j] = 1.0f;
-        if (info[r] == 0) info[r] = k0 + j + 1;
+        d = 1.0f;
+        if (lane == 0 && info[r] == 0) info[r] = k0 + j + 1;
+      }
+      if (lane == j) row[j] = d;
+      if (lane > j) row[j] /= d;
+#pragma unroll
+      for (int c = 0; c < NB; ++c) {
+        if (c > j && c < nb) {
+          const float lcj = __shfl(row[j], c, WAVE_SIZE);
+          if (lane >= c) row[c] -= row[j] * lcj;
+        }
       }
     }
-    __syncthreads();
-    const float dj = diag[j][j];
-    for (int i = j + 1 + tid; i < nb; i += CB) diag[i][j] /= dj;
-    __syncthreads();
-    const int rem = nb - j - 1;
-    for (int e = tid; e < rem * rem; e += CB) {
-      const int i = j + 1 + e / rem;
-      const int c = j + 1 + e % rem;
-      if (c <= i) diag[i][c] -= diag[i][j] * diag[c][j];
+#pragma unroll
+    for (int c = 0; c < NB; ++c) {
+      if (lane < nb && c < nb) diag[lane][c] = row[c];
     }
-    __syncthreads();
   }
+  __syncthreads();
   for (int e = tid; e < nb * nb; e += CB) {
     const int i = e / nb, c = e % nb;
     M[(long)(k0 + i) * n + k0 + c] = (c <= i) ? diag[i][c] : 0.0f;
