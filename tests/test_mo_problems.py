@@ -1,0 +1,190 @@
+"""Analytic-property tests for the native DTLZ/ZDT/WFG suites.
+
+Each test checks a mathematically known property of the canonical
+problem (Pareto-front identities, optima positions), so the
+implementations are validated without the reference's `optproblems`
+dependency.
+"""
+
+import numpy as np
+import pytest
+
+from vizier_amd import pyvizier as vz
+from vizier_amd._src.benchmarks.experimenters.synthetic import mo_problems
+
+
+def _eval(factory, x):
+  exp = factory()
+  problem = exp.problem_statement()
+  names = [pc.name for pc in problem.search_space.parameters]
+  trial = vz.Trial(parameters={n: float(v) for n, v in zip(names, x)})
+  exp.evaluate([trial])
+  ms = trial.final_measurement
+  return np.array([ms.metrics[m.name].value
+                   for m in problem.metric_information])
+
+
+class TestDTLZ:
+
+  def test_dtlz1_pareto_front_sums_to_half(self):
+    # With all distance params at 0.5, g=0 and sum of objectives = 0.5.
+    fac = mo_problems.DTLZExperimenterFactory(
+        name='DTLZ1', dim=7, num_objectives=3)
+    rng = np.random.default_rng(0)
+    for _ in range(5):
+      x = np.concatenate([rng.random(2), np.full(5, 0.5)])
+      f = _eval(fac, x)
+      assert np.all(f >= -1e-12)
+      np.testing.assert_allclose(f.sum(), 0.5, atol=1e-9)
+
+  @pytest.mark.parametrize('name', ['DTLZ2', 'DTLZ3', 'DTLZ4'])
+  def test_spherical_front_unit_norm(self, name):
+    # On the front (distance params at 0.5 for DTLZ2/3; any position),
+    # sum f_m^2 == 1.
+    fac = mo_problems.DTLZExperimenterFactory(
+        name=name, dim=6, num_objectives=3)
+    rng = np.random.default_rng(1)
+    for _ in range(5):
+      x = np.concatenate([rng.random(2), np.full(4, 0.5)])
+      f = _eval(fac, x)
+      np.testing.assert_allclose(np.sum(f ** 2), 1.0, atol=1e-9)
+
+  def test_dtlz5_degenerate_front(self):
+    fac = mo_problems.DTLZExperimenterFactory(
+        name='DTLZ5', dim=6, num_objectives=3)
+    f = _eval(fac, np.concatenate([[0.3, 0.9], np.full(4, 0.5)]))
+    np.testing.assert_allclose(np.sum(f ** 2), 1.0, atol=1e-9)
+
+  def test_dtlz6_off_front_positive_g(self):
+    fac = mo_problems.DTLZExperimenterFactory(
+        name='DTLZ6', dim=6, num_objectives=3)
+    f = _eval(fac, np.full(6, 0.8))
+    assert np.sum(f ** 2) > 1.0  # g > 0 pushes off the unit sphere
+
+  def test_dtlz7_last_objective_formula(self):
+    fac = mo_problems.DTLZExperimenterFactory(
+        name='DTLZ7', dim=6, num_objectives=3)
+    x = np.array([0.2, 0.7, 0.0, 0.0, 0.0, 0.0])
+    f = _eval(fac, x)
+    np.testing.assert_allclose(f[:2], [0.2, 0.7], atol=1e-12)
+    g = 1.0
+    h = 3 - sum(fi / (1 + g) * (1 + np.sin(3 * np.pi * fi))
+                for fi in (0.2, 0.7))
+    np.testing.assert_allclose(f[2], (1 + g) * h, atol=1e-9)
+
+  def test_unknown_name_raises(self):
+    fac = mo_problems.DTLZExperimenterFactory(
+        name='DTLZ9', dim=4, num_objectives=2)
+    with pytest.raises(ValueError, match='not a valid DTLZ'):
+      _eval(fac, np.full(4, 0.5))
+
+
+class TestZDT:
+
+  def test_zdt1_front(self):
+    fac = mo_problems.ZDTExperimenterFactory(name='ZDT1', dim=5)
+    for f1 in (0.0, 0.25, 1.0):
+      f = _eval(fac, np.array([f1, 0, 0, 0, 0]))
+      np.testing.assert_allclose(f, [f1, 1 - np.sqrt(f1)], atol=1e-9)
+
+  def test_zdt2_front(self):
+    fac = mo_problems.ZDTExperimenterFactory(name='ZDT2', dim=5)
+    f = _eval(fac, np.array([0.5, 0, 0, 0, 0]))
+    np.testing.assert_allclose(f, [0.5, 1 - 0.25], atol=1e-9)
+
+  def test_zdt3_matches_formula(self):
+    fac = mo_problems.ZDTExperimenterFactory(name='ZDT3', dim=4)
+    f1 = 0.3
+    f = _eval(fac, np.array([f1, 0, 0, 0]))
+    expected = 1 - np.sqrt(f1) - f1 * np.sin(10 * np.pi * f1)
+    # atol covers the float32 round-trip in the trial->array converter
+    # (sin(10*pi*f1) amplifies the 1e-8 parameter quantization ~30x).
+    np.testing.assert_allclose(f, [f1, expected], atol=1e-5)
+
+  def test_zdt4_optimum_at_half(self):
+    # Normalized 0.5 maps to canonical x_i = 0, where g = 1.
+    fac = mo_problems.ZDTExperimenterFactory(name='ZDT4', dim=3)
+    f = _eval(fac, np.array([0.09, 0.5, 0.5]))
+    np.testing.assert_allclose(f, [0.09, 1 - 0.3], atol=1e-9)
+
+  def test_zdt6_front(self):
+    fac = mo_problems.ZDTExperimenterFactory(name='ZDT6', dim=4)
+    x1 = 0.35
+    f = _eval(fac, np.array([x1, 0, 0, 0]))
+    f1 = 1 - np.exp(-4 * x1) * np.sin(6 * np.pi * x1) ** 6
+    np.testing.assert_allclose(f, [f1, 1 - f1 ** 2], atol=1e-9)
+
+  def test_zdt5_rejected(self):
+    fac = mo_problems.ZDTExperimenterFactory(name='ZDT5', dim=4)
+    with pytest.raises(ValueError, match='ZDT5'):
+      _eval(fac, np.full(4, 0.5))
+
+
+class TestWFG:
+
+  @pytest.mark.parametrize('name', [f'WFG{i}' for i in range(1, 10)])
+  def test_bounded_and_finite(self, name):
+    fac = mo_problems.WFGExperimenterFactory(
+        name=name, dim=7, num_objectives=2)
+    rng = np.random.default_rng(3)
+    for _ in range(20):
+      f = _eval(fac, rng.random(7))
+      assert np.all(np.isfinite(f))
+      # f_m in [0, 2m + D] for the canonical toolkit (S_m = 2m, D=1).
+      assert np.all(f >= -1e-9)
+      assert f[0] <= 3.0 + 1e-9 and f[1] <= 5.0 + 1e-9
+
+  @pytest.mark.parametrize('name', ['WFG4', 'WFG5'])
+  def test_concave_front_identity(self, name):
+    # Separable concave problems: distance params at the optimum
+    # (0.35 for s_multi/s_decept fixed points) put f on the ellipse
+    # sum (f_m / (2m))^2 == 1.
+    fac = mo_problems.WFGExperimenterFactory(
+        name=name, dim=7, num_objectives=2)
+    rng = np.random.default_rng(4)
+    for _ in range(5):
+      x = np.concatenate([[rng.random()], np.full(6, 0.35)])
+      f = _eval(fac, x)
+      np.testing.assert_allclose(
+          (f[0] / 2.0) ** 2 + (f[1] / 4.0) ** 2, 1.0, atol=1e-5)
+
+  def test_wfg6_front_identity(self):
+    fac = mo_problems.WFGExperimenterFactory(
+        name='WFG6', dim=7, num_objectives=2)
+    x = np.concatenate([[0.4], np.full(6, 0.35)])
+    f = _eval(fac, x)
+    np.testing.assert_allclose(
+        (f[0] / 2.0) ** 2 + (f[1] / 4.0) ** 2, 1.0, atol=1e-6)
+
+  def test_wfg_three_objectives(self):
+    fac = mo_problems.WFGExperimenterFactory(
+        name='WFG4', dim=8, num_objectives=3)
+    f = _eval(fac, np.concatenate([[0.2, 0.8], np.full(6, 0.35)]))
+    np.testing.assert_allclose(
+        np.sum((f / np.array([2.0, 4.0, 6.0])) ** 2), 1.0, atol=1e-6)
+
+  def test_odd_distance_dim_rejected(self):
+    with pytest.raises(ValueError, match='must be even'):
+      mo_problems.WFGExperimenterFactory(name='WFG1', dim=4,
+                                         num_objectives=2)
+
+
+class TestFactorySerialization:
+
+  @pytest.mark.parametrize('fac', [
+      mo_problems.DTLZExperimenterFactory(name='DTLZ2', dim=6,
+                                          num_objectives=3),
+      mo_problems.ZDTExperimenterFactory(name='ZDT3', dim=5),
+      mo_problems.WFGExperimenterFactory(name='WFG7', dim=8,
+                                         num_objectives=3),
+  ])
+  def test_dump_recover_roundtrip(self, fac):
+    recovered = type(fac).recover(fac.dump())
+    x = np.full(fac.dim, 0.4)
+    np.testing.assert_allclose(_eval(fac, x), _eval(recovered, x))
+
+  def test_public_api(self):
+    from vizier_amd.benchmarks import experimenters as pub
+    assert pub.DTLZExperimenterFactory is mo_problems.DTLZExperimenterFactory
+    assert pub.WFGExperimenterFactory is mo_problems.WFGExperimenterFactory
+    assert pub.ZDTExperimenterFactory is mo_problems.ZDTExperimenterFactory

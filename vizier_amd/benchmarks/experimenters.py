@@ -48,6 +48,11 @@ from vizier_amd._src.benchmarks.experimenters.synthetic.classic import (
     FixedMultiArmExperimenter,
     HartmannExperimenter,
 )
+from vizier_amd._src.benchmarks.experimenters.synthetic.mo_problems import (
+    DTLZExperimenterFactory,
+    WFGExperimenterFactory,
+    ZDTExperimenterFactory,
+)
 from vizier_amd._src.benchmarks.experimenters.synthetic.simplekd import (
     SimpleKDExperimenter,
 )
