@@ -178,12 +178,19 @@ batched_potrf_panel_kernel(float* __restrict__ A, int* __restrict__ info,
   __shared__ float diag[NB][NB + 1];
   const int r = blockIdx.x;
   if (r >= r_count) return;
+  // Grid (R, row-tiles): each workgroup REDUNDANTLY factors the 32x32
+  // diagonal block (deterministic, cheap — wave-synchronous with shfl
+  // broadcasts) then solves its OWN CB-row slice of the column panel
+  // L21 = A21 L11^-T. The r2 single-workgroup-per-matrix version put
+  // ~70 us x 32 panels on 3 CUs at the headline shape (R=3, N=1000,
+  // profiles/fit_kernels_headline.txt); slicing rows across blockIdx.y
+  // is what lets the panel use the chip.
+  const int row0 = k0 + NB + blockIdx.y * CB;
+  const bool has_rows = row0 < n;
+  if (blockIdx.y > 0 && !has_rows) return;
   float* M = A + (long)r * n * n;
   const int tid = threadIdx.x;
   const int nb = min(NB, n - k0);
-  // Wave-synchronous 32x32 diagonal factor: lane l owns row l in
-  // registers; pivots broadcast with shfl — ZERO block syncs (the
-  // LDS/block-sync version cost ~25 of the panel kernel's 50 us).
   if (tid < WAVE_SIZE) {
     const int lane = tid;
     float row[NB];
@@ -201,7 +208,9 @@ batched_potrf_panel_kernel(float* __restrict__ A, int* __restrict__ info,
         d = sqrtf(piv);
       } else {
         d = 1.0f;
-        if (lane == 0 && info[r] == 0) info[r] = k0 + j + 1;
+        if (blockIdx.y == 0 && lane == 0 && info[r] == 0) {
+          info[r] = k0 + j + 1;
+        }
       }
       if (lane == j) row[j] = d;
       if (lane > j) row[j] /= d;
@@ -219,12 +228,14 @@ batched_potrf_panel_kernel(float* __restrict__ A, int* __restrict__ info,
     }
   }
   __syncthreads();
-  for (int e = tid; e < nb * nb; e += CB) {
-    const int i = e / nb, c = e % nb;
-    M[(long)(k0 + i) * n + k0 + c] = (c <= i) ? diag[i][c] : 0.0f;
+  if (blockIdx.y == 0) {
+    for (int e = tid; e < nb * nb; e += CB) {
+      const int i = e / nb, c = e % nb;
+      M[(long)(k0 + i) * n + k0 + c] = (c <= i) ? diag[i][c] : 0.0f;
+    }
   }
-  __syncthreads();
-  for (int i = k0 + nb + tid; i < n; i += CB) {
+  const int i = row0 + tid;  // one row per thread in this tile
+  if (i < n) {
     float v[NB];
 #pragma unroll
     for (int j = 0; j < NB; ++j) {
@@ -281,10 +292,12 @@ batched_potrf_trailing_kernel(float* __restrict__ A, int r_count, int n,
 extern "C" void launch_batched_potrf(float* A, int* info, int r, int n,
                                      hipStream_t stream) {
   for (int k0 = 0; k0 < n; k0 += NB) {
-    hipLaunchKernelGGL(batched_potrf_panel_kernel, dim3(r), dim3(CB), 0,
-                       stream, A, info, r, n, k0);
     const int nb = (n - k0) < NB ? (n - k0) : NB;
-    const int jtiles = (n - (k0 + nb) + 63) / 64;
+    const int rows = n - (k0 + nb);
+    const int rtiles = rows > 0 ? (rows + CB - 1) / CB : 1;
+    hipLaunchKernelGGL(batched_potrf_panel_kernel, dim3(r, rtiles),
+                       dim3(CB), 0, stream, A, info, r, n, k0);
+    const int jtiles = (rows + 63) / 64;
     if (jtiles > 0) {
       hipLaunchKernelGGL(batched_potrf_trailing_kernel,
                          dim3(r, jtiles), dim3(CB), 0, stream, A, r, n,
