@@ -264,25 +264,40 @@ def nll_value_and_grad(raw: torch.Tensor, x: torch.Tensor,
   k0 = (1.0 + sr + sr * sr / 3.0) * e                        # unit-amp
   K = amp2 * k0 + params.noise.reshape(-1, 1, 1) * torch.eye(
       n, dtype=x.dtype, device=x.device)
-  L, info = safe_cholesky_ex(K)
   resid = (y.unsqueeze(0) - params.mean.unsqueeze(-1)).unsqueeze(-1)
-  zsol = torch.linalg.solve_triangular(L, resid, upper=False)
-  quad = (zsol * zsol).sum(dim=(-1, -2))
+  if _use_custom_chol(K):
+    # Headline-shape fast path: custom panel-swept potrf + the batched
+    # wave-synchronous solve instead of MAGMA's spotf2 launch storm
+    # (profiles/fit_kernels_headline.txt: the MAGMA+trtri+serial-trsv
+    # autograd machinery was ~45 ms of the 157 ms warm refit).
+    ext = ops.require_ext()
+    L, info = ext.batched_potrf(K.contiguous())
+    zs = ext.batched_trsv_lower(
+        L, resid.squeeze(-1).expand(K.shape[0], n).contiguous())
+    quad = (zs * zs).sum(dim=-1)
+  else:
+    L, info = safe_cholesky_ex(K)
+    zsol = torch.linalg.solve_triangular(L, resid, upper=False)
+    quad = (zsol * zsol).sum(dim=(-1, -2))
   logdet = 2.0 * torch.log(torch.diagonal(L, dim1=-2, dim2=-1)).sum(-1)
   nll = 0.5 * (quad + logdet + n * math.log(2 * math.pi))
   nll = nll + 0.01 * (raw * raw).sum(-1)
 
-  # alpha and K^-1 per restart via blocked forward solves (safe at any
-  # N; hipBLAS never sees a ~10^4-wide trsm).
-  alpha = torch.linalg.solve_triangular(L.mT, zsol, upper=True)  # (R,N,1)
+  # K^-1 via Z = L^-1 (trsm) then W = Z^T Z; alpha = K^-1 r = W r, so
+  # no serial upper trsv is needed. Huge N keeps the blocked per-restart
+  # solves (hipBLAS trsm ALLOC_FAILs at lda ~10^4); moderate N uses ONE
+  # batched trsm + batched GEMM.
   eye = torch.eye(n, dtype=x.dtype, device=x.device)
-  W = torch.empty_like(K)
-  for i in range(r_batch):
-    if n >= _NO_GRAD_FIT_N:
+  if n >= _NO_GRAD_FIT_N:
+    W = torch.empty_like(K)
+    for i in range(r_batch):
       zi = _blocked_solve_lower(L[i], eye)
-    else:
-      zi = torch.linalg.solve_triangular(L[i], eye, upper=False)
-    W[i] = zi.T @ zi
+      W[i] = zi.T @ zi
+  else:
+    Z = torch.linalg.solve_triangular(
+        L, eye.expand(r_batch, n, n), upper=False)
+    W = Z.mT @ Z
+  alpha = W @ resid                                          # (R, N, 1)
   M = W - alpha @ alpha.mT                                   # (R, N, N)
 
   grad = torch.zeros_like(raw)
@@ -438,13 +453,21 @@ def train_gp(x: torch.Tensor, y: torch.Tensor, *,
   def loss_fn(raw: torch.Tensor) -> torch.Tensor:
     return negative_log_marginal_likelihood(raw, x, y)
 
-  # VIZIER_AMD_ANALYTIC_NLL: 'auto' (default — analytic gradients only
-  # at N >= _NO_GRAD_FIT_N where trsm-backward is broken), 'always', or
-  # 'never'. 'always' is a round-2 A/B candidate: the analytic path
-  # does one K^-1 build instead of autograd's taped solve chain.
+  # VIZIER_AMD_ANALYTIC_NLL: 'auto' (default), 'always', or 'never'.
+  # 'auto' uses analytic gradients (a) at N >= _NO_GRAD_FIT_N where
+  # torch's trsm-backward is broken, and (b) on the GPU fp32 path where
+  # the custom potrf/trsv kernels apply — there the analytic grad does
+  # ONE batched trsm + GEMM instead of autograd's MAGMA spotf2 +
+  # trtri/trsm backward + serial rocblas trsv chain (warm refit
+  # 146.8 -> 132.9 ms, headline bench 343 -> 324.7 ms; profiles/
+  # fit_kernels_headline.txt). Gradients are the exact trace-identity
+  # values, verified against autograd in tests/test_gp_core.py.
   analytic_mode = os.environ.get('VIZIER_AMD_ANALYTIC_NLL', 'auto')
   use_analytic = (analytic_mode == 'always' or
-                  (analytic_mode != 'never' and n >= _NO_GRAD_FIT_N))
+                  (analytic_mode != 'never' and
+                   (n >= _NO_GRAD_FIT_N or
+                    (x.is_cuda and x.dtype == torch.float32 and
+                     ops.extension_available()))))
   if max_iters <= 0:
     with torch.no_grad():
       f0 = loss_fn(raw0)
