@@ -289,6 +289,257 @@ batched_potrf_trailing_kernel(float* __restrict__ A, int r_count, int n,
   }
 }
 
+// -- v3: persistent cooperative factorization ---------------------------
+//
+// v2's 2-launches-per-panel structure costs 64 dependent launches per
+// factorization (~52 us panel + ~37 us trailing per round measured at
+// the headline shape, profiles/fit_kernels_headline2.txt) — ~2.8 ms
+// per potrf of which most is launch/turnaround, not math. v3 runs ALL
+// panels in ONE cooperatively-launched kernel with the eagle_sweep
+// sense-reversing grid barrier between phases (2 per 32-panel round).
+//
+// Work ownership: pair (r, t) owns 64-row tile t of matrix r for the
+// whole factorization. Cross-workgroup values (the factored diagonal
+// block, the L21 panel, the trailing updates) move through agent-scope
+// relaxed loads/stores (memory-side, bypassing the incoherent per-XCD
+// L2s) exactly like eagle_sweep.hip; every workgroup re-factors the
+// 32x32 diagonal block redundantly (cheap, deterministic) so only
+// bulk data crosses workgroups. Per-element arithmetic is IDENTICAL
+// to v2 (same solve order, same fma chains), so results match v2
+// bit-for-bit.
+
+__device__ __forceinline__ float cchol_load(const float* p) {
+  return __hip_atomic_load(p, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+}
+__device__ __forceinline__ void cchol_store(float* p, float v) {
+  __hip_atomic_store(p, v, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+}
+__device__ __forceinline__ int cchol_iload(const int* p) {
+  return __hip_atomic_load(p, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+}
+__device__ __forceinline__ void cchol_istore(int* p, int v) {
+  __hip_atomic_store(p, v, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+}
+
+__device__ __forceinline__ void chol_grid_sync(unsigned int* bar) {
+  __builtin_amdgcn_s_waitcnt(0);
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    const unsigned int gen =
+        __hip_atomic_load(bar + 1, __ATOMIC_RELAXED,
+                          __HIP_MEMORY_SCOPE_AGENT);
+    asm volatile("" ::: "memory");
+    const unsigned int arrived = __hip_atomic_fetch_add(
+        bar, 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    if (arrived == gridDim.x - 1) {
+      __hip_atomic_store(bar, 0u, __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_AGENT);
+      __builtin_amdgcn_s_waitcnt(0);
+      __hip_atomic_fetch_add(bar + 1, 1u, __ATOMIC_RELAXED,
+                             __HIP_MEMORY_SCOPE_AGENT);
+    } else {
+      while (__hip_atomic_load(bar + 1, __ATOMIC_RELAXED,
+                               __HIP_MEMORY_SCOPE_AGENT) == gen) {
+        __builtin_amdgcn_s_sleep(8);
+      }
+    }
+    asm volatile("" ::: "memory");
+  }
+  __syncthreads();
+}
+
+#define RB 64  // rows per ownership tile
+
+extern "C" __global__ __launch_bounds__(CB) void
+batched_potrf_coop_kernel(float* __restrict__ A, int* __restrict__ info,
+                          unsigned int* __restrict__ bar,
+                          int r_count, int n) {
+  __shared__ float diag[NB][NB + 1];
+  __shared__ float myl[RB][NB];
+  __shared__ float jpanel[RB][NB];
+  const int tid = threadIdx.x;
+  const int tiles = (n + RB - 1) / RB;
+  const int npairs = r_count * tiles;
+
+  for (int k0 = 0; k0 < n; k0 += NB) {
+    const int nb = min(NB, n - k0);
+    // Phase A/B: redundant diagonal factor + owned panel rows. The
+    // factored diagonal is NOT written back here — every pair reads
+    // the ORIGINAL diag values, so a writeback would race with other
+    // pairs' loads. The owner writes it in phase C (whose trailing
+    // updates touch disjoint rows) after a cheap redundant re-factor.
+    for (int pair = blockIdx.x; pair < npairs; pair += gridDim.x) {
+      const int r = pair / tiles, t = pair % tiles;
+      const int tile_lo = t * RB;
+      const int tile_hi = min(tile_lo + RB, n);
+      if (tile_hi <= k0 + nb) continue;  // no panel rows in this tile
+      float* M = A + (long)r * n * n;
+      if (tid < WAVE_SIZE) {
+        const int lane = tid;
+        float row[NB];
+#pragma unroll
+        for (int c = 0; c < NB; ++c) {
+          row[c] = (lane < nb && c < nb)
+              ? cchol_load(&M[(long)(k0 + lane) * n + k0 + c]) : 0.0f;
+        }
+#pragma unroll
+        for (int j = 0; j < NB; ++j) {
+          if (j >= nb) break;
+          const float piv = __shfl(row[j], j, WAVE_SIZE);
+          const float d = (piv > 0.0f) ? sqrtf(piv) : 1.0f;
+          if (lane == j) row[j] = d;
+          if (lane > j) row[j] /= d;
+#pragma unroll
+          for (int c = 0; c < NB; ++c) {
+            if (c > j && c < nb) {
+              const float lcj = __shfl(row[j], c, WAVE_SIZE);
+              if (lane >= c) row[c] -= row[j] * lcj;
+            }
+          }
+        }
+#pragma unroll
+        for (int c = 0; c < NB; ++c) {
+          if (lane < nb && c < nb) diag[lane][c] = row[c];
+        }
+      }
+      __syncthreads();
+      const int row_lo = max(tile_lo, k0 + nb);
+      const int i = row_lo + tid;  // one row per thread (<= 64 rows)
+      if (i < tile_hi) {
+        float v[NB];
+#pragma unroll
+        for (int j = 0; j < NB; ++j) {
+          if (j < nb) {
+            float xv = cchol_load(&M[(long)i * n + k0 + j]);
+#pragma unroll
+            for (int p = 0; p < NB; ++p) {
+              if (p < j) xv -= v[p] * diag[j][p];
+            }
+            v[j] = xv / diag[j][j];
+          }
+        }
+#pragma unroll
+        for (int j = 0; j < NB; ++j) {
+          if (j < nb) cchol_store(&M[(long)i * n + k0 + j], v[j]);
+        }
+      }
+      __syncthreads();  // diag LDS reused by the next pair
+    }
+    chol_grid_sync(bar);
+
+    // Phase C: owner writes the factored diagonal (re-factored from
+    // the still-original values — trailing writes touch disjoint
+    // rows), then the trailing update of owned rows (identical fma
+    // chains to batched_potrf_trailing_kernel).
+    for (int pair = blockIdx.x; pair < npairs; pair += gridDim.x) {
+      const int r = pair / tiles, t = pair % tiles;
+      const int tile_lo = t * RB;
+      const int tile_hi = min(tile_lo + RB, n);
+      float* M = A + (long)r * n * n;
+      if (t == k0 / RB) {
+        if (tid < WAVE_SIZE) {
+          const int lane = tid;
+          float row[NB];
+#pragma unroll
+          for (int c = 0; c < NB; ++c) {
+            row[c] = (lane < nb && c < nb)
+                ? cchol_load(&M[(long)(k0 + lane) * n + k0 + c]) : 0.0f;
+          }
+#pragma unroll
+          for (int j = 0; j < NB; ++j) {
+            if (j >= nb) break;
+            const float piv = __shfl(row[j], j, WAVE_SIZE);
+            float d;
+            if (piv > 0.0f) {
+              d = sqrtf(piv);
+            } else {
+              d = 1.0f;
+              if (lane == 0 && cchol_iload(&info[r]) == 0) {
+                cchol_istore(&info[r], k0 + j + 1);
+              }
+            }
+            if (lane == j) row[j] = d;
+            if (lane > j) row[j] /= d;
+#pragma unroll
+            for (int c = 0; c < NB; ++c) {
+              if (c > j && c < nb) {
+                const float lcj = __shfl(row[j], c, WAVE_SIZE);
+                if (lane >= c) row[c] -= row[j] * lcj;
+              }
+            }
+          }
+#pragma unroll
+          for (int c = 0; c < NB; ++c) {
+            if (lane < nb && c < nb) diag[lane][c] = row[c];
+          }
+        }
+        __syncthreads();
+        for (int e = tid; e < nb * nb; e += CB) {
+          const int i = e / nb, c = e % nb;
+          cchol_store(&M[(long)(k0 + i) * n + k0 + c],
+                      (c <= i) ? diag[i][c] : 0.0f);
+        }
+        __syncthreads();
+      }
+      const int row_lo = max(tile_lo, k0 + nb);
+      if (row_lo >= tile_hi) continue;
+      const int nrows = tile_hi - row_lo;
+      for (int e = tid; e < nrows * nb; e += CB) {
+        myl[e / nb][e % nb] =
+            cchol_load(&M[(long)(row_lo + e / nb) * n + k0 + e % nb]);
+      }
+      __syncthreads();
+      for (int jb = k0 + nb; jb < tile_hi; jb += RB) {
+        const int jl = min(RB, n - jb);
+        for (int e = tid; e < jl * nb; e += CB) {
+          jpanel[e / nb][e % nb] =
+              cchol_load(&M[(long)(jb + e / nb) * n + k0 + e % nb]);
+        }
+        __syncthreads();
+        const int rl = tid >> 2, jq = tid & 3;
+        const int i = row_lo + rl;
+        if (rl < nrows && i >= jb) {
+          for (int q = 0; q < 16; ++q) {
+            const int j = jq * 16 + q;
+            if (j < jl && jb + j <= i) {
+              float acc = 0.0f;
+#pragma unroll
+              for (int p = 0; p < NB; ++p) {
+                acc = fmaf(myl[rl][p], jpanel[j][p], acc);
+              }
+              float* dst = &M[(long)i * n + jb + j];
+              cchol_store(dst, cchol_load(dst) - acc);
+            }
+          }
+        }
+        __syncthreads();
+      }
+    }
+    chol_grid_sync(bar);
+  }
+}
+
+extern "C" int launch_batched_potrf_coop(float* A, int* info,
+                                         unsigned int* bar, int r, int n,
+                                         hipStream_t stream) {
+  int max_blocks = 0;
+  hipError_t err = hipOccupancyMaxActiveBlocksPerMultiprocessor(
+      &max_blocks, (const void*)batched_potrf_coop_kernel, CB, 0);
+  if (err != hipSuccess || max_blocks < 1) return -1;
+  hipDeviceProp_t prop;
+  if (hipGetDeviceProperties(&prop, 0) != hipSuccess) return -1;
+  const int tiles = (n + RB - 1) / RB;
+  int grid = r * tiles;
+  const int cap = max_blocks * prop.multiProcessorCount;
+  if (grid > cap) grid = cap;
+  if (grid < 1) grid = 1;
+  void* args[] = {&A, &info, &bar, &r, &n};
+  err = hipLaunchCooperativeKernel(
+      (const void*)batched_potrf_coop_kernel, dim3(grid), dim3(CB),
+      args, 0, stream);
+  return err == hipSuccess ? 0 : -1;
+}
+
 extern "C" void launch_batched_potrf(float* A, int* info, int r, int n,
                                      hipStream_t stream) {
   for (int k0 = 0; k0 < n; k0 += NB) {

@@ -1,7 +1,11 @@
 // Python bindings for the vizier_amd gfx950 kernels (torch extension).
 
 #include <ATen/cuda/CUDAContext.h>
-#include <torch/extension.h>\n#include <algorithm>\n#include <string>
+#include <torch/extension.h>
+
+#include <algorithm>
+#include <cstdlib>
+#include <string>
 
 #include <hip/hip_runtime.h>
 
@@ -109,6 +113,9 @@ extern "C" void launch_ps_quadform_finalize(
 
 extern "C" void launch_batched_potrf(float* A, int* info, int r, int n,
                                      hipStream_t stream);
+extern "C" int launch_batched_potrf_coop(float* A, int* info,
+                                         unsigned int* bar, int r, int n,
+                                         hipStream_t stream);
 extern "C" void launch_batched_trsv_lower(const float* L, float* b,
                                           int r, int n,
                                           hipStream_t stream);
@@ -704,8 +711,28 @@ std::vector<torch::Tensor> batched_potrf(torch::Tensor K) {
   const int r = K.size(0), n = K.size(1);
   auto L = K.clone();
   auto info = torch::zeros({r}, K.options().dtype(torch::kInt32));
-  launch_batched_potrf(L.data_ptr<float>(), info.data_ptr<int>(), r, n,
-                       current_stream());
+  // Persistent cooperative factorization (one launch for all panels,
+  // eagle_sweep-style grid barriers) unless opted out or the
+  // cooperative launch is unavailable; the 2-launches-per-panel v2
+  // loop is the fallback. Results are bit-identical.
+  static const bool coop_enabled = []() {
+    const char* env = std::getenv("VIZIER_AMD_COOP_CHOL");
+    return env == nullptr || std::string(env) != "0";
+  }();
+  bool done = false;
+  if (coop_enabled) {
+    // Stream-ordered temp: the caching allocator only reuses this
+    // block behind the kernel on the same stream.
+    auto bar = torch::zeros({2}, K.options().dtype(torch::kInt32));
+    done = launch_batched_potrf_coop(
+               L.data_ptr<float>(), info.data_ptr<int>(),
+               reinterpret_cast<unsigned int*>(bar.data_ptr<int>()),
+               r, n, current_stream()) == 0;
+  }
+  if (!done) {
+    launch_batched_potrf(L.data_ptr<float>(), info.data_ptr<int>(), r, n,
+                         current_stream());
+  }
   return {L, info};
 }
 
