@@ -321,9 +321,9 @@ class TestBatchedCholesky:
     assert int(info[0]) == 0 and int(info[2]) == 0
     assert int(info[1]) == 11  # 1-based failing column
 
-  def test_potrf_coop_shapes(self, ext):
-    """Cooperative-kernel edge shapes: single matrix, odd N, batch 12
-    (the line-search ladder), and N crossing many 64-row tiles."""
+  def test_potrf_edge_shapes(self, ext):
+    """Edge shapes: single matrix, odd N, batch 12 (the line-search
+    ladder), and N crossing many panel rounds."""
     g = torch.Generator().manual_seed(17)
     for r, n in ((1, 257), (3, 999), (12, 1000), (2, 64), (4, 65)):
       A = torch.randn(r, n, 32, generator=g)
@@ -334,38 +334,13 @@ class TestBatchedCholesky:
       err = (torch.tril(L) - want).abs().max()
       assert float(err) < 5e-4, f'r={r} n={n}: {err}'
 
-  def test_potrf_coop_flags_non_pd_across_tiles(self, ext):
-    # Failure column inside a later 64-row ownership tile.
+  def test_potrf_flags_non_pd_across_panels(self, ext):
+    # Failure column inside a later panel round.
     K = torch.eye(300).repeat(2, 1, 1).cuda().contiguous()
     K[0, 170, 170] = -2.0
     _, info = ext.batched_potrf(K)
     assert int(info[1]) == 0
     assert int(info[0]) == 171
-
-  def test_potrf_coop_matches_v2_bitwise(self, ext, tmp_path):
-    """The cooperative kernel claims BIT-identical results to the
-    2-launches-per-panel v2 loop; verify via a subprocess with
-    VIZIER_AMD_COOP_CHOL=0 (the switch is latched per process)."""
-    import os
-    import subprocess
-    import sys
-    g = torch.Generator().manual_seed(19)
-    A = torch.randn(5, 777, 32, generator=g)
-    K = (A @ A.mT / 32 + torch.eye(777)).cuda().contiguous()
-    L, info = ext.batched_potrf(K)
-    kpath, lpath = str(tmp_path / 'k.pt'), str(tmp_path / 'l.pt')
-    torch.save(K.cpu(), kpath)
-    env = dict(os.environ, VIZIER_AMD_COOP_CHOL='0')
-    code = (
-        'import torch; from vizier_amd._src import ops;'
-        f'K = torch.load({kpath!r}).cuda().contiguous();'
-        'L, info = ops.require_ext().batched_potrf(K);'
-        f'torch.save((L.cpu(), info.cpu()), {lpath!r})')
-    subprocess.run([sys.executable, '-c', code], check=True, env=env,
-                   timeout=300)
-    L2, info2 = torch.load(lpath)
-    assert torch.equal(info.cpu(), info2)
-    assert torch.equal(torch.tril(L).cpu(), torch.tril(L2))
 
   def test_trsv_matches_torch(self, ext):
     g = torch.Generator().manual_seed(12)

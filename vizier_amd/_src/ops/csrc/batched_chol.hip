@@ -289,24 +289,20 @@ batched_potrf_trailing_kernel(float* __restrict__ A, int r_count, int n,
   }
 }
 
-// -- v3: persistent cooperative factorization ---------------------------
+// -- v3: persistent cooperative factorization (OPT-IN, measured slower)
 //
-// v2's 2-launches-per-panel structure costs 64 dependent launches per
-// factorization (~52 us panel + ~37 us trailing per round measured at
-// the headline shape, profiles/fit_kernels_headline2.txt) — ~2.8 ms
-// per potrf of which most is launch/turnaround, not math. v3 runs ALL
-// panels in ONE cooperatively-launched kernel with the eagle_sweep
-// sense-reversing grid barrier between phases (2 per 32-panel round).
-//
-// Work ownership: pair (r, t) owns 64-row tile t of matrix r for the
-// whole factorization. Cross-workgroup values (the factored diagonal
-// block, the L21 panel, the trailing updates) move through agent-scope
-// relaxed loads/stores (memory-side, bypassing the incoherent per-XCD
-// L2s) exactly like eagle_sweep.hip; every workgroup re-factors the
-// 32x32 diagonal block redundantly (cheap, deterministic) so only
-// bulk data crosses workgroups. Per-element arithmetic is IDENTICAL
-// to v2 (same solve order, same fma chains), so results match v2
-// bit-for-bit.
+// One cooperative launch for all panels with eagle_sweep-style grid
+// barriers, pair (r, t) owning 64-row tile t. MEASURED RESULT at the
+// headline shape: warm refit 132.9 -> 240.9 ms — SLOWER than v2's 64
+// dependent launches. Root cause: a right-looking factorization
+// rewrites the whole trailing matrix every round, and cross-workgroup
+// coherence inside one kernel forces those reads+writes through
+// agent-scope (memory-side, uncached) accesses — ~1.3 GB of L2-bypass
+// traffic per N=1000 factorization, which costs more than the launch
+// turnarounds it saves. (eagle_sweep wins with this pattern because
+// its bulk data — K^-1, x, alpha — is READ-ONLY and stays L2-cached;
+// only small buffers cross workgroups.) Kept opt-in
+// (VIZIER_AMD_COOP_CHOL=1) as a documented experiment.
 
 __device__ __forceinline__ float cchol_load(const float* p) {
   return __hip_atomic_load(p, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);

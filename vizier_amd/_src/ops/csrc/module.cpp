@@ -712,12 +712,17 @@ std::vector<torch::Tensor> batched_potrf(torch::Tensor K) {
   auto L = K.clone();
   auto info = torch::zeros({r}, K.options().dtype(torch::kInt32));
   // Persistent cooperative factorization (one launch for all panels,
-  // eagle_sweep-style grid barriers) unless opted out or the
-  // cooperative launch is unavailable; the 2-launches-per-panel v2
-  // loop is the fallback. Results are bit-identical.
+  // eagle_sweep-style grid barriers) — OPT-IN via
+  // VIZIER_AMD_COOP_CHOL=1. Measured SLOWER than the v2 multi-launch
+  // loop at the headline shape (warm refit 132.9 -> 240.9 ms): unlike
+  // eagle_sweep, whose bulk data is read-only and L2-cached, a
+  // right-looking factorization rewrites the whole trailing matrix
+  // every round, and routing that through agent-scope (memory-side)
+  // accesses for cross-workgroup coherence costs more than the 64
+  // launch turnarounds it saves. Kept for reference/experiments.
   static const bool coop_enabled = []() {
     const char* env = std::getenv("VIZIER_AMD_COOP_CHOL");
-    return env == nullptr || std::string(env) != "0";
+    return env != nullptr && std::string(env) == "1";
   }();
   bool done = false;
   if (coop_enabled) {
