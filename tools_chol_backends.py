@@ -12,7 +12,7 @@ import time
 import torch
 
 sys.path.insert(0, '.')
-from vizier_amd._src import ops  # noqa: E402
+from vizier_amd._src.ops import dispatch as ops  # noqa: E402
 
 
 def bench(fn, n_iter=20):
