@@ -180,13 +180,16 @@ def _use_custom_chol(K: torch.Tensor) -> bool:
 def _use_custom_trsv(K: torch.Tensor) -> bool:
   """MAGMA factorization + the batched wave-synchronous solve: applies
   whenever the custom solve is enabled (any non-'0' mode) but the full
-  custom factorization is not (e.g. 1024 < N <= 2048)."""
+  custom factorization is not. N <= 1200: the one-workgroup-per-matrix
+  solve wins at N=1000 (0.48 vs 0.71 ms rocblas) but LOSES at N=2000
+  (3.1 vs 1.5 ms — 63 serial panel rounds on one CU;
+  tools_chol_backends.py probe)."""
   if os.environ.get('VIZIER_AMD_CUSTOM_CHOL', 'both') not in (
       'trsv', 'both', '1'):
     return False
   return (K.is_cuda and K.dtype == torch.float32 and
           not K.requires_grad and K.shape[0] > 1 and
-          K.shape[-1] <= 2048 and ops.extension_available())
+          K.shape[-1] <= 1200 and ops.extension_available())
 
 
 def negative_log_marginal_likelihood(raw: torch.Tensor, x: torch.Tensor,

@@ -11,6 +11,9 @@
 // line-search (torch.no_grad) calls these.
 
 #include <hip/hip_runtime.h>
+
+#include <cstdlib>
+
 #include "common.h"
 
 #define CB 256
@@ -536,8 +539,178 @@ extern "C" int launch_batched_potrf_coop(float* A, int* info,
   return err == hipSuccess ? 0 : -1;
 }
 
-extern "C" void launch_batched_potrf(float* A, int* info, int r, int n,
-                                     hipStream_t stream) {
+// -- v4: fused round kernel (default) -----------------------------------
+//
+// ONE launch per 32-panel round instead of v2's two (panel + trailing),
+// with NO cross-workgroup coherence traffic: round k's kernel never
+// writes anything another workgroup of the same launch reads. The
+// trick is write deferral + recomputation:
+//   - the factored 32x32 diagonal stays OUT of A until the final
+//     cleanup kernel (each consumer re-factors it from A's stable
+//     values in-wave — ~1 us, deterministic);
+//   - L21 panel rows are recomputed from A's original column values by
+//     every workgroup that needs them (solve vs the LDS diagonal), and
+//     only written back for round k-1 (whose columns nothing in round
+//     k reads);
+//   - the trailing update is in-place on disjoint (i, j >= jb) tiles.
+// Per-element fma chains are IDENTICAL to v2, so v4 results are
+// bit-identical to v2's. Measured: the A/B that motivated this is the
+// ~44 us launch+L2-flush turnaround BETWEEN dependent kernels (potrf
+// R=3 == R=12 == 2.85 ms on v2 — launch-bound, not work-bound).
+
+__device__ __forceinline__ void wave_factor_diag(
+    const float* __restrict__ M, int n, int k0, int nb,
+    float diag_out[NB][NB + 1], int* fail_col /* -1 if ok */) {
+  const int lane = threadIdx.x;
+  float row[NB];
+  int fc = -1;
+#pragma unroll
+  for (int c = 0; c < NB; ++c) {
+    row[c] = (lane < nb && c < nb)
+        ? M[(long)(k0 + lane) * n + k0 + c] : 0.0f;
+  }
+#pragma unroll
+  for (int j = 0; j < NB; ++j) {
+    if (j >= nb) break;
+    const float piv = __shfl(row[j], j, WAVE_SIZE);
+    float d;
+    if (piv > 0.0f) {
+      d = sqrtf(piv);
+    } else {
+      d = 1.0f;
+      if (fc < 0) fc = k0 + j + 1;
+    }
+    if (lane == j) row[j] = d;
+    if (lane > j) row[j] /= d;
+#pragma unroll
+    for (int c = 0; c < NB; ++c) {
+      if (c > j && c < nb) {
+        const float lcj = __shfl(row[j], c, WAVE_SIZE);
+        if (lane >= c) row[c] -= row[j] * lcj;
+      }
+    }
+  }
+#pragma unroll
+  for (int c = 0; c < NB; ++c) {
+    if (lane < nb && c < nb) diag_out[lane][c] = row[c];
+  }
+  *fail_col = fc;
+}
+
+__device__ __forceinline__ void solve_panel_row(
+    const float* __restrict__ M, int n, int i, int k0, int nb,
+    const float diag[NB][NB + 1], float v[NB]) {
+#pragma unroll
+  for (int j = 0; j < NB; ++j) {
+    if (j < nb) {
+      float xv = M[(long)i * n + k0 + j];
+#pragma unroll
+      for (int p = 0; p < NB; ++p) {
+        if (p < j) xv -= v[p] * diag[j][p];
+      }
+      v[j] = xv / diag[j][j];
+    }
+  }
+}
+
+extern "C" __global__ __launch_bounds__(CB) void
+batched_potrf_round_kernel(float* __restrict__ A, int* __restrict__ info,
+                           int r_count, int n, int k0) {
+  __shared__ float pdiag[NB][NB + 1];  // factored diag of round k-1
+  __shared__ float diag[NB][NB + 1];   // factored diag of round k
+  __shared__ float jpanel[64][NB];     // L21^k rows of this j-tile
+  const int r = blockIdx.x;
+  if (r >= r_count) return;
+  float* M = A + (long)r * n * n;
+  const int tid = threadIdx.x;
+  const int nb = min(NB, n - k0);
+  const int k0p = k0 - NB;
+
+  // Step 0: finalize round k-1 — write L21^{k-1} for this tile's rows
+  // (columns k0p, which nothing in round k reads).
+  if (k0p >= 0) {
+    int fcp = -1;
+    if (tid < WAVE_SIZE) {
+      wave_factor_diag(M, n, k0p, NB, pdiag, &fcp);
+    }
+    __syncthreads();
+    if (blockIdx.y == 0 && tid == 0 && fcp > 0 && info[r] == 0) {
+      info[r] = fcp;  // unique writer per launch; launches are ordered
+    }
+    const int w0 = k0 + (int)blockIdx.y * 64;
+    for (int i = w0 + tid; i < min(w0 + 64, n); i += CB) {
+      float v[NB];
+      solve_panel_row(M, n, i, k0p, NB, pdiag, v);
+#pragma unroll
+      for (int j = 0; j < NB; ++j) {
+        M[(long)i * n + k0p + j] = v[j];
+      }
+    }
+  }
+
+  // Step 1: trailing update for this j-tile (jb >= k0+nb), using L21^k
+  // recomputed from A's still-original panel columns.
+  const int jb = k0 + nb + (int)blockIdx.y * 64;
+  if (jb >= n || nb < NB) return;  // last round has no trailing
+  int fc = -1;
+  if (tid < WAVE_SIZE) {
+    wave_factor_diag(M, n, k0, nb, diag, &fc);
+  }
+  __syncthreads();
+  const int jl = min(64, n - jb);
+  // Recompute L21 rows of this j-tile into LDS (identical solve chain).
+  for (int j = tid; j < jl; j += CB) {
+    float v[NB];
+    solve_panel_row(M, n, jb + j, k0, nb, diag, v);
+#pragma unroll
+    for (int p = 0; p < NB; ++p) jpanel[j][p] = v[p];
+  }
+  __syncthreads();
+  for (int i = jb + tid; i < n; i += CB) {
+    float v[NB];
+    solve_panel_row(M, n, i, k0, nb, diag, v);
+    for (int j = 0; j < jl; ++j) {
+      if (jb + j > i) break;
+      float acc = 0.0f;
+#pragma unroll
+      for (int p = 0; p < NB; ++p) {
+        acc = fmaf(v[p], jpanel[j][p], acc);
+      }
+      M[(long)i * n + jb + j] -= acc;
+    }
+  }
+}
+
+// Final cleanup: factor every diagonal block in place (their inputs
+// are stable once the preceding trailing rounds ran), zero the upper
+// part, and record a failure in the LAST block (earlier blocks were
+// recorded by their successor round's step 0).
+extern "C" __global__ __launch_bounds__(CB) void
+batched_potrf_final_kernel(float* __restrict__ A, int* __restrict__ info,
+                           int r_count, int n) {
+  __shared__ float diag[NB][NB + 1];
+  const int r = blockIdx.x;
+  if (r >= r_count) return;
+  float* M = A + (long)r * n * n;
+  const int tid = threadIdx.x;
+  const int k0 = (int)blockIdx.y * NB;
+  const int nb = min(NB, n - k0);
+  int fc = -1;
+  if (tid < WAVE_SIZE) {
+    wave_factor_diag(M, n, k0, nb, diag, &fc);
+  }
+  __syncthreads();
+  if (tid == 0 && fc > 0 && k0 + NB >= n && info[r] == 0) {
+    info[r] = fc;
+  }
+  for (int e = tid; e < nb * nb; e += CB) {
+    const int i = e / nb, c = e % nb;
+    M[(long)(k0 + i) * n + k0 + c] = (c <= i) ? diag[i][c] : 0.0f;
+  }
+}
+
+extern "C" void launch_batched_potrf_v2(float* A, int* info, int r,
+                                        int n, hipStream_t stream) {
   for (int k0 = 0; k0 < n; k0 += NB) {
     const int nb = (n - k0) < NB ? (n - k0) : NB;
     const int rows = n - (k0 + nb);
@@ -551,6 +724,32 @@ extern "C" void launch_batched_potrf(float* A, int* info, int r, int n,
                          k0);
     }
   }
+}
+
+extern "C" void launch_batched_potrf(float* A, int* info, int r, int n,
+                                     hipStream_t stream) {
+  static int use_v2 = -1;
+  if (use_v2 < 0) {
+    const char* env = std::getenv("VIZIER_AMD_CHOL_IMPL");
+    use_v2 = (env != nullptr && env[0] == 'v' && env[1] == '2') ? 1 : 0;
+  }
+  if (use_v2) {
+    launch_batched_potrf_v2(A, info, r, n, stream);
+    return;
+  }
+  for (int k0 = 0; k0 < n; k0 += NB) {
+    const int nb = (n - k0) < NB ? (n - k0) : NB;
+    const int step0_tiles = (n - k0 + 63) / 64;
+    const int jtiles = (n - (k0 + nb) + 63) / 64;
+    int tiles = step0_tiles > jtiles ? step0_tiles : jtiles;
+    if (tiles < 1) tiles = 1;
+    if (k0 == 0 && jtiles == 0) break;  // single block: final only
+    hipLaunchKernelGGL(batched_potrf_round_kernel, dim3(r, tiles),
+                       dim3(CB), 0, stream, A, info, r, n, k0);
+  }
+  const int blocks = (n + NB - 1) / NB;
+  hipLaunchKernelGGL(batched_potrf_final_kernel, dim3(r, blocks),
+                     dim3(CB), 0, stream, A, info, r, n);
 }
 
 extern "C" void launch_batched_trsv_lower(const float* L, float* b,
