@@ -46,6 +46,17 @@ for r, n in ((3, 1000), (12, 1000), (3, 2000)):
     torch.backends.cuda.preferred_linalg_library('cusolver')
     ms = bench(lambda: torch.linalg.cholesky_ex(k))
     print(f'R={r:3d} N={n}: torch-hipsolver{ms:8.3f} ms', flush=True)
+  if which == 'empty':
+    # Launch-floor probe: 63 dependent tiny torch kernels (same count
+    # as v2's per-potrf launches) on a 1-element tensor.
+    x = torch.zeros(1, device='cuda')
+
+    def chain():
+      for _ in range(63):
+        x.add_(1.0)
+    ms = bench(chain)
+    print(f'R={r:3d} N={n}: 63-launch chain{ms:8.3f} ms '
+          f'({ms / 63 * 1e3:.1f} us/launch)', flush=True)
   if which == 'trsv':
     L = torch.linalg.cholesky(k).contiguous()
     b = torch.randn(r, n).cuda().contiguous()

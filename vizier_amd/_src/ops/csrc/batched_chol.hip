@@ -728,12 +728,16 @@ extern "C" void launch_batched_potrf_v2(float* A, int* info, int r,
 
 extern "C" void launch_batched_potrf(float* A, int* info, int r, int n,
                                      hipStream_t stream) {
-  static int use_v2 = -1;
-  if (use_v2 < 0) {
+  // v4 measured SLOWER than v2 at the headline shape (3.63 vs 2.85 ms,
+  // R=3 N=1000): the fused kernel's three inlined 32-wide solves blow
+  // its register budget (VGPR 256 + 248 AGPRs, occupancy 1 wave/SIMD
+  // vs v2's 4). Opt-in via VIZIER_AMD_CHOL_IMPL=v4 for experiments.
+  static int use_v4 = -1;
+  if (use_v4 < 0) {
     const char* env = std::getenv("VIZIER_AMD_CHOL_IMPL");
-    use_v2 = (env != nullptr && env[0] == 'v' && env[1] == '2') ? 1 : 0;
+    use_v4 = (env != nullptr && env[0] == 'v' && env[1] == '4') ? 1 : 0;
   }
-  if (use_v2) {
+  if (!use_v4) {
     launch_batched_potrf_v2(A, info, r, n, stream);
     return;
   }
