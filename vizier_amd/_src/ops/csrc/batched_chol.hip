@@ -129,6 +129,7 @@ batched_trsv_lower_kernel(const float* __restrict__ L,  // (R, N, N)
                           float* __restrict__ b,        // (R, N)
                           int r_count, int n) {
   __shared__ float zseg[NB];
+  __shared__ float rows_lds[CB][NB + 1];
   const int r = blockIdx.x;
   if (r >= r_count) return;
   const float* M = L + (long)r * n * n;
@@ -154,15 +155,25 @@ batched_trsv_lower_kernel(const float* __restrict__ L,  // (R, N, N)
       if (lane < nb) rhs[k0 + lane] = zseg[lane];
     }
     __syncthreads();
-    for (int i = k0 + nb + tid; i < n; i += CB) {
-      float acc = rhs[i];
-#pragma unroll
-      for (int j = 0; j < NB; ++j) {
-        if (j < nb) acc = fmaf(-M[(long)i * n + k0 + j], zseg[j], acc);
+    // Row blocks staged through LDS for coalesced L-column reads.
+    for (int i0 = k0 + nb; i0 < n; i0 += CB) {
+      const int blk = min(CB, n - i0);
+      for (int e = tid; e < blk * nb; e += CB) {
+        rows_lds[e / nb][e % nb] =
+            M[(long)(i0 + e / nb) * n + k0 + e % nb];
       }
-      rhs[i] = acc;
+      __syncthreads();
+      const int i = i0 + tid;
+      if (i < n) {
+        float acc = rhs[i];
+#pragma unroll
+        for (int j = 0; j < NB; ++j) {
+          if (j < nb) acc = fmaf(-rows_lds[tid][j], zseg[j], acc);
+        }
+        rhs[i] = acc;
+      }
+      __syncthreads();
     }
-    __syncthreads();
   }
 }
 
@@ -179,6 +190,7 @@ extern "C" __global__ __launch_bounds__(CB) void
 batched_potrf_panel_kernel(float* __restrict__ A, int* __restrict__ info,
                            int r_count, int n, int k0) {
   __shared__ float diag[NB][NB + 1];
+  __shared__ float rows_lds[CB][NB + 1];
   const int r = blockIdx.x;
   if (r >= r_count) return;
   // Grid (R, row-tiles): each workgroup REDUNDANTLY factors the 32x32
@@ -187,7 +199,10 @@ batched_potrf_panel_kernel(float* __restrict__ A, int* __restrict__ info,
   // L21 = A21 L11^-T. The r2 single-workgroup-per-matrix version put
   // ~70 us x 32 panels on 3 CUs at the headline shape (R=3, N=1000,
   // profiles/fit_kernels_headline.txt); slicing rows across blockIdx.y
-  // is what lets the panel use the chip.
+  // lets the panel use the chip, and the panel block is staged through
+  // LDS so global accesses are COALESCED (one thread per row reads a
+  // 128-byte segment strided n*4 from its neighbor's — measured ~52 us
+  // per launch before staging, launch floor is ~4.5 us).
   const int row0 = k0 + NB + blockIdx.y * CB;
   const bool has_rows = row0 < n;
   if (blockIdx.y > 0 && !has_rows) return;
@@ -237,13 +252,17 @@ batched_potrf_panel_kernel(float* __restrict__ A, int* __restrict__ info,
       M[(long)(k0 + i) * n + k0 + c] = (c <= i) ? diag[i][c] : 0.0f;
     }
   }
-  const int i = row0 + tid;  // one row per thread in this tile
-  if (i < n) {
+  const int tile_rows = has_rows ? min(CB, n - row0) : 0;
+  for (int e = tid; e < tile_rows * nb; e += CB) {
+    rows_lds[e / nb][e % nb] = M[(long)(row0 + e / nb) * n + k0 + e % nb];
+  }
+  __syncthreads();
+  if (tid < tile_rows) {  // one row per thread in this tile
     float v[NB];
 #pragma unroll
     for (int j = 0; j < NB; ++j) {
       if (j < nb) {
-        float xv = M[(long)i * n + k0 + j];
+        float xv = rows_lds[tid][j];
 #pragma unroll
         for (int p = 0; p < NB; ++p) {
           if (p < j) xv -= v[p] * diag[j][p];
@@ -253,8 +272,12 @@ batched_potrf_panel_kernel(float* __restrict__ A, int* __restrict__ info,
     }
 #pragma unroll
     for (int j = 0; j < NB; ++j) {
-      if (j < nb) M[(long)i * n + k0 + j] = v[j];
+      if (j < nb) rows_lds[tid][j] = v[j];
     }
+  }
+  __syncthreads();
+  for (int e = tid; e < tile_rows * nb; e += CB) {
+    M[(long)(row0 + e / nb) * n + k0 + e % nb] = rows_lds[e / nb][e % nb];
   }
 }
 
@@ -262,6 +285,7 @@ extern "C" __global__ __launch_bounds__(CB) void
 batched_potrf_trailing_kernel(float* __restrict__ A, int r_count, int n,
                               int k0) {
   __shared__ float jpanel[64][NB];
+  __shared__ float rows_lds[CB][NB + 1];
   const int r = blockIdx.x;
   if (r >= r_count) return;
   float* M = A + (long)r * n * n;
@@ -274,21 +298,33 @@ batched_potrf_trailing_kernel(float* __restrict__ A, int r_count, int n,
     jpanel[e / nb][e % nb] = M[(long)(jb + e / nb) * n + k0 + e % nb];
   }
   __syncthreads();
-  for (int i = jb + tid; i < n; i += CB) {
-    float row[NB];
-#pragma unroll
-    for (int p = 0; p < NB; ++p) {
-      row[p] = (p < nb) ? M[(long)i * n + k0 + p] : 0.0f;
+  // Row blocks staged through LDS for coalesced panel reads (the
+  // per-thread row loads are 4-KB-strided otherwise).
+  for (int i0 = jb; i0 < n; i0 += CB) {
+    const int blk = min(CB, n - i0);
+    for (int e = tid; e < blk * nb; e += CB) {
+      rows_lds[e / nb][e % nb] =
+          M[(long)(i0 + e / nb) * n + k0 + e % nb];
     }
-    for (int j = 0; j < jl; ++j) {
-      if (jb + j > i) break;
-      float acc = 0.0f;
+    __syncthreads();
+    const int i = i0 + tid;
+    if (i < n) {
+      float row[NB];
 #pragma unroll
       for (int p = 0; p < NB; ++p) {
-        acc = fmaf(row[p], jpanel[j][p], acc);
+        row[p] = (p < nb) ? rows_lds[tid][p] : 0.0f;
       }
-      M[(long)i * n + jb + j] -= acc;
+      for (int j = 0; j < jl; ++j) {
+        if (jb + j > i) break;
+        float acc = 0.0f;
+#pragma unroll
+        for (int p = 0; p < NB; ++p) {
+          acc = fmaf(row[p], jpanel[j][p], acc);
+        }
+        M[(long)i * n + jb + j] -= acc;
+      }
     }
+    __syncthreads();
   }
 }
 
