@@ -745,6 +745,262 @@ batched_potrf_final_kernel(float* __restrict__ A, int* __restrict__ info,
   }
 }
 
+// -- v5: persistent LEFT-looking cooperative factorization --------------
+//
+// Why a 5th design: the per-kernel cost of v2's 63 launches is NOT
+// launch overhead (a 63-deep chain of trivial kernels runs at 4.4
+// us/launch) and NOT uncoalesced access (LDS staging changed nothing)
+// — it is the end-of-kernel release fence, which writes back and
+// invalidates the per-XCD L2s (~40-65 us when megabytes are dirty;
+// same effect eagle_sweep.hip documents for device-scope fences). The
+// fix must therefore run the whole factorization in ONE kernel while
+// keeping the bulk data CACHED — which a right-looking update cannot
+// do (v3: the trailing matrix is cross-workgroup mutable).
+//
+// Left-looking solves it:
+//   - the INPUT matrix A is never written (output is separate), so
+//     its reads are normal cached loads;
+//   - finalized L panels live in a PANEL-MAJOR scratch (32 floats per
+//     row per panel, 128-byte aligned) so no cache line ever spans
+//     two panels; a line is cstore'd once at finalization and only
+//     normal-loaded AFTER that, so caches can never hold a stale
+//     pre-write copy;
+//   - only the factored 32x32 diagonal block crosses workgroups
+//     mid-round, through a tiny agent-scope scratch.
+// A plain kernel afterwards copies panel-major L into the dense
+// output with the upper triangle zeroed.
+//
+// Round k, pair (r, t) owning 64-row tile t (one pair per workgroup,
+// enforced by the launcher):
+//   P1: rows_acc[i][c] = A[i, k0+c] - sum_{kb<k} L[i,:] L[k0+c,:]
+//       (4 threads x 8 columns per row; LDS-staged L tiles);
+//       the diagonal owner wave-factors rows_acc, records info, and
+//       publishes the factored diagonal via agent-scope scratch.
+//   barrier.
+//   P2: every tile solves its own rows >= k0+32 against the diagonal
+//       and finalizes them into the panel scratch.
+//   barrier.
+
+#define V5_RB 64
+
+extern "C" __global__ __launch_bounds__(CB) void
+batched_potrf_left_kernel(
+    const float* __restrict__ A,     // (R, N, N) input, read-only
+    float* __restrict__ panels,      // (R, ceil(N/32), N, 32) scratch
+    float* __restrict__ dscratch,    // (R, 32, 32) diag broadcast
+    int* __restrict__ info,          // (R,)
+    unsigned int* __restrict__ bar,  // (2,) zeroed
+    int r_count, int n) {
+  __shared__ float rows_acc[V5_RB][NB + 1];
+  __shared__ float lrows[V5_RB][NB];   // own-tile L rows, one kb panel
+  __shared__ float ldiag[NB][NB + 1];  // diag-rows L, one kb panel
+  __shared__ float fdiag[NB][NB + 1];  // factored diagonal this round
+  const int tid = threadIdx.x;
+  const int tiles = (n + V5_RB - 1) / V5_RB;
+  const int r = blockIdx.x / tiles;
+  const int t = blockIdx.x % tiles;
+  const int tile_lo = t * V5_RB;
+  const int tile_hi = min(tile_lo + V5_RB, n);
+  const float* Ain = A + (long)r * n * n;
+  float* P = panels + (long)r * ((n + NB - 1) / NB) * n * NB;
+  float* ds = dscratch + (long)r * NB * NB;
+  const int n_rounds = (n + NB - 1) / NB;
+
+  for (int k = 0; k < n_rounds; ++k) {
+    const int k0 = k * NB;
+    const int nb = min(NB, n - k0);
+    const int row_lo = tile_lo > k0 ? tile_lo : k0;
+    const bool active = tile_hi > k0;
+    const bool owner = (k0 >= tile_lo && k0 < tile_hi);
+
+    if (active) {
+      // P1: update own rows' panel columns (left-looking GEMM).
+      const int nrows = tile_hi - row_lo;
+      const int rl = tid >> 2, cq = (tid & 3) * 8;
+      float acc[8];
+      const int i = row_lo + rl;
+      if (rl < nrows) {
+#pragma unroll
+        for (int c = 0; c < 8; ++c) {
+          acc[c] = (cq + c < nb) ? Ain[(long)i * n + k0 + cq + c] : 0.0f;
+        }
+      }
+      for (int kb = 0; kb < k; ++kb) {
+        const float* Pk = P + (long)kb * n * NB;
+        for (int e = tid; e < nrows * NB; e += CB) {
+          lrows[e / NB][e % NB] = Pk[(long)(row_lo + e / NB) * NB
+                                     + e % NB];
+        }
+        for (int e = tid; e < nb * NB; e += CB) {
+          ldiag[e / NB][e % NB] = Pk[(long)(k0 + e / NB) * NB + e % NB];
+        }
+        __syncthreads();
+        if (rl < nrows) {
+#pragma unroll
+          for (int c = 0; c < 8; ++c) {
+            float a = acc[c];
+            if (cq + c < nb) {
+#pragma unroll
+              for (int p = 0; p < NB; ++p) {
+                a = fmaf(-lrows[rl][p], ldiag[cq + c][p], a);
+              }
+            }
+            acc[c] = a;
+          }
+        }
+        __syncthreads();
+      }
+      if (rl < nrows) {
+#pragma unroll
+        for (int c = 0; c < 8; ++c) {
+          rows_acc[rl][cq + c] = acc[c];
+        }
+      }
+      __syncthreads();
+
+      if (owner) {
+        // Wave-factor the updated diagonal block (rows k0..k0+nb are
+        // inside this tile) and publish it.
+        const int d0 = k0 - row_lo;
+        int fc = -1;
+        if (tid < WAVE_SIZE) {
+          const int lane = tid;
+          float row[NB];
+#pragma unroll
+          for (int c = 0; c < NB; ++c) {
+            row[c] = (lane < nb && c < nb) ? rows_acc[d0 + lane][c]
+                                           : 0.0f;
+          }
+#pragma unroll
+          for (int j = 0; j < NB; ++j) {
+            if (j >= nb) break;
+            const float piv = __shfl(row[j], j, WAVE_SIZE);
+            float d;
+            if (piv > 0.0f) {
+              d = sqrtf(piv);
+            } else {
+              d = 1.0f;
+              if (fc < 0) fc = k0 + j + 1;
+            }
+            if (lane == j) row[j] = d;
+            if (lane > j) row[j] /= d;
+#pragma unroll
+            for (int c = 0; c < NB; ++c) {
+              if (c > j && c < nb) {
+                const float lcj = __shfl(row[j], c, WAVE_SIZE);
+                if (lane >= c) row[c] -= row[j] * lcj;
+              }
+            }
+          }
+#pragma unroll
+          for (int c = 0; c < NB; ++c) {
+            if (lane < nb && c < nb) fdiag[lane][c] = row[c];
+          }
+          if (lane == 0 && fc > 0 && cchol_iload(&info[r]) == 0) {
+            cchol_istore(&info[r], fc);
+          }
+        }
+        __syncthreads();
+        // Publish diag: agent-scope scratch for other tiles + the
+        // finalized panel rows for later rounds' GEMMs.
+        float* Pk = P + (long)k * n * NB;
+        for (int e = tid; e < nb * nb; e += CB) {
+          const int ii = e / nb, c = e % nb;
+          cchol_store(&ds[ii * NB + c], fdiag[ii][c]);
+          cchol_store(&Pk[(long)(k0 + ii) * NB + c],
+                      (c <= ii) ? fdiag[ii][c] : 0.0f);
+        }
+      }
+    }
+    chol_grid_sync(bar);
+
+    // P2: solve own rows below the diagonal block and finalize them.
+    if (active) {
+      const int row_lo2 = tile_lo > k0 + nb ? tile_lo : k0 + nb;
+      if (row_lo2 < tile_hi) {
+        if (!owner) {
+          for (int e = tid; e < nb * nb; e += CB) {
+            fdiag[e / nb][e % nb] = cchol_load(&ds[(e / nb) * NB
+                                                   + e % nb]);
+          }
+        }
+        __syncthreads();
+        const int base = row_lo2 - (tile_lo > k0 ? tile_lo : k0);
+        const int i = row_lo2 + tid;
+        float* Pk = P + (long)k * n * NB;
+        if (i < tile_hi) {
+          float v[NB];
+#pragma unroll
+          for (int j = 0; j < NB; ++j) {
+            if (j < nb) {
+              float xv = rows_acc[base + tid][j];
+#pragma unroll
+              for (int p = 0; p < NB; ++p) {
+                if (p < j) xv -= v[p] * fdiag[j][p];
+              }
+              v[j] = xv / fdiag[j][j];
+            } else {
+              v[j] = 0.0f;
+            }
+          }
+#pragma unroll
+          for (int j = 0; j < NB; ++j) {
+            cchol_store(&Pk[(long)i * NB + j], v[j]);
+          }
+        }
+      }
+      __syncthreads();
+    }
+    chol_grid_sync(bar);
+  }
+}
+
+// Copies panel-major L into the dense (R, N, N) output, zeroing the
+// upper triangle (and the never-written rows above each panel).
+extern "C" __global__ __launch_bounds__(CB) void
+potrf_copy_out_kernel(const float* __restrict__ panels,
+                      float* __restrict__ L, int r_count, int n) {
+  const int r = blockIdx.x;
+  if (r >= r_count) return;
+  const int pk = blockIdx.y;
+  const int p0 = pk * NB;
+  const float* Pk = panels + ((long)r * ((n + NB - 1) / NB) + pk)
+                    * (long)n * NB;
+  float* M = L + (long)r * n * n;
+  for (int e = threadIdx.x; e < n * NB; e += CB) {
+    const int i = e / NB, p = e % NB;
+    const int j = p0 + p;
+    if (j < n) {
+      M[(long)i * n + j] = (j <= i) ? Pk[(long)i * NB + p] : 0.0f;
+    }
+  }
+}
+
+extern "C" int launch_batched_potrf_v5(const float* A, float* panels,
+                                       float* dscratch, float* L,
+                                       int* info, unsigned int* bar,
+                                       int r, int n,
+                                       hipStream_t stream) {
+  int max_blocks = 0;
+  hipError_t err = hipOccupancyMaxActiveBlocksPerMultiprocessor(
+      &max_blocks, (const void*)batched_potrf_left_kernel, CB, 0);
+  if (err != hipSuccess || max_blocks < 1) return -1;
+  hipDeviceProp_t prop;
+  if (hipGetDeviceProperties(&prop, 0) != hipSuccess) return -1;
+  const int tiles = (n + V5_RB - 1) / V5_RB;
+  const int grid = r * tiles;
+  if (grid > max_blocks * prop.multiProcessorCount) return -1;
+  void* args[] = {&A, &panels, &dscratch, &info, &bar, &r, &n};
+  err = hipLaunchCooperativeKernel(
+      (const void*)batched_potrf_left_kernel, dim3(grid), dim3(CB),
+      args, 0, stream);
+  if (err != hipSuccess) return -1;
+  const int pk = (n + NB - 1) / NB;
+  hipLaunchKernelGGL(potrf_copy_out_kernel, dim3(r, pk), dim3(CB), 0,
+                     stream, panels, L, r, n);
+  return 0;
+}
+
 extern "C" void launch_batched_potrf_v2(float* A, int* info, int r,
                                         int n, hipStream_t stream) {
   for (int k0 = 0; k0 < n; k0 += NB) {

@@ -116,6 +116,10 @@ extern "C" void launch_batched_potrf(float* A, int* info, int r, int n,
 extern "C" int launch_batched_potrf_coop(float* A, int* info,
                                          unsigned int* bar, int r, int n,
                                          hipStream_t stream);
+extern "C" int launch_batched_potrf_v5(const float* A, float* panels,
+                                       float* dscratch, float* L,
+                                       int* info, unsigned int* bar,
+                                       int r, int n, hipStream_t stream);
 extern "C" void launch_batched_trsv_lower(const float* L, float* b,
                                           int r, int n,
                                           hipStream_t stream);
@@ -709,35 +713,55 @@ std::vector<torch::Tensor> batched_potrf(torch::Tensor K) {
   TORCH_CHECK(K.dim() == 3 && K.size(1) == K.size(2),
               "K must be (R, N, N)");
   const int r = K.size(0), n = K.size(1);
-  auto L = K.clone();
   auto info = torch::zeros({r}, K.options().dtype(torch::kInt32));
-  // Persistent cooperative factorization (one launch for all panels,
-  // eagle_sweep-style grid barriers) — OPT-IN via
-  // VIZIER_AMD_COOP_CHOL=1. Measured SLOWER than the v2 multi-launch
-  // loop at the headline shape (warm refit 132.9 -> 240.9 ms): unlike
-  // eagle_sweep, whose bulk data is read-only and L2-cached, a
-  // right-looking factorization rewrites the whole trailing matrix
-  // every round, and routing that through agent-scope (memory-side)
-  // accesses for cross-workgroup coherence costs more than the 64
-  // launch turnarounds it saves. Kept for reference/experiments.
-  static const bool coop_enabled = []() {
-    const char* env = std::getenv("VIZIER_AMD_COOP_CHOL");
-    return env != nullptr && std::string(env) == "1";
+  // Implementation ladder (VIZIER_AMD_CHOL_IMPL=v2|v3|v5, default v5):
+  //   v5: ONE persistent left-looking cooperative kernel — the input
+  //       stays read-only (cached), finalized panels live in a
+  //       panel-major scratch, only the 32x32 diagonal crosses
+  //       workgroups via agent-scope scratch. Built because v2's 63
+  //       launches pay an end-of-kernel L2 writeback each (~45 us
+  //       with megabytes dirty; a trivial-kernel chain runs at 4.4
+  //       us/launch, so it is the flush, not the launch).
+  //   v2: 2-launches-per-panel fallback (also used when the
+  //       cooperative grid would exceed co-residency).
+  //   v3: right-looking cooperative experiment — SLOWER (trailing
+  //       matrix is cross-workgroup mutable, so its rewrites go
+  //       memory-side; fit 132.9 -> 240.9 ms). Kept for reference.
+  static const int impl = []() {
+    const char* env = std::getenv("VIZIER_AMD_CHOL_IMPL");
+    if (env == nullptr) return 5;
+    const std::string s(env);
+    if (s == "v2" || s == "v4") return 2;  // v4 re-checked downstream
+    if (s == "v3") return 3;
+    return 5;
   }();
-  bool done = false;
-  if (coop_enabled) {
-    // Stream-ordered temp: the caching allocator only reuses this
-    // block behind the kernel on the same stream.
+  if (impl == 5) {
+    auto L = torch::empty_like(K);
+    const long pk = (n + 31) / 32;
+    auto panels = torch::empty({(long)r * pk * n * 32}, K.options());
+    auto dscratch = torch::empty({(long)r * 32 * 32}, K.options());
     auto bar = torch::zeros({2}, K.options().dtype(torch::kInt32));
-    done = launch_batched_potrf_coop(
-               L.data_ptr<float>(), info.data_ptr<int>(),
-               reinterpret_cast<unsigned int*>(bar.data_ptr<int>()),
-               r, n, current_stream()) == 0;
+    if (launch_batched_potrf_v5(
+            K.data_ptr<float>(), panels.data_ptr<float>(),
+            dscratch.data_ptr<float>(), L.data_ptr<float>(),
+            info.data_ptr<int>(),
+            reinterpret_cast<unsigned int*>(bar.data_ptr<int>()),
+            r, n, current_stream()) == 0) {
+      return {L, info};
+    }
   }
-  if (!done) {
-    launch_batched_potrf(L.data_ptr<float>(), info.data_ptr<int>(), r, n,
-                         current_stream());
+  auto L = K.clone();
+  if (impl == 3) {
+    auto bar = torch::zeros({2}, K.options().dtype(torch::kInt32));
+    if (launch_batched_potrf_coop(
+            L.data_ptr<float>(), info.data_ptr<int>(),
+            reinterpret_cast<unsigned int*>(bar.data_ptr<int>()),
+            r, n, current_stream()) == 0) {
+      return {L, info};
+    }
   }
+  launch_batched_potrf(L.data_ptr<float>(), info.data_ptr<int>(), r, n,
+                       current_stream());
   return {L, info};
 }
 
