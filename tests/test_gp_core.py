@@ -442,6 +442,51 @@ class TestAnalyticNLLGradient:
     assert torch.allclose(nll_a, nll_t.detach(), rtol=1e-10)
     assert torch.allclose(grad_a, grad_t, atol=1e-7, rtol=1e-6)
 
+  def test_chol_hint_matches_no_hint(self):
+    """The line-search ladder hands its (L, info) to the gradient eval
+    (lbfgs ladder_fn); with the factors of the SAME raw values the
+    result must be identical to factoring from scratch."""
+    g = torch.Generator().manual_seed(5)
+    x = torch.rand(40, 3, generator=g).double()
+    y = torch.randn(40, generator=g).double()
+    raw = torch.randn(3, 6, generator=g).double() * 0.5
+    nll_ref, grad_ref = gp_model.nll_value_and_grad(raw, x, y)
+    _, L, info = gp_model.nll_values_with_chol(raw, x, y)
+    nll_h, grad_h = gp_model.nll_value_and_grad(
+        raw, x, y, chol_hint=(L, info))
+    # Not bitwise: the ladder builds K via gram_matern52 while the
+    # gradient eval builds it from explicit distances, so L differs in
+    # the last ulp. Equal to fp64 rounding.
+    assert torch.allclose(nll_h, nll_ref, rtol=1e-12)
+    assert torch.allclose(grad_h, grad_ref, rtol=1e-8, atol=1e-8)
+
+  def test_ladder_fn_end_to_end_matches_plain(self):
+    """train_gp's ladder-cache plumbing must not change the optimum:
+    run minimize_batched with and without ladder_fn."""
+    from vizier_amd._src.gp import lbfgs
+    g = torch.Generator().manual_seed(7)
+    x = torch.rand(30, 2, generator=g).double()
+    y = torch.sin(4 * x[:, 0]).double()
+    raw0 = torch.randn(2, 5, generator=g).double() * 0.4
+
+    def loss_fn(raw):
+      return gp_model.negative_log_marginal_likelihood(raw, x, y)
+
+    def vag(raw, hint=None):
+      return gp_model.nll_value_and_grad(raw, x, y, chol_hint=hint)
+
+    def ladder(raw):
+      nll_v, L, info = gp_model.nll_values_with_chol(raw, x, y)
+      return nll_v, (L, info)
+
+    best_a, f_a = lbfgs.minimize_batched(
+        loss_fn, raw0.clone(), max_iters=15, value_and_grad_fn=vag)
+    best_b, f_b = lbfgs.minimize_batched(
+        loss_fn, raw0.clone(), max_iters=15, value_and_grad_fn=vag,
+        ladder_fn=ladder)
+    assert torch.allclose(f_a, f_b, rtol=1e-6)
+    assert torch.allclose(best_a, best_b, rtol=1e-4, atol=1e-4)
+
   def test_float32_accuracy_vs_f64_truth(self):
     """fp32 analytic grads are as accurate as fp32 autograd grads
     (both measured against the f64 analytic ground truth)."""

@@ -192,9 +192,17 @@ def _use_custom_trsv(K: torch.Tensor) -> bool:
           K.shape[-1] <= 1200 and ops.extension_available())
 
 
-def negative_log_marginal_likelihood(raw: torch.Tensor, x: torch.Tensor,
-                                     y: torch.Tensor) -> torch.Tensor:
-  """Batched NLL over restarts. raw (R, D+3); x (N, D); y (N,)."""
+def nll_values_with_chol(raw: torch.Tensor, x: torch.Tensor,
+                         y: torch.Tensor):
+  """Batched NLL that also returns its Cholesky factors.
+
+  Returns (nll (R,), L (R, N, N), info (R,) int). The factors let the
+  L-BFGS line search hand the ACCEPTED candidate's factorization to
+  the gradient evaluation (lbfgs.py `ladder_fn`), skipping one full
+  batched potrf per iteration — x_new is bit-identical to the selected
+  ladder row, so L here factors the same K up to the last-ulp
+  difference between this K builder (gram_matern52) and the gradient
+  eval's explicit-distance build."""
   params = GPParams.from_raw(raw)
   n = x.shape[0]
   K = gram_matern52(x.unsqueeze(0), None, params.lengthscales,
@@ -230,11 +238,18 @@ def negative_log_marginal_likelihood(raw: torch.Tensor, x: torch.Tensor,
   nll = nll + 0.01 * (raw * raw).sum(-1)
   # Failed factorizations get +inf so the optimizer backs off.
   nll = torch.where(info == 0, nll, torch.full_like(nll, float('inf')))
+  return nll, L, info
+
+
+def negative_log_marginal_likelihood(raw: torch.Tensor, x: torch.Tensor,
+                                     y: torch.Tensor) -> torch.Tensor:
+  """Batched NLL over restarts. raw (R, D+3); x (N, D); y (N,)."""
+  nll, _, _ = nll_values_with_chol(raw, x, y)
   return nll
 
 
 def nll_value_and_grad(raw: torch.Tensor, x: torch.Tensor,
-                       y: torch.Tensor
+                       y: torch.Tensor, chol_hint=None,
                        ) -> Tuple[torch.Tensor, torch.Tensor]:
   """Batched NLL and its ANALYTIC gradient (no autograd).
 
@@ -268,7 +283,21 @@ def nll_value_and_grad(raw: torch.Tensor, x: torch.Tensor,
   K = amp2 * k0 + params.noise.reshape(-1, 1, 1) * torch.eye(
       n, dtype=x.dtype, device=x.device)
   resid = (y.unsqueeze(0) - params.mean.unsqueeze(-1)).unsqueeze(-1)
-  if _use_custom_chol(K):
+  if chol_hint is not None:
+    # The line-search ladder already factored K at these exact raw
+    # values (bit-identical x_new); reuse its (L, info) and skip the
+    # batched potrf entirely.
+    L, info = chol_hint
+    L = L.contiguous()
+    if _use_custom_chol(K) or _use_custom_trsv(K):
+      ext = ops.require_ext()
+      zs = ext.batched_trsv_lower(
+          L, resid.squeeze(-1).expand(K.shape[0], n).contiguous())
+      quad = (zs * zs).sum(dim=-1)
+    else:
+      zsol = torch.linalg.solve_triangular(L, resid, upper=False)
+      quad = (zsol * zsol).sum(dim=(-1, -2))
+  elif _use_custom_chol(K):
     # Headline-shape fast path: custom panel-swept potrf + the batched
     # wave-synchronous solve instead of MAGMA's spotf2 launch storm
     # (profiles/fit_kernels_headline.txt: the MAGMA+trtri+serial-trsv
@@ -480,9 +509,21 @@ def train_gp(x: torch.Tensor, y: torch.Tensor, *,
     # ROCm 7.2 (even unbatched), so huge studies cannot autograd the
     # NLL. Fit with ANALYTIC gradients instead (forward-only solves,
     # verified against autograd to 1e-9 in tests/test_gp_core.py).
+    # Ladder-cache: the line search returns its (L, info) per trial so
+    # the gradient eval at the accepted point (bit-identical raw)
+    # skips one full batched potrf per iteration. Gated on cache size
+    # (the (S*R, N, N) factors are ~6 GB at config-4 scale).
+    r_ladder = raw0.shape[0] * 4  # ls_steps ladder width
+    ladder = None
+    if r_ladder * n * n * 4 <= (1 << 30):
+      def ladder(raw):
+        nll_v, L, information = nll_values_with_chol(raw, x, y)
+        return nll_v, (L, information)
     best_raw, best_f = lbfgs.minimize_batched(
         loss_fn, raw0, max_iters=max_iters, check_every=5,
-        value_and_grad_fn=lambda raw: nll_value_and_grad(raw, x, y))
+        value_and_grad_fn=lambda raw, hint=None: nll_value_and_grad(
+            raw, x, y, chol_hint=hint),
+        ladder_fn=ladder)
   else:
     best_raw, best_f = lbfgs.minimize_batched(loss_fn, raw0,
                                               max_iters=max_iters,

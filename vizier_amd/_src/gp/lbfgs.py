@@ -27,9 +27,10 @@ def minimize_batched(
     grad_tol: float = 1e-7,
     ls_steps: Tuple[float, ...] = (1.0, 0.3, 0.08, 0.02),
     check_every: int = 10,
-    value_and_grad_fn: Optional[Callable[[torch.Tensor],
-                                         Tuple[torch.Tensor,
-                                               torch.Tensor]]] = None,
+    value_and_grad_fn: Optional[Callable[..., Tuple[torch.Tensor,
+                                                    torch.Tensor]]] = None,
+    ladder_fn: Optional[Callable[[torch.Tensor], Tuple[
+        torch.Tensor, Tuple[torch.Tensor, ...]]]] = None,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
   """Minimizes loss_fn over a batch of R independent parameter vectors.
 
@@ -53,12 +54,17 @@ def minimize_batched(
   S = steps.numel()
   inf = torch.tensor(float('inf'), dtype=dtype, device=device)
 
-  def value_and_grad(params: torch.Tensor
+  def value_and_grad(params: torch.Tensor, hint=None
                      ) -> Tuple[torch.Tensor, torch.Tensor]:
     if value_and_grad_fn is not None:
       # Analytic gradients (e.g. gp_model.nll_value_and_grad) — used
       # where autograd is unavailable (huge-N trsm-backward failures).
-      loss, grad = value_and_grad_fn(params.detach())
+      # `hint` forwards the line-search ladder's cached factorization
+      # for the accepted candidates (bit-identical inputs).
+      if hint is not None:
+        loss, grad = value_and_grad_fn(params.detach(), hint)
+      else:
+        loss, grad = value_and_grad_fn(params.detach())
     else:
       params = params.detach().requires_grad_(True)
       loss = loss_fn(params)
@@ -107,8 +113,13 @@ def minimize_batched(
 
     # Parallel line search: one (S*R, P) loss call.
     trials = x.unsqueeze(0) + steps.reshape(S, 1, 1) * d.unsqueeze(0)
+    cache = None
     with torch.no_grad():
-      f_trials = loss_fn(trials.reshape(S * R, P)).reshape(S, R)
+      if ladder_fn is not None:
+        f_flat, cache = ladder_fn(trials.reshape(S * R, P))
+        f_trials = f_flat.reshape(S, R)
+      else:
+        f_trials = loss_fn(trials.reshape(S * R, P)).reshape(S, R)
     f_trials = torch.where(torch.isfinite(f_trials), f_trials, inf)
     armijo = f_trials <= f.unsqueeze(0) + \
         1e-4 * steps.reshape(S, 1) * dg.unsqueeze(0)
@@ -118,8 +129,17 @@ def minimize_batched(
     step_sel = steps[s_idx] * moved.to(dtype)
     x_new = x + step_sel.unsqueeze(1) * d
 
+    hint = None
+    if cache is not None:
+      # Ladder rows are laid out s-major: row for (step s, restart r)
+      # is s*R + r. x_new[r] is bit-identical to the selected row, so
+      # its cached factorization applies verbatim. Unmoved rows pass a
+      # stale factor, but their value/grad results are masked below.
+      sel = s_idx * R + torch.arange(R, device=device)
+      hint = tuple(c[sel] for c in cache)
+
     f_prev, g_prev, x_prev = f, g, x
-    f, g = value_and_grad(x_new)
+    f, g = value_and_grad(x_new, hint)
     f = torch.where(moved, f, f_prev)
     g = torch.where(moved.unsqueeze(1), g, g_prev)
     x = torch.where(moved.unsqueeze(1), x_new, x_prev)
