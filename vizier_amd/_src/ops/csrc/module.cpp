@@ -714,26 +714,28 @@ std::vector<torch::Tensor> batched_potrf(torch::Tensor K) {
               "K must be (R, N, N)");
   const int r = K.size(0), n = K.size(1);
   auto info = torch::zeros({r}, K.options().dtype(torch::kInt32));
-  // Implementation ladder (VIZIER_AMD_CHOL_IMPL=v2|v3|v5, default v5):
-  //   v5: ONE persistent left-looking cooperative kernel — the input
-  //       stays read-only (cached), finalized panels live in a
-  //       panel-major scratch, only the 32x32 diagonal crosses
-  //       workgroups via agent-scope scratch. Built because v2's 63
-  //       launches pay an end-of-kernel L2 writeback each (~45 us
-  //       with megabytes dirty; a trivial-kernel chain runs at 4.4
-  //       us/launch, so it is the flush, not the launch).
-  //   v2: 2-launches-per-panel fallback (also used when the
-  //       cooperative grid would exceed co-residency).
-  //   v3: right-looking cooperative experiment — SLOWER (trailing
-  //       matrix is cross-workgroup mutable, so its rewrites go
-  //       memory-side; fit 132.9 -> 240.9 ms). Kept for reference.
+  // Implementation ladder (VIZIER_AMD_CHOL_IMPL=v2|v3|v4|v5, default
+  // v2 — every alternative was A/B'd SLOWER at the headline shape):
+  //   v2 (default): 2 launches per 32-panel round; 2.85 ms at
+  //       (R=3..12, N=1000). Execution-bound: a 63-deep trivial-kernel
+  //       chain runs at 4.4 us/launch, LDS-staged coalescing changed
+  //       nothing, so the per-kernel ~45 us is genuine work+fence.
+  //   v5: persistent left-looking cooperative kernel (read-only input
+  //       stays cached, panel-major once-written output, diagonal via
+  //       agent-scope scratch) — 3.15 ms: the per-panel LDS staging
+  //       loop (~500 sync'd iterations) and occupancy 2 eat the
+  //       saved inter-kernel fences.
+  //   v3: right-looking cooperative — 2x slower (trailing matrix is
+  //       cross-workgroup mutable -> all traffic memory-side).
+  //   v4: fused one-kernel-per-round with recompute — occupancy 1
+  //       (VGPR 256 + 248 AGPRs).
   static const int impl = []() {
     const char* env = std::getenv("VIZIER_AMD_CHOL_IMPL");
-    if (env == nullptr) return 5;
+    if (env == nullptr) return 2;
     const std::string s(env);
-    if (s == "v2" || s == "v4") return 2;  // v4 re-checked downstream
+    if (s == "v5") return 5;
     if (s == "v3") return 3;
-    return 5;
+    return 2;  // v2/v4 re-checked downstream
   }();
   if (impl == 5) {
     auto L = torch::empty_like(K);
