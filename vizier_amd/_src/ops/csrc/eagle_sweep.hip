@@ -414,7 +414,23 @@ extern "C" int launch_eagle_sweep(
       &blocks_per_cu, (const void*)eagle_sweep_kernel, BLOCK, 0);
   if (err != hipSuccess || blocks_per_cu < 1) return 0;
   const int tiles_n = (n + VZ_QF_TILE - 1) / VZ_QF_TILE;
+  // Grid size is a pure throughput knob (every phase is a grid-stride
+  // loop and partial-sum layout is grid-independent, so results are
+  // bit-identical at any size). Measured at the headline shape
+  // (N=1000, batch 25): tiles_n^2=256 WGs -> 61 us/iter but 64-128
+  // WGs -> ~42 us/iter — the sense-reversing barrier's cost grows
+  // with arrival count, so cap the default at 128 and floor at 64
+  // (tiles_n^2 starves phase A at small N: 4 WGs at N=125 ran at
+  // 107 us/iter). Override with VIZIER_AMD_SWEEP_GRID.
   int grid = tiles_n * tiles_n;                   // fills phase B
+  if (grid > 128) grid = 128;
+  if (grid < 64) grid = 64;
+  static int grid_env = -1;
+  if (grid_env < 0) {
+    const char* env = getenv("VIZIER_AMD_SWEEP_GRID");
+    grid_env = (env != nullptr) ? atoi(env) : 0;
+  }
+  if (grid_env > 0) grid = grid_env;
   const int max_grid = blocks_per_cu * num_cu;
   if (grid > max_grid) grid = max_grid;
   if (grid < 1) grid = 1;
