@@ -126,7 +126,11 @@ class VectorizedEagleStrategy:
     if self.device.type == 'cuda' and dtype == torch.float32:
       from vizier_amd._src.ops import dispatch as ops
       self._ext = ops.require_ext()
-      self._iter_t = torch.zeros(2, dtype=torch.long, device=self.device)
+      # [0:2] device iteration counters; [2:10] megakernel per-phase
+      # cycle accumulators (A, barA, B, barB, B2, barB2, C, barC) —
+      # written by workgroup 0 of eagle_sweep for profiling.
+      self._iter_t = torch.zeros(12, dtype=torch.long,
+                                 device=self.device)
       self._out_cont = torch.empty(self.batch_size, n_parallel,
                                    n_continuous, dtype=dtype,
                                    device=self.device)
@@ -242,8 +246,9 @@ class VectorizedEagleStrategy:
       return CandidateBatch(batch.continuous.clone(),
                             batch.categorical.clone())
     if self._ext is not None and state.iterations == n_batches:
-      # Entering steady state: align the device iteration counter once.
-      self._iter_t.fill_(state.iterations)
+      # Entering steady state: align the device iteration counter once
+      # (first two slots only — the rest are profiling accumulators).
+      self._iter_t[:2].fill_(state.iterations)
     if self._ext is not None:
       cfg = self.config
       cat_factor = (cfg.pure_categorical_perturbation_factor

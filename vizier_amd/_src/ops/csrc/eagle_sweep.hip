@@ -157,10 +157,24 @@ eagle_sweep_kernel(
   auto fmin_ = [](float a, float c) { return fminf(a, c); };
   auto fmax_ = [](float a, float c) { return fmaxf(a, c); };
 
+  // Per-phase cycle profiling (workgroup 0's view, including its
+  // barrier-wait slack): accumulated into iter_ptr[2..9] as
+  // A, barA, B, barB, B2, barB2, C, barC. Reads are a handful of
+  // s_memtime per iteration on one lane — no measurable perturbation.
+  const bool profwg = (blockIdx.x == 0 && threadIdx.x == 0);
+  unsigned long long tmark = profwg ? wall_clock64() : 0;
+#define VZ_SWEEP_PROF(slot)                                        \
+  if (profwg) {                                                    \
+    const unsigned long long tnow = wall_clock64();                \
+    iter_ptr[2 + (slot)] += tnow - tmark;                          \
+    tmark = tnow;                                                  \
+  }
+
   for (long long it = it_start; it < it_start + iterations; ++it) {
     const unsigned long long offset = (unsigned long long)it;
     const int batch_start =
         (int)(offset % (unsigned long long)n_batches) * batch_size;
+    if (profwg) tmark = wall_clock64();
 
     // ---- Phase A: suggest + k-vec (workgroup owns candidate b) ----
     for (int b = blockIdx.x; b < batch_size; b += G) {
@@ -269,7 +283,9 @@ eagle_sweep_kernel(
       if (tid == 0) cstore(dist_ws + b, dist);
       __syncthreads();
     }
+    VZ_SWEEP_PROF(0)
     grid_sync(barrier_buf);
+    VZ_SWEEP_PROF(1)
 
     // ---- Phase B: K^-1 quadform (64x64 tiles, Kinv read ONCE) ----
     // Shared template with the standalone chunked scorer
@@ -291,7 +307,9 @@ eagle_sweep_kernel(
             });
       }
     }
+    VZ_SWEEP_PROF(2)
     grid_sync(barrier_buf);
+    VZ_SWEEP_PROF(3)
 
     // ---- Phase B2: reduce tile partials (order == the standalone
     // ps_reduce_parts_kernel) ----
@@ -311,7 +329,9 @@ eagle_sweep_kernel(
         __syncthreads();
       }
     }
+    VZ_SWEEP_PROF(4)
     grid_sync(barrier_buf);
+    VZ_SWEEP_PROF(5)
 
     // ---- Phase C: finalize scores (inline) + update ----
     for (int i = blockIdx.x; i < batch_size; i += G) {
@@ -375,7 +395,9 @@ eagle_sweep_kernel(
       }
       __syncthreads();
     }
+    VZ_SWEEP_PROF(6)
     grid_sync(barrier_buf);
+    VZ_SWEEP_PROF(7)
   }
 
   if (blockIdx.x == 0 && tid == 0) {
