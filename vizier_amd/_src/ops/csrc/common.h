@@ -131,12 +131,37 @@ __device__ __forceinline__ void vz_quadform_tile(
 #pragma unroll
   for (int qq = 0; qq < 8; ++qq) acc[qq] = 0.0f;
   if (lane < jlen) {
-    for (int i = 0; i < ilen; ++i) {
-      const float kv = kinv[(long)(i0 + i) * n + j0 + lane];
+    if (ilen == VZ_QF_TILE) {
+      // Full tile: batch the Kinv row loads 8 at a time so they are
+      // all in flight together. The rolled one-load-per-iteration
+      // form serializes 64 dependent L2 round-trips (~300 ns each),
+      // which made phase B ~12 us per tile in the megakernel
+      // (profiles: tools_sweep_phases.py). Same fma order per
+      // candidate chain — results are bit-identical.
+      for (int ib = 0; ib < VZ_QF_TILE; ib += 8) {
+        float kvv[8];
 #pragma unroll
-      for (int qq = 0; qq < 8; ++qq) {
-        acc[qq] = fmaf(k_i_lds[(wave * 8 + qq) * VZ_QF_TILE + i], kv,
-                       acc[qq]);
+        for (int u = 0; u < 8; ++u) {
+          kvv[u] = kinv[(long)(i0 + ib + u) * n + j0 + lane];
+        }
+#pragma unroll
+        for (int u = 0; u < 8; ++u) {
+#pragma unroll
+          for (int qq = 0; qq < 8; ++qq) {
+            acc[qq] = fmaf(
+                k_i_lds[(wave * 8 + qq) * VZ_QF_TILE + ib + u],
+                kvv[u], acc[qq]);
+          }
+        }
+      }
+    } else {
+      for (int i = 0; i < ilen; ++i) {
+        const float kv = kinv[(long)(i0 + i) * n + j0 + lane];
+#pragma unroll
+        for (int qq = 0; qq < 8; ++qq) {
+          acc[qq] = fmaf(k_i_lds[(wave * 8 + qq) * VZ_QF_TILE + i], kv,
+                         acc[qq]);
+        }
       }
     }
   }
