@@ -35,15 +35,16 @@ fac = VectorizedOptimizerFactory(max_evaluations=75000,
 opt = fac(n_continuous=20, categorical_sizes=[], seed=0, device='cuda')
 opt.optimize(score_fn, count=1)  # warm
 strat = opt.strategy
-base = strat._iter_t[2:10].clone()
+base = strat._iter_t[2:12].clone()
 torch.cuda.synchronize()
 t0 = time.perf_counter()
 opt.optimize(score_fn, count=1)
 torch.cuda.synchronize()
 wall = time.perf_counter() - t0
-delta = (strat._iter_t[2:10] - base).cpu().numpy().astype(float)
-names = ['A', 'barA', 'B', 'barB', 'B2', 'barB2', 'C', 'barC']
-total = delta.sum()
+delta = (strat._iter_t[2:12] - base).cpu().numpy().astype(float)
+names = ['A', 'barA', 'B', 'barB', 'B2', 'barB2', 'C', 'barC',
+         'B.stage', 'B.main']
+total = delta[:8].sum()
 iters = 3000 - (strat.pool_size // strat.batch_size + 2)
 print(f'sweep wall: {wall*1e3:.1f} ms over ~{iters} megakernel iters '
       f'({wall/3000*1e6:.1f} us/iter incl. eager head)')

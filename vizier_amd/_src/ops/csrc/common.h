@@ -108,8 +108,11 @@ __device__ __forceinline__ float vz_normal_pdf(float z) {
 template <typename LoadK, typename StoreP>
 __device__ __forceinline__ void vz_quadform_tile(
     const float* __restrict__ kinv, int b, int n, int tile, int tiles_n,
-    float* k_i_lds, float* k_j_lds, LoadK loadk, StoreP store) {
+    float* k_i_lds, float* k_j_lds, LoadK loadk, StoreP store,
+    unsigned long long* prof = nullptr) {
   const int tid = threadIdx.x;
+  const bool profme = (prof != nullptr && tid == 0);
+  unsigned long long tq0 = profme ? wall_clock64() : 0;
   const int ti = tile / tiles_n, tj = tile % tiles_n;
   const int i0 = ti * VZ_QF_TILE, j0 = tj * VZ_QF_TILE;
   const int ilen = min(VZ_QF_TILE, n - i0);
@@ -122,6 +125,11 @@ __device__ __forceinline__ void vz_quadform_tile(
                                      : 0.0f;
   }
   __syncthreads();
+  if (profme) {
+    const unsigned long long tq1 = wall_clock64();
+    prof[0] += tq1 - tq0;
+    tq0 = tq1;
+  }
   // Wave w owns candidates [8w, 8w+8); lane = tile column j. Kinv rows
   // are read coalesced across lanes (the 4 waves share each row
   // segment through L2).
@@ -173,4 +181,5 @@ __device__ __forceinline__ void vz_quadform_tile(
     if (lane == 0 && q < b) store(q, v);
   }
   __syncthreads();
+  if (profme) prof[1] += wall_clock64() - tq0;
 }

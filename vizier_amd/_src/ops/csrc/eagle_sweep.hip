@@ -304,7 +304,8 @@ eagle_sweep_kernel(
             [&](long idx) { return cload(k_ws + idx); },
             [&](int q, float v) {
               cstore(var_ws + (long)q * (total_tiles + 1) + t, v);
-            });
+            },
+            blockIdx.x == 0 ? iter_ptr + 10 : nullptr);
       }
     }
     VZ_SWEEP_PROF(2)
