@@ -179,9 +179,7 @@ class VectorizedOptimizer:
       # (the megakernel's phase B/B2 layout; see eagle_sweep.hip).
       var_ws = _torch.empty(b, tiles_n * tiles_n + 1,
                             dtype=_torch.float32, device=dev)
-      # 2 global (count, generation) + up to 256 per-8-WG group
-      # arrival counters for the hierarchical grid barrier.
-      barrier_buf = _torch.zeros(258, dtype=_torch.int32, device=dev)
+      barrier_buf = _torch.zeros(2, dtype=_torch.int32, device=dev)
       from vizier_amd._src.ops import dispatch as ops
       amp2 = scoring._amp * scoring._amp
       tr_radius = (scoring._tr_radius

@@ -83,36 +83,15 @@ __device__ __forceinline__ void grid_sync(unsigned int* bar) {
         __hip_atomic_load(bar + 1, __ATOMIC_RELAXED,
                           __HIP_MEMORY_SCOPE_AGENT);
     asm volatile("" ::: "memory");
-    // Two-level arrival: fetch_adds on ONE line serialize at the
-    // memory side (~128 x 25 ns = ~3 us per barrier at grid 128), so
-    // workgroups first arrive at a per-8-WG group counter (bar[2+g],
-    // parallel across lines) and only each group's last arrival
-    // touches the global counter (bar[0]).
-    const unsigned int g = blockIdx.x >> 3;
-    const unsigned int ngroups = (gridDim.x + 7) >> 3;
-    const unsigned int gsize =
-        (g == ngroups - 1) ? (gridDim.x - (g << 3)) : 8u;
-    const unsigned int garrived = __hip_atomic_fetch_add(
-        bar + 2 + g, 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-    if (garrived == gsize - 1) {
-      __hip_atomic_store(bar + 2 + g, 0u, __ATOMIC_RELAXED,
+    const unsigned int arrived = __hip_atomic_fetch_add(
+        bar, 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    if (arrived == gridDim.x - 1) {
+      __hip_atomic_store(bar, 0u, __ATOMIC_RELAXED,
                          __HIP_MEMORY_SCOPE_AGENT);
-      // The group reset must land before the global arrival: a peer
-      // released by this round could otherwise re-arrive at the group
-      // counter before the reset and corrupt the count.
-      __builtin_amdgcn_s_waitcnt(0);
-      const unsigned int arrived = __hip_atomic_fetch_add(
-          bar, 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-      if (arrived == ngroups - 1) {
-        __hip_atomic_store(bar, 0u, __ATOMIC_RELAXED,
-                           __HIP_MEMORY_SCOPE_AGENT);
-        __builtin_amdgcn_s_waitcnt(0);  // resets land before gen++
-        __hip_atomic_fetch_add(bar + 1, 1u, __ATOMIC_RELAXED,
-                               __HIP_MEMORY_SCOPE_AGENT);
-      }
-    }
-    if (__hip_atomic_load(bar + 1, __ATOMIC_RELAXED,
-                          __HIP_MEMORY_SCOPE_AGENT) == gen) {
+      __builtin_amdgcn_s_waitcnt(0);  // count reset lands before gen++
+      __hip_atomic_fetch_add(bar + 1, 1u, __ATOMIC_RELAXED,
+                             __HIP_MEMORY_SCOPE_AGENT);
+    } else {
       while (__hip_atomic_load(bar + 1, __ATOMIC_RELAXED,
                                __HIP_MEMORY_SCOPE_AGENT) == gen) {
         __builtin_amdgcn_s_sleep(8);

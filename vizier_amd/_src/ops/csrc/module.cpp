@@ -865,9 +865,7 @@ int64_t eagle_sweep(
   alpha = check_f32(alpha, "alpha");
   kinv = check_f32(kinv, "kinv");
   TORCH_CHECK(barrier_buf.scalar_type() == torch::kInt32 &&
-              barrier_buf.numel() >= 258,
-              "barrier_buf must be int32 (258,): 2 global + 256 "
-              "hierarchical group counters");
+              barrier_buf.numel() >= 2, "barrier_buf must be int32 (2,)");
   const int dc = x.size(1);
   const int n = x.size(0);
   TORCH_CHECK(pool_cont.numel() == pool_size * dc,
