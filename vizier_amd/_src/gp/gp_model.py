@@ -349,13 +349,17 @@ def nll_value_and_grad(raw: torch.Tensor, x: torch.Tensor,
   g_mean = -alpha.sum(dim=(-1, -2))
   grad[:, 2] = g_mean * (_MEAN_BOUNDS[1] - _MEAN_BOUNDS[0]) *       dbound[:, 2]
   # lengthscales (cols 3:): A = 1/2 M * G with
-  # G = amp^2 * (5/3)(1 + sqrt5 r) e^{-sqrt5 r}; symmetric.
+  # G = amp^2 * (5/3)(1 + sqrt5 r) e^{-sqrt5 r}; symmetric. All D
+  # traces in ONE batched GEMM: z_d^T A z_d = sum_i z_id (A z)_id and
+  # sum_ij A_ij z_id^2 = sum_i z_id^2 rowsum(A)_i (a per-d einsum loop
+  # re-read the (R, N, N) A tensor D times).
   A = 0.5 * M * (amp2 * (5.0 / 3.0) * (1.0 + sr) * e)
   arow = A.sum(-1)                                           # (R, N)
-  for dd in range(d):
-    zd = z[:, :, dd]                                          # (R, N)
-    t = 2.0 * (zd * zd * arow).sum(-1) -         2.0 * torch.einsum('ri,rij,rj->r', zd, A, zd)
-    grad[:, 3 + dd] = t * (_LOG_LS_BOUNDS[1] - _LOG_LS_BOUNDS[0]) *         dbound[:, 3 + dd]
+  az = torch.bmm(A, z)                                       # (R, N, D)
+  t_all = 2.0 * ((z * z) * arow.unsqueeze(-1)).sum(dim=1) - \
+      2.0 * (z * az).sum(dim=1)                              # (R, D)
+  grad[:, 3:] = t_all * (_LOG_LS_BOUNDS[1] - _LOG_LS_BOUNDS[0]) * \
+      dbound[:, 3:]
 
   grad = grad + 0.02 * raw
   bad = info != 0
