@@ -140,16 +140,28 @@ batched_trsv_lower_kernel(const float* __restrict__ L,  // (R, N, N)
   for (int k0 = 0; k0 < n; k0 += NB) {
     const int nb = min(NB, n - k0);
     if (tid < WAVE_SIZE) {
+      // Lane l preloads ITS row of the 32x32 diagonal block into
+      // registers up front (one parallel burst), so the serial
+      // 32-step substitution touches only registers + shfl. The
+      // previous form read M inside the chain — 32 dependent L2
+      // round trips per panel, ~800 us per 1000-row solve
+      // (profiles/fit_kernels_headline3.txt).
+      float row[NB];
+#pragma unroll
+      for (int c = 0; c < NB; ++c) {
+        row[c] = (lane < nb && c < nb && c <= lane)
+            ? M[(long)(k0 + lane) * n + k0 + c] : 0.0f;
+      }
       float myv = (lane < nb) ? rhs[k0 + lane] : 0.0f;
       for (int j = 0; j < nb; ++j) {
         float zj = 0.0f;
         if (j == lane) {
-          zj = myv / M[(long)(k0 + j) * n + k0 + j];
+          zj = myv / row[j];
           zseg[j] = zj;
         }
         zj = __shfl(zj, j, WAVE_SIZE);
         if (lane > j && lane < nb) {
-          myv = fmaf(-M[(long)(k0 + lane) * n + k0 + j], zj, myv);
+          myv = fmaf(-row[j], zj, myv);
         }
       }
       if (lane < nb) rhs[k0 + lane] = zseg[lane];
