@@ -25,14 +25,34 @@ def gp_bandit(problem, seed):
       max_evaluations=10000, device=device), seed=seed)
 
 
+def gp_ucb_pe(problem, seed):
+  from vizier_amd._src.algorithms.designers.gp_ucb_pe import (
+      UCBPEConfig,
+      VizierGPUCBPEBandit,
+  )
+  return VizierGPUCBPEBandit(problem, UCBPEConfig(device=device),
+                             seed=seed)
+
+
+FULL = len(sys.argv) > 1 and sys.argv[1] == 'full'
+functions = [('Sphere', bbob.Sphere), ('Rastrigin', bbob.Rastrigin)]
+algos = [('gp_bandit_ucb', gp_bandit)]
+seeds = 3
+if FULL:
+  functions += [('SharpRidge', bbob.SharpRidge),
+                ('Rosenbrock', bbob.Rosenbrock)]
+  algos += [('gp_ucb_pe', gp_ucb_pe)]
+  seeds = 2
+
 out = {}
-for fname, fn in (('Sphere', bbob.Sphere), ('Rastrigin', bbob.Rastrigin)):
-  bests = []
-  for seed in range(3):
-    t0 = time.time()
-    best = rb.run(gp_bandit, fn, 20, 100, seed)
-    bests.append(best)
-    print(f'{fname} seed={seed} best@100={best:.4f} '
-          f'({time.time()-t0:.1f}s)', flush=True)
-  out[f'{fname}/gp_bandit_ucb'] = bests
+for fname, fn in functions:
+  for aname, factory in algos:
+    bests = []
+    for seed in range(seeds):
+      t0 = time.time()
+      best = rb.run(factory, fn, 20, 100, seed)
+      bests.append(best)
+      print(f'{fname} {aname} seed={seed} best@100={best:.4f} '
+            f'({time.time()-t0:.1f}s)', flush=True)
+    out[f'{fname}/{aname}'] = bests
 print(json.dumps(out))
