@@ -205,10 +205,20 @@ def nll_values_with_chol(raw: torch.Tensor, x: torch.Tensor,
   eval's explicit-distance build."""
   params = GPParams.from_raw(raw)
   n = x.shape[0]
-  K = gram_matern52(x.unsqueeze(0), None, params.lengthscales,
-                    params.amplitude)
-  noise = params.noise.reshape(-1, 1, 1)
-  K = K + noise * torch.eye(n, dtype=x.dtype, device=x.device)
+  if (x.is_cuda and x.dtype == torch.float32 and not raw.requires_grad
+      and ops.extension_available()):
+    # One fused kernel builds K for every restart with noise*I folded
+    # in, replacing the 7-pass torch chain (scale, GEMM d2, clamp,
+    # sqrt, exp, mul, add-eye) over (R, N, N).
+    K = ops.require_ext().gram_matern52_batched(
+        x.contiguous(), params.lengthscales.contiguous(),
+        params.amplitude.contiguous(), params.noise.contiguous(),
+        False)[0]
+  else:
+    K = gram_matern52(x.unsqueeze(0), None, params.lengthscales,
+                      params.amplitude)
+    noise = params.noise.reshape(-1, 1, 1)
+    K = K + noise * torch.eye(n, dtype=x.dtype, device=x.device)
   resid = (y.unsqueeze(0) - params.mean.unsqueeze(-1)).unsqueeze(-1)
   if _use_custom_chol(K):
     ext = ops.require_ext()
@@ -273,15 +283,26 @@ def nll_value_and_grad(raw: torch.Tensor, x: torch.Tensor,
   amp2 = (params.amplitude ** 2).reshape(-1, 1, 1)          # (R,1,1)
   ls = params.lengthscales                                   # (R, D)
   z = x.unsqueeze(0) / ls.unsqueeze(1)                       # (R, N, D)
-  d2 = ((z * z).sum(-1, keepdim=True) +
-        (z * z).sum(-1).unsqueeze(-2) -
-        2.0 * z @ z.transpose(-1, -2)).clamp_min(1e-18)
-  r = d2.sqrt()
-  sr = math.sqrt(5.0) * r
-  e = torch.exp(-sr)
-  k0 = (1.0 + sr + sr * sr / 3.0) * e                        # unit-amp
-  K = amp2 * k0 + params.noise.reshape(-1, 1, 1) * torch.eye(
-      n, dtype=x.dtype, device=x.device)
+  fused_gram = (x.is_cuda and x.dtype == torch.float32 and
+                ops.extension_available())
+  if fused_gram:
+    # One kernel emits K (+noise*I) AND the gradient factor
+    # G = amp^2 (5/3)(1+sr)e from the same distance pass.
+    K, G = ops.require_ext().gram_matern52_batched(
+        x.contiguous(), ls.contiguous(),
+        params.amplitude.contiguous(), params.noise.contiguous(), True)
+    k0 = None
+  else:
+    d2 = ((z * z).sum(-1, keepdim=True) +
+          (z * z).sum(-1).unsqueeze(-2) -
+          2.0 * z @ z.transpose(-1, -2)).clamp_min(1e-18)
+    r = d2.sqrt()
+    sr = math.sqrt(5.0) * r
+    e = torch.exp(-sr)
+    k0 = (1.0 + sr + sr * sr / 3.0) * e                      # unit-amp
+    G = amp2 * (5.0 / 3.0) * (1.0 + sr) * e
+    K = amp2 * k0 + params.noise.reshape(-1, 1, 1) * torch.eye(
+        n, dtype=x.dtype, device=x.device)
   resid = (y.unsqueeze(0) - params.mean.unsqueeze(-1)).unsqueeze(-1)
   if chol_hint is not None:
     # The line-search ladder already factored K at these exact raw
@@ -338,11 +359,16 @@ def nll_value_and_grad(raw: torch.Tensor, x: torch.Tensor,
   dbound = sig * (1 - sig)
 
   # amplitude (raw col 0): dK/dv = 2 * amp^2 * k0, v = log amp.
-  g_amp = 0.5 * (M * (2.0 * amp2 * k0)).sum(dim=(-1, -2))
+  tr_m = torch.diagonal(M, dim1=-2, dim2=-1).sum(-1)         # (R,)
+  if k0 is not None:
+    g_amp = 0.5 * (M * (2.0 * amp2 * k0)).sum(dim=(-1, -2))
+  else:
+    # amp^2 k0 == K - noise*I, so the trace rewrites without k0:
+    # 0.5 tr(M * 2(K - noise I)) = sum(M*K) - noise * tr(M).
+    g_amp = (M * K).sum(dim=(-1, -2)) - params.noise * tr_m
   grad[:, 0] = g_amp * (_LOG_AMP_BOUNDS[1] - _LOG_AMP_BOUNDS[0]) *       dbound[:, 0]
   # noise (col 1): dK/dv = noise * I.
-  g_noise = 0.5 * params.noise * torch.diagonal(
-      M, dim1=-2, dim2=-1).sum(-1)
+  g_noise = 0.5 * params.noise * tr_m
   grad[:, 1] = g_noise * (_LOG_NOISE_BOUNDS[1] -
                           _LOG_NOISE_BOUNDS[0]) * dbound[:, 1]
   # mean (col 2): dNLL/dm = -sum(alpha).
@@ -353,7 +379,7 @@ def nll_value_and_grad(raw: torch.Tensor, x: torch.Tensor,
   # traces in ONE batched GEMM: z_d^T A z_d = sum_i z_id (A z)_id and
   # sum_ij A_ij z_id^2 = sum_i z_id^2 rowsum(A)_i (a per-d einsum loop
   # re-read the (R, N, N) A tensor D times).
-  A = 0.5 * M * (amp2 * (5.0 / 3.0) * (1.0 + sr) * e)
+  A = 0.5 * M * G
   arow = A.sum(-1)                                           # (R, N)
   az = torch.bmm(A, z)                                       # (R, N, D)
   t_all = 2.0 * ((z * z) * arow.unsqueeze(-1)).sum(dim=1) - \

@@ -14,6 +14,11 @@ extern "C" void launch_gram_matern52(const float* x1, const float* x2,
                                      int n, int m, int d, float amp2,
                                      int sym, hipStream_t stream);
 
+extern "C" void launch_gram_matern52_batched(
+    const float* x, const float* inv_ls, const float* amp2,
+    const float* noise, float* kout, float* gout, int r, int n, int d,
+    hipStream_t stream);
+
 extern "C" void launch_gram_matern52_bf16(
     const unsigned short* z1, const unsigned short* z2, const float* n1,
     const float* n2, float* out, int n, int m, int dp, float amp2,
@@ -185,6 +190,39 @@ torch::Tensor gram_matern52(torch::Tensor x1, torch::Tensor x2,
                        m, d, (float)(amplitude * amplitude), sym,
                        current_stream());
   return out;
+}
+
+std::vector<torch::Tensor> gram_matern52_batched(
+    torch::Tensor x, torch::Tensor lengthscales, torch::Tensor amplitude,
+    torch::Tensor noise, bool want_grad_factor) {
+  // (N, D) x; (R, D) lengthscales; (R,) amplitude, noise. Returns
+  // [K (R, N, N) with noise*I folded in] or [K, G] with the
+  // Matern-5/2 gradient factor.
+  x = check_f32(x, "x");
+  lengthscales = check_f32(lengthscales, "lengthscales");
+  amplitude = check_f32(amplitude, "amplitude");
+  noise = check_f32(noise, "noise");
+  const int n = x.size(0), d = x.size(1);
+  const int r = lengthscales.size(0);
+  TORCH_CHECK(lengthscales.dim() == 2 && lengthscales.size(1) == d,
+              "lengthscales must be (R, D)");
+  TORCH_CHECK(amplitude.numel() == r && noise.numel() == r,
+              "amplitude/noise must be (R,)");
+  auto inv_ls = (1.0f / lengthscales).contiguous();
+  auto amp2 = (amplitude * amplitude).contiguous();
+  auto K = torch::empty({r, n, n}, x.options());
+  torch::Tensor G;
+  float* gptr = nullptr;
+  if (want_grad_factor) {
+    G = torch::empty({r, n, n}, x.options());
+    gptr = G.data_ptr<float>();
+  }
+  launch_gram_matern52_batched(
+      x.data_ptr<float>(), inv_ls.data_ptr<float>(),
+      amp2.data_ptr<float>(), noise.data_ptr<float>(),
+      K.data_ptr<float>(), gptr, r, n, d, current_stream());
+  if (want_grad_factor) return {K, G};
+  return {K};
 }
 
 torch::Tensor gram_matern52_bf16_impl(torch::Tensor x1, torch::Tensor x2,
@@ -904,6 +942,9 @@ int64_t eagle_sweep(
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("gram_matern52_batched", &gram_matern52_batched,
+        "Batched-restart fused Matern-5/2 gram (+noise*I, optional "
+        "gradient factor)");
   m.def("gram_matern52", &gram_matern52,
         "Fused Matern-5/2 ARD Gram matrix (gfx950)");
   m.def("gram_matern52_bf16", &gram_matern52_bf16,
