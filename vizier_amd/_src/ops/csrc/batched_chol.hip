@@ -1030,28 +1030,18 @@ extern "C" void launch_batched_potrf_v2(float* A, int* info, int r,
   }
 }
 
-extern "C" void launch_batched_potrf_v6(float* A, int* info, int r,
-                                        int n, hipStream_t stream);
-
 extern "C" void launch_batched_potrf(float* A, int* info, int r, int n,
                                      hipStream_t stream) {
-  // Default: v6 (64-row panels, half the launches) for N >= 192,
-  // else v2 (32-row). v4 measured SLOWER (3.63 vs 2.85 ms, R=3
-  // N=1000: its three inlined solves blow the register budget to
-  // VGPR 256 + 248 AGPRs, occupancy 1) and stays opt-in via
-  // VIZIER_AMD_CHOL_IMPL=v4; =v2 forces 32-row panels everywhere.
-  static int impl = -1;
-  if (impl < 0) {
+  // v4 measured SLOWER than v2 at the headline shape (3.63 vs 2.85 ms,
+  // R=3 N=1000): the fused kernel's three inlined 32-wide solves blow
+  // its register budget (VGPR 256 + 248 AGPRs, occupancy 1 wave/SIMD
+  // vs v2's 4). Opt-in via VIZIER_AMD_CHOL_IMPL=v4 for experiments.
+  static int use_v4 = -1;
+  if (use_v4 < 0) {
     const char* env = std::getenv("VIZIER_AMD_CHOL_IMPL");
-    if (env != nullptr && env[0] == 'v' && env[1] == '4') impl = 4;
-    else if (env != nullptr && env[0] == 'v' && env[1] == '2') impl = 2;
-    else impl = 6;
+    use_v4 = (env != nullptr && env[0] == 'v' && env[1] == '4') ? 1 : 0;
   }
-  if (impl == 6 && n >= 192) {
-    launch_batched_potrf_v6(A, info, r, n, stream);
-    return;
-  }
-  if (impl != 4) {
+  if (!use_v4) {
     launch_batched_potrf_v2(A, info, r, n, stream);
     return;
   }
@@ -1075,150 +1065,4 @@ extern "C" void launch_batched_trsv_lower(const float* L, float* b,
                                           hipStream_t stream) {
   hipLaunchKernelGGL(batched_trsv_lower_kernel, dim3(r), dim3(CB), 0,
                      stream, L, b, r, n);
-}
-
-// -- v6: 64-wide panels (default for N >= 192) --------------------------
-//
-// The per-launch cost of the dependent panel/trailing chain is a
-// ~35-45 us FLOOR regardless of work (measured: N=64's three launches
-// run 35 us each; a trivial-kernel chain runs 4.4 us/launch, and five
-// kernel redesigns failed to beat the flat cost — it tracks the
-// footprint-scaled end-of-kernel L2 writeback). So the remaining
-// lever is LAUNCH COUNT: 64-row panels halve the rounds (16 panels,
-// 31 launches at N=1000 vs 63) at the same total flops. The 64x64
-// diagonal factor maps exactly onto one 64-lane wavefront (lane l
-// owns row l; pivots broadcast with shfl).
-
-#define NB64 64
-
-extern "C" __global__ __launch_bounds__(CB, 1) void
-batched_potrf_panel64_kernel(float* __restrict__ A,
-                             int* __restrict__ info,
-                             int r_count, int n, int k0) {
-  __shared__ float diag[NB64][NB64 + 1];
-  const int r = blockIdx.x;
-  if (r >= r_count) return;
-  const int row0 = k0 + NB64 + blockIdx.y * CB;
-  if (blockIdx.y > 0 && row0 >= n) return;
-  float* M = A + (long)r * n * n;
-  const int tid = threadIdx.x;
-  const int nb = min(NB64, n - k0);
-  if (tid < WAVE_SIZE) {
-    const int lane = tid;
-    // Every loop over row[] is FULLY unrolled with runtime guards:
-    // partial unrolls leave dynamic indexing and the compiler spills
-    // the 64-float array to scratch (measured 272 B/lane).
-    float row[NB64];
-#pragma unroll
-    for (int c = 0; c < NB64; ++c) {
-      row[c] = (lane < nb && c < nb)
-          ? M[(long)(k0 + lane) * n + k0 + c] : 0.0f;
-    }
-#pragma unroll
-    for (int j = 0; j < NB64; ++j) {
-      if (j < nb) {
-        const float piv = __shfl(row[j], j, WAVE_SIZE);
-        float d;
-        if (piv > 0.0f) {
-          d = sqrtf(piv);
-        } else {
-          d = 1.0f;
-          if (blockIdx.y == 0 && lane == 0 && info[r] == 0) {
-            info[r] = k0 + j + 1;
-          }
-        }
-        if (lane == j) row[j] = d;
-        if (lane > j) row[j] /= d;
-        const float lj = row[j];
-#pragma unroll
-        for (int c = 0; c < NB64; ++c) {
-          if (c > j && c < nb) {
-            const float lcj = __shfl(lj, c, WAVE_SIZE);
-            if (lane >= c) row[c] -= lj * lcj;
-          }
-        }
-      }
-    }
-#pragma unroll
-    for (int c = 0; c < NB64; ++c) {
-      if (lane < nb && c < nb) diag[lane][c] = row[c];
-    }
-  }
-  __syncthreads();
-  if (blockIdx.y == 0) {
-    for (int e = tid; e < nb * nb; e += CB) {
-      const int i = e / nb, c = e % nb;
-      M[(long)(k0 + i) * n + k0 + c] = (c <= i) ? diag[i][c] : 0.0f;
-    }
-  }
-  const int i = row0 + tid;  // one row per thread
-  if (i < n) {
-    float v[NB64];
-#pragma unroll
-    for (int j = 0; j < NB64; ++j) {
-      if (j < nb) {
-        float xv = M[(long)i * n + k0 + j];
-#pragma unroll
-        for (int p = 0; p < NB64; ++p) {
-          if (p < j) xv -= v[p] * diag[j][p];
-        }
-        v[j] = xv / diag[j][j];
-      }
-    }
-#pragma unroll
-    for (int j = 0; j < NB64; ++j) {
-      if (j < nb) M[(long)i * n + k0 + j] = v[j];
-    }
-  }
-}
-
-extern "C" __global__ __launch_bounds__(CB) void
-batched_potrf_trailing64_kernel(float* __restrict__ A, int r_count,
-                                int n, int k0) {
-  __shared__ float jpanel[NB64][NB64 + 1];
-  const int r = blockIdx.x;
-  if (r >= r_count) return;
-  float* M = A + (long)r * n * n;
-  const int tid = threadIdx.x;
-  const int nb = min(NB64, n - k0);
-  const int jb = k0 + nb + blockIdx.y * NB64;
-  if (jb >= n) return;
-  const int jl = min(NB64, n - jb);
-  for (int e = tid; e < jl * nb; e += CB) {
-    jpanel[e / nb][e % nb] = M[(long)(jb + e / nb) * n + k0 + e % nb];
-  }
-  __syncthreads();
-  for (int i = jb + tid; i < n; i += CB) {
-    float row[NB64];
-#pragma unroll
-    for (int p = 0; p < NB64; ++p) {
-      row[p] = (p < nb) ? M[(long)i * n + k0 + p] : 0.0f;
-    }
-    for (int j = 0; j < jl; ++j) {
-      if (jb + j > i) break;
-      float acc = 0.0f;
-#pragma unroll
-      for (int p = 0; p < NB64; ++p) {
-        acc = fmaf(row[p], jpanel[j][p], acc);
-      }
-      M[(long)i * n + jb + j] -= acc;
-    }
-  }
-}
-
-extern "C" void launch_batched_potrf_v6(float* A, int* info, int r,
-                                        int n, hipStream_t stream) {
-  for (int k0 = 0; k0 < n; k0 += NB64) {
-    const int nb = (n - k0) < NB64 ? (n - k0) : NB64;
-    const int rows = n - (k0 + nb);
-    const int rtiles = rows > 0 ? (rows + CB - 1) / CB : 1;
-    hipLaunchKernelGGL(batched_potrf_panel64_kernel, dim3(r, rtiles),
-                       dim3(CB), 0, stream, A, info, r, n, k0);
-    const int jtiles = (rows + NB64 - 1) / NB64;
-    if (jtiles > 0) {
-      hipLaunchKernelGGL(batched_potrf_trailing64_kernel,
-                         dim3(r, jtiles), dim3(CB), 0, stream, A, r,
-                         n, k0);
-    }
-  }
 }
