@@ -188,3 +188,58 @@ class TestFactorySerialization:
     assert pub.DTLZExperimenterFactory is mo_problems.DTLZExperimenterFactory
     assert pub.WFGExperimenterFactory is mo_problems.WFGExperimenterFactory
     assert pub.ZDTExperimenterFactory is mo_problems.ZDTExperimenterFactory
+
+
+class TestPropertyInvariants:
+  """Hypothesis-driven invariants over the native suites."""
+
+  def _hyp(self):
+    import hypothesis
+    import hypothesis.strategies as st
+    return hypothesis, st
+
+  def test_dtlz2_front_identity_random_positions(self):
+    hypothesis, st = self._hyp()
+
+    @hypothesis.given(st.integers(2, 4), st.integers(0, 10**6))
+    @hypothesis.settings(max_examples=20, deadline=None)
+    def check(m, seed):
+      rng = np.random.default_rng(seed)
+      dim = m + 4
+      fac = mo_problems.DTLZExperimenterFactory(
+          name='DTLZ2', dim=dim, num_objectives=m)
+      x = np.concatenate([rng.random(m - 1), np.full(dim - m + 1, 0.5)])
+      f = _eval(fac, x)
+      np.testing.assert_allclose(np.sum(f ** 2), 1.0, atol=1e-5)
+
+    check()
+
+  def test_zdt1_g_monotone_in_distance_params(self):
+    hypothesis, st = self._hyp()
+
+    @hypothesis.given(st.floats(0.05, 0.95), st.floats(0.0, 0.5),
+                      st.floats(0.5, 1.0))
+    @hypothesis.settings(max_examples=20, deadline=None)
+    def check(f1, lo, hi):
+      fac = mo_problems.ZDTExperimenterFactory(name='ZDT1', dim=4)
+      f_lo = _eval(fac, np.array([f1, lo, lo, lo]))
+      f_hi = _eval(fac, np.array([f1, hi, hi, hi]))
+      assert f_hi[1] >= f_lo[1] - 1e-6  # g grows with distance params
+
+    check()
+
+  def test_wfg_outputs_in_canonical_box(self):
+    hypothesis, st = self._hyp()
+
+    @hypothesis.given(st.integers(1, 9), st.integers(0, 10**6))
+    @hypothesis.settings(max_examples=25, deadline=None)
+    def check(which, seed):
+      rng = np.random.default_rng(seed)
+      fac = mo_problems.WFGExperimenterFactory(
+          name=f'WFG{which}', dim=5, num_objectives=2)
+      f = _eval(fac, rng.random(5))
+      assert np.all(np.isfinite(f))
+      assert np.all(f >= -1e-9)
+      assert f[0] <= 3.0 + 1e-6 and f[1] <= 5.0 + 1e-6
+
+    check()
